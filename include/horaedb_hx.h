@@ -1,0 +1,186 @@
+/* horaedb_hx.h — C-ABI drop-in boundary for HoraeDB's metric-engine scan/
+ * aggregate hot path, MI355X-native (gfx950 HIP kernels behind this ABI).
+ *
+ * Each entry point names the reference interface it replaces (apache/horaedb
+ * `main`, paths relative to /root/reference/src). The reference surface kept
+ * (SURVEY.md §8(b)):
+ *   trait ColumnarStorage { schema/write/scan/compact }   columnar_storage/src/storage.rs:76-89
+ *   ScanRequest { range, predicate, projections }         columnar_storage/src/storage.rs:65-70
+ *   trait MergeOperator { merge(batch) -> batch }         columnar_storage/src/operator.rs:30-34
+ *
+ * Conventions: all structs are POD; caller owns all inputs; callee allocates
+ * outputs, freed by the matching hx_*_free; status codes + hx_last_error()
+ * (thread-local string). A handle is usable from many threads; each call set
+ * uses its own HIP stream set per device. All compute is GPU-resident — if no
+ * MI355X/HIP runtime is available every scan entry returns HX_ERR_NO_GPU
+ * (there is no CPU fallback in this library).
+ */
+#ifndef HORAEDB_HX_H
+#define HORAEDB_HX_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef int32_t hx_status;
+enum {
+    HX_OK             = 0,
+    HX_ERR_IO         = 1,   /* file missing/unreadable */
+    HX_ERR_FORMAT     = 2,   /* not a parquet SST / unsupported layout */
+    HX_ERR_UNSUPPORTED= 3,   /* encoding/codec outside round-1 scope */
+    HX_ERR_NO_GPU     = 4,   /* HIP runtime/device unavailable */
+    HX_ERR_HIP        = 5,   /* HIP call failed (see hx_last_error) */
+    HX_ERR_INVALID    = 6,   /* bad argument */
+    HX_ERR_SCHEMA     = 7,   /* SST schema violates the metric contract */
+};
+
+/* Thread-local description of the last error from this thread. */
+const char* hx_last_error(void);
+
+/* ---- time range: [start, end), ms — types.rs:46-133 (TimeRange) -------- */
+typedef struct { int64_t start; int64_t end; } hx_time_range;
+
+/* ---- SST descriptor — sst.rs:154-160 (SstFile/FileMeta) ---------------- */
+typedef struct {
+    const char* path;       /* parquet file written by the reference writer */
+    uint64_t    sequence;   /* __seq__ of every row in the file (= file id,
+                               sst.rs:39-46); dedup winner = max sequence    */
+} hx_sst_desc;
+
+/* ---- predicates — ScanRequest.predicate (storage.rs:65-70) ------------- */
+typedef enum {
+    HX_PRED_SERIES_IN = 1,  /* series_id ∈ set (tag predicate materialized as
+                               a series-id set, BASELINE config 3)           */
+} hx_pred_kind;
+
+typedef struct {
+    int32_t kind;                   /* hx_pred_kind */
+    const uint64_t* series_ids;     /* HX_PRED_SERIES_IN: sorted or not */
+    size_t          n_series;
+} hx_pred;
+
+/* ---- scan spec — ScanRequest (storage.rs:65-70) ------------------------ */
+typedef struct {
+    hx_time_range   range;      /* ts-range predicate + SST pruning          */
+    const hx_pred*  preds;      /* extra predicates, may be NULL             */
+    size_t          n_preds;
+    const hx_sst_desc* ssts;    /* explicit SST list; NULL => all SSTs the
+                                   handle discovered under {store}/data that
+                                   overlap `range` (Manifest::find_ssts,
+                                   manifest/mod.rs:165-172)                  */
+    size_t          n_ssts;
+    const int32_t*  projection; /* hx_scan only: user-column indices, NULL =
+                                   all user columns (builtins stripped —
+                                   types.rs:203-239)                         */
+    size_t          n_projection;
+} hx_scan_spec;
+
+/* ---- aggregate spec — no reference counterpart;
+ * semantics pinned by BASELINE.json configs + rfc:218-231 ----------------- */
+enum {
+    HX_AGG_SUM   = 1u << 0,
+    HX_AGG_COUNT = 1u << 1,
+    HX_AGG_MIN   = 1u << 2,
+    HX_AGG_MAX   = 1u << 3,
+    HX_AGG_AVG   = 1u << 4,   /* finalized as sum/count */
+};
+typedef struct {
+    uint32_t ops;         /* OR of HX_AGG_*  (over the value column) */
+    int64_t  bucket_ms;   /* 0: group by series_id; >0: by (series_id,
+                             timestamp/bucket_ms) — config 5 downsample */
+} hx_agg_spec;
+
+typedef struct {
+    const int32_t* device_ids;  /* HIP device ordinals; NULL => {0} */
+    int32_t        n_devices;
+} hx_device_set;
+
+/* ---- results ----------------------------------------------------------- */
+/* Aggregate result: one row per group, sorted by (series_id, bucket) —
+ * the reference stream's PK order (read.rs:479-494). Arrays are owned by the
+ * table; NULL when the op was not requested. */
+typedef struct hx_result_table {
+    size_t          n_groups;
+    const uint64_t* series_id;
+    const int64_t*  bucket;     /* bucket index (ts/bucket_ms); NULL if bucket_ms==0 */
+    const double*   sum;
+    const uint64_t* count;
+    const double*   vmin;
+    const double*   vmax;
+    const double*   avg;
+} hx_result_table;
+
+/* Streaming columnar batch for the non-aggregating parity mode — the
+ * SendableRecordBatchStream analog (storage.rs:84, read.rs:349-385).
+ * Columns follow the projection order; builtin __seq__/__reserved__ are
+ * stripped as MergeStream does (read.rs:330-343). Column buffers are valid
+ * only during the callback. */
+typedef struct {
+    size_t   n_rows;
+    size_t   n_cols;
+    const void* const* cols;      /* col i: n_rows × 8B elements            */
+    const int32_t*     col_types; /* 0=u64, 1=i64(ts ms), 2=f64             */
+} hx_col_batch;
+typedef int32_t (*hx_batch_cb)(void* ctx, const hx_col_batch* batch); /* nonzero => stop */
+
+/* ---- lifecycle --------------------------------------------------------- */
+typedef struct hx_handle   hx_handle;
+typedef struct hx_prepared hx_prepared;
+
+/* Replaces ObjectBasedStorage::try_new (storage.rs:138-187): opens a store
+ * rooted at `store_path` ({store}/data/*.sst), reads every SST footer into
+ * an in-memory catalog (FileMeta: rows, size, ts min/max — sst.rs:154-160).
+ * segment_duration_ms partitions SSTs into time segments exactly as
+ * storage.rs:106-136 does (server default 12h, server/config.rs:53).
+ * Sequence of each file = numeric file stem ({id}.sst, sst.rs:193-205). */
+hx_status hx_open(const char* store_path, int64_t segment_duration_ms,
+                  hx_handle** out);
+void      hx_close(hx_handle*);
+
+/* Catalog introspection (Manifest::find_ssts, manifest/mod.rs:165-172). */
+hx_status hx_find_ssts(hx_handle*, hx_time_range range,
+                       const hx_sst_desc** out, size_t* n_out);
+
+/* Stage the scan's column chunks into device HBM (file IO + footer/page
+ * parse + row-group pruning + PCIe upload). Untimed prep: the timed hot
+ * path starts at hx_exec_agg with inputs resident in HBM. */
+hx_status hx_prepare(hx_handle*, const hx_scan_spec*, const hx_device_set*,
+                     hx_prepared** out);
+void      hx_prepared_free(hx_prepared*);
+
+/* The hot path: decode + filter + dedup-merge + group-by aggregate, fully
+ * on-GPU. Replaces scan (storage.rs:335-370) + read plan (read.rs:429-494)
+ * + MergeExec (read.rs:100-391) + the aggregate the north star adds. */
+hx_status hx_exec_agg(hx_prepared*, const hx_agg_spec*, hx_result_table** out);
+void      hx_result_free(hx_result_table*);
+
+/* One-shot convenience: prepare + exec_agg + release staging. */
+hx_status hx_scan_agg(hx_handle*, const hx_scan_spec*, const hx_agg_spec*,
+                      const hx_device_set*, hx_result_table** out);
+
+/* Non-aggregating parity mode: the merged, deduplicated, PK-sorted row
+ * stream itself (ColumnarStorage::scan semantics, storage.rs:335-370),
+ * delivered as columnar batches through `cb`. GPU decode/filter/merge with
+ * host readback; for parity testing, not the bench path. */
+hx_status hx_scan(hx_handle*, const hx_scan_spec*, const hx_device_set*,
+                  hx_batch_cb cb, void* ctx);
+
+/* ---- introspection for bench/tests ------------------------------------ */
+typedef struct {
+    double  exec_ms;          /* wall of last hx_exec_agg (HIP events)      */
+    double  agg_kernel_ms;    /* k_scan_agg launch time (HIP events)        */
+    double  decode_kernel_ms; /* delta/snappy decode kernels, if any        */
+    int64_t rows_scanned;     /* rows examined (pre-filter)                 */
+    int64_t rows_matched;     /* rows passing filter+dedup                  */
+    int64_t bytes_staged;     /* page payload bytes resident in HBM         */
+    double  stage_ms;         /* last hx_prepare wall (IO+parse+PCIe)       */
+} hx_exec_stats;
+hx_status hx_get_stats(hx_prepared*, hx_exec_stats* out);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* HORAEDB_HX_H */
