@@ -1,0 +1,193 @@
+#!/usr/bin/env python3
+# tools/gen_ssts.py — synthetic SST generator (shared by tests and bench.py).
+#
+# Writes metric-shaped Parquet SSTs exactly as the reference writer contract
+# produces them (storage.rs:193-213; defaults config.rs:120-133: row-group
+# 8192, PLAIN, dict off; encodings enum config.rs:54-75; Snappy or
+# uncompressed per config):
+#   (series_id u64 PK, timestamp i64 PK, value f64, __seq__ u64, __reserved__ u64)
+#   rows sorted by (series_id, timestamp); one __seq__ per file = file id;
+#   file path {out}/data/{seq}.sst (sst.rs:193-205).
+#
+# Dataset shape per BASELINE.json / SURVEY §8(d): n_series ids (sorted unique
+# u64), n_points = n_rows/n_series timestamps per series at 10s steps from
+# ts_start; SST w holds the w-th contiguous time-index window (disjoint ts
+# ranges => no cross-SST duplicate PKs, the compacted layout; overlap
+# scenarios for dedup tests are built with gen_sst_from_arrays).
+#
+# value(s_idx, t) is a counter-based splitmix64 hash -> U[0,1) so any row's
+# value is reproducible independent of chunking or worker count.
+import argparse
+import os
+import json
+import numpy as np
+
+TS_START_DEFAULT = 1735689600000  # 2025-01-01T00:00:00Z, SURVEY §8(d)
+STEP_MS_DEFAULT = 10_000
+
+SCHEMA_COLS = ["series_id", "timestamp", "value", "__seq__", "__reserved__"]
+
+
+def _pa_schema():
+    import pyarrow as pa
+    return pa.schema([
+        pa.field("series_id", pa.uint64(), nullable=False),
+        pa.field("timestamp", pa.int64(), nullable=False),
+        pa.field("value", pa.float64(), nullable=False),
+        pa.field("__seq__", pa.uint64(), nullable=False),
+        pa.field("__reserved__", pa.uint64(), nullable=False),
+    ])
+
+
+def splitmix64(x):
+    """Vectorized splitmix64 on uint64 arrays (wrapping arithmetic)."""
+    x = x.astype(np.uint64, copy=True)
+    x += np.uint64(0x9E3779B97F4A7C15)
+    z = x
+    z = (z ^ (z >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+    z = (z ^ (z >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    return z ^ (z >> np.uint64(31))
+
+
+def value_of(seed, global_idx):
+    """value = U[0,1) from splitmix64(seed*GOLDEN ^ global_idx)."""
+    mixed_seed = np.uint64((int(seed) * 0x9E3779B97F4A7C15) & 0xFFFFFFFFFFFFFFFF)
+    h = splitmix64(mixed_seed ^ global_idx.astype(np.uint64))
+    return (h >> np.uint64(11)).astype(np.float64) * (2.0 ** -53)
+
+
+def make_series_ids(n_series, seed):
+    """n_series sorted unique u64 ids (seahash-of-labels style)."""
+    rng = np.random.default_rng(np.random.PCG64(seed))
+    ids = rng.integers(0, np.iinfo(np.uint64).max, n_series, dtype=np.uint64)
+    ids = np.unique(ids)
+    while len(ids) < n_series:  # collision top-up (vanishingly rare)
+        extra = rng.integers(0, np.iinfo(np.uint64).max,
+                             n_series - len(ids) + 16, dtype=np.uint64)
+        ids = np.unique(np.concatenate([ids, extra]))
+    return ids[:n_series]
+
+
+def write_sst(path, series, ts, value, seq, row_group=8192,
+              compression="none", ts_encoding="PLAIN"):
+    """Write one SST from explicit row arrays (must be (series,ts)-sorted)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    n = len(series)
+    tbl = pa.table({
+        "series_id": np.ascontiguousarray(series, dtype=np.uint64),
+        "timestamp": np.ascontiguousarray(ts, dtype=np.int64),
+        "value": np.ascontiguousarray(value, dtype=np.float64),
+        "__seq__": np.full(n, seq, dtype=np.uint64),
+        "__reserved__": np.zeros(n, dtype=np.uint64),
+    }, schema=_pa_schema())
+    enc = {c: "PLAIN" for c in SCHEMA_COLS}
+    enc["timestamp"] = ts_encoding
+    pq.write_table(
+        tbl, path, row_group_size=row_group, use_dictionary=False,
+        compression="NONE" if compression == "none" else compression.upper(),
+        data_page_version="1.0", column_encoding=enc,
+        write_statistics=True)
+    return n
+
+
+def gen_sst_from_arrays(store_dir, seq, series, ts, value, sort=True, **kw):
+    """Test scenario helper: build {store}/data/{seq}.sst from explicit rows.
+    sort=True applies the writer's (pk...)-stable sort (storage.rs:244-256)."""
+    series = np.asarray(series, dtype=np.uint64)
+    ts = np.asarray(ts, dtype=np.int64)
+    value = np.asarray(value, dtype=np.float64)
+    if sort:
+        order = np.lexsort((ts, series))  # stable: equal PKs keep input order
+        series, ts, value = series[order], ts[order], value[order]
+    ddir = os.path.join(store_dir, "data")
+    os.makedirs(ddir, exist_ok=True)
+    path = os.path.join(ddir, f"{seq}.sst")
+    write_sst(path, series, ts, value, seq, **kw)
+    return path
+
+
+def _gen_one(args):
+    (w, out_dir, ids_path, n_series, n_points, n_ssts, seed, ts_start,
+     step_ms, row_group, compression, ts_encoding) = args
+    ids = np.load(ids_path, mmap_mode="r")
+    t0 = w * n_points // n_ssts
+    t1 = (w + 1) * n_points // n_ssts
+    k = t1 - t0
+    if k == 0:
+        return None
+    series = np.repeat(ids, k)
+    t_idx = np.tile(np.arange(t0, t1, dtype=np.int64), n_series)
+    ts = ts_start + t_idx * step_ms
+    gidx = (np.repeat(np.arange(n_series, dtype=np.uint64), k) *
+            np.uint64(n_points)) + t_idx.astype(np.uint64)
+    value = value_of(seed, gidx)
+    seq = w + 1
+    path = os.path.join(out_dir, "data", f"{seq}.sst")
+    n = write_sst(path, series, ts, value, seq, row_group=row_group,
+                  compression=compression, ts_encoding=ts_encoding)
+    return {"seq": seq, "path": path, "rows": int(n),
+            "ts_min": int(ts_start + t0 * step_ms),
+            "ts_max": int(ts_start + (t1 - 1) * step_ms)}
+
+
+def gen_dataset(out_dir, n_rows, n_series, n_ssts, seed=42,
+                ts_start=TS_START_DEFAULT, step_ms=STEP_MS_DEFAULT,
+                row_group=8192, compression="none", ts_encoding="PLAIN",
+                workers=1):
+    assert n_rows % n_series == 0, "n_rows must be a multiple of n_series"
+    n_points = n_rows // n_series
+    os.makedirs(os.path.join(out_dir, "data"), exist_ok=True)
+    ids = make_series_ids(n_series, seed)
+    ids_path = os.path.join(out_dir, "series_ids.npy")
+    np.save(ids_path, ids)
+    jobs = [(w, out_dir, ids_path, n_series, n_points, n_ssts, seed,
+             ts_start, step_ms, row_group, compression, ts_encoding)
+            for w in range(n_ssts)]
+    if workers > 1:
+        import multiprocessing as mp
+        with mp.get_context("spawn").Pool(workers) as pool:
+            metas = pool.map(_gen_one, jobs)
+    else:
+        metas = [_gen_one(j) for j in jobs]
+    metas = [m for m in metas if m]
+    manifest = {
+        "n_rows": n_rows, "n_series": n_series, "n_points": n_points,
+        "n_ssts": len(metas), "seed": seed, "ts_start": ts_start,
+        "step_ms": step_ms, "row_group": row_group,
+        "compression": compression, "ts_encoding": ts_encoding,
+        "ts_end": ts_start + n_points * step_ms, "ssts": metas,
+    }
+    with open(os.path.join(out_dir, "dataset.json"), "w") as f:
+        json.dump(manifest, f, indent=1)
+    return manifest
+
+
+def middle_range(manifest, frac=0.5):
+    """The benchmark's ts-range: middle `frac` of the dataset span."""
+    span = manifest["ts_end"] - manifest["ts_start"]
+    lo = manifest["ts_start"] + int(span * (0.5 - frac / 2))
+    hi = manifest["ts_start"] + int(span * (0.5 + frac / 2))
+    return lo, hi
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("out_dir")
+    p.add_argument("--rows", type=int, default=10_000_000)
+    p.add_argument("--series", type=int, default=100_000)
+    p.add_argument("--ssts", type=int, default=1)
+    p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--compression", default="none", choices=["none", "snappy"])
+    p.add_argument("--ts-encoding", default="PLAIN",
+                   choices=["PLAIN", "DELTA_BINARY_PACKED"])
+    p.add_argument("--workers", type=int, default=1)
+    args = p.parse_args()
+    m = gen_dataset(args.out_dir, args.rows, args.series, args.ssts,
+                    seed=args.seed, compression=args.compression,
+                    ts_encoding=args.ts_encoding, workers=args.workers)
+    print(json.dumps({k: v for k, v in m.items() if k != "ssts"}))
+
+
+if __name__ == "__main__":
+    main()
