@@ -1,0 +1,1245 @@
+// engine.cpp — C-ABI implementation (include/horaedb_hx.h): catalog (replaces
+// ObjectBasedStorage::try_new + Manifest::find_ssts), staging (replaces the
+// ParquetExec reader factory, read.rs:78-93, incl. the reference's row-group
+// pruning pushdown read.rs:459-470), and the GPU execution path (DESIGN.md
+// §3-§5). Compute is GPU-only; no CPU fallback exists here.
+#include "../../include/horaedb_hx.h"
+#include "parquet_meta.h"
+#include "hx_device.h"
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <atomic>
+#include <chrono>
+#include <cstdio>
+#include <cstring>
+#include <dirent.h>
+#include <fcntl.h>
+#include <functional>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <unistd.h>
+#include <vector>
+
+#include "hx_kernels.h"
+
+// ---------------------------------------------------------------------------
+// error plumbing
+// ---------------------------------------------------------------------------
+static thread_local std::string g_last_error;
+
+extern "C" const char* hx_last_error(void) { return g_last_error.c_str(); }
+
+static hx_status fail(hx_status code, const std::string& msg) {
+    g_last_error = msg;
+    return code;
+}
+
+#define HIP_TRY(expr)                                                         \
+    do {                                                                      \
+        hipError_t _e = (expr);                                               \
+        if (_e != hipSuccess)                                                 \
+            return fail(HX_ERR_HIP, std::string(#expr) + ": " +               \
+                                        hipGetErrorString(_e));               \
+    } while (0)
+
+// ---------------------------------------------------------------------------
+// catalog (hx_open)
+// ---------------------------------------------------------------------------
+namespace {
+
+struct ChunkRef {
+    int32_t codec = 0;
+    int64_t chunk_start = 0;
+    int64_t comp_size = 0;
+    int64_t num_values = 0;
+};
+
+struct CatRg {
+    int64_t n_rows = 0;
+    int64_t ts_min = 0, ts_max = 0;
+    bool has_ts_stats = false;
+    ChunkRef cols[3];  // series, ts, value
+};
+
+struct CatSst {
+    std::string path;
+    uint64_t seq = 0;
+    int64_t n_rows = 0;
+    int64_t ts_min = 0, ts_max = 0;
+    std::vector<CatRg> rgs;
+};
+
+}  // namespace
+
+struct hx_handle {
+    std::string store;
+    int64_t segment_ms = 0;
+    std::vector<CatSst> ssts;            // ascending seq
+    std::vector<hx_sst_desc> find_out;   // scratch for hx_find_ssts
+};
+
+static hx_status read_file_meta(const std::string& path, uint64_t seq,
+                                CatSst& out) {
+    int fd = open(path.c_str(), O_RDONLY);
+    if (fd < 0) return fail(HX_ERR_IO, "open " + path);
+    off_t fsize = lseek(fd, 0, SEEK_END);
+    uint8_t tail8[8];
+    if (fsize < 12 || pread(fd, tail8, 8, fsize - 8) != 8) {
+        close(fd);
+        return fail(HX_ERR_FORMAT, path + ": too small");
+    }
+    if (std::memcmp(tail8 + 4, "PAR1", 4) != 0) {
+        close(fd);
+        return fail(HX_ERR_FORMAT, path + ": missing PAR1 magic");
+    }
+    uint32_t flen;
+    std::memcpy(&flen, tail8, 4);
+    if ((int64_t)flen + 8 > fsize) {
+        close(fd);
+        return fail(HX_ERR_FORMAT, path + ": bad footer length");
+    }
+    std::vector<uint8_t> tail(flen + 8);
+    if (pread(fd, tail.data(), flen + 8, fsize - 8 - flen) != (ssize_t)(flen + 8)) {
+        close(fd);
+        return fail(HX_ERR_IO, path + ": footer read failed");
+    }
+    close(fd);
+
+    hx::FileMetadata m;
+    try {
+        m = hx::parse_footer(tail.data(), tail.size(), fsize);
+    } catch (const std::exception& e) {
+        return fail(HX_ERR_FORMAT, path + ": " + e.what());
+    }
+
+    // schema contract (types.rs:150-240): series_id/timestamp/value present,
+    // INT64/INT64/DOUBLE physical
+    int ci[3] = {-1, -1, -1};
+    for (size_t i = 0; i < m.columns.size(); i++) {
+        const auto& c = m.columns[i];
+        if (c.name == "series_id") ci[0] = (int)i;
+        else if (c.name == "timestamp") ci[1] = (int)i;
+        else if (c.name == "value") ci[2] = (int)i;
+    }
+    if (ci[0] < 0 || ci[1] < 0 || ci[2] < 0)
+        return fail(HX_ERR_SCHEMA, path + ": metric schema columns missing");
+    if (m.columns[ci[0]].physical_type != hx::PT_INT64 ||
+        m.columns[ci[1]].physical_type != hx::PT_INT64 ||
+        m.columns[ci[2]].physical_type != hx::PT_DOUBLE)
+        return fail(HX_ERR_SCHEMA, path + ": unexpected physical types");
+
+    out.path = path;
+    out.seq = seq;
+    out.n_rows = m.num_rows;
+    bool first = true;
+    for (const auto& rg : m.row_groups) {
+        CatRg cr;
+        cr.n_rows = rg.num_rows;
+        for (int k = 0; k < 3; k++) {
+            if ((size_t)ci[k] >= rg.columns.size())
+                return fail(HX_ERR_FORMAT, path + ": column chunk missing");
+            const auto& cc = rg.columns[ci[k]];
+            cr.cols[k].codec = cc.codec;
+            cr.cols[k].chunk_start = cc.chunk_start();
+            cr.cols[k].comp_size = cc.total_compressed_size;
+            cr.cols[k].num_values = cc.num_values;
+        }
+        const auto& tscc = rg.columns[ci[1]];
+        if (tscc.has_stats && tscc.stat_min.size() == 8 &&
+            tscc.stat_max.size() == 8) {
+            cr.has_ts_stats = true;
+            cr.ts_min = hx::stat_i64(tscc.stat_min);
+            cr.ts_max = hx::stat_i64(tscc.stat_max);
+            if (first || cr.ts_min < out.ts_min) out.ts_min = cr.ts_min;
+            if (first || cr.ts_max > out.ts_max) out.ts_max = cr.ts_max;
+            first = false;
+        }
+        out.rgs.push_back(std::move(cr));
+    }
+    if (first) {  // no stats anywhere: unbounded range (never pruned)
+        out.ts_min = INT64_MIN;
+        out.ts_max = INT64_MAX;
+    }
+    return HX_OK;
+}
+
+extern "C" hx_status hx_open(const char* store_path, int64_t segment_duration_ms,
+                             hx_handle** out) {
+    if (!store_path || !out) return fail(HX_ERR_INVALID, "null argument");
+    auto h = std::make_unique<hx_handle>();
+    h->store = store_path;
+    h->segment_ms = segment_duration_ms > 0 ? segment_duration_ms
+                                            : 12ll * 3600 * 1000;  // server/config.rs:53
+    std::string data_dir = h->store + "/data";
+    DIR* d = opendir(data_dir.c_str());
+    if (!d) return fail(HX_ERR_IO, "no data dir: " + data_dir);
+    std::vector<std::pair<uint64_t, std::string>> files;
+    while (dirent* e = readdir(d)) {
+        std::string name = e->d_name;
+        if (name.size() < 5 || name.substr(name.size() - 4) != ".sst") continue;
+        errno = 0;
+        char* endp = nullptr;
+        uint64_t seq = strtoull(name.c_str(), &endp, 10);
+        if (errno || !endp || std::string(endp) != ".sst") continue;  // sst.rs:193-205
+        files.emplace_back(seq, data_dir + "/" + name);
+    }
+    closedir(d);
+    std::sort(files.begin(), files.end());
+    for (auto& [seq, path] : files) {
+        CatSst c;
+        hx_status st = read_file_meta(path, seq, c);
+        if (st != HX_OK) return st;
+        h->ssts.push_back(std::move(c));
+    }
+    *out = h.release();
+    return HX_OK;
+}
+
+extern "C" void hx_close(hx_handle* h) { delete h; }
+
+// TimeRange::overlaps (types.rs:125-127): [start,end) vs file [min,max]
+static bool overlaps(const CatSst& s, hx_time_range r) {
+    return s.ts_min < r.end && s.ts_max >= r.start;
+}
+
+extern "C" hx_status hx_find_ssts(hx_handle* h, hx_time_range range,
+                                  const hx_sst_desc** out, size_t* n_out) {
+    if (!h || !out || !n_out) return fail(HX_ERR_INVALID, "null argument");
+    h->find_out.clear();
+    for (const auto& s : h->ssts)
+        if (overlaps(s, range))
+            h->find_out.push_back({s.path.c_str(), s.seq});
+    *out = h->find_out.data();
+    *n_out = h->find_out.size();
+    return HX_OK;
+}
+
+// ---------------------------------------------------------------------------
+// prepared scan (staging)
+// ---------------------------------------------------------------------------
+namespace {
+
+struct DevPlan {
+    int device = 0;
+    std::vector<hx::RgDesc> rgs;
+    std::vector<hx::SstDev> ssts;
+    std::vector<hx::ClusterDev> clusters;
+    std::vector<int32_t> cluster_members;
+    std::vector<hx::DeltaPageDesc> delta_pages;
+    std::vector<hx::CopyDesc> copies;
+    size_t blob_bytes = 0;
+    size_t dec_bytes = 0;
+    int64_t rows_scanned = 0;
+
+    // host staging
+    uint8_t* h_blob = nullptr;
+    bool h_blob_pinned = false;
+
+    // device memory
+    uint8_t* d_blob = nullptr;
+    uint8_t* d_dec = nullptr;
+    hx::RgDesc* d_rgs = nullptr;
+    hx::SstDev* d_ssts = nullptr;
+    hx::ClusterDev* d_clusters = nullptr;
+    int32_t* d_members = nullptr;
+    hx::DeltaPageDesc* d_delta = nullptr;
+    hx::CopyDesc* d_copies = nullptr;
+
+    // aggregate table (lazily sized)
+    uint32_t slots = 0;
+    uint64_t* t_series = nullptr;
+    int64_t* t_bucket = nullptr;
+    uint32_t* t_state = nullptr;
+    double* t_sum = nullptr;
+    unsigned long long* t_cnt = nullptr;
+    unsigned long long* t_min = nullptr;
+    unsigned long long* t_max = nullptr;
+    unsigned long long* d_counters = nullptr;  // fill, overflow, matched, n_out
+    uint64_t* d_sset = nullptr;
+    size_t sset_cap = 0;
+    uint64_t sset_empty = ~0ull;
+    uint32_t sset_mask = 0;
+
+    bool decoded = false;       // delta/copy kernels already ran
+    double decode_ms = 0;
+
+    // cached result scratch (device)
+    void* d_scratch = nullptr;
+    size_t scratch_cap = 0;
+    void* d_sort_temp = nullptr;
+    size_t sort_temp_cap = 0;
+
+    hipStream_t stream = nullptr;
+};
+
+static hx_status ensure_dev(void** p, size_t* cap, size_t need) {
+    if (*cap >= need) return HX_OK;
+    if (*p) hipFree(*p);
+    *p = nullptr;
+    *cap = 0;
+    HIP_TRY(hipMalloc(p, need));
+    *cap = need;
+    return HX_OK;
+}
+
+struct StagedSst {  // host-side bookkeeping per prepared SST
+    const CatSst* cat;
+    std::vector<int> rg_idx;   // selected row groups (ascending)
+    int64_t staged_rows = 0;
+    int32_t cluster = -1;
+    int32_t rank = 0;
+    int dev_slot = -1;         // index into DevPlan.ssts
+    int plan = -1;             // which DevPlan
+};
+
+}  // namespace
+
+struct hx_prepared {
+    hx_handle* h = nullptr;
+    hx_scan_spec spec{};
+    std::vector<uint64_t> sset_keys;   // owned copy of series-set predicate
+    std::vector<DevPlan> plans;
+    int64_t rows_scanned_total = 0;
+    int64_t bytes_staged = 0;
+    double stage_ms = 0;
+    hx_exec_stats last_stats{};
+};
+
+static int hip_device_count() {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+}
+
+static size_t align64(size_t x) { return (x + 63) & ~size_t(63); }
+
+static hx_status stage_device(hx_prepared* P, DevPlan& plan,
+                              std::vector<StagedSst*>& members) {
+    // ---- layout pass ----------------------------------------------------
+    struct PageJob {  // one (rg, col) data page to read+pack
+        StagedSst* ss;
+        int rg_cat;         // catalog rg index
+        int col;            // 0 series 1 ts 2 value
+        int64_t chunk_start, comp_size, num_values;
+        int32_t codec;
+        size_t dst_off;     // blob offset reserved (chunk_size upper bound)
+        // filled after page walk:
+        uint64_t final_off = 0;  // RgDesc offset value (blob or OFF_DEC)
+    };
+
+    std::vector<PageJob> jobs;
+    size_t blob_off = 0, dec_off = 0;
+
+    for (StagedSst* ss : members) {
+        hx::SstDev sd{};
+        sd.cluster = ss->cluster;
+        sd.rank = ss->rank;
+        sd.dense_series = 0;
+        sd.dense_ts = 0;
+        sd.n_staged = ss->staged_rows;
+        bool need_dense = ss->cluster >= 0;
+        if (need_dense) {
+            sd.dense_series = hx::OFF_DEC | dec_off;
+            dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
+            sd.dense_ts = hx::OFF_DEC | dec_off;
+            dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
+        }
+        ss->dev_slot = (int)plan.ssts.size();
+        plan.ssts.push_back(sd);
+
+        int64_t row_base = 0;
+        int prev_rg_desc = -1;
+        for (int rgi : ss->rg_idx) {
+            const CatRg& rg = ss->cat->rgs[rgi];
+            hx::RgDesc rd{};
+            rd.n_rows = (uint32_t)rg.n_rows;
+            rd.sst_id = (uint32_t)ss->dev_slot;
+            rd.row_base = row_base;
+            rd.next_rg = -1;
+            int this_desc = (int)plan.rgs.size();
+            if (prev_rg_desc >= 0) plan.rgs[prev_rg_desc].next_rg = this_desc;
+            prev_rg_desc = this_desc;
+            plan.rgs.push_back(rd);
+            for (int c = 0; c < 3; c++) {
+                const ChunkRef& cr = rg.cols[c];
+                PageJob j;
+                j.ss = ss;
+                j.rg_cat = rgi;
+                j.col = c;
+                j.chunk_start = cr.chunk_start;
+                j.comp_size = cr.comp_size;
+                j.num_values = cr.num_values;
+                j.codec = cr.codec;
+                j.dst_off = blob_off;
+                blob_off = align64(blob_off + size_t(cr.comp_size));
+                jobs.push_back(j);
+            }
+            row_base += rg.n_rows;
+            plan.rows_scanned += rg.n_rows;
+        }
+    }
+    // dec region for non-overlap delta pages is appended after the page walk
+    // (we do not know which chunks are delta until headers are read); reserve
+    // lazily via an atomic cursor.
+    plan.blob_bytes = blob_off;
+
+    // ---- read + page walk (parallel over jobs) --------------------------
+    if (plan.h_blob == nullptr && blob_off > 0) {
+        if (hipHostMalloc((void**)&plan.h_blob, blob_off,
+                          hipHostMallocDefault) == hipSuccess) {
+            plan.h_blob_pinned = true;
+        } else {
+            plan.h_blob = (uint8_t*)malloc(blob_off);
+            plan.h_blob_pinned = false;
+            if (!plan.h_blob) return fail(HX_ERR_IO, "staging alloc failed");
+        }
+    }
+
+    std::atomic<size_t> next{0};
+    std::atomic<int> err_flag{0};
+    std::mutex mu;  // guards delta/copy/dec_off bookkeeping + error string
+    std::string err_msg;
+    // per-rg descriptor index: jobs are 3 per rg in plan order
+    auto worker = [&]() {
+        std::vector<uint8_t> tmp;
+        int last_fd = -1;
+        const CatSst* last_cat = nullptr;
+        for (;;) {
+            size_t i = next.fetch_add(1);
+            if (i >= jobs.size() || err_flag.load()) break;
+            PageJob& j = jobs[i];
+            if (last_cat != j.ss->cat) {
+                if (last_fd >= 0) close(last_fd);
+                last_fd = open(j.ss->cat->path.c_str(), O_RDONLY);
+                last_cat = j.ss->cat;
+            }
+            if (last_fd < 0) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = "open " + j.ss->cat->path;
+                err_flag = 1;
+                break;
+            }
+            tmp.resize(j.comp_size);
+            if (pread(last_fd, tmp.data(), j.comp_size, j.chunk_start) !=
+                (ssize_t)j.comp_size) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = "chunk read " + j.ss->cat->path;
+                err_flag = 1;
+                break;
+            }
+            std::vector<hx::PageDesc> pages;
+            try {
+                pages = hx::walk_pages(tmp.data(), tmp.size(), j.chunk_start,
+                                       j.num_values);
+            } catch (const std::exception& e) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": " + e.what();
+                err_flag = 1;
+                break;
+            }
+            // round-1 layout contract: exactly one v1/v2 data page per chunk,
+            // no dictionary page, uncompressed (Snappy next)
+            const hx::PageDesc* dp = nullptr;
+            int n_data = 0;
+            for (const auto& p : pages) {
+                if (p.page_type == 0 || p.page_type == 3) {
+                    dp = &p;
+                    n_data++;
+                } else if (p.page_type == 2) {
+                    std::lock_guard<std::mutex> g(mu);
+                    err_msg = j.ss->cat->path + ": dictionary pages unsupported (round 1)";
+                    err_flag = 1;
+                }
+            }
+            if (err_flag.load()) break;
+            if (n_data != 1 || !dp || dp->num_values != j.num_values) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": expected one data page per chunk "
+                          "(row-group 8192 writer contract)";
+                err_flag = 1;
+                break;
+            }
+            if (j.codec != hx::CODEC_UNCOMPRESSED) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": codec unsupported in round 1 "
+                          "(uncompressed only; Snappy is next)";
+                err_flag = 1;
+                break;
+            }
+            size_t in_chunk = size_t(dp->payload_off - j.chunk_start);
+            size_t payload = size_t(dp->compressed_size) - size_t(dp->def_level_bytes);
+            const uint8_t* src = tmp.data() + in_chunk + dp->def_level_bytes;
+            std::memcpy(plan.h_blob + j.dst_off, src, payload);
+
+            if (dp->encoding == hx::ENC_PLAIN) {
+                if (payload != size_t(j.num_values) * 8) {
+                    std::lock_guard<std::mutex> g(mu);
+                    err_msg = j.ss->cat->path + ": PLAIN payload size mismatch";
+                    err_flag = 1;
+                    break;
+                }
+                j.final_off = j.dst_off;
+            } else if (dp->encoding == hx::ENC_DELTA_BINARY_PACKED && j.col != 2) {
+                std::lock_guard<std::mutex> g(mu);
+                hx::DeltaPageDesc dd{};
+                dd.src_off = j.dst_off;
+                dd.src_len = (uint32_t)payload;
+                dd.n_values = (uint32_t)j.num_values;
+                if (j.num_values > 8192) {
+                    err_msg = j.ss->cat->path + ": delta page > 8192 values";
+                    err_flag = 1;
+                    break;
+                }
+                dd.dst_off = hx::OFF_DEC | dec_off;
+                dec_off = align64(dec_off + size_t(j.num_values) * 8);
+                plan.delta_pages.push_back(dd);
+                j.final_off = dd.dst_off;
+            } else {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": encoding " +
+                          std::to_string(dp->encoding) + " unsupported (round 1)";
+                err_flag = 1;
+                break;
+            }
+        }
+        if (last_fd >= 0) close(last_fd);
+    };
+    unsigned n_threads = std::min<unsigned>(16, std::max(1u, std::thread::hardware_concurrency()));
+    {
+        std::vector<std::thread> ts;
+        for (unsigned t = 0; t < n_threads; t++) ts.emplace_back(worker);
+        for (auto& t : ts) t.join();
+    }
+    if (err_flag.load()) return fail(HX_ERR_UNSUPPORTED, err_msg);
+
+    // patch RgDesc offsets from job results; build dense copies for overlap
+    size_t job_i = 0;
+    size_t rg_i = 0;
+    for (StagedSst* ss : members) {
+        hx::SstDev& sd = plan.ssts[ss->dev_slot];
+        int64_t row_base = 0;
+        for (size_t k = 0; k < ss->rg_idx.size(); k++, rg_i++) {
+            hx::RgDesc& rd = plan.rgs[rg_i];
+            uint64_t offs[3];
+            for (int c = 0; c < 3; c++, job_i++) offs[c] = jobs[job_i].final_off;
+            rd.series_off = offs[0];
+            rd.ts_off = offs[1];
+            rd.val_off = offs[2];
+            if (ss->cluster >= 0) {
+                // dense (series, ts) arrays for binary-search dedup
+                hx::CopyDesc cs{};
+                cs.src_off = offs[0];
+                cs.dst_off = (sd.dense_series & hx::OFF_MASK) + uint64_t(row_base) * 8;
+                cs.n_values = rd.n_rows;
+                plan.copies.push_back(cs);
+                hx::CopyDesc ct{};
+                ct.src_off = offs[1];
+                ct.dst_off = (sd.dense_ts & hx::OFF_MASK) + uint64_t(row_base) * 8;
+                ct.n_values = rd.n_rows;
+                plan.copies.push_back(ct);
+            }
+            row_base += rd.n_rows;
+        }
+    }
+    plan.dec_bytes = dec_off;
+
+    // ---- upload ----------------------------------------------------------
+    HIP_TRY(hipSetDevice(plan.device));
+    if (!plan.stream) HIP_TRY(hipStreamCreate(&plan.stream));
+    if (plan.blob_bytes) {
+        HIP_TRY(hipMalloc((void**)&plan.d_blob, plan.blob_bytes));
+        HIP_TRY(hipMemcpyAsync(plan.d_blob, plan.h_blob, plan.blob_bytes,
+                               hipMemcpyHostToDevice, plan.stream));
+    }
+    if (plan.dec_bytes)
+        HIP_TRY(hipMalloc((void**)&plan.d_dec, plan.dec_bytes));
+    auto upload = [&](auto*& dptr, const auto& vec) -> hipError_t {
+        using T = std::remove_reference_t<decltype(vec[0])>;
+        if (vec.empty()) { dptr = nullptr; return hipSuccess; }
+        hipError_t e = hipMalloc((void**)&dptr, vec.size() * sizeof(T));
+        if (e != hipSuccess) return e;
+        return hipMemcpyAsync(dptr, vec.data(), vec.size() * sizeof(T),
+                              hipMemcpyHostToDevice, plan.stream);
+    };
+    HIP_TRY(upload(plan.d_rgs, plan.rgs));
+    HIP_TRY(upload(plan.d_ssts, plan.ssts));
+    HIP_TRY(upload(plan.d_clusters, plan.clusters));
+    HIP_TRY(upload(plan.d_members, plan.cluster_members));
+    HIP_TRY(upload(plan.d_delta, plan.delta_pages));
+    HIP_TRY(upload(plan.d_copies, plan.copies));
+    HIP_TRY(hipMalloc((void**)&plan.d_counters, 4 * sizeof(unsigned long long)));
+    HIP_TRY(hipStreamSynchronize(plan.stream));
+    return HX_OK;
+}
+
+extern "C" hx_status hx_prepare(hx_handle* h, const hx_scan_spec* spec,
+                                const hx_device_set* devs, hx_prepared** out) {
+    if (!h || !spec || !out) return fail(HX_ERR_INVALID, "null argument");
+    auto t0 = std::chrono::steady_clock::now();
+    int n_gpu = hip_device_count();
+    if (n_gpu <= 0)
+        return fail(HX_ERR_NO_GPU, "no HIP device visible (the scan path is "
+                                   "GPU-only; there is no CPU fallback)");
+    std::vector<int> device_ids;
+    if (devs && devs->device_ids && devs->n_devices > 0)
+        device_ids.assign(devs->device_ids, devs->device_ids + devs->n_devices);
+    else
+        device_ids = {0};
+    for (int d : device_ids)
+        if (d < 0 || d >= n_gpu)
+            return fail(HX_ERR_INVALID, "bad device id");
+
+    auto P = std::make_unique<hx_prepared>();
+    P->h = h;
+    P->spec = *spec;
+    for (size_t i = 0; i < spec->n_preds; i++) {
+        const hx_pred& p = spec->preds[i];
+        if (p.kind != HX_PRED_SERIES_IN)
+            return fail(HX_ERR_UNSUPPORTED, "unknown predicate kind");
+        P->sset_keys.insert(P->sset_keys.end(), p.series_ids,
+                            p.series_ids + p.n_series);
+    }
+
+    // resolve SSTs
+    std::vector<const CatSst*> chosen;
+    if (spec->ssts && spec->n_ssts) {
+        for (size_t i = 0; i < spec->n_ssts; i++) {
+            const CatSst* found = nullptr;
+            for (const auto& s : h->ssts)
+                if (s.path == spec->ssts[i].path) { found = &s; break; }
+            if (!found)
+                return fail(HX_ERR_INVALID,
+                            std::string("sst not in catalog: ") + spec->ssts[i].path);
+            chosen.push_back(found);
+        }
+    } else {
+        for (const auto& s : h->ssts)
+            if (overlaps(s, spec->range)) chosen.push_back(&s);
+    }
+
+    // select row groups (reference pushdown pruning read.rs:459-470)
+    std::vector<StagedSst> staged;
+    staged.reserve(chosen.size());
+    for (const CatSst* c : chosen) {
+        StagedSst ss;
+        ss.cat = c;
+        for (size_t r = 0; r < c->rgs.size(); r++) {
+            const CatRg& rg = c->rgs[r];
+            if (rg.has_ts_stats &&
+                !(rg.ts_min < spec->range.end && rg.ts_max >= spec->range.start))
+                continue;
+            ss.rg_idx.push_back((int)r);
+            ss.staged_rows += rg.n_rows;
+        }
+        if (!ss.rg_idx.empty()) staged.push_back(std::move(ss));
+    }
+
+    // overlap clusters: connected components of ts-range
+    // intersection; members sorted by seq => dense ranks. Clusters are the
+    // unit of device assignment (binary-search dedup needs co-location).
+    size_t n = staged.size();
+    std::vector<int> parent(n);
+    for (size_t i = 0; i < n; i++) parent[i] = (int)i;
+    std::function<int(int)> find = [&](int x) {
+        while (parent[x] != x) { parent[x] = parent[parent[x]]; x = parent[x]; }
+        return x;
+    };
+    for (size_t i = 0; i < n; i++)
+        for (size_t j = i + 1; j < n; j++) {
+            const CatSst* a = staged[i].cat;
+            const CatSst* b = staged[j].cat;
+            if (a->ts_min <= b->ts_max && b->ts_min <= a->ts_max)
+                parent[find((int)i)] = find((int)j);
+        }
+    // group indexes by root; singleton groups get cluster = -1
+    std::vector<std::vector<int>> groups;
+    {
+        std::vector<int> root_to_group((int)n, -1);
+        for (size_t i = 0; i < n; i++) {
+            int r = find((int)i);
+            if (root_to_group[r] < 0) {
+                root_to_group[r] = (int)groups.size();
+                groups.emplace_back();
+            }
+            groups[root_to_group[r]].push_back((int)i);
+        }
+    }
+
+    // assign groups to devices (least-loaded by staged rows)
+    P->plans.resize(device_ids.size());
+    for (size_t d = 0; d < device_ids.size(); d++)
+        P->plans[d].device = device_ids[d];
+    std::vector<int64_t> load(device_ids.size(), 0);
+    std::vector<std::vector<StagedSst*>> members_per_plan(device_ids.size());
+    // big groups first for balance
+    std::sort(groups.begin(), groups.end(), [&](const auto& a, const auto& b) {
+        int64_t ra = 0, rb = 0;
+        for (int i : a) ra += staged[i].staged_rows;
+        for (int i : b) rb += staged[i].staged_rows;
+        return ra > rb;
+    });
+    for (auto& g : groups) {
+        size_t best = 0;
+        for (size_t d = 1; d < load.size(); d++)
+            if (load[d] < load[best]) best = d;
+        DevPlan& plan = P->plans[best];
+        // sort group members by seq => ranks
+        std::sort(g.begin(), g.end(), [&](int a, int b) {
+            return staged[a].cat->seq < staged[b].cat->seq;
+        });
+        int32_t cluster_id = -1;
+        if (g.size() > 1) {
+            cluster_id = (int32_t)plan.clusters.size();
+            hx::ClusterDev cd{(int32_t)plan.cluster_members.size(), (int32_t)g.size()};
+            plan.clusters.push_back(cd);
+        }
+        int32_t rank = 0;
+        for (int idx : g) {
+            staged[idx].cluster = cluster_id;
+            staged[idx].rank = rank++;
+            staged[idx].plan = (int)best;
+            members_per_plan[best].push_back(&staged[idx]);
+            load[best] += staged[idx].staged_rows;
+        }
+        if (cluster_id >= 0) {
+            // dev_slot not known yet; fill members after staging assigns slots
+        }
+    }
+
+    for (size_t d = 0; d < P->plans.size(); d++) {
+        hx_status st = stage_device(P.get(), P->plans[d], members_per_plan[d]);
+        if (st != HX_OK) return st;
+        // cluster member lists (dev slots) — members were pushed in group
+        // order, so slots ascend within each cluster
+        DevPlan& plan = P->plans[d];
+        plan.cluster_members.assign(plan.ssts.size(), 0);
+        {
+            std::vector<int32_t> cursor(plan.clusters.size());
+            for (size_t c = 0; c < plan.clusters.size(); c++)
+                cursor[c] = plan.clusters[c].first;
+            for (StagedSst* ss : members_per_plan[d])
+                if (ss->cluster >= 0)
+                    plan.cluster_members[cursor[ss->cluster]++] = ss->dev_slot;
+        }
+        if (!plan.cluster_members.empty()) {
+            HIP_TRY(hipSetDevice(plan.device));
+            if (plan.d_members) HIP_TRY(hipFree(plan.d_members));
+            HIP_TRY(hipMalloc((void**)&plan.d_members,
+                              plan.cluster_members.size() * sizeof(int32_t)));
+            HIP_TRY(hipMemcpy(plan.d_members, plan.cluster_members.data(),
+                              plan.cluster_members.size() * sizeof(int32_t),
+                              hipMemcpyHostToDevice));
+        }
+        P->rows_scanned_total += plan.rows_scanned;
+        P->bytes_staged += (int64_t)plan.blob_bytes;
+    }
+    P->stage_ms = std::chrono::duration<double, std::milli>(
+                      std::chrono::steady_clock::now() - t0).count();
+    *out = P.release();
+    return HX_OK;
+}
+
+extern "C" void hx_prepared_free(hx_prepared* P) {
+    if (!P) return;
+    for (auto& plan : P->plans) {
+        hipSetDevice(plan.device);
+        if (plan.h_blob) {
+            if (plan.h_blob_pinned) hipHostFree(plan.h_blob);
+            else free(plan.h_blob);
+        }
+        for (void* p : {(void*)plan.d_blob, (void*)plan.d_dec, (void*)plan.d_rgs,
+                        (void*)plan.d_ssts, (void*)plan.d_clusters,
+                        (void*)plan.d_members, (void*)plan.d_delta,
+                        (void*)plan.d_copies, (void*)plan.t_series,
+                        (void*)plan.t_bucket, (void*)plan.t_state,
+                        (void*)plan.t_sum, (void*)plan.t_cnt, (void*)plan.t_min,
+                        (void*)plan.t_max, (void*)plan.d_counters,
+                        (void*)plan.d_sset})
+            if (p) hipFree(p);
+        if (plan.stream) hipStreamDestroy(plan.stream);
+    }
+    delete P;
+}
+
+// ---------------------------------------------------------------------------
+// execution (hx_exec_agg): the timed hot path — DESIGN.md §4/§8
+// ---------------------------------------------------------------------------
+namespace {
+
+struct PartialResult {  // one device's sorted aggregate table, on host
+    size_t n = 0;
+    std::vector<uint64_t> series;
+    std::vector<int64_t> bucket;
+    std::vector<double> sum;
+    std::vector<unsigned long long> cnt;
+    std::vector<double> vmin, vmax;
+};
+
+uint32_t next_pow2_u32(uint64_t x) {
+    uint32_t p = 1;
+    while (p < x && p < (1u << 30)) p <<= 1;
+    return p;
+}
+
+hx_status alloc_table(DevPlan& plan, uint32_t slots, uint32_t ops, bool bucket) {
+    if (plan.slots == slots && plan.t_series) return HX_OK;
+    for (void** p : {(void**)&plan.t_series, (void**)&plan.t_bucket,
+                     (void**)&plan.t_state, (void**)&plan.t_sum,
+                     (void**)&plan.t_cnt, (void**)&plan.t_min,
+                     (void**)&plan.t_max})
+        if (*p) { hipFree(*p); *p = nullptr; }
+    plan.slots = slots;
+    HIP_TRY(hipMalloc((void**)&plan.t_series, size_t(slots) * 8));
+    HIP_TRY(hipMalloc((void**)&plan.t_state, size_t(slots) * 4));
+    if (bucket) HIP_TRY(hipMalloc((void**)&plan.t_bucket, size_t(slots) * 8));
+    if (ops & (HX_AGG_SUM | HX_AGG_AVG))
+        HIP_TRY(hipMalloc((void**)&plan.t_sum, size_t(slots) * 8));
+    if (ops & (HX_AGG_COUNT | HX_AGG_AVG))
+        HIP_TRY(hipMalloc((void**)&plan.t_cnt, size_t(slots) * 8));
+    if (ops & HX_AGG_MIN)
+        HIP_TRY(hipMalloc((void**)&plan.t_min, size_t(slots) * 8));
+    if (ops & HX_AGG_MAX)
+        HIP_TRY(hipMalloc((void**)&plan.t_max, size_t(slots) * 8));
+    return HX_OK;
+}
+
+// ops mask for the kernel: AVG implies SUM+COUNT accumulation
+uint32_t kernel_ops(uint32_t ops) {
+    uint32_t k = ops;
+    if (ops & HX_AGG_AVG) k |= HX_AGG_SUM | HX_AGG_COUNT;
+    return k;
+}
+
+hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
+                    PartialResult& out, double* agg_kernel_ms,
+                    unsigned long long* matched_out) {
+    HIP_TRY(hipSetDevice(plan.device));
+    hipStream_t s = plan.stream;
+    const uint32_t ops = kernel_ops(agg->ops);
+    const bool bucket = agg->bucket_ms > 0;
+
+    // series-set predicate upload (once per prepared)
+    if (!P->sset_keys.empty() && !plan.d_sset) {
+        // open hash set; sentinel = a value not in the set
+        std::vector<uint64_t> keys(P->sset_keys);
+        std::sort(keys.begin(), keys.end());
+        keys.erase(std::unique(keys.begin(), keys.end()), keys.end());
+        uint64_t empty = ~0ull;
+        while (std::binary_search(keys.begin(), keys.end(), empty)) empty--;
+        uint32_t cap = next_pow2_u32(keys.size() * 2 + 16);
+        std::vector<uint64_t> table(cap, empty);
+        auto mix = [](uint64_t x) {
+            x += 0x9E3779B97F4A7C15ull;
+            x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+            x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+            return x ^ (x >> 31);
+        };
+        for (uint64_t k : keys) {
+            uint32_t i = (uint32_t)mix(k) & (cap - 1);
+            while (table[i] != empty) i = (i + 1) & (cap - 1);
+            table[i] = k;
+        }
+        HIP_TRY(hipMalloc((void**)&plan.d_sset, size_t(cap) * 8));
+        HIP_TRY(hipMemcpy(plan.d_sset, table.data(), size_t(cap) * 8,
+                          hipMemcpyHostToDevice));
+        plan.sset_mask = cap - 1;
+        plan.sset_empty = empty;
+    }
+
+    // decode passes (persist across exec calls)
+    if (!plan.decoded) {
+        hipEvent_t d0, d1;
+        HIP_TRY(hipEventCreate(&d0));
+        HIP_TRY(hipEventCreate(&d1));
+        HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
+        HIP_TRY(hipEventRecord(d0, s));
+        if (!plan.delta_pages.empty())
+            HIP_TRY(hx::launch_decode_delta(s, plan.d_blob, plan.d_dec,
+                                            plan.d_delta,
+                                            (uint32_t)plan.delta_pages.size(),
+                                            plan.d_counters + 1));
+        if (!plan.copies.empty())
+            HIP_TRY(hx::launch_copy_u64(s, plan.d_blob, plan.d_dec,
+                                        plan.d_copies,
+                                        (uint32_t)plan.copies.size()));
+        HIP_TRY(hipEventRecord(d1, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        float ms = 0;
+        HIP_TRY(hipEventElapsedTime(&ms, d0, d1));
+        plan.decode_ms = ms;
+        unsigned long long decode_err = 0;
+        HIP_TRY(hipMemcpy(&decode_err, plan.d_counters + 1, 8,
+                          hipMemcpyDeviceToHost));
+        hipEventDestroy(d0);
+        hipEventDestroy(d1);
+        if (decode_err)
+            return fail(HX_ERR_FORMAT, "delta page decode failed (malformed "
+                                       "or >8192-value page)");
+        plan.decoded = true;
+    }
+
+    // table size heuristic; grows on overflow
+    uint32_t slots = plan.slots;
+    if (!slots) {
+        const char* env = getenv("HX_TABLE_SLOTS");
+        if (env) slots = next_pow2_u32(strtoull(env, nullptr, 10));
+        else slots = next_pow2_u32(std::max<uint64_t>(1 << 16,
+                                       (uint64_t)plan.rows_scanned / 4));
+        if (slots > (1u << 27)) slots = 1u << 27;
+    }
+
+    unsigned long long counters[4];
+    for (int attempt = 0; attempt < 4; attempt++) {
+        hx_status st = alloc_table(plan, slots, ops, bucket);
+        if (st != HX_OK) return st;
+        // reset table + counters (part of the step)
+        HIP_TRY(hipMemsetAsync(plan.t_state, 0, size_t(slots) * 4, s));
+        if (plan.t_sum) HIP_TRY(hipMemsetAsync(plan.t_sum, 0, size_t(slots) * 8, s));
+        if (plan.t_cnt) HIP_TRY(hipMemsetAsync(plan.t_cnt, 0, size_t(slots) * 8, s));
+        if (plan.t_min) HIP_TRY(hipMemsetAsync(plan.t_min, 0xFF, size_t(slots) * 8, s));
+        if (plan.t_max) HIP_TRY(hipMemsetAsync(plan.t_max, 0, size_t(slots) * 8, s));
+        HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
+
+        hx::AggParams A{};
+        A.rgs = plan.d_rgs;
+        A.n_rgs = (uint32_t)plan.rgs.size();
+        A.ssts = plan.d_ssts;
+        A.clusters = plan.d_clusters;
+        A.cluster_members = plan.d_members;
+        A.blob = plan.d_blob;
+        A.dec = plan.d_dec;
+        A.ts_lo = P->spec.range.start;
+        A.ts_hi = P->spec.range.end;
+        A.sset = plan.d_sset;
+        A.sset_mask = plan.sset_mask;
+        A.sset_empty = plan.sset_empty;
+        A.use_sset = plan.d_sset ? 1 : 0;
+        A.bucket_ms = bucket ? agg->bucket_ms : 0;
+        A.ops = ops;
+        A.table = {plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
+                   plan.t_cnt, plan.t_min, plan.t_max, slots - 1};
+        A.fill = plan.d_counters + 0;
+        A.overflow = plan.d_counters + 1;
+        A.matched = plan.d_counters + 2;
+
+        hipEvent_t e0, e1;
+        HIP_TRY(hipEventCreate(&e0));
+        HIP_TRY(hipEventCreate(&e1));
+        HIP_TRY(hipEventRecord(e0, s));
+        HIP_TRY(hx::launch_scan_agg(s, A, 0));
+        HIP_TRY(hipEventRecord(e1, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        float ms = 0;
+        HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
+        hipEventDestroy(e0);
+        hipEventDestroy(e1);
+        HIP_TRY(hipMemcpy(counters, plan.d_counters, 32, hipMemcpyDeviceToHost));
+        if (counters[1] == 0) {  // no overflow
+            *agg_kernel_ms = ms;
+            break;
+        }
+        if (attempt == 3)
+            return fail(HX_ERR_HIP, "aggregate table overflow persisted");
+        slots = slots >= (1u << 28) ? slots : slots * 4;
+        plan.slots = 0;  // force realloc
+    }
+    *matched_out = counters[2];
+    unsigned long long fill = counters[0];
+
+    // ---- compact + sort + gather ---------------------------------------
+    out.n = fill;
+    if (fill == 0) return HX_OK;
+    const uint32_t n = (uint32_t)fill;
+    // scratch layout: up to 10 arrays of n x 8B + 2 perm arrays n x 4B
+    const int n_val_arrays = 1 /*series*/ + (bucket ? 1 : 0) +
+                             (plan.t_sum ? 1 : 0) + (plan.t_cnt ? 1 : 0) +
+                             (plan.t_min ? 1 : 0) + (plan.t_max ? 1 : 0);
+    size_t need = size_t(n) * 8 * (2 * n_val_arrays + 2) + size_t(n) * 4 * 3 + 256;
+    hx_status st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
+    if (st != HX_OK) return st;
+    uint8_t* base = (uint8_t*)plan.d_scratch;
+    auto carve8 = [&](uint32_t count) {
+        uint8_t* p = base;
+        base += size_t(count) * 8;
+        return p;
+    };
+    uint64_t* c_series = (uint64_t*)carve8(n);
+    long long* c_bucket = bucket ? (long long*)carve8(n) : nullptr;
+    double* c_sum = plan.t_sum ? (double*)carve8(n) : nullptr;
+    unsigned long long* c_cnt = plan.t_cnt ? (unsigned long long*)carve8(n) : nullptr;
+    double* c_min = plan.t_min ? (double*)carve8(n) : nullptr;
+    double* c_max = plan.t_max ? (double*)carve8(n) : nullptr;
+    uint64_t* keys_tmp = (uint64_t*)carve8(n);       // sort keys in/out
+    uint64_t* vals_tmp = (uint64_t*)carve8(n);       // gather staging
+    // remaining val arrays share vals_tmp sequentially (gather one at a time)
+    uint32_t* perm_a = (uint32_t*)base; base += size_t(n) * 4;
+    uint32_t* perm_b = (uint32_t*)base; base += size_t(n) * 4;
+    uint32_t* perm_c = (uint32_t*)base; base += size_t(n) * 4;
+    unsigned long long* d_nout = (unsigned long long*)base;
+
+    HIP_TRY(hipMemsetAsync(d_nout, 0, 8, s));
+    hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
+    hx::AggTable T{plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
+                   plan.t_cnt, plan.t_min, plan.t_max, plan.slots - 1};
+    HIP_TRY(hx::launch_compact(s, T, plan.slots, ops,
+                               bucket ? agg->bucket_ms : 0, co));
+
+    // sort: LSD-stable — by bucket first (if any), then by series
+    HIP_TRY(hx::launch_iota(s, perm_a, n));
+    const uint32_t* perm_in = perm_a;
+    uint32_t* perm_out = perm_b;
+    if (bucket) {
+        // biased bucket keys for signed order
+        HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_bucket,
+                                      perm_a, (unsigned long long*)keys_tmp, n));
+        // bias: XOR sign bit (monotone i64->u64). gather wrote bucket[perm]=bucket (iota)
+        // reuse avg kernel slot: do bias inside gather? simpler: small kernel not
+        // available — bias via sort on raw then fix order: instead bias on host? n large.
+        // Use trick: radix sort i64 keys as u64 after XOR with sign bit — do the
+        // XOR with a gather variant below.
+        HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)keys_tmp, n));
+        HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, (uint64_t*)vals_tmp, perm_a,
+                                   perm_out, n, &plan.d_sort_temp,
+                                   &plan.sort_temp_cap));
+        perm_in = perm_out;
+        perm_out = perm_c;
+    }
+    // series pass (stable keeps bucket order within equal series)
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_series, perm_in,
+                                  (unsigned long long*)keys_tmp, n));
+    HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, (uint64_t*)vals_tmp, perm_in,
+                               perm_out, n, &plan.d_sort_temp,
+                               &plan.sort_temp_cap));
+    const uint32_t* perm = perm_out;
+
+    // gather each array by final perm and copy D2H
+    out.series.resize(n);
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_series, perm,
+                                  (unsigned long long*)vals_tmp, n));
+    HIP_TRY(hipMemcpyAsync(out.series.data(), vals_tmp, size_t(n) * 8,
+                           hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    auto fetch = [&](void* host_dst, const void* dev_src) -> hx_status {
+        HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)dev_src, perm,
+                                      (unsigned long long*)vals_tmp, n));
+        HIP_TRY(hipMemcpyAsync(host_dst, vals_tmp, size_t(n) * 8,
+                               hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        return HX_OK;
+    };
+    if (bucket) {
+        out.bucket.resize(n);
+        st = fetch(out.bucket.data(), c_bucket);
+        if (st != HX_OK) return st;
+    }
+    if (c_sum) {
+        out.sum.resize(n);
+        st = fetch(out.sum.data(), c_sum);
+        if (st != HX_OK) return st;
+    }
+    if (c_cnt) {
+        out.cnt.resize(n);
+        st = fetch(out.cnt.data(), c_cnt);
+        if (st != HX_OK) return st;
+    }
+    if (c_min) {
+        out.vmin.resize(n);
+        st = fetch(out.vmin.data(), c_min);
+        if (st != HX_OK) return st;
+    }
+    if (c_max) {
+        out.vmax.resize(n);
+        st = fetch(out.vmax.data(), c_max);
+        if (st != HX_OK) return st;
+    }
+    return HX_OK;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// result assembly + public entry points
+// ---------------------------------------------------------------------------
+namespace {
+
+struct ResultStorage {  // backs hx_result_table arrays
+    std::vector<uint64_t> series;
+    std::vector<int64_t> bucket;
+    std::vector<double> sum;
+    std::vector<unsigned long long> cnt;
+    std::vector<double> vmin, vmax, avg;
+};
+
+// n-way merge of per-device sorted partials, combining equal (series,bucket)
+// groups: the host partial-aggregate merge of SURVEY §8(e) (partials are
+// O(groups), not O(rows)).
+void merge_partials(std::vector<PartialResult>& parts, uint32_t ops,
+                    bool bucket, ResultStorage& out) {
+    const uint32_t kops = kernel_ops(ops);
+    std::vector<size_t> idx(parts.size(), 0);
+    for (;;) {
+        // find smallest key among cursors
+        bool any = false;
+        uint64_t ks = 0;
+        int64_t kb = 0;
+        for (size_t p = 0; p < parts.size(); p++) {
+            if (idx[p] >= parts[p].n) continue;
+            uint64_t s = parts[p].series[idx[p]];
+            int64_t b = bucket ? parts[p].bucket[idx[p]] : 0;
+            if (!any || s < ks || (s == ks && b < kb)) { ks = s; kb = b; any = true; }
+        }
+        if (!any) break;
+        double sum = 0, vmin = 0, vmax = 0;
+        unsigned long long cnt = 0;
+        bool first = true;
+        for (size_t p = 0; p < parts.size(); p++) {
+            size_t i = idx[p];
+            if (i >= parts[p].n || parts[p].series[i] != ks ||
+                (bucket && parts[p].bucket[i] != kb))
+                continue;
+            if (kops & HX_AGG_SUM) sum += parts[p].sum[i];
+            if (kops & HX_AGG_COUNT) cnt += parts[p].cnt[i];
+            if (kops & HX_AGG_MIN)
+                vmin = first ? parts[p].vmin[i] : std::min(vmin, parts[p].vmin[i]);
+            if (kops & HX_AGG_MAX)
+                vmax = first ? parts[p].vmax[i] : std::max(vmax, parts[p].vmax[i]);
+            first = false;
+            idx[p]++;
+        }
+        out.series.push_back(ks);
+        if (bucket) out.bucket.push_back(kb);
+        if (ops & HX_AGG_SUM) out.sum.push_back(sum);
+        if (ops & HX_AGG_COUNT) out.cnt.push_back(cnt);
+        if (ops & HX_AGG_MIN) out.vmin.push_back(vmin);
+        if (ops & HX_AGG_MAX) out.vmax.push_back(vmax);
+        if (ops & HX_AGG_AVG) out.avg.push_back(sum / double(cnt));
+    }
+}
+
+// single-device fast path: move arrays, compute avg
+void finish_single(PartialResult& p, uint32_t ops, bool bucket,
+                   ResultStorage& out) {
+    out.series = std::move(p.series);
+    if (bucket) out.bucket = std::move(p.bucket);
+    if (ops & HX_AGG_AVG) {
+        out.avg.resize(p.n);
+        for (size_t i = 0; i < p.n; i++)
+            out.avg[i] = p.sum[i] / double(p.cnt[i]);
+    }
+    if (ops & HX_AGG_SUM) out.sum = std::move(p.sum);
+    if (ops & HX_AGG_COUNT) out.cnt = std::move(p.cnt);
+    if (ops & HX_AGG_MIN) out.vmin = std::move(p.vmin);
+    if (ops & HX_AGG_MAX) out.vmax = std::move(p.vmax);
+}
+
+}  // namespace
+
+struct hx_result_impl {
+    hx_result_table pub_{};
+    ResultStorage store;
+};
+
+extern "C" hx_status hx_exec_agg(hx_prepared* P, const hx_agg_spec* agg,
+                                 hx_result_table** out) {
+    if (!P || !agg || !out) return fail(HX_ERR_INVALID, "null argument");
+    if (agg->ops == 0) return fail(HX_ERR_INVALID, "no aggregate ops requested");
+    if (agg->bucket_ms < 0) return fail(HX_ERR_INVALID, "negative bucket_ms");
+    auto t0 = std::chrono::steady_clock::now();
+    const bool bucket = agg->bucket_ms > 0;
+
+    std::vector<PartialResult> parts(P->plans.size());
+    double agg_ms_max = 0, decode_ms_max = 0;
+    unsigned long long matched = 0;
+    for (size_t d = 0; d < P->plans.size(); d++) {
+        double agg_ms = 0;
+        unsigned long long m = 0;
+        hx_status st = exec_plan(P, P->plans[d], agg, parts[d], &agg_ms, &m);
+        if (st != HX_OK) return st;
+        agg_ms_max = std::max(agg_ms_max, agg_ms);
+        decode_ms_max = std::max(decode_ms_max, P->plans[d].decode_ms);
+        matched += m;
+    }
+
+    auto R = std::make_unique<hx_result_impl>();
+    if (parts.size() == 1)
+        finish_single(parts[0], agg->ops, bucket, R->store);
+    else
+        merge_partials(parts, agg->ops, bucket, R->store);
+
+    hx_result_table& T = R->pub_;
+    T.n_groups = R->store.series.size();
+    T.series_id = R->store.series.data();
+    T.bucket = bucket ? R->store.bucket.data() : nullptr;
+    T.sum = (agg->ops & HX_AGG_SUM) ? R->store.sum.data() : nullptr;
+    T.count = (agg->ops & HX_AGG_COUNT) ? (const uint64_t*)R->store.cnt.data() : nullptr;
+    T.vmin = (agg->ops & HX_AGG_MIN) ? R->store.vmin.data() : nullptr;
+    T.vmax = (agg->ops & HX_AGG_MAX) ? R->store.vmax.data() : nullptr;
+    T.avg = (agg->ops & HX_AGG_AVG) ? R->store.avg.data() : nullptr;
+
+    P->last_stats.exec_ms = std::chrono::duration<double, std::milli>(
+                                std::chrono::steady_clock::now() - t0).count();
+    P->last_stats.agg_kernel_ms = agg_ms_max;
+    P->last_stats.decode_kernel_ms = decode_ms_max;
+    P->last_stats.rows_scanned = P->rows_scanned_total;
+    P->last_stats.rows_matched = (int64_t)matched;
+    P->last_stats.bytes_staged = P->bytes_staged;
+    P->last_stats.stage_ms = P->stage_ms;
+
+    *out = &R.release()->pub_;
+    return HX_OK;
+}
+
+extern "C" void hx_result_free(hx_result_table* t) {
+    if (!t) return;
+    // pub_ is the first member of hx_result_impl
+    delete reinterpret_cast<hx_result_impl*>(t);
+}
+
+extern "C" hx_status hx_get_stats(hx_prepared* P, hx_exec_stats* out) {
+    if (!P || !out) return fail(HX_ERR_INVALID, "null argument");
+    *out = P->last_stats;
+    return HX_OK;
+}
+
+extern "C" hx_status hx_scan_agg(hx_handle* h, const hx_scan_spec* spec,
+                                 const hx_agg_spec* agg,
+                                 const hx_device_set* devs,
+                                 hx_result_table** out) {
+    hx_prepared* prep = nullptr;
+    hx_status st = hx_prepare(h, spec, devs, &prep);
+    if (st != HX_OK) return st;
+    st = hx_exec_agg(prep, agg, out);
+    hx_prepared_free(prep);
+    return st;
+}
+
+extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
+                             const hx_device_set* devs, hx_batch_cb cb,
+                             void* ctx) {
+    (void)h; (void)spec; (void)devs; (void)cb; (void)ctx;
+    return fail(HX_ERR_UNSUPPORTED,
+                "hx_scan streaming parity mode: implemented later this round");
+}
+
+// introspection used by CPU tests (no GPU needed): per-SST catalog entries
+extern "C" hx_status hx_catalog_size(hx_handle* h, size_t* n) {
+    if (!h || !n) return fail(HX_ERR_INVALID, "null");
+    *n = h->ssts.size();
+    return HX_OK;
+}
+extern "C" hx_status hx_catalog_entry(hx_handle* h, size_t i, uint64_t* seq,
+                                      int64_t* n_rows, int64_t* ts_min,
+                                      int64_t* ts_max, int64_t* n_rgs) {
+    if (!h || i >= h->ssts.size()) return fail(HX_ERR_INVALID, "bad index");
+    const CatSst& s = h->ssts[i];
+    *seq = s.seq;
+    *n_rows = s.n_rows;
+    *ts_min = s.ts_min;
+    *ts_max = s.ts_max;
+    *n_rgs = (int64_t)s.rgs.size();
+    return HX_OK;
+}

@@ -1,0 +1,109 @@
+// hx_device.h — structures shared between the host engine and the gfx950
+// kernels. All offsets are byte offsets; bit 63 of an
+// offset selects the decode blob (dec) over the staged page blob (blob).
+#pragma once
+#include <cstdint>
+
+namespace hx {
+
+// aggregate op bits (must mirror HX_AGG_* in include/horaedb_hx.h)
+enum : uint32_t { HXK_SUM = 1u, HXK_COUNT = 2u, HXK_MIN = 4u, HXK_MAX = 8u,
+                  HXK_AVG = 16u };
+
+constexpr uint64_t OFF_DEC = 1ull << 63;   // offset lives in decode blob
+constexpr uint64_t OFF_MASK = OFF_DEC - 1;
+
+// One (SST, row group) unit: the fused kernel's workgroup granule.
+// Row groups of one SST are staged in ascending row order; next_rg gives the
+// SAME SST's next staged row group (for the adjacent-dedup boundary row),
+// -1 if none. Grid order of rgs is a speed-only choice (host interleaves by
+// row ordinal across SSTs for L3 locality of the group table).
+struct RgDesc {
+    uint64_t series_off;
+    uint64_t ts_off;
+    uint64_t val_off;
+    uint32_t n_rows;
+    uint32_t sst_id;
+    int64_t  row_base;     // staged-row index base within the SST (dense)
+    int32_t  next_rg;      // index into rg array, -1 = last staged rg of sst
+    int32_t  _pad;
+};
+
+// Per-SST device info. overlap==1 => this SST belongs to a ts-overlap
+// cluster: rows must check higher-seq SSTs of the cluster for
+// equal PKs via binary search over their dense (series, ts) arrays.
+struct SstDev {
+    int32_t  cluster;        // overlap-cluster id, -1 = none
+    int32_t  rank;           // seq rank within cluster (ascending), dense
+    uint64_t dense_series;   // byte offset (dec blob) of staged series array
+    uint64_t dense_ts;       // byte offset (dec blob) of staged ts array
+    int64_t  n_staged;       // staged rows (dense array length)
+};
+
+// Overlap cluster: member SSTs sorted by ascending seq; members[] indexes
+// the SstDev array. Flattened: cluster c owns members[first..first+n).
+struct ClusterDev {
+    int32_t first;
+    int32_t n;
+};
+
+// Aggregate hash table: open addressing, linear probe.
+// state: 0 empty, 1 claim in flight, 2 ready (key words valid).
+struct AggTable {
+    uint64_t* series;
+    int64_t*  bucket;        // null if bucket_ms == 0
+    uint32_t* state;
+    double*   sum;
+    unsigned long long* cnt;
+    unsigned long long* vmin;  // ordered-u64 mapped f64 (bit-exact min/max)
+    unsigned long long* vmax;
+    uint32_t  mask;            // slots-1 (power of two)
+};
+
+struct AggParams {
+    const RgDesc* rgs;
+    uint32_t n_rgs;
+    const SstDev* ssts;
+    const ClusterDev* clusters;
+    const int32_t* cluster_members;
+    const uint8_t* blob;
+    const uint8_t* dec;
+    int64_t ts_lo, ts_hi;          // [lo, hi)
+    // optional series-membership set (open hash, EMPTY sentinel chosen by host)
+    const uint64_t* sset;
+    uint32_t sset_mask;
+    uint64_t sset_empty;
+    int32_t use_sset;
+    int64_t bucket_ms;             // 0 = group by series only
+    uint32_t ops;                  // HX_AGG_* mask
+    AggTable table;
+    unsigned long long* fill;      // claimed slots
+    unsigned long long* overflow;  // !=0 => rerun with a larger table
+    unsigned long long* matched;   // rows surviving filter+dedup
+};
+
+// DELTA_BINARY_PACKED decode unit: one page -> dense i64 at dst_off (dec).
+struct DeltaPageDesc {
+    uint64_t src_off;     // page payload (bit63: in dec blob, e.g. post-snappy)
+    uint64_t dst_off;     // dec blob byte offset for n_values i64
+    uint32_t n_values;
+    uint32_t src_len;
+};
+
+// Snappy decompress unit: one page.
+struct SnappyPageDesc {
+    uint64_t src_off;     // blob
+    uint64_t dst_off;     // dec blob
+    uint32_t comp_len;
+    uint32_t uncomp_len;
+};
+
+// Plain copy unit (materialize dense arrays for the overlap path).
+struct CopyDesc {
+    uint64_t src_off;
+    uint64_t dst_off;
+    uint32_t n_values;    // 8-byte values
+    uint32_t _pad;
+};
+
+}  // namespace hx

@@ -1,0 +1,42 @@
+// hx_kernels.h — host-side launcher API implemented in kernels.hip (keeps
+// device code in one TU; no -fgpu-rdc needed).
+#pragma once
+#include <hip/hip_runtime.h>
+#include "hx_device.h"
+
+namespace hx {
+
+struct CompactOut {
+    uint64_t* series;
+    long long* bucket;
+    double* sum;
+    unsigned long long* cnt;
+    double* vmin;
+    double* vmax;
+    unsigned long long* n_out;
+};
+
+hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                               const DeltaPageDesc* pages, uint32_t n_pages,
+                               unsigned long long* err_flag);
+hipError_t launch_copy_u64(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                           const CopyDesc* descs, uint32_t n_descs);
+hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid);
+hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
+                          uint32_t ops, int64_t bucket_ms, const CompactOut& o);
+hipError_t launch_gather_u64(hipStream_t s, const unsigned long long* in,
+                             const uint32_t* perm, unsigned long long* out,
+                             uint32_t n);
+hipError_t launch_avg(hipStream_t s, const double* sum,
+                      const unsigned long long* cnt, double* avg, uint32_t n);
+hipError_t launch_iota(hipStream_t s, uint32_t* out, uint32_t n);
+hipError_t launch_xor_sign(hipStream_t s, unsigned long long* buf, uint32_t n);
+
+// rocPRIM stable LSD radix sort: sorts values (u32 perm) by u64 keys.
+// temp buffer managed internally on the stream (hipMallocAsync-free impl).
+hipError_t sort_pairs_u64(hipStream_t s, const uint64_t* keys_in,
+                          uint64_t* keys_out, const uint32_t* vals_in,
+                          uint32_t* vals_out, size_t n, void** d_temp,
+                          size_t* temp_bytes);
+
+}  // namespace hx

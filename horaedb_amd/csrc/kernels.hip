@@ -1,0 +1,537 @@
+// kernels.hip — gfx950 (MI355X, CDNA4) kernels for the scan/aggregate hot
+// path. HBM-bandwidth-bound: wave = 64, coalesced 8B loads,
+// one workgroup per 8192-row row group; group table updated with device
+// atomics (distinct addresses; L2/L3-resident by grid ordering).
+// Replaces the decode/filter/merge arithmetic the reference runs in
+// parquet-rs/arrow-rs/datafusion (SURVEY §2 third-party row) plus the
+// north-star aggregate (no reference counterpart).
+#include <hip/hip_runtime.h>
+#include "hx_device.h"
+
+namespace hx {
+
+#define RLX __ATOMIC_RELAXED
+#define AGT __HIP_MEMORY_SCOPE_AGENT
+
+__device__ __forceinline__ const uint8_t* hx_ptr(const uint8_t* blob,
+                                                 const uint8_t* dec,
+                                                 uint64_t off) {
+    return ((off & OFF_DEC) ? dec : blob) + (off & OFF_MASK);
+}
+
+__device__ __forceinline__ uint64_t mix64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+// monotone f64 -> u64 map: preserves total order, so u64 atomicMin/Max give
+// bit-exact f64 min/max.
+__device__ __forceinline__ unsigned long long f64_ordered(double v) {
+    unsigned long long u = __double_as_longlong(v);
+    return (u >> 63) ? ~u : (u | 0x8000000000000000ull);
+}
+__device__ __forceinline__ double ordered_f64(unsigned long long u) {
+    return __longlong_as_double((u >> 63) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u);
+}
+
+__device__ __forceinline__ int64_t floordiv(int64_t a, int64_t b) {
+    int64_t q = a / b;
+    if ((a % b) != 0 && ((a < 0) != (b < 0))) q--;
+    return q;
+}
+
+__device__ __forceinline__ bool sset_has(const AggParams& P, uint64_t s) {
+    if (s == P.sset_empty) return false;  // host guarantees sentinel ∉ set
+    uint32_t i = (uint32_t)mix64(s) & P.sset_mask;
+    for (uint32_t probes = 0; probes <= P.sset_mask; ++probes) {
+        uint64_t k = P.sset[i];
+        if (k == s) return true;
+        if (k == P.sset_empty) return false;
+        i = (i + 1) & P.sset_mask;
+    }
+    return false;
+}
+
+// Cross-SST dedup: is row (s,t) of SST `me` shadowed by an
+// equal PK in a higher-seq SST of the same ts-overlap cluster? SSTs are
+// PK-sorted (storage.rs:244-256), so a binary search per higher-seq member.
+__device__ bool shadowed(const AggParams& P, const SstDev& me,
+                         uint64_t s, int64_t t) {
+    ClusterDev c = P.clusters[me.cluster];
+    for (int32_t j = c.first; j < c.first + c.n; ++j) {
+        const SstDev o = P.ssts[P.cluster_members[j]];
+        if (o.rank <= me.rank) continue;
+        const uint64_t* S = (const uint64_t*)(P.dec + (o.dense_series & OFF_MASK));
+        const int64_t* T = (const int64_t*)(P.dec + (o.dense_ts & OFF_MASK));
+        int64_t lo = 0, hi = o.n_staged;
+        while (lo < hi) {
+            int64_t mid = (lo + hi) >> 1;
+            uint64_t sm = S[mid];
+            if (sm < s || (sm == s && T[mid] < t)) lo = mid + 1;
+            else hi = mid;
+        }
+        if (lo < o.n_staged && S[lo] == s && T[lo] == t) return true;
+    }
+    return false;
+}
+
+// Open-addressing claim + update. state: 0 empty / 1 claiming / 2 ready.
+// Claim via CAS on the state word (works for any 8/16-byte key — no key
+// sentinel needed); claimer stores key words then releases state=2; readers
+// are gated by the control dependency on state==2 (all table words accessed
+// with agent-scope atomics => L2-coherent, no L1 staleness).
+__device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
+                                           int64_t b, double v) {
+    const bool use_b = P.bucket_ms != 0;
+    uint64_t h = mix64(s ^ ((uint64_t)b * 0xD1B54A32D192ED03ull));
+    uint32_t i = (uint32_t)h & P.table.mask;
+    for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
+        uint32_t st = __hip_atomic_load(&P.table.state[i], RLX, AGT);
+        if (st == 0) {
+            uint32_t expected = 0;
+            if (__hip_atomic_compare_exchange_strong(&P.table.state[i],
+                    &expected, 1u, RLX, RLX, AGT)) {
+                __hip_atomic_store(&P.table.series[i], s, RLX, AGT);
+                if (use_b)
+                    __hip_atomic_store((unsigned long long*)&P.table.bucket[i],
+                                       (unsigned long long)b, RLX, AGT);
+                __hip_atomic_fetch_add(P.fill, 1ull, RLX, AGT);
+                __hip_atomic_store(&P.table.state[i], 2u, __ATOMIC_RELEASE, AGT);
+                st = 2;
+            } else {
+                st = expected;
+            }
+        }
+        if (st == 1) {  // claimer is storing the key; bounded spin
+            uint32_t spins = 0;
+            do {
+                st = __hip_atomic_load(&P.table.state[i], RLX, AGT);
+                if (++spins > (1u << 22)) {
+                    __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+                    return;
+                }
+            } while (st != 2);
+        }
+        if (__hip_atomic_load(&P.table.series[i], RLX, AGT) == s &&
+            (!use_b || (int64_t)__hip_atomic_load(
+                           (unsigned long long*)&P.table.bucket[i], RLX, AGT) == b)) {
+            if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd(&P.table.sum[i], v);
+            if (P.ops & (HXK_COUNT | HXK_AVG)) atomicAdd(&P.table.cnt[i], 1ull);
+            if (P.ops & HXK_MIN) atomicMin(&P.table.vmin[i], f64_ordered(v));
+            if (P.ops & HXK_MAX) atomicMax(&P.table.vmax[i], f64_ordered(v));
+            return;
+        }
+        i = (i + 1) & P.table.mask;
+    }
+    __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+}
+
+// ---------------------------------------------------------------------------
+// The headline fused kernel: decode(PLAIN in place) + ts-range/series-set
+// filter + MergeExec dedup + hash group-by aggregate. One workgroup per row
+// group, grid-stride; each thread strides rows (coalesced 8B column loads).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+k_scan_agg(AggParams P) {
+    unsigned long long my_matched = 0;
+    for (uint32_t rgi = blockIdx.x; rgi < P.n_rgs; rgi += gridDim.x) {
+        const RgDesc rg = P.rgs[rgi];
+        const uint64_t* S = (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+        const int64_t* T = (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+        const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+        const SstDev sst = P.ssts[rg.sst_id];
+        const uint32_t n = rg.n_rows;
+        for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
+            int64_t t = T[r];
+            if (t < P.ts_lo || t >= P.ts_hi) continue;
+            uint64_t s = S[r];
+            if (P.use_sset && !sset_has(P, s)) continue;
+            // within-SST dedup: the LAST row of an equal-PK run survives
+            // (LastValueOperator, operator.rs:37-44; plan order read.rs:456-480)
+            bool dup = false;
+            if (r + 1 < n) {
+                dup = (S[r + 1] == s) & (T[r + 1] == t);
+            } else if (rg.next_rg >= 0) {
+                const RgDesc nx = P.rgs[rg.next_rg];
+                uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec, nx.series_off);
+                int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
+                dup = (s2 == s) & (t2 == t);
+            }
+            if (dup) continue;
+            if (sst.cluster >= 0 && shadowed(P, sst, s, t)) continue;
+            double v = V[r];
+            int64_t b = P.bucket_ms ? floordiv(t, P.bucket_ms) : 0;
+            agg_update(P, s, b, v);
+            my_matched++;
+        }
+    }
+    // one atomic per wave for the matched counter (guideline 12)
+    for (int off = 32; off > 0; off >>= 1)
+        my_matched += __shfl_down(my_matched, off, 64);
+    if ((threadIdx.x & 63) == 0 && my_matched)
+        atomicAdd(P.matched, my_matched);
+}
+
+// ---------------------------------------------------------------------------
+// Result compaction: live slots -> dense arrays (unsorted; host sorts with
+// rocPRIM then gathers).
+// ---------------------------------------------------------------------------
+struct CompactParams {
+    AggTable table;
+    uint32_t n_slots;
+    uint32_t ops;
+    int64_t bucket_ms;
+    uint64_t* out_series;
+    long long* out_bucket;
+    double* out_sum;
+    unsigned long long* out_cnt;
+    double* out_min;
+    double* out_max;
+    unsigned long long* n_out;
+};
+
+extern "C" __global__ void __launch_bounds__(256)
+k_compact(CompactParams C) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < C.n_slots;
+         i += blockDim.x * gridDim.x) {
+        if (C.table.state[i] != 2u) continue;
+        unsigned long long j = atomicAdd(C.n_out, 1ull);
+        C.out_series[j] = C.table.series[i];
+        if (C.bucket_ms) C.out_bucket[j] = C.table.bucket[i];
+        if (C.out_sum) C.out_sum[j] = C.table.sum[i];
+        if (C.out_cnt) C.out_cnt[j] = C.table.cnt[i];
+        if (C.out_min) C.out_min[j] = ordered_f64(C.table.vmin[i]);
+        if (C.out_max) C.out_max[j] = ordered_f64(C.table.vmax[i]);
+    }
+}
+
+// Gather 8-byte elements by permutation (applies the sort order).
+extern "C" __global__ void __launch_bounds__(256)
+k_gather_u64(const unsigned long long* in, const uint32_t* perm,
+             unsigned long long* out, uint32_t n) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += blockDim.x * gridDim.x)
+        out[i] = in[perm[i]];
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_avg(const double* sum, const unsigned long long* cnt, double* avg, uint32_t n) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += blockDim.x * gridDim.x)
+        avg[i] = sum[i] / (double)cnt[i];
+}
+
+// ---------------------------------------------------------------------------
+// Dense materialization (overlap path / streaming mode): copy staged PLAIN
+// page payloads into dense 8B arrays. One unit per page, grid-stride.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+k_copy_u64(const uint8_t* blob, const uint8_t* dec_in, uint8_t* dec,
+           const CopyDesc* descs, uint32_t n_descs) {
+    for (uint32_t d = blockIdx.x; d < n_descs; d += gridDim.x) {
+        CopyDesc c = descs[d];
+        const uint64_t* src = (const uint64_t*)hx_ptr(blob, dec_in, c.src_off);
+        uint64_t* dst = (uint64_t*)(dec + (c.dst_off & OFF_MASK));
+        for (uint32_t i = threadIdx.x; i < c.n_values; i += blockDim.x)
+            dst[i] = src[i];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// DELTA_BINARY_PACKED i64 decode (parquet-format Encodings.md; the encoding
+// the reference's config enables for ts, config.rs:54-75). One workgroup per
+// page: lane-serial header walk, parallel miniblock bit-unpack, hierarchical
+// prefix sum in LDS, dense i64 output. Page <= 8192 values (the writer
+// contract: row group 8192, one data page per chunk — engine validates).
+// ---------------------------------------------------------------------------
+#define DELTA_MAX_VALUES 8192
+#define DELTA_MAX_BLOCKS 128   // >= max_values / min_block_size(128)
+
+struct DeltaHdr {
+    int32_t values_per_mb;
+    int32_t n_mb_per_block;
+    int32_t total_count;
+    int64_t first_value;
+    int32_t n_blocks;
+    int32_t err;
+};
+
+__device__ inline uint64_t d_varint(const uint8_t* p, uint32_t len, uint32_t& pos,
+                                    int32_t* err) {
+    uint64_t r = 0;
+    int s = 0;
+    for (;;) {
+        if (pos >= len || s > 63) { *err = 1; return 0; }
+        uint8_t b = p[pos++];
+        r |= (uint64_t)(b & 0x7f) << s;
+        if (!(b & 0x80)) return r;
+        s += 7;
+    }
+}
+__device__ inline int64_t d_zigzag(const uint8_t* p, uint32_t len, uint32_t& pos,
+                                   int32_t* err) {
+    uint64_t u = d_varint(p, len, pos, err);
+    return (int64_t)(u >> 1) ^ -(int64_t)(u & 1);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_decode_delta_i64(const uint8_t* blob, uint8_t* dec,
+                   const DeltaPageDesc* pages, uint32_t n_pages,
+                   unsigned long long* err_flag) {
+    __shared__ DeltaHdr H;
+    __shared__ int64_t vals[DELTA_MAX_VALUES];          // deltas -> prefix
+    __shared__ int64_t blk_min_delta[DELTA_MAX_BLOCKS];
+    __shared__ uint32_t mb_off[DELTA_MAX_BLOCKS * 8];   // per-miniblock data offset
+    __shared__ uint8_t mb_w[DELTA_MAX_BLOCKS * 8];
+    __shared__ int64_t tsum[256 / 64];                  // per-wave partials
+    __shared__ int64_t chunk_sum[256];
+
+    for (uint32_t pg = blockIdx.x; pg < n_pages; pg += gridDim.x) {
+        const DeltaPageDesc pd = pages[pg];
+        const uint8_t* src = ((pd.src_off & OFF_DEC) ? dec : blob) +
+                             (pd.src_off & OFF_MASK);
+        const uint32_t len = pd.src_len;
+
+        if (threadIdx.x == 0) {
+            int32_t err = 0;
+            uint32_t pos = 0;
+            int32_t block_size = (int32_t)d_varint(src, len, pos, &err);
+            int32_t n_mb = (int32_t)d_varint(src, len, pos, &err);
+            int32_t total = (int32_t)d_varint(src, len, pos, &err);
+            int64_t first = d_zigzag(src, len, pos, &err);
+            H.err = err;
+            if (!err && (n_mb <= 0 || block_size <= 0 || n_mb > 8 ||
+                         block_size % n_mb != 0 || total > DELTA_MAX_VALUES ||
+                         (uint32_t)total != pd.n_values ||
+                         (block_size / n_mb) % 32 != 0)) {
+                H.err = 2;
+            }
+            if (!H.err) {
+                H.values_per_mb = block_size / n_mb;
+                H.n_mb_per_block = n_mb;
+                H.total_count = total;
+                H.first_value = first;
+                int32_t deltas = total - 1;
+                int32_t nblk = deltas <= 0 ? 0 : (deltas + block_size - 1) / block_size;
+                H.n_blocks = nblk;
+                // serial walk of block headers (varints force it; 64 blocks max)
+                int32_t remaining = deltas;
+                for (int32_t b = 0; b < nblk && !H.err; b++) {
+                    blk_min_delta[b] = d_zigzag(src, len, pos, &err);
+                    if (err) { H.err = 3; break; }
+                    uint32_t wpos = pos;
+                    pos += n_mb;  // bit width bytes
+                    if (pos > len) { H.err = 3; break; }
+                    for (int32_t m = 0; m < n_mb; m++) {
+                        uint8_t w = src[wpos + m];
+                        mb_w[b * 8 + m] = w;
+                        mb_off[b * 8 + m] = pos;
+                        // data present for miniblocks that hold any values
+                        if (remaining > 0) pos += (uint32_t)H.values_per_mb * w / 8;
+                        remaining -= H.values_per_mb;
+                    }
+                    if (pos > len) { H.err = 3; }
+                }
+            }
+        }
+        __syncthreads();
+        if (H.err) {
+            if (threadIdx.x == 0) atomicAdd(err_flag, 1ull);
+            __syncthreads();
+            continue;
+        }
+        const int32_t total = H.total_count;
+        const int32_t vpm = H.values_per_mb;
+        const int32_t deltas = total - 1;
+
+        // parallel unpack: delta j (0-based, value index j+1)
+        for (int32_t j = threadIdx.x; j < deltas; j += blockDim.x) {
+            int32_t blk = j / (vpm * H.n_mb_per_block);
+            int32_t inblk = j - blk * vpm * H.n_mb_per_block;
+            int32_t mb = inblk / vpm;
+            int32_t inmb = inblk - mb * vpm;
+            uint8_t w = mb_w[blk * 8 + mb];
+            uint64_t raw = 0;
+            if (w > 0) {
+                uint64_t bitpos = (uint64_t)inmb * w;
+                const uint8_t* base = src + mb_off[blk * 8 + mb] + (bitpos >> 3);
+                int shift = (int)(bitpos & 7);
+                // assemble enough bytes for w bits + shift (<= 9 bytes)
+                uint64_t lo = 0;
+                for (int k = 0; k < 8; k++) lo |= (uint64_t)base[k] << (8 * k);
+                raw = lo >> shift;
+                if (w + shift > 64) {
+                    uint64_t hi = base[8];
+                    raw |= hi << (64 - shift);
+                }
+                if (w < 64) raw &= ((1ull << w) - 1);
+            }
+            vals[j + 1] = (int64_t)raw + blk_min_delta[blk];
+        }
+        if (threadIdx.x == 0) vals[0] = 0;
+        __syncthreads();
+
+        // hierarchical inclusive prefix sum over vals[0..total)
+        // each thread scans a contiguous chunk of 32, then chunk offsets
+        const int32_t CH = 32;
+        int32_t c0 = threadIdx.x * CH;
+        int64_t acc = 0;
+        for (int32_t j = c0; j < min(c0 + CH, total); j++) {
+            acc += vals[j];
+            vals[j] = acc;
+        }
+        chunk_sum[threadIdx.x] = acc;
+        __syncthreads();
+        if (threadIdx.x < 64) {  // single wave scans the 256 chunk sums
+            int64_t s0 = 0;
+            for (int k = 0; k < 4; k++) {
+                int idx = threadIdx.x * 4 + k;
+                s0 += chunk_sum[idx];
+            }
+            // wave inclusive scan of s0 over 64 lanes
+            int64_t sc = s0;
+            for (int off = 1; off < 64; off <<= 1) {
+                int64_t up = __shfl_up(sc, off, 64);
+                if ((int)(threadIdx.x) >= off) sc += up;
+            }
+            tsum[0] = 0;  // unused; keep LDS referenced
+            // exclusive base for each 4-chunk group of this lane
+            int64_t base4 = sc - s0;
+            int64_t run = base4;
+            for (int k = 0; k < 4; k++) {
+                int idx = threadIdx.x * 4 + k;
+                int64_t cs = chunk_sum[idx];
+                chunk_sum[idx] = run;  // exclusive chunk base
+                run += cs;
+            }
+        }
+        __syncthreads();
+        int64_t base = chunk_sum[threadIdx.x] + H.first_value;
+        for (int32_t j = c0; j < min(c0 + CH, total); j++) vals[j] += base;
+        // note: vals[0] = 0 + base = first_value ✓
+        __syncthreads();
+
+        int64_t* dst = (int64_t*)(dec + (pd.dst_off & OFF_MASK));
+        for (int32_t j = threadIdx.x; j < total; j += blockDim.x)
+            dst[j] = vals[j];
+        __syncthreads();
+    }
+}
+
+}  // namespace hx
+
+// ---------------------------------------------------------------------------
+// host-side launchers (hx_kernels.h)
+// ---------------------------------------------------------------------------
+#include "hx_kernels.h"
+#include <cstring>
+#include <rocprim/device/device_radix_sort.hpp>
+
+namespace hx {
+
+extern "C" __global__ void __launch_bounds__(256)
+k_iota(uint32_t* out, uint32_t n) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += blockDim.x * gridDim.x)
+        out[i] = i;
+}
+
+static uint32_t grid_for(uint32_t work, uint32_t per_block) {
+    uint32_t g = (work + per_block - 1) / per_block;
+    return g == 0 ? 1 : (g > 2048u ? 2048u : g);
+}
+
+hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                               const DeltaPageDesc* pages, uint32_t n_pages,
+                               unsigned long long* err_flag) {
+    uint32_t grid = n_pages > 4096 ? 4096 : n_pages;
+    hipLaunchKernelGGL(k_decode_delta_i64, dim3(grid), dim3(256), 0, s,
+                       blob, dec, pages, n_pages, err_flag);
+    return hipGetLastError();
+}
+
+hipError_t launch_copy_u64(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                           const CopyDesc* descs, uint32_t n_descs) {
+    uint32_t grid = n_descs > 4096 ? 4096 : n_descs;
+    hipLaunchKernelGGL(k_copy_u64, dim3(grid), dim3(256), 0, s,
+                       blob, dec, dec, descs, n_descs);
+    return hipGetLastError();
+}
+
+hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid) {
+    if (grid == 0) grid = p.n_rgs > 65535 ? 65535 : (p.n_rgs ? p.n_rgs : 1);
+    hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(256), 0, s, p);
+    return hipGetLastError();
+}
+
+hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
+                          uint32_t ops, int64_t bucket_ms, const CompactOut& o) {
+    CompactParams C;
+    C.table = t;
+    C.n_slots = n_slots;
+    C.ops = ops;
+    C.bucket_ms = bucket_ms;
+    C.out_series = o.series;
+    C.out_bucket = o.bucket;
+    C.out_sum = o.sum;
+    C.out_cnt = o.cnt;
+    C.out_min = o.vmin;
+    C.out_max = o.vmax;
+    C.n_out = o.n_out;
+    hipLaunchKernelGGL(k_compact, dim3(grid_for(n_slots, 256)), dim3(256), 0, s, C);
+    return hipGetLastError();
+}
+
+hipError_t launch_gather_u64(hipStream_t s, const unsigned long long* in,
+                             const uint32_t* perm, unsigned long long* out,
+                             uint32_t n) {
+    hipLaunchKernelGGL(k_gather_u64, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       in, perm, out, n);
+    return hipGetLastError();
+}
+
+hipError_t launch_avg(hipStream_t s, const double* sum,
+                      const unsigned long long* cnt, double* avg, uint32_t n) {
+    hipLaunchKernelGGL(k_avg, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       sum, cnt, avg, n);
+    return hipGetLastError();
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_xor_sign(unsigned long long* buf, uint32_t n) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += blockDim.x * gridDim.x)
+        buf[i] ^= 0x8000000000000000ull;
+}
+
+hipError_t launch_xor_sign(hipStream_t s, unsigned long long* buf, uint32_t n) {
+    hipLaunchKernelGGL(k_xor_sign, dim3(grid_for(n, 256)), dim3(256), 0, s, buf, n);
+    return hipGetLastError();
+}
+
+hipError_t launch_iota(hipStream_t s, uint32_t* out, uint32_t n) {
+    hipLaunchKernelGGL(k_iota, dim3(grid_for(n, 256)), dim3(256), 0, s, out, n);
+    return hipGetLastError();
+}
+
+hipError_t sort_pairs_u64(hipStream_t s, const uint64_t* keys_in,
+                          uint64_t* keys_out, const uint32_t* vals_in,
+                          uint32_t* vals_out, size_t n, void** d_temp,
+                          size_t* temp_bytes) {
+    size_t need = 0;
+    hipError_t e = rocprim::radix_sort_pairs(nullptr, need, keys_in, keys_out,
+                                             vals_in, vals_out, n, 0, 64, s);
+    if (e != hipSuccess) return e;
+    if (need > *temp_bytes) {
+        if (*d_temp) hipFree(*d_temp);
+        e = hipMalloc(d_temp, need);
+        if (e != hipSuccess) { *temp_bytes = 0; *d_temp = nullptr; return e; }
+        *temp_bytes = need;
+    }
+    return rocprim::radix_sort_pairs(*d_temp, *temp_bytes, keys_in, keys_out,
+                                     vals_in, vals_out, n, 0, 64, s);
+}
+
+}  // namespace hx

@@ -1,0 +1,248 @@
+# horaedb_amd/store.py — ctypes binding of the C-ABI (include/horaedb_hx.h).
+# Store mirrors the reference's ColumnarStorage trait surface
+# (storage.rs:76-89): schema contract fixed by the metric engine, scan via
+# scan_agg (hot path) and scan (streaming parity mode).
+import ctypes as C
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB = os.path.join(_HERE, "libhoraedb_hx.so")
+
+AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 1, 2, 4, 8, 16
+
+_ERRNAMES = {0: "OK", 1: "IO", 2: "FORMAT", 3: "UNSUPPORTED", 4: "NO_GPU",
+             5: "HIP", 6: "INVALID", 7: "SCHEMA"}
+
+
+class HxError(RuntimeError):
+    def __init__(self, code, msg):
+        super().__init__(f"hx error {_ERRNAMES.get(code, code)}: {msg}")
+        self.code = code
+
+
+class _TimeRange(C.Structure):
+    _fields_ = [("start", C.c_int64), ("end", C.c_int64)]
+
+
+class _SstDesc(C.Structure):
+    _fields_ = [("path", C.c_char_p), ("sequence", C.c_uint64)]
+
+
+class _Pred(C.Structure):
+    _fields_ = [("kind", C.c_int32), ("series_ids", C.POINTER(C.c_uint64)),
+                ("n_series", C.c_size_t)]
+
+
+class _ScanSpec(C.Structure):
+    _fields_ = [("range", _TimeRange), ("preds", C.POINTER(_Pred)),
+                ("n_preds", C.c_size_t), ("ssts", C.POINTER(_SstDesc)),
+                ("n_ssts", C.c_size_t), ("projection", C.POINTER(C.c_int32)),
+                ("n_projection", C.c_size_t)]
+
+
+class _AggSpec(C.Structure):
+    _fields_ = [("ops", C.c_uint32), ("bucket_ms", C.c_int64)]
+
+
+class _DeviceSet(C.Structure):
+    _fields_ = [("device_ids", C.POINTER(C.c_int32)), ("n_devices", C.c_int32)]
+
+
+class _ResultTable(C.Structure):
+    _fields_ = [("n_groups", C.c_size_t),
+                ("series_id", C.POINTER(C.c_uint64)),
+                ("bucket", C.POINTER(C.c_int64)),
+                ("sum", C.POINTER(C.c_double)),
+                ("count", C.POINTER(C.c_uint64)),
+                ("vmin", C.POINTER(C.c_double)),
+                ("vmax", C.POINTER(C.c_double)),
+                ("avg", C.POINTER(C.c_double))]
+
+
+class _ExecStats(C.Structure):
+    _fields_ = [("exec_ms", C.c_double), ("agg_kernel_ms", C.c_double),
+                ("decode_kernel_ms", C.c_double), ("rows_scanned", C.c_int64),
+                ("rows_matched", C.c_int64), ("bytes_staged", C.c_int64),
+                ("stage_ms", C.c_double)]
+
+
+def _load():
+    if not os.path.exists(_LIB):
+        raise ImportError(
+            f"{_LIB} missing — build it with `make -C horaedb_amd/csrc` "
+            "(the HIP product path; no CPU fallback exists)")
+    lib = C.CDLL(_LIB)
+    lib.hx_last_error.restype = C.c_char_p
+    lib.hx_open.argtypes = [C.c_char_p, C.c_int64, C.POINTER(C.c_void_p)]
+    lib.hx_close.argtypes = [C.c_void_p]
+    lib.hx_find_ssts.argtypes = [C.c_void_p, _TimeRange,
+                                 C.POINTER(C.POINTER(_SstDesc)),
+                                 C.POINTER(C.c_size_t)]
+    lib.hx_prepare.argtypes = [C.c_void_p, C.POINTER(_ScanSpec),
+                               C.POINTER(_DeviceSet), C.POINTER(C.c_void_p)]
+    lib.hx_prepared_free.argtypes = [C.c_void_p]
+    lib.hx_exec_agg.argtypes = [C.c_void_p, C.POINTER(_AggSpec),
+                                C.POINTER(C.POINTER(_ResultTable))]
+    lib.hx_result_free.argtypes = [C.POINTER(_ResultTable)]
+    lib.hx_scan_agg.argtypes = [C.c_void_p, C.POINTER(_ScanSpec),
+                                C.POINTER(_AggSpec), C.POINTER(_DeviceSet),
+                                C.POINTER(C.POINTER(_ResultTable))]
+    lib.hx_get_stats.argtypes = [C.c_void_p, C.POINTER(_ExecStats)]
+    lib.hx_catalog_size.argtypes = [C.c_void_p, C.POINTER(C.c_size_t)]
+    lib.hx_catalog_entry.argtypes = [C.c_void_p, C.c_size_t,
+                                     C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
+                                     C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+                                     C.POINTER(C.c_int64)]
+    return lib
+
+
+_lib = _load()
+
+
+def lib_path():
+    return _LIB
+
+
+def _check(code):
+    if code != 0:
+        raise HxError(code, _lib.hx_last_error().decode())
+
+
+def _np(ptr, n, dtype):
+    if not ptr or n == 0:
+        return np.empty(0, dtype)
+    return np.ctypeslib.as_array(ptr, shape=(n,)).astype(dtype, copy=True)
+
+
+class Prepared:
+    """A staged scan (column chunks resident in HBM). exec_agg is the timed
+    hot path."""
+
+    def __init__(self, store, handle):
+        self._store = store
+        self._h = handle
+
+    def exec_agg(self, ops=AGG_SUM | AGG_COUNT, bucket_ms=0):
+        spec = _AggSpec(ops=ops, bucket_ms=bucket_ms)
+        out = C.POINTER(_ResultTable)()
+        _check(_lib.hx_exec_agg(self._h, C.byref(spec), C.byref(out)))
+        t = out.contents
+        n = t.n_groups
+        res = {"series_id": _np(t.series_id, n, np.uint64)}
+        if t.bucket:
+            res["bucket"] = _np(t.bucket, n, np.int64)
+        if ops & AGG_SUM:
+            res["sum"] = _np(t.sum, n, np.float64)
+        if ops & AGG_COUNT:
+            res["count"] = _np(t.count, n, np.uint64)
+        if ops & AGG_MIN:
+            res["vmin"] = _np(t.vmin, n, np.float64)
+        if ops & AGG_MAX:
+            res["vmax"] = _np(t.vmax, n, np.float64)
+        if ops & AGG_AVG:
+            res["avg"] = _np(t.avg, n, np.float64)
+        _lib.hx_result_free(out)
+        return res
+
+    def stats(self):
+        st = _ExecStats()
+        _check(_lib.hx_get_stats(self._h, C.byref(st)))
+        return {f: getattr(st, f) for f, _ in st._fields_}
+
+    def close(self):
+        if self._h:
+            _lib.hx_prepared_free(self._h)
+            self._h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+
+class Store:
+    """MI355X-native ColumnarStorage (scan side). storage.rs:76-89."""
+
+    def __init__(self, store_path, segment_duration_ms=0):
+        h = C.c_void_p()
+        _check(_lib.hx_open(store_path.encode(), segment_duration_ms,
+                            C.byref(h)))
+        self._h = h
+
+    def close(self):
+        if self._h:
+            _lib.hx_close(self._h)
+            self._h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    def catalog(self):
+        n = C.c_size_t()
+        _check(_lib.hx_catalog_size(self._h, C.byref(n)))
+        out = []
+        for i in range(n.value):
+            seq = C.c_uint64()
+            rows = C.c_int64()
+            tmin = C.c_int64()
+            tmax = C.c_int64()
+            nrg = C.c_int64()
+            _check(_lib.hx_catalog_entry(self._h, i, C.byref(seq), C.byref(rows),
+                                         C.byref(tmin), C.byref(tmax),
+                                         C.byref(nrg)))
+            out.append({"seq": seq.value, "n_rows": rows.value,
+                        "ts_min": tmin.value, "ts_max": tmax.value,
+                        "n_row_groups": nrg.value})
+        return out
+
+    def find_ssts(self, ts_range):
+        """Manifest::find_ssts (manifest/mod.rs:165-172)."""
+        out = C.POINTER(_SstDesc)()
+        n = C.c_size_t()
+        _check(_lib.hx_find_ssts(self._h, _TimeRange(*ts_range), C.byref(out),
+                                 C.byref(n)))
+        return [(out[i].path.decode(), out[i].sequence) for i in range(n.value)]
+
+    def _spec(self, ts_range, series_in=None):
+        spec = _ScanSpec()
+        spec.range = _TimeRange(*ts_range)
+        self._keepalive = []
+        if series_in is not None:
+            arr = np.ascontiguousarray(np.asarray(sorted(set(int(s) for s in series_in)),
+                                                  dtype=np.uint64))
+            pred = _Pred(kind=1,
+                         series_ids=arr.ctypes.data_as(C.POINTER(C.c_uint64)),
+                         n_series=len(arr))
+            preds = (_Pred * 1)(pred)
+            spec.preds = preds
+            spec.n_preds = 1
+            self._keepalive += [arr, preds]
+        return spec
+
+    def _devset(self, devices):
+        if devices is None:
+            return None
+        ids = (C.c_int32 * len(devices))(*devices)
+        ds = _DeviceSet(device_ids=ids, n_devices=len(devices))
+        self._keepalive += [ids, ds]
+        return ds
+
+    def prepare(self, ts_range, series_in=None, devices=None):
+        spec = self._spec(ts_range, series_in)
+        ds = self._devset(devices)
+        out = C.c_void_p()
+        _check(_lib.hx_prepare(self._h, C.byref(spec),
+                               C.byref(ds) if ds else None, C.byref(out)))
+        return Prepared(self, out)
+
+    def scan_agg(self, ts_range, ops=AGG_SUM | AGG_COUNT, bucket_ms=0,
+                 series_in=None, devices=None):
+        """One-shot scan+aggregate (prepare + exec + release)."""
+        with self.prepare(ts_range, series_in, devices) as p:
+            return p.exec_agg(ops=ops, bucket_ms=bucket_ms)

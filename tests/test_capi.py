@@ -1,0 +1,106 @@
+# CPU-side tests of the C-ABI library: it loads, exports every symbol
+# include/horaedb_hx.h declares, the catalog (footer parse) matches pyarrow,
+# and GPU-requiring calls fail loudly without a GPU (no CPU fallback).
+import ctypes
+import os
+import re
+import subprocess
+
+import numpy as np
+import pytest
+
+import horaedb_amd
+from horaedb_amd import Store, HxError
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _has_gpu():
+    try:
+        out = subprocess.run(["rocm-smi", "--showid"], capture_output=True,
+                             timeout=10)
+        return out.returncode == 0 and b"GPU" in out.stdout
+    except Exception:
+        return False
+
+
+def test_library_exports_header_symbols():
+    header = open(os.path.join(REPO, "include", "horaedb_hx.h")).read()
+    decls = re.findall(r"^(?:hx_status|void|const char\*)\s+(hx_\w+)\(", header,
+                       re.M)
+    assert len(decls) >= 10, "header symbol scrape failed"
+    lib = ctypes.CDLL(horaedb_amd.lib_path())
+    for sym in decls:
+        assert hasattr(lib, sym), f"symbol {sym} missing from libhoraedb_hx.so"
+
+
+@pytest.fixture(scope="module")
+def store_dir(tmp_path_factory):
+    from tools.gen_ssts import gen_dataset
+    out = str(tmp_path_factory.mktemp("capi"))
+    m = gen_dataset(out, n_rows=50_000, n_series=250, n_ssts=4, seed=11)
+    return out, m
+
+
+def test_catalog_matches_pyarrow(store_dir):
+    import pyarrow.parquet as pq
+    out, m = store_dir
+    with Store(out) as st:
+        cat = st.catalog()
+    assert len(cat) == len(m["ssts"])
+    for entry, sst in zip(cat, sorted(m["ssts"], key=lambda s: s["seq"])):
+        pf = pq.ParquetFile(sst["path"])
+        assert entry["seq"] == sst["seq"]
+        assert entry["n_rows"] == pf.metadata.num_rows
+        assert entry["n_row_groups"] == pf.metadata.num_row_groups
+        # ts min/max from our thrift stats parse == pyarrow statistics
+        tmins, tmaxs = [], []
+        for rg in range(pf.metadata.num_row_groups):
+            c = pf.metadata.row_group(rg).column(1)
+            tmins.append(c.statistics.min)
+            tmaxs.append(c.statistics.max)
+        assert entry["ts_min"] == min(tmins)
+        assert entry["ts_max"] == max(tmaxs)
+
+
+def test_find_ssts_time_overlap(store_dir):
+    out, m = store_dir
+    with Store(out) as st:
+        # full range: all SSTs
+        assert len(st.find_ssts((0, 2**62))) == len(m["ssts"])
+        # empty range before data
+        assert st.find_ssts((0, m["ts_start"])) == []
+        # range covering exactly the first SST's window
+        s0 = m["ssts"][0]
+        hits = st.find_ssts((s0["ts_min"], s0["ts_max"] + 1))
+        assert any(seq == s0["seq"] for _, seq in hits)
+        # half-open end: range ending AT an sst's ts_min excludes it
+        # (TimeRange semantics types.rs:125-127)
+        s1 = m["ssts"][1]
+        hits = st.find_ssts((m["ts_start"] - 10, s1["ts_min"]))
+        assert all(seq != s1["seq"] for _, seq in hits)
+
+
+@pytest.mark.skipif(_has_gpu(), reason="GPU present: NO_GPU path not testable")
+def test_scan_fails_loudly_without_gpu(store_dir):
+    out, m = store_dir
+    with Store(out) as st:
+        with pytest.raises(HxError) as ei:
+            st.scan_agg((0, 2**62))
+        assert ei.value.code == 4  # HX_ERR_NO_GPU
+        assert "no CPU fallback" in str(ei.value)
+
+
+def test_open_missing_store():
+    with pytest.raises(HxError) as ei:
+        Store("/nonexistent/path/xyz")
+    assert ei.value.code == 1  # IO
+
+
+def test_open_rejects_non_parquet(tmp_path):
+    ddir = tmp_path / "data"
+    ddir.mkdir()
+    (ddir / "1.sst").write_bytes(b"this is not parquet at all........")
+    with pytest.raises(HxError) as ei:
+        Store(str(tmp_path))
+    assert ei.value.code == 2  # FORMAT
