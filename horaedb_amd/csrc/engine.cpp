@@ -674,6 +674,7 @@ extern "C" hx_status hx_prepare(hx_handle* h, const hx_scan_spec* spec,
     for (size_t d = 0; d < device_ids.size(); d++)
         P->plans[d].device = device_ids[d];
     std::vector<int64_t> load(device_ids.size(), 0);
+    std::vector<int32_t> member_count(device_ids.size(), 0);
     std::vector<std::vector<StagedSst*>> members_per_plan(device_ids.size());
     // big groups first for balance
     std::sort(groups.begin(), groups.end(), [&](const auto& a, const auto& b) {
@@ -694,7 +695,8 @@ extern "C" hx_status hx_prepare(hx_handle* h, const hx_scan_spec* spec,
         int32_t cluster_id = -1;
         if (g.size() > 1) {
             cluster_id = (int32_t)plan.clusters.size();
-            hx::ClusterDev cd{(int32_t)plan.cluster_members.size(), (int32_t)g.size()};
+            hx::ClusterDev cd{member_count[best], (int32_t)g.size()};
+            member_count[best] += (int32_t)g.size();
             plan.clusters.push_back(cd);
         }
         int32_t rank = 0;
@@ -964,7 +966,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     uint8_t* base = (uint8_t*)plan.d_scratch;
     auto carve8 = [&](uint32_t count) {
         uint8_t* p = base;
-        base += size_t(count) * 8;
+        base += (size_t(count) * 8 + 7) & ~size_t(7);  // keep 8B alignment
         return p;
     };
     uint64_t* c_series = (uint64_t*)carve8(n);
@@ -975,11 +977,12 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     double* c_max = plan.t_max ? (double*)carve8(n) : nullptr;
     uint64_t* keys_tmp = (uint64_t*)carve8(n);       // sort keys in/out
     uint64_t* vals_tmp = (uint64_t*)carve8(n);       // gather staging
-    // remaining val arrays share vals_tmp sequentially (gather one at a time)
+    unsigned long long* d_nout = (unsigned long long*)carve8(1);
+    // 4-byte perm arrays go last (they would break 8B alignment otherwise)
     uint32_t* perm_a = (uint32_t*)base; base += size_t(n) * 4;
     uint32_t* perm_b = (uint32_t*)base; base += size_t(n) * 4;
     uint32_t* perm_c = (uint32_t*)base; base += size_t(n) * 4;
-    unsigned long long* d_nout = (unsigned long long*)base;
+    (void)perm_c;
 
     HIP_TRY(hipMemsetAsync(d_nout, 0, 8, s));
     hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
