@@ -70,6 +70,8 @@ struct CatSst {
     uint64_t seq = 0;
     int64_t n_rows = 0;
     int64_t ts_min = 0, ts_max = 0;
+    bool series_stats_ok = true;    // every chunk has series stats
+    uint64_t series_max = 0;        // max over chunks (unsigned order)
     std::vector<CatRg> rgs;
 };
 
@@ -147,6 +149,14 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
             cr.cols[k].chunk_start = cc.chunk_start();
             cr.cols[k].comp_size = cc.total_compressed_size;
             cr.cols[k].num_values = cc.num_values;
+        }
+        const auto& secc = rg.columns[ci[0]];
+        if (secc.has_stats && secc.stat_max.size() == 8) {
+            uint64_t mx;
+            std::memcpy(&mx, secc.stat_max.data(), 8);
+            if (mx > out.series_max) out.series_max = mx;
+        } else {
+            out.series_stats_ok = false;
         }
         const auto& tscc = rg.columns[ci[1]];
         if (tscc.has_stats && tscc.stat_min.size() == 8 &&
@@ -300,6 +310,7 @@ struct StagedSst {  // host-side bookkeeping per prepared SST
 
 struct hx_prepared {
     hx_handle* h = nullptr;
+    bool key_claim_safe = true;   // no staged series_id == ~0 (proven by stats)
     hx_scan_spec spec{};
     std::vector<uint64_t> sset_keys;   // owned copy of series-set predicate
     std::vector<DevPlan> plans;
@@ -547,6 +558,39 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     }
     plan.dec_bytes = dec_off;
 
+    // ---- grid order: interleave row groups by ordinal across SSTs --------
+    // Rows are PK-sorted within each SST, so the k-th row group of every SST
+    // covers roughly the same series range. Interleaving makes concurrently
+    // resident workgroups touch a narrow series window, keeping the group
+    // table's hot lines in L2/L3 instead of HBM (speed only — dispatch
+    // order is never relied on for correctness).
+    {
+        const size_t nrg = plan.rgs.size();
+        std::vector<uint32_t> ordinal(nrg);
+        {
+            size_t ri = 0;
+            for (StagedSst* ss : members) {
+                for (size_t k = 0; k < ss->rg_idx.size(); k++, ri++)
+                    ordinal[ri] = (uint32_t)k;
+            }
+        }
+        std::vector<uint32_t> order(nrg);
+        for (size_t i = 0; i < nrg; i++) order[i] = (uint32_t)i;
+        std::stable_sort(order.begin(), order.end(),
+                         [&](uint32_t a, uint32_t b) {
+                             return ordinal[a] < ordinal[b];
+                         });
+        std::vector<int32_t> inv(nrg);
+        for (size_t i = 0; i < nrg; i++) inv[order[i]] = (int32_t)i;
+        std::vector<hx::RgDesc> reordered(nrg);
+        for (size_t i = 0; i < nrg; i++) {
+            reordered[i] = plan.rgs[order[i]];
+            if (reordered[i].next_rg >= 0)
+                reordered[i].next_rg = inv[reordered[i].next_rg];
+        }
+        plan.rgs.swap(reordered);
+    }
+
     // ---- upload ----------------------------------------------------------
     HIP_TRY(hipSetDevice(plan.device));
     if (!plan.stream) HIP_TRY(hipStreamCreate(&plan.stream));
@@ -620,6 +664,10 @@ extern "C" hx_status hx_prepare(hx_handle* h, const hx_scan_spec* spec,
         for (const auto& s : h->ssts)
             if (overlaps(s, spec->range)) chosen.push_back(&s);
     }
+
+    for (const CatSst* c : chosen)
+        if (!c->series_stats_ok || c->series_max == ~0ull)
+            P->key_claim_safe = false;
 
     // select row groups (reference pushdown pruning read.rs:459-470)
     std::vector<StagedSst> staged;
@@ -884,13 +932,16 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         plan.decoded = true;
     }
 
+    // one-CAS key claim: series-only grouping with a stats-proven sentinel
+    const int32_t key_claim = (!bucket && P->key_claim_safe) ? 1 : 0;
+
     // table size heuristic; grows on overflow
     uint32_t slots = plan.slots;
     if (!slots) {
         const char* env = getenv("HX_TABLE_SLOTS");
         if (env) slots = next_pow2_u32(strtoull(env, nullptr, 10));
         else slots = next_pow2_u32(std::max<uint64_t>(1 << 16,
-                                       (uint64_t)plan.rows_scanned / 4));
+                                       (uint64_t)plan.rows_scanned / 32));
         if (slots > (1u << 27)) slots = 1u << 27;
     }
 
@@ -899,7 +950,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         hx_status st = alloc_table(plan, slots, ops, bucket);
         if (st != HX_OK) return st;
         // reset table + counters (part of the step)
-        HIP_TRY(hipMemsetAsync(plan.t_state, 0, size_t(slots) * 4, s));
+        if (key_claim)
+            HIP_TRY(hipMemsetAsync(plan.t_series, 0xFF, size_t(slots) * 8, s));
+        else
+            HIP_TRY(hipMemsetAsync(plan.t_state, 0, size_t(slots) * 4, s));
         if (plan.t_sum) HIP_TRY(hipMemsetAsync(plan.t_sum, 0, size_t(slots) * 8, s));
         if (plan.t_cnt) HIP_TRY(hipMemsetAsync(plan.t_cnt, 0, size_t(slots) * 8, s));
         if (plan.t_min) HIP_TRY(hipMemsetAsync(plan.t_min, 0xFF, size_t(slots) * 8, s));
@@ -922,6 +976,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.use_sset = plan.d_sset ? 1 : 0;
         A.bucket_ms = bucket ? agg->bucket_ms : 0;
         A.ops = ops;
+        A.key_claim = key_claim;
         A.table = {plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
                    plan.t_cnt, plan.t_min, plan.t_max, slots - 1};
         A.fill = plan.d_counters + 0;
@@ -988,7 +1043,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
     hx::AggTable T{plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
                    plan.t_cnt, plan.t_min, plan.t_max, plan.slots - 1};
-    HIP_TRY(hx::launch_compact(s, T, plan.slots, ops,
+    HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,
                                bucket ? agg->bucket_ms : 0, co));
 
     // sort: LSD-stable — by bucket first (if any), then by series

@@ -76,6 +76,7 @@ struct AggParams {
     int32_t use_sset;
     int64_t bucket_ms;             // 0 = group by series only
     uint32_t ops;                  // HX_AGG_* mask
+    int32_t key_claim;             // 1 = one-CAS key-claim mode (no state word)
     AggTable table;
     unsigned long long* fill;      // claimed slots
     unsigned long long* overflow;  // !=0 => rerun with a larger table

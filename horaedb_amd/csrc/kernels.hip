@@ -13,6 +13,10 @@ namespace hx {
 #define RLX __ATOMIC_RELAXED
 #define AGT __HIP_MEMORY_SCOPE_AGENT
 
+// sentinel for the one-CAS key-claim table mode (host proves, from column
+// statistics, that no staged series_id equals it before enabling the mode)
+#define KEY_EMPTY 0xFFFFFFFFFFFFFFFFull
+
 __device__ __forceinline__ const uint8_t* hx_ptr(const uint8_t* blob,
                                                  const uint8_t* dec,
                                                  uint64_t off) {
@@ -82,8 +86,56 @@ __device__ bool shadowed(const AggParams& P, const SstDev& me,
 // sentinel needed); claimer stores key words then releases state=2; readers
 // are gated by the control dependency on state==2 (all table words accessed
 // with agent-scope atomics => L2-coherent, no L1 staleness).
+__device__ __forceinline__ void table_add(const AggParams& P, uint32_t i,
+                                          double vsum, unsigned long long cnt,
+                                          double mn, double mx) {
+    if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd(&P.table.sum[i], vsum);
+    if (P.ops & (HXK_COUNT | HXK_AVG)) atomicAdd(&P.table.cnt[i], cnt);
+    if (P.ops & HXK_MIN) atomicMin(&P.table.vmin[i], f64_ordered(mn));
+    if (P.ops & HXK_MAX) atomicMax(&P.table.vmax[i], f64_ordered(mx));
+}
+
+// Fast claim path: series_id IS the slot word, claimed by one CAS against
+// KEY_EMPTY (host proves via column statistics that no series == KEY_EMPTY;
+// series-only grouping). One L3 load per probe, no state word.
+__device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s,
+                                                  double vsum,
+                                                  unsigned long long cnt,
+                                                  double mn, double mx) {
+    uint32_t i = (uint32_t)mix64(s) & P.table.mask;
+    for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
+        uint64_t k = __hip_atomic_load(&P.table.series[i], RLX, AGT);
+        if (k == KEY_EMPTY) {
+            uint64_t expected = KEY_EMPTY;
+            if (__hip_atomic_compare_exchange_strong(&P.table.series[i],
+                    &expected, s, RLX, RLX, AGT)) {
+                __hip_atomic_fetch_add(P.fill, 1ull, RLX, AGT);
+                k = s;
+            } else {
+                k = expected;
+            }
+        }
+        if (k == s) {
+            table_add(P, i, vsum, cnt, mn, mx);
+            return;
+        }
+        i = (i + 1) & P.table.mask;
+    }
+    __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+}
+
+// General path (16-byte keys, e.g. (series,bucket)): claim via a CAS on a
+// state word (0 empty / 1 claiming / 2 ready), then key words; readers are
+// gated by the control dependency on state==2 (agent-scope atomics =>
+// L2-bypassing, coherent at the memory side).
 __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
-                                           int64_t b, double v) {
+                                           int64_t b, double vsum,
+                                           unsigned long long cnt,
+                                           double mn, double mx) {
+    if (P.key_claim) {
+        agg_update_keycas(P, s, vsum, cnt, mn, mx);
+        return;
+    }
     const bool use_b = P.bucket_ms != 0;
     uint64_t h = mix64(s ^ ((uint64_t)b * 0xD1B54A32D192ED03ull));
     uint32_t i = (uint32_t)h & P.table.mask;
@@ -117,10 +169,7 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
         if (__hip_atomic_load(&P.table.series[i], RLX, AGT) == s &&
             (!use_b || (int64_t)__hip_atomic_load(
                            (unsigned long long*)&P.table.bucket[i], RLX, AGT) == b)) {
-            if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd(&P.table.sum[i], v);
-            if (P.ops & (HXK_COUNT | HXK_AVG)) atomicAdd(&P.table.cnt[i], 1ull);
-            if (P.ops & HXK_MIN) atomicMin(&P.table.vmin[i], f64_ordered(v));
-            if (P.ops & HXK_MAX) atomicMax(&P.table.vmax[i], f64_ordered(v));
+            table_add(P, i, vsum, cnt, mn, mx);
             return;
         }
         i = (i + 1) & P.table.mask;
@@ -136,6 +185,7 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
 extern "C" __global__ void __launch_bounds__(256)
 k_scan_agg(AggParams P) {
     unsigned long long my_matched = 0;
+    const int lane = threadIdx.x & 63;
     for (uint32_t rgi = blockIdx.x; rgi < P.n_rgs; rgi += gridDim.x) {
         const RgDesc rg = P.rgs[rgi];
         const uint64_t* S = (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
@@ -143,28 +193,69 @@ k_scan_agg(AggParams P) {
         const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
         const SstDev sst = P.ssts[rg.sst_id];
         const uint32_t n = rg.n_rows;
-        for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
-            int64_t t = T[r];
-            if (t < P.ts_lo || t >= P.ts_hi) continue;
-            uint64_t s = S[r];
-            if (P.use_sset && !sset_has(P, s)) continue;
-            // within-SST dedup: the LAST row of an equal-PK run survives
-            // (LastValueOperator, operator.rs:37-44; plan order read.rs:456-480)
-            bool dup = false;
-            if (r + 1 < n) {
-                dup = (S[r + 1] == s) & (T[r + 1] == t);
-            } else if (rg.next_rg >= 0) {
-                const RgDesc nx = P.rgs[rg.next_rg];
-                uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec, nx.series_off);
-                int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
-                dup = (s2 == s) & (t2 == t);
+        for (uint32_t base = 0; base < n; base += blockDim.x) {
+            const uint32_t r = base + threadIdx.x;
+            const bool inb = r < n;
+            uint64_t s = KEY_EMPTY;
+            int64_t t = 0;
+            double v = 0.0;
+            bool alive = false;
+            if (inb) {
+                t = T[r];
+                s = S[r];
+                alive = (t >= P.ts_lo) & (t < P.ts_hi);
+                if (alive && P.use_sset) alive = sset_has(P, s);
+                if (alive) {
+                    // within-SST dedup: the LAST row of an equal-PK run
+                    // survives (LastValueOperator, operator.rs:37-44; plan
+                    // order read.rs:456-480)
+                    bool dup = false;
+                    if (r + 1 < n) {
+                        dup = (S[r + 1] == s) & (T[r + 1] == t);
+                    } else if (rg.next_rg >= 0) {
+                        const RgDesc nx = P.rgs[rg.next_rg];
+                        uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
+                                                               nx.series_off);
+                        int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec,
+                                                             nx.ts_off);
+                        dup = (s2 == s) & (t2 == t);
+                    }
+                    if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+                    alive = !dup;
+                }
+                if (alive) v = V[r];
             }
-            if (dup) continue;
-            if (sst.cluster >= 0 && shadowed(P, sst, s, t)) continue;
-            double v = V[r];
-            int64_t b = P.bucket_ms ? floordiv(t, P.bucket_ms) : 0;
-            agg_update(P, s, b, v);
-            my_matched++;
+            int64_t b = (inb && P.bucket_ms) ? floordiv(t, P.bucket_ms) : 0;
+            // Wave-level segmented pre-reduction over the sorted rows: lanes
+            // hold 64 consecutive rows; equal (series,bucket) runs are
+            // contiguous (SSTs are PK-sorted), so each run head accumulates
+            // its run and issues ONE table update (guideline 12).
+            unsigned long long c = alive ? 1ull : 0ull;
+            double vv = alive ? v : 0.0;
+            double mn = alive ? v : HUGE_VAL;
+            double mx = alive ? v : -HUGE_VAL;
+            my_matched += c;
+            const uint64_t sp = __shfl_up(s, 1, 64);
+            const int64_t bp = __shfl_up((long long)b, 1, 64);
+            const bool head = (lane == 0) || sp != s || bp != b;
+            bool done = false;
+            for (int d = 1; d < 64; d++) {
+                const uint64_t s2 = __shfl_down(s, d, 64);
+                const long long b2 = __shfl_down((long long)b, d, 64);
+                const double v2 = __shfl_down(vv, d, 64);
+                const unsigned long long c2 = __shfl_down(c, d, 64);
+                const double mn2 = __shfl_down(mn, d, 64);
+                const double mx2 = __shfl_down(mx, d, 64);
+                done = done || (lane + d >= 64) || s2 != s || b2 != b;
+                if (head && !done) {
+                    vv += v2;
+                    c += c2;
+                    mn = fmin(mn, mn2);
+                    mx = fmax(mx, mx2);
+                }
+                if (__all(done)) break;
+            }
+            if (head && c > 0) agg_update(P, s, b, vv, c, mn, mx);
         }
     }
     // one atomic per wave for the matched counter (guideline 12)
@@ -182,6 +273,7 @@ struct CompactParams {
     AggTable table;
     uint32_t n_slots;
     uint32_t ops;
+    int32_t key_claim;
     int64_t bucket_ms;
     uint64_t* out_series;
     long long* out_bucket;
@@ -194,16 +286,31 @@ struct CompactParams {
 
 extern "C" __global__ void __launch_bounds__(256)
 k_compact(CompactParams C) {
-    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < C.n_slots;
-         i += blockDim.x * gridDim.x) {
-        if (C.table.state[i] != 2u) continue;
-        unsigned long long j = atomicAdd(C.n_out, 1ull);
-        C.out_series[j] = C.table.series[i];
-        if (C.bucket_ms) C.out_bucket[j] = C.table.bucket[i];
-        if (C.out_sum) C.out_sum[j] = C.table.sum[i];
-        if (C.out_cnt) C.out_cnt[j] = C.table.cnt[i];
-        if (C.out_min) C.out_min[j] = ordered_f64(C.table.vmin[i]);
-        if (C.out_max) C.out_max[j] = ordered_f64(C.table.vmax[i]);
+    const int lane = threadIdx.x & 63;
+    const uint32_t stride = blockDim.x * gridDim.x;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+         __any(i < C.n_slots); i += stride) {
+        const bool live = (i < C.n_slots) &&
+                          (C.key_claim ? (C.table.series[i] != KEY_EMPTY)
+                                       : (C.table.state[i] == 2u));
+        const unsigned long long mask = __ballot(live);
+        if (!mask) continue;
+        // one atomic per wave (guideline 12): first live lane reserves
+        const int leader = __ffsll((unsigned long long)mask) - 1;
+        unsigned long long wave_base = 0;
+        if (lane == leader)
+            wave_base = atomicAdd(C.n_out, (unsigned long long)__popcll(mask));
+        wave_base = __shfl(wave_base, leader, 64);
+        if (live) {
+            const unsigned long long j =
+                wave_base + __popcll(mask & ((1ull << lane) - 1ull));
+            C.out_series[j] = C.table.series[i];
+            if (C.bucket_ms) C.out_bucket[j] = C.table.bucket[i];
+            if (C.out_sum) C.out_sum[j] = C.table.sum[i];
+            if (C.out_cnt) C.out_cnt[j] = C.table.cnt[i];
+            if (C.out_min) C.out_min[j] = ordered_f64(C.table.vmin[i]);
+            if (C.out_max) C.out_max[j] = ordered_f64(C.table.vmax[i]);
+        }
     }
 }
 
@@ -467,11 +574,13 @@ hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid) {
 }
 
 hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
-                          uint32_t ops, int64_t bucket_ms, const CompactOut& o) {
+                          uint32_t ops, int32_t key_claim, int64_t bucket_ms,
+                          const CompactOut& o) {
     CompactParams C;
     C.table = t;
     C.n_slots = n_slots;
     C.ops = ops;
+    C.key_claim = key_claim;
     C.bucket_ms = bucket_ms;
     C.out_series = o.series;
     C.out_bucket = o.bucket;
