@@ -820,13 +820,25 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
 // ---------------------------------------------------------------------------
 namespace {
 
-struct PartialResult {  // one device's sorted aggregate table, on host
+struct HostTable {  // one device's sorted aggregate table, on host
     size_t n = 0;
-    std::vector<uint64_t> series;
-    std::vector<int64_t> bucket;
-    std::vector<double> sum;
-    std::vector<unsigned long long> cnt;
-    std::vector<double> vmin, vmax;
+    void* buf = nullptr;   // pinned when possible (single D2H copy)
+    bool pinned = false;
+    uint64_t* series = nullptr;
+    int64_t* bucket = nullptr;
+    double* sum = nullptr;
+    unsigned long long* cnt = nullptr;
+    double* vmin = nullptr;
+    double* vmax = nullptr;
+    double* avg = nullptr;
+    void release() {
+        if (buf) {
+            if (pinned) hipHostFree(buf);
+            else free(buf);
+            buf = nullptr;
+        }
+        n = 0;
+    }
 };
 
 uint32_t next_pow2_u32(uint64_t x) {
@@ -865,7 +877,7 @@ uint32_t kernel_ops(uint32_t ops) {
 }
 
 hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
-                    PartialResult& out, double* agg_kernel_ms,
+                    HostTable& out, double* agg_kernel_ms,
                     unsigned long long* matched_out) {
     HIP_TRY(hipSetDevice(plan.device));
     hipStream_t s = plan.stream;
@@ -987,7 +999,16 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipEventCreate(&e0));
         HIP_TRY(hipEventCreate(&e1));
         HIP_TRY(hipEventRecord(e0, s));
-        HIP_TRY(hx::launch_scan_agg(s, A, 0));
+        const bool use_gang = !bucket && !getenv("HX_NO_GANG");
+        if (use_gang) {
+            uint32_t gang = 32;
+            if (const char* ge = getenv("HX_GANG"))
+                gang = (uint32_t)strtoul(ge, nullptr, 10);
+            HIP_TRY(hx::launch_scan_agg_gang(s, A, gang,
+                                             (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
+        } else {
+            HIP_TRY(hx::launch_scan_agg(s, A, 0));
+        }
         HIP_TRY(hipEventRecord(e1, s));
         HIP_TRY(hipStreamSynchronize(s));
         float ms = 0;
@@ -1007,37 +1028,44 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     *matched_out = counters[2];
     unsigned long long fill = counters[0];
 
-    // ---- compact + sort + gather ---------------------------------------
+    // ---- compact + sort + gather + single D2H ---------------------------
+    out.release();
     out.n = fill;
     if (fill == 0) return HX_OK;
     const uint32_t n = (uint32_t)fill;
-    // scratch layout: up to 10 arrays of n x 8B + 2 perm arrays n x 4B
-    const int n_val_arrays = 1 /*series*/ + (bucket ? 1 : 0) +
-                             (plan.t_sum ? 1 : 0) + (plan.t_cnt ? 1 : 0) +
-                             (plan.t_min ? 1 : 0) + (plan.t_max ? 1 : 0);
-    size_t need = size_t(n) * 8 * (2 * n_val_arrays + 2) + size_t(n) * 4 * 3 + 256;
+    const bool has_sum = plan.t_sum != nullptr;
+    const bool has_cnt = plan.t_cnt != nullptr;
+    const bool has_min = plan.t_min != nullptr;
+    const bool has_max = plan.t_max != nullptr;
+    const bool has_avg = (agg->ops & HX_AGG_AVG) != 0;
+    const uint32_t n_core = 1 + (bucket ? 1 : 0) + has_sum + has_cnt +
+                            has_min + has_max;
+    const uint32_t n_total = n_core + (has_avg ? 1 : 0);
+    // scratch: compact arrays (n_core) + sort keys in/out (2) + dst (n_total)
+    // + 3 perm arrays + counter
+    size_t need = size_t(n) * 8 * (n_core + 2 + n_total) +
+                  size_t(n) * 4 * 3 + 256;
     hx_status st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
     if (st != HX_OK) return st;
     uint8_t* base = (uint8_t*)plan.d_scratch;
-    auto carve8 = [&](uint32_t count) {
+    auto carve8 = [&](size_t count) {
         uint8_t* p = base;
-        base += (size_t(count) * 8 + 7) & ~size_t(7);  // keep 8B alignment
+        base += (count * 8 + 7) & ~size_t(7);
         return p;
     };
     uint64_t* c_series = (uint64_t*)carve8(n);
     long long* c_bucket = bucket ? (long long*)carve8(n) : nullptr;
-    double* c_sum = plan.t_sum ? (double*)carve8(n) : nullptr;
-    unsigned long long* c_cnt = plan.t_cnt ? (unsigned long long*)carve8(n) : nullptr;
-    double* c_min = plan.t_min ? (double*)carve8(n) : nullptr;
-    double* c_max = plan.t_max ? (double*)carve8(n) : nullptr;
-    uint64_t* keys_tmp = (uint64_t*)carve8(n);       // sort keys in/out
-    uint64_t* vals_tmp = (uint64_t*)carve8(n);       // gather staging
+    double* c_sum = has_sum ? (double*)carve8(n) : nullptr;
+    unsigned long long* c_cnt = has_cnt ? (unsigned long long*)carve8(n) : nullptr;
+    double* c_min = has_min ? (double*)carve8(n) : nullptr;
+    double* c_max = has_max ? (double*)carve8(n) : nullptr;
+    uint64_t* keys_tmp = (uint64_t*)carve8(n);
+    uint64_t* keys_out = (uint64_t*)carve8(n);
+    unsigned long long* d_dst = (unsigned long long*)carve8(size_t(n) * n_total);
     unsigned long long* d_nout = (unsigned long long*)carve8(1);
-    // 4-byte perm arrays go last (they would break 8B alignment otherwise)
     uint32_t* perm_a = (uint32_t*)base; base += size_t(n) * 4;
     uint32_t* perm_b = (uint32_t*)base; base += size_t(n) * 4;
     uint32_t* perm_c = (uint32_t*)base; base += size_t(n) * 4;
-    (void)perm_c;
 
     HIP_TRY(hipMemsetAsync(d_nout, 0, 8, s));
     hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
@@ -1051,69 +1079,58 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     const uint32_t* perm_in = perm_a;
     uint32_t* perm_out = perm_b;
     if (bucket) {
-        // biased bucket keys for signed order
         HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_bucket,
                                       perm_a, (unsigned long long*)keys_tmp, n));
-        // bias: XOR sign bit (monotone i64->u64). gather wrote bucket[perm]=bucket (iota)
-        // reuse avg kernel slot: do bias inside gather? simpler: small kernel not
-        // available — bias via sort on raw then fix order: instead bias on host? n large.
-        // Use trick: radix sort i64 keys as u64 after XOR with sign bit — do the
-        // XOR with a gather variant below.
         HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)keys_tmp, n));
-        HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, (uint64_t*)vals_tmp, perm_a,
-                                   perm_out, n, &plan.d_sort_temp,
-                                   &plan.sort_temp_cap));
+        HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, keys_out, perm_a, perm_out, n,
+                                   &plan.d_sort_temp, &plan.sort_temp_cap));
         perm_in = perm_out;
         perm_out = perm_c;
     }
-    // series pass (stable keeps bucket order within equal series)
     HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_series, perm_in,
                                   (unsigned long long*)keys_tmp, n));
-    HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, (uint64_t*)vals_tmp, perm_in,
-                               perm_out, n, &plan.d_sort_temp,
-                               &plan.sort_temp_cap));
+    HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, keys_out, perm_in, perm_out, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
     const uint32_t* perm = perm_out;
 
-    // gather each array by final perm and copy D2H
-    out.series.resize(n);
-    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_series, perm,
-                                  (unsigned long long*)vals_tmp, n));
-    HIP_TRY(hipMemcpyAsync(out.series.data(), vals_tmp, size_t(n) * 8,
-                           hipMemcpyDeviceToHost, s));
+    // one multi-array gather into the contiguous dst, then ONE D2H copy
+    const unsigned long long* srcs[8];
+    uint32_t na = 0;
+    srcs[na++] = (const unsigned long long*)c_series;
+    size_t off_bucket = bucket ? na : 0;
+    if (bucket) srcs[na++] = (const unsigned long long*)c_bucket;
+    size_t off_sum = has_sum ? na : 0;
+    if (has_sum) srcs[na++] = (const unsigned long long*)c_sum;
+    size_t off_cnt = has_cnt ? na : 0;
+    if (has_cnt) srcs[na++] = (const unsigned long long*)c_cnt;
+    size_t off_min = has_min ? na : 0;
+    if (has_min) srcs[na++] = (const unsigned long long*)c_min;
+    size_t off_max = has_max ? na : 0;
+    if (has_max) srcs[na++] = (const unsigned long long*)c_max;
+    HIP_TRY(hx::launch_gather_multi(s, srcs, na, perm, d_dst, n));
+    if (has_avg)
+        HIP_TRY(hx::launch_avg(s, (const double*)(d_dst + size_t(off_sum) * n),
+                               d_dst + size_t(off_cnt) * n,
+                               (double*)(d_dst + size_t(n_core) * n), n));
+
+    const size_t bytes = size_t(n) * 8 * n_total;
+    if (hipHostMalloc(&out.buf, bytes, hipHostMallocDefault) == hipSuccess) {
+        out.pinned = true;
+    } else {
+        out.buf = malloc(bytes);
+        out.pinned = false;
+        if (!out.buf) return fail(HX_ERR_IO, "result alloc failed");
+    }
+    HIP_TRY(hipMemcpyAsync(out.buf, d_dst, bytes, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
-    auto fetch = [&](void* host_dst, const void* dev_src) -> hx_status {
-        HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)dev_src, perm,
-                                      (unsigned long long*)vals_tmp, n));
-        HIP_TRY(hipMemcpyAsync(host_dst, vals_tmp, size_t(n) * 8,
-                               hipMemcpyDeviceToHost, s));
-        HIP_TRY(hipStreamSynchronize(s));
-        return HX_OK;
-    };
-    if (bucket) {
-        out.bucket.resize(n);
-        st = fetch(out.bucket.data(), c_bucket);
-        if (st != HX_OK) return st;
-    }
-    if (c_sum) {
-        out.sum.resize(n);
-        st = fetch(out.sum.data(), c_sum);
-        if (st != HX_OK) return st;
-    }
-    if (c_cnt) {
-        out.cnt.resize(n);
-        st = fetch(out.cnt.data(), c_cnt);
-        if (st != HX_OK) return st;
-    }
-    if (c_min) {
-        out.vmin.resize(n);
-        st = fetch(out.vmin.data(), c_min);
-        if (st != HX_OK) return st;
-    }
-    if (c_max) {
-        out.vmax.resize(n);
-        st = fetch(out.vmax.data(), c_max);
-        if (st != HX_OK) return st;
-    }
+    uint64_t* hb = (uint64_t*)out.buf;
+    out.series = hb;
+    out.bucket = bucket ? (int64_t*)(hb + off_bucket * n) : nullptr;
+    out.sum = has_sum ? (double*)(hb + off_sum * n) : nullptr;
+    out.cnt = has_cnt ? (unsigned long long*)(hb + off_cnt * n) : nullptr;
+    out.vmin = has_min ? (double*)(hb + off_min * n) : nullptr;
+    out.vmax = has_max ? (double*)(hb + off_max * n) : nullptr;
+    out.avg = has_avg ? (double*)(hb + size_t(n_core) * n) : nullptr;
     return HX_OK;
 }
 
@@ -1135,7 +1152,7 @@ struct ResultStorage {  // backs hx_result_table arrays
 // n-way merge of per-device sorted partials, combining equal (series,bucket)
 // groups: the host partial-aggregate merge of SURVEY §8(e) (partials are
 // O(groups), not O(rows)).
-void merge_partials(std::vector<PartialResult>& parts, uint32_t ops,
+void merge_partials(std::vector<HostTable>& parts, uint32_t ops,
                     bool bucket, ResultStorage& out) {
     const uint32_t kops = kernel_ops(ops);
     std::vector<size_t> idx(parts.size(), 0);
@@ -1178,27 +1195,13 @@ void merge_partials(std::vector<PartialResult>& parts, uint32_t ops,
     }
 }
 
-// single-device fast path: move arrays, compute avg
-void finish_single(PartialResult& p, uint32_t ops, bool bucket,
-                   ResultStorage& out) {
-    out.series = std::move(p.series);
-    if (bucket) out.bucket = std::move(p.bucket);
-    if (ops & HX_AGG_AVG) {
-        out.avg.resize(p.n);
-        for (size_t i = 0; i < p.n; i++)
-            out.avg[i] = p.sum[i] / double(p.cnt[i]);
-    }
-    if (ops & HX_AGG_SUM) out.sum = std::move(p.sum);
-    if (ops & HX_AGG_COUNT) out.cnt = std::move(p.cnt);
-    if (ops & HX_AGG_MIN) out.vmin = std::move(p.vmin);
-    if (ops & HX_AGG_MAX) out.vmax = std::move(p.vmax);
-}
-
 }  // namespace
 
 struct hx_result_impl {
     hx_result_table pub_{};
-    ResultStorage store;
+    HostTable owned;       // single-device fast path: pub_ points into owned
+    ResultStorage store;   // multi-device merge path
+    ~hx_result_impl() { owned.release(); }
 };
 
 extern "C" hx_status hx_exec_agg(hx_prepared* P, const hx_agg_spec* agg,
@@ -1209,7 +1212,7 @@ extern "C" hx_status hx_exec_agg(hx_prepared* P, const hx_agg_spec* agg,
     auto t0 = std::chrono::steady_clock::now();
     const bool bucket = agg->bucket_ms > 0;
 
-    std::vector<PartialResult> parts(P->plans.size());
+    std::vector<HostTable> parts(P->plans.size());
     double agg_ms_max = 0, decode_ms_max = 0;
     unsigned long long matched = 0;
     for (size_t d = 0; d < P->plans.size(); d++) {
@@ -1223,20 +1226,31 @@ extern "C" hx_status hx_exec_agg(hx_prepared* P, const hx_agg_spec* agg,
     }
 
     auto R = std::make_unique<hx_result_impl>();
-    if (parts.size() == 1)
-        finish_single(parts[0], agg->ops, bucket, R->store);
-    else
-        merge_partials(parts, agg->ops, bucket, R->store);
-
     hx_result_table& T = R->pub_;
-    T.n_groups = R->store.series.size();
-    T.series_id = R->store.series.data();
-    T.bucket = bucket ? R->store.bucket.data() : nullptr;
-    T.sum = (agg->ops & HX_AGG_SUM) ? R->store.sum.data() : nullptr;
-    T.count = (agg->ops & HX_AGG_COUNT) ? (const uint64_t*)R->store.cnt.data() : nullptr;
-    T.vmin = (agg->ops & HX_AGG_MIN) ? R->store.vmin.data() : nullptr;
-    T.vmax = (agg->ops & HX_AGG_MAX) ? R->store.vmax.data() : nullptr;
-    T.avg = (agg->ops & HX_AGG_AVG) ? R->store.avg.data() : nullptr;
+    if (parts.size() == 1) {
+        R->owned = parts[0];
+        parts[0].buf = nullptr;  // ownership moved
+        const HostTable& H = R->owned;
+        T.n_groups = H.n;
+        T.series_id = H.series;
+        T.bucket = bucket ? H.bucket : nullptr;
+        T.sum = (agg->ops & HX_AGG_SUM) ? H.sum : nullptr;
+        T.count = (agg->ops & HX_AGG_COUNT) ? (const uint64_t*)H.cnt : nullptr;
+        T.vmin = (agg->ops & HX_AGG_MIN) ? H.vmin : nullptr;
+        T.vmax = (agg->ops & HX_AGG_MAX) ? H.vmax : nullptr;
+        T.avg = (agg->ops & HX_AGG_AVG) ? H.avg : nullptr;
+    } else {
+        merge_partials(parts, agg->ops, bucket, R->store);
+        for (auto& p : parts) p.release();
+        T.n_groups = R->store.series.size();
+        T.series_id = R->store.series.data();
+        T.bucket = bucket ? R->store.bucket.data() : nullptr;
+        T.sum = (agg->ops & HX_AGG_SUM) ? R->store.sum.data() : nullptr;
+        T.count = (agg->ops & HX_AGG_COUNT) ? (const uint64_t*)R->store.cnt.data() : nullptr;
+        T.vmin = (agg->ops & HX_AGG_MIN) ? R->store.vmin.data() : nullptr;
+        T.vmax = (agg->ops & HX_AGG_MAX) ? R->store.vmax.data() : nullptr;
+        T.avg = (agg->ops & HX_AGG_AVG) ? R->store.avg.data() : nullptr;
+    }
 
     P->last_stats.exec_ms = std::chrono::duration<double, std::milli>(
                                 std::chrono::steady_clock::now() - t0).count();

@@ -22,6 +22,12 @@ hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
 hipError_t launch_copy_u64(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                            const CopyDesc* descs, uint32_t n_descs);
 hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid);
+hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
+                                uint32_t gang_size, bool minmax);
+hipError_t launch_gather_multi(hipStream_t s,
+                               const unsigned long long* const* srcs,
+                               uint32_t n_arrays, const uint32_t* perm,
+                               unsigned long long* dst, uint32_t n);
 hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
                           uint32_t ops, int32_t key_claim, int64_t bucket_ms,
                           const CompactOut& o);

@@ -104,7 +104,10 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
                                                   double mn, double mx) {
     uint32_t i = (uint32_t)mix64(s) & P.table.mask;
     for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
-        uint64_t k = __hip_atomic_load(&P.table.series[i], RLX, AGT);
+        // plain (cached) probe: a slot transitions KEY_EMPTY -> key exactly
+        // once per exec, so a stale read can only be KEY_EMPTY — which falls
+        // through to the CAS (coherent) and learns the truth.
+        uint64_t k = P.table.series[i];
         if (k == KEY_EMPTY) {
             uint64_t expected = KEY_EMPTY;
             if (__hip_atomic_compare_exchange_strong(&P.table.series[i],
@@ -177,6 +180,32 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
     __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
 }
 
+// Per-row filter + dedup (shared by both aggregate kernels).
+// Returns alive; s/t are the row's PK. DESIGN.md §5.
+__device__ __forceinline__ bool row_alive(const AggParams& P, const RgDesc& rg,
+                                          const SstDev& sst, const uint64_t* S,
+                                          const int64_t* T, uint32_t r,
+                                          uint32_t n, uint64_t s, int64_t t) {
+    bool alive = (t >= P.ts_lo) & (t < P.ts_hi);
+    if (alive && P.use_sset) alive = sset_has(P, s);
+    if (alive) {
+        // within-SST dedup: the LAST row of an equal-PK run survives
+        // (LastValueOperator, operator.rs:37-44; plan order read.rs:456-480)
+        bool dup = false;
+        if (r + 1 < n) {
+            dup = (S[r + 1] == s) & (T[r + 1] == t);
+        } else if (rg.next_rg >= 0) {
+            const RgDesc nx = P.rgs[rg.next_rg];
+            uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec, nx.series_off);
+            int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
+            dup = (s2 == s) & (t2 == t);
+        }
+        if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+        alive = !dup;
+    }
+    return alive;
+}
+
 // ---------------------------------------------------------------------------
 // The headline fused kernel: decode(PLAIN in place) + ts-range/series-set
 // filter + MergeExec dedup + hash group-by aggregate. One workgroup per row
@@ -203,26 +232,7 @@ k_scan_agg(AggParams P) {
             if (inb) {
                 t = T[r];
                 s = S[r];
-                alive = (t >= P.ts_lo) & (t < P.ts_hi);
-                if (alive && P.use_sset) alive = sset_has(P, s);
-                if (alive) {
-                    // within-SST dedup: the LAST row of an equal-PK run
-                    // survives (LastValueOperator, operator.rs:37-44; plan
-                    // order read.rs:456-480)
-                    bool dup = false;
-                    if (r + 1 < n) {
-                        dup = (S[r + 1] == s) & (T[r + 1] == t);
-                    } else if (rg.next_rg >= 0) {
-                        const RgDesc nx = P.rgs[rg.next_rg];
-                        uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
-                                                               nx.series_off);
-                        int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec,
-                                                             nx.ts_off);
-                        dup = (s2 == s) & (t2 == t);
-                    }
-                    if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
-                    alive = !dup;
-                }
+                alive = row_alive(P, rg, sst, S, T, r, n, s, t);
                 if (alive) v = V[r];
             }
             int64_t b = (inb && P.bucket_ms) ? floordiv(t, P.bucket_ms) : 0;
@@ -259,6 +269,103 @@ k_scan_agg(AggParams P) {
         }
     }
     // one atomic per wave for the matched counter (guideline 12)
+    for (int off = 32; off > 0; off >>= 1)
+        my_matched += __shfl_down(my_matched, off, 64);
+    if ((threadIdx.x & 63) == 0 && my_matched)
+        atomicAdd(P.matched, my_matched);
+}
+
+// ---------------------------------------------------------------------------
+// Gang kernel (series-only grouping): one workgroup aggregates a GANG of
+// same-ordinal row groups from many SSTs into an LDS hash table, then
+// flushes distinct keys to the global table. Same-ordinal row groups cover
+// roughly the same series window (SSTs are PK-sorted over one id universe),
+// so a series touched by G SSTs costs ONE global update instead of G —
+// global atomic traffic drops by ~G (guideline 12 at gang scale).
+// ---------------------------------------------------------------------------
+struct GangParams {
+    AggParams P;
+    uint32_t gang_size;
+    uint32_t n_gangs;
+    uint32_t ne;        // LDS hash entries (power of two)
+    uint32_t has_mm;    // min/max tracked
+};
+
+extern "C" __global__ void __launch_bounds__(512)
+k_scan_agg_gang(GangParams G) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const AggParams& P = G.P;
+    const uint32_t ne = G.ne;
+    uint64_t* lkey = (uint64_t*)smem;
+    double* lsum = (double*)(smem + (size_t)ne * 8);
+    unsigned long long* lmin =
+        (unsigned long long*)(smem + (size_t)ne * 16);
+    unsigned long long* lmax =
+        (unsigned long long*)(smem + (size_t)ne * (G.has_mm ? 24 : 16));
+    unsigned int* lcnt =
+        (unsigned int*)(smem + (size_t)ne * (G.has_mm ? 32 : 16));
+
+    unsigned long long my_matched = 0;
+    for (uint32_t gang = blockIdx.x; gang < G.n_gangs; gang += gridDim.x) {
+        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+            lkey[i] = KEY_EMPTY;
+            lsum[i] = 0.0;
+            lcnt[i] = 0u;
+            if (G.has_mm) {
+                lmin[i] = ~0ull;
+                lmax[i] = 0ull;
+            }
+        }
+        __syncthreads();
+        const uint32_t rg_end = min(gang * G.gang_size + G.gang_size, P.n_rgs);
+        for (uint32_t rgi = gang * G.gang_size; rgi < rg_end; rgi++) {
+            const RgDesc rg = P.rgs[rgi];
+            const uint64_t* S =
+                (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+            const int64_t* T = (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+            const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+            const SstDev sst = P.ssts[rg.sst_id];
+            const uint32_t n = rg.n_rows;
+            for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
+                const int64_t t = T[r];
+                const uint64_t s = S[r];
+                if (!row_alive(P, rg, sst, S, T, r, n, s, t)) continue;
+                const double v = V[r];
+                my_matched++;
+                // LDS hash insert-or-update; overflow spills to global
+                uint32_t i = (uint32_t)mix64(s) & (ne - 1);
+                bool done = false;
+                for (int probes = 0; probes < 24; probes++) {
+                    uint64_t k = lkey[i];
+                    if (k == KEY_EMPTY) {
+                        uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, s);
+                        k = (old == KEY_EMPTY) ? s : old;
+                    }
+                    if (k == s) {
+                        atomicAdd(&lsum[i], v);
+                        atomicAdd(&lcnt[i], 1u);
+                        if (G.has_mm) {
+                            atomicMin(&lmin[i], f64_ordered(v));
+                            atomicMax(&lmax[i], f64_ordered(v));
+                        }
+                        done = true;
+                        break;
+                    }
+                    i = (i + 1) & (ne - 1);
+                }
+                if (!done)  // LDS table full: direct global update
+                    agg_update(P, s, 0, v, 1ull, v, v);
+            }
+        }
+        __syncthreads();
+        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+            if (lkey[i] == KEY_EMPTY) continue;
+            agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
+                       G.has_mm ? ordered_f64(lmin[i]) : 0.0,
+                       G.has_mm ? ordered_f64(lmax[i]) : 0.0);
+        }
+        __syncthreads();
+    }
     for (int off = 32; off > 0; off >>= 1)
         my_matched += __shfl_down(my_matched, off, 64);
     if ((threadIdx.x & 63) == 0 && my_matched)
@@ -315,6 +422,24 @@ k_compact(CompactParams C) {
 }
 
 // Gather 8-byte elements by permutation (applies the sort order).
+struct GatherMulti {
+    const unsigned long long* src[8];
+    unsigned long long* dst;   // dst array a at dst + a*n
+    const uint32_t* perm;
+    uint32_t n;
+    uint32_t n_arrays;
+};
+
+extern "C" __global__ void __launch_bounds__(256)
+k_gather_multi(GatherMulti g) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < g.n;
+         i += blockDim.x * gridDim.x) {
+        const uint32_t p = g.perm[i];
+        for (uint32_t a = 0; a < g.n_arrays; a++)
+            g.dst[(size_t)a * g.n + i] = g.src[a][p];
+    }
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 k_gather_u64(const unsigned long long* in, const uint32_t* perm,
              unsigned long long* out, uint32_t n) {
@@ -570,6 +695,34 @@ hipError_t launch_copy_u64(hipStream_t s, const uint8_t* blob, uint8_t* dec,
 hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid) {
     if (grid == 0) grid = p.n_rgs > 65535 ? 65535 : (p.n_rgs ? p.n_rgs : 1);
     hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(256), 0, s, p);
+    return hipGetLastError();
+}
+
+hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
+                                uint32_t gang_size, bool minmax) {
+    GangParams G;
+    G.P = p;
+    G.gang_size = gang_size;
+    G.n_gangs = (p.n_rgs + gang_size - 1) / gang_size;
+    G.has_mm = minmax ? 1u : 0u;
+    G.ne = minmax ? 2048u : 4096u;
+    size_t lds = (size_t)G.ne * (minmax ? 36 : 20);
+    uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
+    hipLaunchKernelGGL(k_scan_agg_gang, dim3(grid), dim3(512), lds, s, G);
+    return hipGetLastError();
+}
+
+hipError_t launch_gather_multi(hipStream_t s,
+                               const unsigned long long* const* srcs,
+                               uint32_t n_arrays, const uint32_t* perm,
+                               unsigned long long* dst, uint32_t n) {
+    GatherMulti g{};
+    for (uint32_t a = 0; a < n_arrays && a < 8; a++) g.src[a] = srcs[a];
+    g.dst = dst;
+    g.perm = perm;
+    g.n = n;
+    g.n_arrays = n_arrays;
+    hipLaunchKernelGGL(k_gather_multi, dim3(grid_for(n, 256)), dim3(256), 0, s, g);
     return hipGetLastError();
 }
 

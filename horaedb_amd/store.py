@@ -124,12 +124,17 @@ class Prepared:
         self._store = store
         self._h = handle
 
-    def exec_agg(self, ops=AGG_SUM | AGG_COUNT, bucket_ms=0):
+    def exec_agg(self, ops=AGG_SUM | AGG_COUNT, bucket_ms=0, copy=True):
+        """copy=False: run the full query (result lands in host memory) but
+        skip the numpy materialization — for benchmarking loops."""
         spec = _AggSpec(ops=ops, bucket_ms=bucket_ms)
         out = C.POINTER(_ResultTable)()
         _check(_lib.hx_exec_agg(self._h, C.byref(spec), C.byref(out)))
         t = out.contents
         n = t.n_groups
+        if not copy:
+            _lib.hx_result_free(out)
+            return {"n_groups": n}
         res = {"series_id": _np(t.series_id, n, np.uint64)}
         if t.bucket:
             res["bucket"] = _np(t.bucket, n, np.int64)
