@@ -999,16 +999,16 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipEventCreate(&e0));
         HIP_TRY(hipEventCreate(&e1));
         HIP_TRY(hipEventRecord(e0, s));
-        const bool use_gang = !bucket && !getenv("HX_NO_GANG");
+        bool use_gang = !bucket && !getenv("HX_NO_GANG");
         if (use_gang) {
             uint32_t gang = 32;
             if (const char* ge = getenv("HX_GANG"))
                 gang = (uint32_t)strtoul(ge, nullptr, 10);
-            HIP_TRY(hx::launch_scan_agg_gang(s, A, gang,
-                                             (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
-        } else {
-            HIP_TRY(hx::launch_scan_agg(s, A, 0));
+            hipError_t ge2 = hx::launch_scan_agg_gang(
+                s, A, gang, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0);
+            if (ge2 != hipSuccess) use_gang = false;  // e.g. LDS size rejected
         }
+        if (!use_gang) HIP_TRY(hx::launch_scan_agg(s, A, 0));
         HIP_TRY(hipEventRecord(e1, s));
         HIP_TRY(hipStreamSynchronize(s));
         float ms = 0;

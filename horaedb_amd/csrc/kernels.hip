@@ -291,7 +291,7 @@ struct GangParams {
     uint32_t has_mm;    // min/max tracked
 };
 
-extern "C" __global__ void __launch_bounds__(512)
+extern "C" __global__ void __launch_bounds__(1024)
 k_scan_agg_gang(GangParams G) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const AggParams& P = G.P;
@@ -335,7 +335,7 @@ k_scan_agg_gang(GangParams G) {
                 // LDS hash insert-or-update; overflow spills to global
                 uint32_t i = (uint32_t)mix64(s) & (ne - 1);
                 bool done = false;
-                for (int probes = 0; probes < 24; probes++) {
+                for (int probes = 0; probes < 8; probes++) {
                     uint64_t k = lkey[i];
                     if (k == KEY_EMPTY) {
                         uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, s);
@@ -705,10 +705,12 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     G.gang_size = gang_size;
     G.n_gangs = (p.n_rgs + gang_size - 1) / gang_size;
     G.has_mm = minmax ? 1u : 0u;
-    G.ne = minmax ? 2048u : 4096u;
+    // LDS table must hold the distinct keys of one aligned series window
+    // (~rows_per_rg / points_per_series); 8192 x 20B = the full 160 KiB LDS
+    G.ne = minmax ? 4096u : 8192u;
     size_t lds = (size_t)G.ne * (minmax ? 36 : 20);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
-    hipLaunchKernelGGL(k_scan_agg_gang, dim3(grid), dim3(512), lds, s, G);
+    hipLaunchKernelGGL(k_scan_agg_gang, dim3(grid), dim3(1024), lds, s, G);
     return hipGetLastError();
 }
 
