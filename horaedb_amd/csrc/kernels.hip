@@ -326,35 +326,90 @@ k_scan_agg_gang(GangParams G) {
             const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
             const SstDev sst = P.ssts[rg.sst_id];
             const uint32_t n = rg.n_rows;
-            for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
-                const int64_t t = T[r];
-                const uint64_t s = S[r];
-                if (!row_alive(P, rg, sst, S, T, r, n, s, t)) continue;
-                const double v = V[r];
-                my_matched++;
-                // LDS hash insert-or-update; overflow spills to global
-                uint32_t i = (uint32_t)mix64(s) & (ne - 1);
-                bool done = false;
-                for (int probes = 0; probes < 8; probes++) {
-                    uint64_t k = lkey[i];
-                    if (k == KEY_EMPTY) {
-                        uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, s);
-                        k = (old == KEY_EMPTY) ? s : old;
+            const int lane = threadIdx.x & 63;
+            // 4 rows per thread per iteration: 32B vector loads (coalesced in
+            // 32B chunks) and 4 independent LDS-update chains per thread —
+            // the latency-hiding ILP the 1-block/CU occupancy cannot provide
+            // (SQ_WAIT_ANY was 88% of wave cycles with 1 row per thread).
+            const uint32_t chunk = blockDim.x * 4;
+            for (uint32_t cbase = 0; cbase < n; cbase += chunk) {
+                const uint32_t r0 = cbase + threadIdx.x * 4;
+                uint64_t s4[4];
+                int64_t t4[4];
+                double v4[4];
+                const uint32_t avail = (r0 < n) ? min(4u, n - r0) : 0u;
+                if (avail == 4) {
+                    const ulonglong4 sv = *(const ulonglong4*)(S + r0);
+                    const longlong4 tv = *(const longlong4*)(T + r0);
+                    const double4 vv = *(const double4*)(V + r0);
+                    s4[0] = sv.x; s4[1] = sv.y; s4[2] = sv.z; s4[3] = sv.w;
+                    t4[0] = tv.x; t4[1] = tv.y; t4[2] = tv.z; t4[3] = tv.w;
+                    v4[0] = vv.x; v4[1] = vv.y; v4[2] = vv.z; v4[3] = vv.w;
+                } else {
+                    for (uint32_t k = 0; k < 4; k++) {
+                        const uint32_t r = r0 + k;
+                        s4[k] = (k < avail) ? S[r] : KEY_EMPTY;
+                        t4[k] = (k < avail) ? T[r] : 0;
+                        v4[k] = (k < avail) ? V[r] : 0.0;
                     }
-                    if (k == s) {
-                        atomicAdd(&lsum[i], v);
-                        atomicAdd(&lcnt[i], 1u);
-                        if (G.has_mm) {
-                            atomicMin(&lmin[i], f64_ordered(v));
-                            atomicMax(&lmax[i], f64_ordered(v));
-                        }
-                        done = true;
-                        break;
-                    }
-                    i = (i + 1) & (ne - 1);
                 }
-                if (!done)  // LDS table full: direct global update
-                    agg_update(P, s, 0, v, 1ull, v, v);
+                // successor of this thread's LAST row = next thread's first
+                const uint64_t s_next = __shfl_down(s4[0], 1, 64);
+                const int64_t t_next = __shfl_down(t4[0], 1, 64);
+                for (uint32_t k = 0; k < avail; k++) {
+                    const uint32_t r = r0 + k;
+                    const uint64_t sv = s4[k];
+                    const int64_t tv = t4[k];
+                    bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
+                    if (alive && P.use_sset) alive = sset_has(P, sv);
+                    if (alive) {
+                        bool dup = false;
+                        if (k < 3 && k + 1 < avail) {
+                            dup = (s4[k + 1] == sv) & (t4[k + 1] == tv);
+                        } else if (r + 1 < n) {
+                            if (k == 3 && lane < 63 && r0 + 4 < n) {
+                                dup = (s_next == sv) & (t_next == tv);
+                            } else {
+                                dup = (S[r + 1] == sv) & (T[r + 1] == tv);
+                            }
+                        } else if (rg.next_rg >= 0) {
+                            const RgDesc nx = P.rgs[rg.next_rg];
+                            uint64_t s2 = *(const uint64_t*)hx_ptr(
+                                P.blob, P.dec, nx.series_off);
+                            int64_t t2 = *(const int64_t*)hx_ptr(
+                                P.blob, P.dec, nx.ts_off);
+                            dup = (s2 == sv) & (t2 == tv);
+                        }
+                        if (!dup && sst.cluster >= 0)
+                            dup = shadowed(P, sst, sv, tv);
+                        alive = !dup;
+                    }
+                    if (!alive) continue;
+                    const double v = v4[k];
+                    my_matched++;
+                    uint32_t i = (uint32_t)mix64(sv) & (ne - 1);
+                    bool done = false;
+                    for (int probes = 0; probes < 8; probes++) {
+                        uint64_t kk = lkey[i];
+                        if (kk == KEY_EMPTY) {
+                            uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, sv);
+                            kk = (old == KEY_EMPTY) ? sv : old;
+                        }
+                        if (kk == sv) {
+                            atomicAdd(&lsum[i], v);
+                            atomicAdd(&lcnt[i], 1u);
+                            if (G.has_mm) {
+                                atomicMin(&lmin[i], f64_ordered(v));
+                                atomicMax(&lmax[i], f64_ordered(v));
+                            }
+                            done = true;
+                            break;
+                        }
+                        i = (i + 1) & (ne - 1);
+                    }
+                    if (!done)  // LDS table full: direct global update
+                        agg_update(P, sv, 0, v, 1ull, v, v);
+                }
             }
         }
         __syncthreads();
