@@ -558,33 +558,69 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     }
     plan.dec_bytes = dec_off;
 
-    // ---- grid order: interleave row groups by ordinal across SSTs --------
-    // Rows are PK-sorted within each SST, so the k-th row group of every SST
-    // covers roughly the same series range. Interleaving makes concurrently
-    // resident workgroups touch a narrow series window, keeping the group
-    // table's hot lines in L2/L3 instead of HBM (speed only — dispatch
-    // order is never relied on for correctness).
+    // ---- slice + align row groups across SSTs ----------------------------
+    // SSTs may carry different rows-per-series (time windows of 1 vs 2
+    // points), so the k-th row group of two SSTs can cover very different
+    // series ranges. For the gang kernel's LDS table (and the group table's
+    // L2/L3 locality) the unit list is (a) split into ~half-row-group slices
+    // so one unit's distinct-series count stays under the LDS table size,
+    // and (b) ordered by FRACTIONAL row position within its SST — units at
+    // the same fraction cover the same series quantile of the shared series
+    // universe. Speed-only: dispatch order is never relied on for
+    // correctness; dedup successor links (next_rg) are remapped.
     {
-        const size_t nrg = plan.rgs.size();
-        std::vector<uint32_t> ordinal(nrg);
-        {
-            size_t ri = 0;
-            for (StagedSst* ss : members) {
-                for (size_t k = 0; k < ss->rg_idx.size(); k++, ri++)
-                    ordinal[ri] = (uint32_t)k;
+        uint32_t split = 2;
+        if (const char* se = getenv("HX_RG_SPLIT"))
+            split = (uint32_t)strtoul(se, nullptr, 10);
+        if (split < 1) split = 1;
+        const size_t n_old = plan.rgs.size();
+        std::vector<hx::RgDesc> sliced;
+        std::vector<int32_t> new_first(n_old);
+        std::vector<int32_t> last_slice(n_old);
+        for (size_t i = 0; i < n_old; i++) {
+            const hx::RgDesc& rd = plan.rgs[i];
+            const uint32_t ssize =
+                std::max<uint32_t>(1024, (rd.n_rows + split - 1) / split);
+            new_first[i] = (int32_t)sliced.size();
+            for (uint32_t start = 0; start < rd.n_rows; start += ssize) {
+                hx::RgDesc sl = rd;
+                sl.series_off = rd.series_off + uint64_t(start) * 8;
+                sl.ts_off = rd.ts_off + uint64_t(start) * 8;
+                sl.val_off = rd.val_off + uint64_t(start) * 8;
+                sl.n_rows = std::min(ssize, rd.n_rows - start);
+                sl.row_base = rd.row_base + start;
+                sl.next_rg = (start + ssize < rd.n_rows)
+                                 ? (int32_t)sliced.size() + 1
+                                 : rd.next_rg;  // old id; remapped below
+                last_slice[i] = (int32_t)sliced.size();
+                sliced.push_back(sl);
             }
         }
-        std::vector<uint32_t> order(nrg);
-        for (size_t i = 0; i < nrg; i++) order[i] = (uint32_t)i;
+        for (size_t i = 0; i < n_old; i++) {
+            int32_t nx = plan.rgs[i].next_rg;
+            sliced[last_slice[i]].next_rg = nx >= 0 ? new_first[nx] : -1;
+        }
+        // fractional position of each slice within its SST's staged rows
+        std::vector<int64_t> sst_rows(plan.ssts.size(), 0);
+        for (const auto& sl : sliced) {
+            int64_t end = sl.row_base + sl.n_rows;
+            if (end > sst_rows[sl.sst_id]) sst_rows[sl.sst_id] = end;
+        }
+        std::vector<uint32_t> order(sliced.size());
+        for (size_t i = 0; i < sliced.size(); i++) order[i] = (uint32_t)i;
+        std::vector<double> frac(sliced.size());
+        for (size_t i = 0; i < sliced.size(); i++)
+            frac[i] = sst_rows[sliced[i].sst_id]
+                          ? double(sliced[i].row_base) /
+                                double(sst_rows[sliced[i].sst_id])
+                          : 0.0;
         std::stable_sort(order.begin(), order.end(),
-                         [&](uint32_t a, uint32_t b) {
-                             return ordinal[a] < ordinal[b];
-                         });
-        std::vector<int32_t> inv(nrg);
-        for (size_t i = 0; i < nrg; i++) inv[order[i]] = (int32_t)i;
-        std::vector<hx::RgDesc> reordered(nrg);
-        for (size_t i = 0; i < nrg; i++) {
-            reordered[i] = plan.rgs[order[i]];
+                         [&](uint32_t a, uint32_t b) { return frac[a] < frac[b]; });
+        std::vector<int32_t> inv(sliced.size());
+        for (size_t i = 0; i < sliced.size(); i++) inv[order[i]] = (int32_t)i;
+        std::vector<hx::RgDesc> reordered(sliced.size());
+        for (size_t i = 0; i < sliced.size(); i++) {
+            reordered[i] = sliced[order[i]];
             if (reordered[i].next_rg >= 0)
                 reordered[i].next_rg = inv[reordered[i].next_rg];
         }
