@@ -569,7 +569,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     // universe. Speed-only: dispatch order is never relied on for
     // correctness; dedup successor links (next_rg) are remapped.
     {
-        uint32_t split = 2;
+        uint32_t split = 4;   // 2048-row units: distinct series <= LDS table/2
         if (const char* se = getenv("HX_RG_SPLIT"))
             split = (uint32_t)strtoul(se, nullptr, 10);
         if (split < 1) split = 1;
@@ -1025,6 +1025,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.bucket_ms = bucket ? agg->bucket_ms : 0;
         A.ops = ops;
         A.key_claim = key_claim;
+        A.skip = getenv("HX_SKIP") ? atoi(getenv("HX_SKIP")) : 0;
         A.table = {plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
                    plan.t_cnt, plan.t_min, plan.t_max, slots - 1};
         A.fill = plan.d_counters + 0;
@@ -1037,7 +1038,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipEventRecord(e0, s));
         bool use_gang = !bucket && !getenv("HX_NO_GANG");
         if (use_gang) {
-            uint32_t gang = 32;
+            uint32_t gang = 64;
             if (const char* ge = getenv("HX_GANG"))
                 gang = (uint32_t)strtoul(ge, nullptr, 10);
             hipError_t ge2 = hx::launch_scan_agg_gang(

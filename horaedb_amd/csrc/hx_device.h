@@ -77,6 +77,7 @@ struct AggParams {
     int64_t bucket_ms;             // 0 = group by series only
     uint32_t ops;                  // HX_AGG_* mask
     int32_t key_claim;             // 1 = one-CAS key-claim mode (no state word)
+    int32_t skip;                  // debug bisect: 1 skip table, 2 also skip dedup
     AggTable table;
     unsigned long long* fill;      // claimed slots
     unsigned long long* overflow;  // !=0 => rerun with a larger table

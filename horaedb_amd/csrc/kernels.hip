@@ -291,6 +291,35 @@ struct GangParams {
     uint32_t has_mm;    // min/max tracked
 };
 
+__device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
+                                           double* lsum, unsigned int* lcnt,
+                                           unsigned long long* lmin,
+                                           unsigned long long* lmax,
+                                           uint32_t ne, uint64_t sv, double v,
+                                           uint32_t cnt, double mn, double mx) {
+    uint32_t i = (uint32_t)mix64(sv) & (ne - 1);
+#pragma unroll 1
+    for (int probes = 0; probes < 8; probes++) {
+        uint64_t kk = lkey[i];
+        if (kk == KEY_EMPTY) {
+            uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, sv);
+            kk = (old == KEY_EMPTY) ? sv : old;
+        }
+        if (kk == sv) {
+            atomicAdd(&lsum[i], v);
+            atomicAdd(&lcnt[i], cnt);
+            if (G.has_mm) {
+                atomicMin(&lmin[i], f64_ordered(mn));
+                atomicMax(&lmax[i], f64_ordered(mx));
+            }
+            return;
+        }
+        i = (i + 1) & (ne - 1);
+    }
+    // LDS table full: direct global update
+    agg_update(G.P, sv, 0, v, cnt, mn, mx);
+}
+
 extern "C" __global__ void __launch_bounds__(1024)
 k_scan_agg_gang(GangParams G) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -356,13 +385,18 @@ k_scan_agg_gang(GangParams G) {
                 // successor of this thread's LAST row = next thread's first
                 const uint64_t s_next = __shfl_down(s4[0], 1, 64);
                 const int64_t t_next = __shfl_down(t4[0], 1, 64);
+                bool have_run = false;
+                uint64_t run_key = 0;
+                double run_sum = 0, run_min = 0, run_max = 0;
+                uint32_t run_cnt = 0;
+#pragma unroll 1
                 for (uint32_t k = 0; k < avail; k++) {
                     const uint32_t r = r0 + k;
                     const uint64_t sv = s4[k];
                     const int64_t tv = t4[k];
                     bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
                     if (alive && P.use_sset) alive = sset_has(P, sv);
-                    if (alive) {
+                    if (alive && P.skip < 2) {
                         bool dup = false;
                         if (k < 3 && k + 1 < avail) {
                             dup = (s4[k + 1] == sv) & (t4[k + 1] == tv);
@@ -387,29 +421,30 @@ k_scan_agg_gang(GangParams G) {
                     if (!alive) continue;
                     const double v = v4[k];
                     my_matched++;
-                    uint32_t i = (uint32_t)mix64(sv) & (ne - 1);
-                    bool done = false;
-                    for (int probes = 0; probes < 8; probes++) {
-                        uint64_t kk = lkey[i];
-                        if (kk == KEY_EMPTY) {
-                            uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, sv);
-                            kk = (old == KEY_EMPTY) ? sv : old;
-                        }
-                        if (kk == sv) {
-                            atomicAdd(&lsum[i], v);
-                            atomicAdd(&lcnt[i], 1u);
-                            if (G.has_mm) {
-                                atomicMin(&lmin[i], f64_ordered(v));
-                                atomicMax(&lmax[i], f64_ordered(v));
-                            }
-                            done = true;
-                            break;
-                        }
-                        i = (i + 1) & (ne - 1);
+                    if (P.skip) continue;
+                    // combine consecutive equal-series rows of this batch in
+                    // registers (rows are PK-sorted): fewer LDS probe chains
+                    if (have_run && run_key == sv) {
+                        run_sum += v;
+                        run_cnt++;
+                        run_min = fmin(run_min, v);
+                        run_max = fmax(run_max, v);
+                    } else {
+                        if (have_run)
+                            lds_update(G, lkey, lsum, lcnt, lmin, lmax, ne,
+                                       run_key, run_sum, run_cnt, run_min,
+                                       run_max);
+                        have_run = true;
+                        run_key = sv;
+                        run_sum = v;
+                        run_cnt = 1;
+                        run_min = v;
+                        run_max = v;
                     }
-                    if (!done)  // LDS table full: direct global update
-                        agg_update(P, sv, 0, v, 1ull, v, v);
                 }
+                if (have_run)
+                    lds_update(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
+                               run_sum, run_cnt, run_min, run_max);
             }
         }
         __syncthreads();
@@ -760,9 +795,10 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     G.gang_size = gang_size;
     G.n_gangs = (p.n_rgs + gang_size - 1) / gang_size;
     G.has_mm = minmax ? 1u : 0u;
-    // LDS table must hold the distinct keys of one aligned series window
-    // (~rows_per_rg / points_per_series); 8192 x 20B = the full 160 KiB LDS
-    G.ne = minmax ? 4096u : 8192u;
+    // LDS table holds the distinct keys of one aligned unit window (host
+    // slices units to <= ne/2 distinct); 4096 x 20B = 80 KiB => 2 blocks
+    // (32 waves) per CU for latency hiding.
+    G.ne = minmax ? 2048u : 4096u;
     size_t lds = (size_t)G.ne * (minmax ? 36 : 20);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
     hipLaunchKernelGGL(k_scan_agg_gang, dim3(grid), dim3(1024), lds, s, G);
