@@ -291,6 +291,7 @@ struct GangParams {
     uint32_t has_mm;    // min/max tracked
 };
 
+template <bool MM>
 __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
                                            double* lsum, unsigned int* lcnt,
                                            unsigned long long* lmin,
@@ -308,7 +309,7 @@ __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
         if (kk == sv) {
             atomicAdd(&lsum[i], v);
             atomicAdd(&lcnt[i], cnt);
-            if (G.has_mm) {
+            if (MM) {
                 atomicMin(&lmin[i], f64_ordered(mn));
                 atomicMax(&lmax[i], f64_ordered(mx));
             }
@@ -320,7 +321,8 @@ __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
     agg_update(G.P, sv, 0, v, cnt, mn, mx);
 }
 
-extern "C" __global__ void __launch_bounds__(1024)
+template <bool MM>
+__global__ void __launch_bounds__(1024, 8)
 k_scan_agg_gang(GangParams G) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const AggParams& P = G.P;
@@ -340,7 +342,7 @@ k_scan_agg_gang(GangParams G) {
             lkey[i] = KEY_EMPTY;
             lsum[i] = 0.0;
             lcnt[i] = 0u;
-            if (G.has_mm) {
+            if (MM) {
                 lmin[i] = ~0ull;
                 lmax[i] = 0ull;
             }
@@ -431,7 +433,7 @@ k_scan_agg_gang(GangParams G) {
                         run_max = fmax(run_max, v);
                     } else {
                         if (have_run)
-                            lds_update(G, lkey, lsum, lcnt, lmin, lmax, ne,
+                            lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne,
                                        run_key, run_sum, run_cnt, run_min,
                                        run_max);
                         have_run = true;
@@ -443,7 +445,7 @@ k_scan_agg_gang(GangParams G) {
                     }
                 }
                 if (have_run)
-                    lds_update(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
+                    lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
                                run_sum, run_cnt, run_min, run_max);
             }
         }
@@ -451,8 +453,8 @@ k_scan_agg_gang(GangParams G) {
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             if (lkey[i] == KEY_EMPTY) continue;
             agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
-                       G.has_mm ? ordered_f64(lmin[i]) : 0.0,
-                       G.has_mm ? ordered_f64(lmax[i]) : 0.0);
+                       MM ? ordered_f64(lmin[i]) : 0.0,
+                       MM ? ordered_f64(lmax[i]) : 0.0);
         }
         __syncthreads();
     }
@@ -801,7 +803,10 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     G.ne = minmax ? 2048u : 4096u;
     size_t lds = (size_t)G.ne * (minmax ? 36 : 20);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
-    hipLaunchKernelGGL(k_scan_agg_gang, dim3(grid), dim3(1024), lds, s, G);
+    if (minmax)
+        hipLaunchKernelGGL(k_scan_agg_gang<true>, dim3(grid), dim3(1024), lds, s, G);
+    else
+        hipLaunchKernelGGL(k_scan_agg_gang<false>, dim3(grid), dim3(1024), lds, s, G);
     return hipGetLastError();
 }
 
