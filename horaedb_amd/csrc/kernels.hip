@@ -322,7 +322,7 @@ __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
 }
 
 template <bool MM>
-__global__ void __launch_bounds__(1024, 8)
+__global__ void __launch_bounds__(1024)
 k_scan_agg_gang(GangParams G) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const AggParams& P = G.P;
@@ -751,6 +751,7 @@ k_decode_delta_i64(const uint8_t* blob, uint8_t* dec,
 // ---------------------------------------------------------------------------
 #include "hx_kernels.h"
 #include <cstring>
+#include <cstdlib>
 #include <rocprim/device/device_radix_sort.hpp>
 
 namespace hx {
@@ -801,6 +802,10 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     // slices units to <= ne/2 distinct); 4096 x 20B = 80 KiB => 2 blocks
     // (32 waves) per CU for latency hiding.
     G.ne = minmax ? 2048u : 4096u;
+    if (const char* nee = getenv("HX_NE")) {
+        uint32_t ne = (uint32_t)strtoul(nee, nullptr, 10);
+        if (ne >= 1024 && ne <= 8192 && !(ne & (ne - 1))) G.ne = ne;
+    }
     size_t lds = (size_t)G.ne * (minmax ? 36 : 20);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
     if (minmax)
