@@ -324,20 +324,35 @@ __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
 template <bool MM>
 __global__ void __launch_bounds__(1024)
 k_scan_agg_gang(GangParams G) {
+    // Transposed gang walk: a thread owns TWO row positions of the aligned
+    // series window and visits them across every unit (row-group slice) of
+    // the gang. Same-size SSTs slice into EXACTLY aligned units, so the
+    // series at a fixed row position is constant across them — the thread
+    // accumulates in registers and touches the LDS table only when the key
+    // changes. Loads stay coalesced (lanes = consecutive rows) and the
+    // unit-loop iterations are independent (memory-level parallelism).
+    // Alignment is a speed matter only: register combining merges equal
+    // ADJACENT keys, which is correct for any data.
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const AggParams& P = G.P;
     const uint32_t ne = G.ne;
-    uint64_t* lkey = (uint64_t*)smem;
-    double* lsum = (double*)(smem + (size_t)ne * 8);
+    RgDesc* ldesc = (RgDesc*)smem;
+    const size_t desc_bytes = (size_t)G.gang_size * sizeof(RgDesc);
+    uint64_t* lkey = (uint64_t*)(smem + desc_bytes);
+    double* lsum = (double*)(smem + desc_bytes + (size_t)ne * 8);
     unsigned long long* lmin =
-        (unsigned long long*)(smem + (size_t)ne * 16);
+        (unsigned long long*)(smem + desc_bytes + (size_t)ne * 16);
     unsigned long long* lmax =
-        (unsigned long long*)(smem + (size_t)ne * (G.has_mm ? 24 : 16));
+        (unsigned long long*)(smem + desc_bytes + (size_t)ne * (MM ? 24 : 16));
     unsigned int* lcnt =
-        (unsigned int*)(smem + (size_t)ne * (G.has_mm ? 32 : 16));
+        (unsigned int*)(smem + desc_bytes + (size_t)ne * (MM ? 32 : 16));
 
+    const int lane = threadIdx.x & 63;
     unsigned long long my_matched = 0;
     for (uint32_t gang = blockIdx.x; gang < G.n_gangs; gang += gridDim.x) {
+        const uint32_t rg0 = gang * G.gang_size;
+        const uint32_t rg_end = min(rg0 + G.gang_size, P.n_rgs);
+        const uint32_t nu = rg_end - rg0;
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             lkey[i] = KEY_EMPTY;
             lsum[i] = 0.0;
@@ -347,63 +362,63 @@ k_scan_agg_gang(GangParams G) {
                 lmax[i] = 0ull;
             }
         }
+        // preload unit descriptors (one dynamic-LDS object only — G17)
+        for (uint32_t u = threadIdx.x; u < nu; u += blockDim.x)
+            ldesc[u] = P.rgs[rg0 + u];
         __syncthreads();
-        const uint32_t rg_end = min(gang * G.gang_size + G.gang_size, P.n_rgs);
-        for (uint32_t rgi = gang * G.gang_size; rgi < rg_end; rgi++) {
-            const RgDesc rg = P.rgs[rgi];
-            const uint64_t* S =
-                (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
-            const int64_t* T = (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
-            const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
-            const SstDev sst = P.ssts[rg.sst_id];
-            const uint32_t n = rg.n_rows;
-            const int lane = threadIdx.x & 63;
-            // 4 rows per thread per iteration: 32B vector loads (coalesced in
-            // 32B chunks) and 4 independent LDS-update chains per thread —
-            // the latency-hiding ILP the 1-block/CU occupancy cannot provide
-            // (SQ_WAIT_ANY was 88% of wave cycles with 1 row per thread).
-            const uint32_t chunk = blockDim.x * 4;
-            for (uint32_t cbase = 0; cbase < n; cbase += chunk) {
-                const uint32_t r0 = cbase + threadIdx.x * 4;
-                uint64_t s4[4];
-                int64_t t4[4];
-                double v4[4];
-                const uint32_t avail = (r0 < n) ? min(4u, n - r0) : 0u;
-                if (avail == 4) {
-                    const ulonglong4 sv = *(const ulonglong4*)(S + r0);
-                    const longlong4 tv = *(const longlong4*)(T + r0);
-                    const double4 vv = *(const double4*)(V + r0);
-                    s4[0] = sv.x; s4[1] = sv.y; s4[2] = sv.z; s4[3] = sv.w;
-                    t4[0] = tv.x; t4[1] = tv.y; t4[2] = tv.z; t4[3] = tv.w;
-                    v4[0] = vv.x; v4[1] = vv.y; v4[2] = vv.z; v4[3] = vv.w;
-                } else {
-                    for (uint32_t k = 0; k < 4; k++) {
-                        const uint32_t r = r0 + k;
-                        s4[k] = (k < avail) ? S[r] : KEY_EMPTY;
-                        t4[k] = (k < avail) ? T[r] : 0;
-                        v4[k] = (k < avail) ? V[r] : 0.0;
-                    }
-                }
-                // successor of this thread's LAST row = next thread's first
-                const uint64_t s_next = __shfl_down(s4[0], 1, 64);
-                const int64_t t_next = __shfl_down(t4[0], 1, 64);
-                bool have_run = false;
-                uint64_t run_key = 0;
-                double run_sum = 0, run_min = 0, run_max = 0;
-                uint32_t run_cnt = 0;
+        uint32_t max_n = 0;
+        for (uint32_t u = 0; u < nu; u++) max_n = max(max_n, ldesc[u].n_rows);
+
+        bool have_run = false;
+        uint64_t run_key = 0;
+        double run_sum = 0, run_min = 0, run_max = 0;
+        uint32_t run_cnt = 0;
+        for (uint32_t roff = 0; roff < max_n; roff += blockDim.x * 2) {
+            const uint32_t r0 = roff + threadIdx.x * 2;
 #pragma unroll 1
-                for (uint32_t k = 0; k < avail; k++) {
-                    const uint32_t r = r0 + k;
-                    const uint64_t sv = s4[k];
-                    const int64_t tv = t4[k];
+            for (uint32_t u = 0; u < nu; u++) {
+                const RgDesc rg = ldesc[u];
+                const uint32_t n = rg.n_rows;
+                if (r0 >= n) continue;   // wave-uniform for uniform units
+                const uint64_t* S =
+                    (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+                const int64_t* T =
+                    (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+                const double* V =
+                    (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+                uint64_t s0, s1;
+                int64_t t0, t1;
+                double v0, v1;
+                const bool has2 = r0 + 2 <= n;
+                if (has2) {
+                    const ulonglong2 sv = *(const ulonglong2*)(S + r0);
+                    const longlong2 tv = *(const longlong2*)(T + r0);
+                    const double2 vv = *(const double2*)(V + r0);
+                    s0 = sv.x; s1 = sv.y;
+                    t0 = tv.x; t1 = tv.y;
+                    v0 = vv.x; v1 = vv.y;
+                } else {
+                    s0 = S[r0]; t0 = T[r0]; v0 = V[r0];
+                    s1 = KEY_EMPTY; t1 = 0; v1 = 0;
+                }
+                // successor of this thread's second row = next lane's first
+                const uint64_t s_next = __shfl_down(s0, 1, 64);
+                const int64_t t_next = __shfl_down(t0, 1, 64);
+                const SstDev sst = P.ssts[rg.sst_id];
+#pragma unroll 1
+                for (uint32_t kk = 0; kk < 2; kk++) {
+                    const uint32_t r = r0 + kk;
+                    if (r >= n) break;
+                    const uint64_t sv = kk ? s1 : s0;
+                    const int64_t tv = kk ? t1 : t0;
                     bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
                     if (alive && P.use_sset) alive = sset_has(P, sv);
                     if (alive && P.skip < 2) {
                         bool dup = false;
-                        if (k < 3 && k + 1 < avail) {
-                            dup = (s4[k + 1] == sv) & (t4[k + 1] == tv);
+                        if (kk == 0 && has2) {
+                            dup = (s1 == sv) & (t1 == tv);
                         } else if (r + 1 < n) {
-                            if (k == 3 && lane < 63 && r0 + 4 < n) {
+                            if (kk == 1 && lane < 63 && r0 + 2 < n) {
                                 dup = (s_next == sv) & (t_next == tv);
                             } else {
                                 dup = (S[r + 1] == sv) & (T[r + 1] == tv);
@@ -421,21 +436,21 @@ k_scan_agg_gang(GangParams G) {
                         alive = !dup;
                     }
                     if (!alive) continue;
-                    const double v = v4[k];
+                    const double v = kk ? v1 : v0;
                     my_matched++;
                     if (P.skip) continue;
-                    // combine consecutive equal-series rows of this batch in
-                    // registers (rows are PK-sorted): fewer LDS probe chains
                     if (have_run && run_key == sv) {
                         run_sum += v;
                         run_cnt++;
-                        run_min = fmin(run_min, v);
-                        run_max = fmax(run_max, v);
+                        if (MM) {
+                            run_min = fmin(run_min, v);
+                            run_max = fmax(run_max, v);
+                        }
                     } else {
                         if (have_run)
                             lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne,
-                                       run_key, run_sum, run_cnt, run_min,
-                                       run_max);
+                                           run_key, run_sum, run_cnt, run_min,
+                                           run_max);
                         have_run = true;
                         run_key = sv;
                         run_sum = v;
@@ -444,9 +459,11 @@ k_scan_agg_gang(GangParams G) {
                         run_max = v;
                     }
                 }
-                if (have_run)
-                    lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
+            }
+            if (have_run) {
+                lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
                                run_sum, run_cnt, run_min, run_max);
+                have_run = false;
             }
         }
         __syncthreads();
@@ -806,7 +823,8 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
         uint32_t ne = (uint32_t)strtoul(nee, nullptr, 10);
         if (ne >= 1024 && ne <= 8192 && !(ne & (ne - 1))) G.ne = ne;
     }
-    size_t lds = (size_t)G.ne * (minmax ? 36 : 20);
+    size_t lds = (size_t)G.ne * (minmax ? 36 : 20) +
+                 (size_t)G.gang_size * sizeof(RgDesc);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
     if (minmax)
         hipLaunchKernelGGL(k_scan_agg_gang<true>, dim3(grid), dim3(1024), lds, s, G);
