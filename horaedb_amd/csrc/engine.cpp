@@ -614,8 +614,17 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                           ? double(sliced[i].row_base) /
                                 double(sst_rows[sliced[i].sst_id])
                           : 0.0;
+        // primary key: the SST's staged size class — equal-size SSTs slice
+        // into EXACTLY aligned units, so a gang drawn from one class gives
+        // the transposed walk a constant per-thread series key; secondary:
+        // fractional position (the aligned window)
         std::stable_sort(order.begin(), order.end(),
-                         [&](uint32_t a, uint32_t b) { return frac[a] < frac[b]; });
+                         [&](uint32_t a, uint32_t b) {
+                             int64_t ra = sst_rows[sliced[a].sst_id];
+                             int64_t rb = sst_rows[sliced[b].sst_id];
+                             if (ra != rb) return ra < rb;
+                             return frac[a] < frac[b];
+                         });
         std::vector<int32_t> inv(sliced.size());
         for (size_t i = 0; i < sliced.size(); i++) inv[order[i]] = (int32_t)i;
         std::vector<hx::RgDesc> reordered(sliced.size());
