@@ -17,6 +17,7 @@
 #include <dirent.h>
 #include <fcntl.h>
 #include <functional>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -1047,7 +1048,25 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipEventRecord(e0, s));
         bool use_gang = !bucket && !getenv("HX_NO_GANG");
         if (use_gang) {
-            uint32_t gang = 64;
+            // gang = number of SSTs in the dominant size class: one gang
+            // then covers EXACTLY one aligned series window (the transposed
+            // walk's register-combining + LDS-table-capacity contract)
+            uint32_t gang = 1;
+            {
+                std::map<int64_t, int> class_count;
+                for (const auto& sd : plan.ssts) (void)sd;
+                std::map<int64_t, int64_t> class_rows;
+                std::vector<int64_t> per_sst(plan.ssts.size(), 0);
+                for (const auto& rd : plan.rgs) {
+                    int64_t end = rd.row_base + rd.n_rows;
+                    if (end > per_sst[rd.sst_id]) per_sst[rd.sst_id] = end;
+                }
+                for (int64_t rows : per_sst) class_count[rows]++;
+                int best = 0;
+                for (auto& [rows, cnt] : class_count)
+                    if (cnt > best) best = cnt;
+                gang = best > 0 ? (uint32_t)best : 1;
+            }
             if (const char* ge = getenv("HX_GANG"))
                 gang = (uint32_t)strtoul(ge, nullptr, 10);
             hipError_t ge2 = hx::launch_scan_agg_gang(
