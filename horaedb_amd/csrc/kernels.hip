@@ -373,98 +373,79 @@ k_scan_agg_gang(GangParams G) {
         uint64_t run_key = 0;
         double run_sum = 0, run_min = 0, run_max = 0;
         uint32_t run_cnt = 0;
-        for (uint32_t roff = 0; roff < max_n; roff += blockDim.x * 2) {
-            const uint32_t r0 = roff + threadIdx.x * 2;
+        for (uint32_t roff = 0; roff < max_n; roff += blockDim.x) {
+            const uint32_t r = roff + threadIdx.x;
 #pragma unroll 1
             for (uint32_t u = 0; u < nu; u++) {
                 const RgDesc rg = ldesc[u];
                 const uint32_t n = rg.n_rows;
-                if (r0 >= n) continue;   // wave-uniform for uniform units
+                if (__all(r >= n)) continue;
                 const uint64_t* S =
                     (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
                 const int64_t* T =
                     (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
                 const double* V =
                     (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
-                uint64_t s0, s1;
-                int64_t t0, t1;
-                double v0, v1;
-                const bool has2 = r0 + 2 <= n;
-                if (has2) {
-                    const ulonglong2 sv = *(const ulonglong2*)(S + r0);
-                    const longlong2 tv = *(const longlong2*)(T + r0);
-                    const double2 vv = *(const double2*)(V + r0);
-                    s0 = sv.x; s1 = sv.y;
-                    t0 = tv.x; t1 = tv.y;
-                    v0 = vv.x; v1 = vv.y;
+                const bool inb = r < n;
+                const uint64_t sv = inb ? S[r] : KEY_EMPTY;
+                const int64_t tv = inb ? T[r] : 0;
+                // successor row r+1 lives in the next lane (coalesced rows)
+                const uint64_t s_next = __shfl_down(sv, 1, 64);
+                const int64_t t_next = __shfl_down(tv, 1, 64);
+                if (!inb) continue;
+                bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
+                if (alive && P.use_sset) alive = sset_has(P, sv);
+                if (alive && P.skip < 2) {
+                    bool dup = false;
+                    if (r + 1 < n) {
+                        if (lane < 63) {
+                            dup = (s_next == sv) & (t_next == tv);
+                        } else {
+                            dup = (S[r + 1] == sv) & (T[r + 1] == tv);
+                        }
+                    } else if (rg.next_rg >= 0) {
+                        const RgDesc nx = P.rgs[rg.next_rg];
+                        uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
+                                                               nx.series_off);
+                        int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec,
+                                                             nx.ts_off);
+                        dup = (s2 == sv) & (t2 == tv);
+                    }
+                    if (!dup) {
+                        const SstDev sst = P.ssts[rg.sst_id];
+                        if (sst.cluster >= 0) dup = shadowed(P, sst, sv, tv);
+                    }
+                    alive = !dup;
+                }
+                if (!alive) continue;
+                const double v = V[r];
+                my_matched++;
+                if (P.skip) continue;
+                if (have_run && run_key == sv) {
+                    run_sum += v;
+                    run_cnt++;
+                    if (MM) {
+                        run_min = fmin(run_min, v);
+                        run_max = fmax(run_max, v);
+                    }
                 } else {
-                    s0 = S[r0]; t0 = T[r0]; v0 = V[r0];
-                    s1 = KEY_EMPTY; t1 = 0; v1 = 0;
-                }
-                // successor of this thread's second row = next lane's first
-                const uint64_t s_next = __shfl_down(s0, 1, 64);
-                const int64_t t_next = __shfl_down(t0, 1, 64);
-                const SstDev sst = P.ssts[rg.sst_id];
-#pragma unroll 1
-                for (uint32_t kk = 0; kk < 2; kk++) {
-                    const uint32_t r = r0 + kk;
-                    if (r >= n) break;
-                    const uint64_t sv = kk ? s1 : s0;
-                    const int64_t tv = kk ? t1 : t0;
-                    bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
-                    if (alive && P.use_sset) alive = sset_has(P, sv);
-                    if (alive && P.skip < 2) {
-                        bool dup = false;
-                        if (kk == 0 && has2) {
-                            dup = (s1 == sv) & (t1 == tv);
-                        } else if (r + 1 < n) {
-                            if (kk == 1 && lane < 63 && r0 + 2 < n) {
-                                dup = (s_next == sv) & (t_next == tv);
-                            } else {
-                                dup = (S[r + 1] == sv) & (T[r + 1] == tv);
-                            }
-                        } else if (rg.next_rg >= 0) {
-                            const RgDesc nx = P.rgs[rg.next_rg];
-                            uint64_t s2 = *(const uint64_t*)hx_ptr(
-                                P.blob, P.dec, nx.series_off);
-                            int64_t t2 = *(const int64_t*)hx_ptr(
-                                P.blob, P.dec, nx.ts_off);
-                            dup = (s2 == sv) & (t2 == tv);
-                        }
-                        if (!dup && sst.cluster >= 0)
-                            dup = shadowed(P, sst, sv, tv);
-                        alive = !dup;
-                    }
-                    if (!alive) continue;
-                    const double v = kk ? v1 : v0;
-                    my_matched++;
-                    if (P.skip) continue;
-                    if (have_run && run_key == sv) {
-                        run_sum += v;
-                        run_cnt++;
-                        if (MM) {
-                            run_min = fmin(run_min, v);
-                            run_max = fmax(run_max, v);
-                        }
-                    } else {
-                        if (have_run)
-                            lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne,
-                                           run_key, run_sum, run_cnt, run_min,
-                                           run_max);
-                        have_run = true;
-                        run_key = sv;
-                        run_sum = v;
-                        run_cnt = 1;
-                        run_min = v;
-                        run_max = v;
-                    }
+                    if (have_run)
+                        lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne,
+                                       run_key, run_sum, run_cnt, run_min,
+                                       run_max);
+                    have_run = true;
+                    run_key = sv;
+                    run_sum = v;
+                    run_cnt = 1;
+                    run_min = v;
+                    run_max = v;
                 }
             }
-            if (have_run) {
-                lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
-                               run_sum, run_cnt, run_min, run_max);
-                have_run = false;
-            }
+        }
+        if (have_run) {
+            lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
+                           run_sum, run_cnt, run_min, run_max);
+            have_run = false;
         }
         __syncthreads();
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
