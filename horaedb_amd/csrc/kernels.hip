@@ -349,6 +349,7 @@ k_scan_agg_gang(GangParams G) {
 
     const int lane = threadIdx.x & 63;
     unsigned long long my_matched = 0;
+    unsigned long long my_flushes = 0;
     for (uint32_t gang = blockIdx.x; gang < G.n_gangs; gang += gridDim.x) {
         const uint32_t rg0 = gang * G.gang_size;
         const uint32_t rg_end = min(rg0 + G.gang_size, P.n_rgs);
@@ -375,7 +376,7 @@ k_scan_agg_gang(GangParams G) {
         uint32_t run_cnt = 0;
         for (uint32_t roff = 0; roff < max_n; roff += blockDim.x) {
             const uint32_t r = roff + threadIdx.x;
-#pragma unroll 1
+#pragma unroll 2
             for (uint32_t u = 0; u < nu; u++) {
                 const RgDesc rg = ldesc[u];
                 const uint32_t n = rg.n_rows;
@@ -387,22 +388,20 @@ k_scan_agg_gang(GangParams G) {
                 const double* V =
                     (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
                 const bool inb = r < n;
+                // issue all three column loads together (one wait, not three
+                // serialized dependent waits per unit iteration)
                 const uint64_t sv = inb ? S[r] : KEY_EMPTY;
                 const int64_t tv = inb ? T[r] : 0;
-                // successor row r+1 lives in the next lane (coalesced rows)
-                const uint64_t s_next = __shfl_down(sv, 1, 64);
-                const int64_t t_next = __shfl_down(tv, 1, 64);
+                const double v = inb ? V[r] : 0.0;
                 if (!inb) continue;
                 bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
                 if (alive && P.use_sset) alive = sset_has(P, sv);
                 if (alive && P.skip < 2) {
                     bool dup = false;
                     if (r + 1 < n) {
-                        if (lane < 63) {
-                            dup = (s_next == sv) & (t_next == tv);
-                        } else {
-                            dup = (S[r + 1] == sv) & (T[r + 1] == tv);
-                        }
+                        // successor is the adjacent element (same cache line
+                        // as this lane group's loads; no cross-lane sync)
+                        dup = (S[r + 1] == sv) & (T[r + 1] == tv);
                     } else if (rg.next_rg >= 0) {
                         const RgDesc nx = P.rgs[rg.next_rg];
                         uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
@@ -418,9 +417,8 @@ k_scan_agg_gang(GangParams G) {
                     alive = !dup;
                 }
                 if (!alive) continue;
-                const double v = V[r];
                 my_matched++;
-                if (P.skip) continue;
+                if (P.skip == 1 || P.skip == 2) continue;
                 if (have_run && run_key == sv) {
                     run_sum += v;
                     run_cnt++;
@@ -429,10 +427,12 @@ k_scan_agg_gang(GangParams G) {
                         run_max = fmax(run_max, v);
                     }
                 } else {
-                    if (have_run)
-                        lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne,
-                                       run_key, run_sum, run_cnt, run_min,
-                                       run_max);
+                    if (have_run) {
+                        if (P.skip == 3) my_flushes++;   // bisect: count only
+                        else lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax,
+                                            ne, run_key, run_sum, run_cnt,
+                                            run_min, run_max);
+                    }
                     have_run = true;
                     run_key = sv;
                     run_sum = v;
@@ -456,6 +456,7 @@ k_scan_agg_gang(GangParams G) {
         }
         __syncthreads();
     }
+    if (P.skip == 3) my_matched = my_flushes;  // bisect: report flush count
     for (int off = 32; off > 0; off >>= 1)
         my_matched += __shfl_down(my_matched, off, 64);
     if ((threadIdx.x & 63) == 0 && my_matched)
