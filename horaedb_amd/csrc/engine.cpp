@@ -275,6 +275,8 @@ struct DevPlan {
     uint64_t sset_empty = ~0ull;
     uint32_t sset_mask = 0;
 
+    void* h_params = nullptr;   // pinned staging for kernel param structs
+    void* d_params = nullptr;
     bool decoded = false;       // delta/copy kernels already ran
     double decode_ms = 0;
 
@@ -847,6 +849,8 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
             if (plan.h_blob_pinned) hipHostFree(plan.h_blob);
             else free(plan.h_blob);
         }
+        if (plan.d_params) hipFree(plan.d_params);
+        if (plan.h_params) hipHostFree(plan.h_params);
         for (void* p : {(void*)plan.d_blob, (void*)plan.d_dec, (void*)plan.d_rgs,
                         (void*)plan.d_ssts, (void*)plan.d_clusters,
                         (void*)plan.d_members, (void*)plan.d_delta,
@@ -1069,8 +1073,16 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
             }
             if (const char* ge = getenv("HX_GANG"))
                 gang = (uint32_t)strtoul(ge, nullptr, 10);
+            if (!plan.d_params) {
+                HIP_TRY(hipMalloc(&plan.d_params, 512));
+                if (hipHostMalloc(&plan.h_params, 512, hipHostMallocDefault)
+                    != hipSuccess)
+                    plan.h_params = malloc(512);
+            }
             hipError_t ge2 = hx::launch_scan_agg_gang(
-                s, A, gang, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0);
+                s, A, gang, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0,
+                (hx::GangParams*)plan.h_params,
+                (hx::GangParams*)plan.d_params);
             if (ge2 != hipSuccess) use_gang = false;  // e.g. LDS size rejected
         }
         if (!use_gang) HIP_TRY(hx::launch_scan_agg(s, A, 0));

@@ -22,8 +22,10 @@ hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
 hipError_t launch_copy_u64(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                            const CopyDesc* descs, uint32_t n_descs);
 hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid);
+struct GangParams;  // kernels.hip internal; caller provides param buffers
 hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
-                                uint32_t gang_size, bool minmax);
+                                uint32_t gang_size, bool minmax,
+                                GangParams* h_params, GangParams* d_params);
 hipError_t launch_gather_multi(hipStream_t s,
                                const unsigned long long* const* srcs,
                                uint32_t n_arrays, const uint32_t* perm,

@@ -323,7 +323,8 @@ __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
 
 template <bool MM>
 __global__ void __launch_bounds__(1024)
-k_scan_agg_gang(GangParams G) {
+k_scan_agg_gang(const GangParams* __restrict__ gp) {
+    const GangParams& G = *gp;
     // Transposed gang walk: a thread owns TWO row positions of the aligned
     // series window and visits them across every unit (row-group slice) of
     // the gang. Same-size SSTs slice into EXACTLY aligned units, so the
@@ -791,8 +792,9 @@ hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid) {
 }
 
 hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
-                                uint32_t gang_size, bool minmax) {
-    GangParams G;
+                                uint32_t gang_size, bool minmax,
+                                GangParams* h_params, GangParams* d_params) {
+    GangParams& G = *h_params;
     G.P = p;
     G.gang_size = gang_size;
     G.n_gangs = (p.n_rgs + gang_size - 1) / gang_size;
@@ -808,10 +810,13 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     size_t lds = (size_t)G.ne * (minmax ? 36 : 20) +
                  (size_t)G.gang_size * sizeof(RgDesc);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
+    hipError_t e = hipMemcpyAsync(d_params, h_params, sizeof(GangParams),
+                                  hipMemcpyHostToDevice, s);
+    if (e != hipSuccess) return e;
     if (minmax)
-        hipLaunchKernelGGL(k_scan_agg_gang<true>, dim3(grid), dim3(1024), lds, s, G);
+        hipLaunchKernelGGL(k_scan_agg_gang<true>, dim3(grid), dim3(1024), lds, s, d_params);
     else
-        hipLaunchKernelGGL(k_scan_agg_gang<false>, dim3(grid), dim3(1024), lds, s, G);
+        hipLaunchKernelGGL(k_scan_agg_gang<false>, dim3(grid), dim3(1024), lds, s, d_params);
     return hipGetLastError();
 }
 
