@@ -325,6 +325,16 @@ template <bool MM>
 __global__ void __launch_bounds__(1024)
 k_scan_agg_gang(const GangParams* __restrict__ gp) {
     const GangParams& G = *gp;
+    // hoist hot scalars once (pointer access in the hot loop re-issues
+    // s_loads; by-value params spill SGPRs — this is the middle ground)
+    const AggParams P_hot = G.P;
+    const uint8_t* const blob = P_hot.blob;
+    const uint8_t* const dec = P_hot.dec;
+    const int64_t ts_lo = P_hot.ts_lo, ts_hi = P_hot.ts_hi;
+    const int32_t skip = P_hot.skip;
+    const int32_t use_sset = P_hot.use_sset;
+    const uint32_t gang_size = G.gang_size;
+    const uint32_t n_gangs = G.n_gangs;
     // Transposed gang walk: a thread owns TWO row positions of the aligned
     // series window and visits them across every unit (row-group slice) of
     // the gang. Same-size SSTs slice into EXACTLY aligned units, so the
@@ -335,7 +345,6 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
     // Alignment is a speed matter only: register combining merges equal
     // ADJACENT keys, which is correct for any data.
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    const AggParams& P = G.P;
     const uint32_t ne = G.ne;
     RgDesc* ldesc = (RgDesc*)smem;
     const size_t desc_bytes = (size_t)G.gang_size * sizeof(RgDesc);
@@ -351,9 +360,9 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
     const int lane = threadIdx.x & 63;
     unsigned long long my_matched = 0;
     unsigned long long my_flushes = 0;
-    for (uint32_t gang = blockIdx.x; gang < G.n_gangs; gang += gridDim.x) {
-        const uint32_t rg0 = gang * G.gang_size;
-        const uint32_t rg_end = min(rg0 + G.gang_size, P.n_rgs);
+    for (uint32_t gang = blockIdx.x; gang < n_gangs; gang += gridDim.x) {
+        const uint32_t rg0 = gang * gang_size;
+        const uint32_t rg_end = min(rg0 + gang_size, P_hot.n_rgs);
         const uint32_t nu = rg_end - rg0;
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             lkey[i] = KEY_EMPTY;
@@ -366,7 +375,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
         }
         // preload unit descriptors (one dynamic-LDS object only — G17)
         for (uint32_t u = threadIdx.x; u < nu; u += blockDim.x)
-            ldesc[u] = P.rgs[rg0 + u];
+            ldesc[u] = P_hot.rgs[rg0 + u];
         __syncthreads();
         uint32_t max_n = 0;
         for (uint32_t u = 0; u < nu; u++) max_n = max(max_n, ldesc[u].n_rows);
@@ -383,11 +392,11 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                 const uint32_t n = rg.n_rows;
                 if (__all(r >= n)) continue;
                 const uint64_t* S =
-                    (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+                    (const uint64_t*)hx_ptr(blob, dec, rg.series_off);
                 const int64_t* T =
-                    (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+                    (const int64_t*)hx_ptr(blob, dec, rg.ts_off);
                 const double* V =
-                    (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+                    (const double*)hx_ptr(blob, dec, rg.val_off);
                 const bool inb = r < n;
                 // issue all three column loads together (one wait, not three
                 // serialized dependent waits per unit iteration)
@@ -395,31 +404,31 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                 const int64_t tv = inb ? T[r] : 0;
                 const double v = inb ? V[r] : 0.0;
                 if (!inb) continue;
-                bool alive = (tv >= P.ts_lo) & (tv < P.ts_hi);
-                if (alive && P.use_sset) alive = sset_has(P, sv);
-                if (alive && P.skip < 2) {
+                bool alive = (tv >= ts_lo) & (tv < ts_hi);
+                if (alive && use_sset) alive = sset_has(G.P, sv);
+                if (alive && skip < 2) {
                     bool dup = false;
                     if (r + 1 < n) {
                         // successor is the adjacent element (same cache line
                         // as this lane group's loads; no cross-lane sync)
                         dup = (S[r + 1] == sv) & (T[r + 1] == tv);
                     } else if (rg.next_rg >= 0) {
-                        const RgDesc nx = P.rgs[rg.next_rg];
-                        uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
+                        const RgDesc nx = P_hot.rgs[rg.next_rg];
+                        uint64_t s2 = *(const uint64_t*)hx_ptr(blob, dec,
                                                                nx.series_off);
-                        int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec,
+                        int64_t t2 = *(const int64_t*)hx_ptr(blob, dec,
                                                              nx.ts_off);
                         dup = (s2 == sv) & (t2 == tv);
                     }
                     if (!dup) {
-                        const SstDev sst = P.ssts[rg.sst_id];
-                        if (sst.cluster >= 0) dup = shadowed(P, sst, sv, tv);
+                        const SstDev sst = P_hot.ssts[rg.sst_id];
+                        if (sst.cluster >= 0) dup = shadowed(G.P, sst, sv, tv);
                     }
                     alive = !dup;
                 }
                 if (!alive) continue;
                 my_matched++;
-                if (P.skip == 1 || P.skip == 2) continue;
+                if (skip == 1 || skip == 2) continue;
                 if (have_run && run_key == sv) {
                     run_sum += v;
                     run_cnt++;
@@ -429,7 +438,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                     }
                 } else {
                     if (have_run) {
-                        if (P.skip == 3) my_flushes++;   // bisect: count only
+                        if (skip == 3) my_flushes++;   // bisect: count only
                         else lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax,
                                             ne, run_key, run_sum, run_cnt,
                                             run_min, run_max);
@@ -451,17 +460,17 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
         __syncthreads();
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             if (lkey[i] == KEY_EMPTY) continue;
-            agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
+            agg_update(G.P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
                        MM ? ordered_f64(lmin[i]) : 0.0,
                        MM ? ordered_f64(lmax[i]) : 0.0);
         }
         __syncthreads();
     }
-    if (P.skip == 3) my_matched = my_flushes;  // bisect: report flush count
+    if (skip == 3) my_matched = my_flushes;  // bisect: report flush count
     for (int off = 32; off > 0; off >>= 1)
         my_matched += __shfl_down(my_matched, off, 64);
     if ((threadIdx.x & 63) == 0 && my_matched)
-        atomicAdd(P.matched, my_matched);
+        atomicAdd(P_hot.matched, my_matched);
 }
 
 // ---------------------------------------------------------------------------
