@@ -292,7 +292,7 @@ struct GangParams {
 };
 
 template <bool MM>
-__device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
+__device__ __forceinline__ void lds_update(const AggParams& P, uint64_t* lkey,
                                            double* lsum, unsigned int* lcnt,
                                            unsigned long long* lmin,
                                            unsigned long long* lmax,
@@ -318,23 +318,28 @@ __device__ __forceinline__ void lds_update(const GangParams& G, uint64_t* lkey,
         i = (i + 1) & (ne - 1);
     }
     // LDS table full: direct global update
-    agg_update(G.P, sv, 0, v, cnt, mn, mx);
+    agg_update(P, sv, 0, v, cnt, mn, mx);
 }
 
 template <bool MM>
 __global__ void __launch_bounds__(1024)
 k_scan_agg_gang(const GangParams* __restrict__ gp) {
     const GangParams& G = *gp;
-    // hoist hot scalars once (pointer access in the hot loop re-issues
-    // s_loads; by-value params spill SGPRs — this is the middle ground)
-    const AggParams P_hot = G.P;
-    const uint8_t* const blob = P_hot.blob;
-    const uint8_t* const dec = P_hot.dec;
-    const int64_t ts_lo = P_hot.ts_lo, ts_hi = P_hot.ts_hi;
-    const int32_t skip = P_hot.skip;
-    const int32_t use_sset = P_hot.use_sset;
+    // Param handling: ~12 hot scalars live in SGPRs; everything the cold
+    // paths need (table pointers, cluster arrays, counters) is mirrored
+    // into LDS once per block — neither SGPR spills (by-value struct) nor
+    // per-iteration s_load stalls (pointer chasing).
+    const uint8_t* const blob = G.P.blob;
+    const uint8_t* const dec = G.P.dec;
+    const int64_t ts_lo = G.P.ts_lo, ts_hi = G.P.ts_hi;
+    const int32_t skip = G.P.skip;
+    const int32_t use_sset = G.P.use_sset;
     const uint32_t gang_size = G.gang_size;
     const uint32_t n_gangs = G.n_gangs;
+    const uint32_t n_rgs_all = G.P.n_rgs;
+    const RgDesc* const rgs_all = G.P.rgs;
+    const SstDev* const ssts_all = G.P.ssts;
+    unsigned long long* const matched_ptr = G.P.matched;
     // Transposed gang walk: a thread owns TWO row positions of the aligned
     // series window and visits them across every unit (row-group slice) of
     // the gang. Same-size SSTs slice into EXACTLY aligned units, so the
@@ -346,8 +351,9 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
     // ADJACENT keys, which is correct for any data.
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const uint32_t ne = G.ne;
-    RgDesc* ldesc = (RgDesc*)smem;
-    const size_t desc_bytes = (size_t)G.gang_size * sizeof(RgDesc);
+    AggParams* Pm = (AggParams*)smem;          // LDS mirror for cold paths
+    RgDesc* ldesc = (RgDesc*)(smem + 256);
+    const size_t desc_bytes = 256 + (size_t)gang_size * sizeof(RgDesc);
     uint64_t* lkey = (uint64_t*)(smem + desc_bytes);
     double* lsum = (double*)(smem + desc_bytes + (size_t)ne * 8);
     unsigned long long* lmin =
@@ -356,13 +362,21 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
         (unsigned long long*)(smem + desc_bytes + (size_t)ne * (MM ? 24 : 16));
     unsigned int* lcnt =
         (unsigned int*)(smem + desc_bytes + (size_t)ne * (MM ? 32 : 16));
+    {
+        static_assert(sizeof(AggParams) <= 256, "grow the LDS mirror");
+        const uint64_t* src = (const uint64_t*)&gp->P;
+        uint64_t* dst = (uint64_t*)Pm;
+        for (uint32_t i = threadIdx.x; i < sizeof(AggParams) / 8;
+             i += blockDim.x)
+            dst[i] = src[i];
+    }
 
     const int lane = threadIdx.x & 63;
     unsigned long long my_matched = 0;
     unsigned long long my_flushes = 0;
     for (uint32_t gang = blockIdx.x; gang < n_gangs; gang += gridDim.x) {
         const uint32_t rg0 = gang * gang_size;
-        const uint32_t rg_end = min(rg0 + gang_size, P_hot.n_rgs);
+        const uint32_t rg_end = min(rg0 + gang_size, n_rgs_all);
         const uint32_t nu = rg_end - rg0;
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             lkey[i] = KEY_EMPTY;
@@ -375,7 +389,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
         }
         // preload unit descriptors (one dynamic-LDS object only — G17)
         for (uint32_t u = threadIdx.x; u < nu; u += blockDim.x)
-            ldesc[u] = P_hot.rgs[rg0 + u];
+            ldesc[u] = rgs_all[rg0 + u];
         __syncthreads();
         uint32_t max_n = 0;
         for (uint32_t u = 0; u < nu; u++) max_n = max(max_n, ldesc[u].n_rows);
@@ -405,7 +419,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                 const double v = inb ? V[r] : 0.0;
                 if (!inb) continue;
                 bool alive = (tv >= ts_lo) & (tv < ts_hi);
-                if (alive && use_sset) alive = sset_has(G.P, sv);
+                if (alive && use_sset) alive = sset_has(*Pm, sv);
                 if (alive && skip < 2) {
                     bool dup = false;
                     if (r + 1 < n) {
@@ -413,7 +427,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                         // as this lane group's loads; no cross-lane sync)
                         dup = (S[r + 1] == sv) & (T[r + 1] == tv);
                     } else if (rg.next_rg >= 0) {
-                        const RgDesc nx = P_hot.rgs[rg.next_rg];
+                        const RgDesc nx = rgs_all[rg.next_rg];
                         uint64_t s2 = *(const uint64_t*)hx_ptr(blob, dec,
                                                                nx.series_off);
                         int64_t t2 = *(const int64_t*)hx_ptr(blob, dec,
@@ -421,8 +435,8 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                         dup = (s2 == sv) & (t2 == tv);
                     }
                     if (!dup) {
-                        const SstDev sst = P_hot.ssts[rg.sst_id];
-                        if (sst.cluster >= 0) dup = shadowed(G.P, sst, sv, tv);
+                        const SstDev sst = ssts_all[rg.sst_id];
+                        if (sst.cluster >= 0) dup = shadowed(*Pm, sst, sv, tv);
                     }
                     alive = !dup;
                 }
@@ -439,7 +453,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
                 } else {
                     if (have_run) {
                         if (skip == 3) my_flushes++;   // bisect: count only
-                        else lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax,
+                        else lds_update<MM>(*Pm, lkey, lsum, lcnt, lmin, lmax,
                                             ne, run_key, run_sum, run_cnt,
                                             run_min, run_max);
                     }
@@ -453,14 +467,14 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
             }
         }
         if (have_run) {
-            lds_update<MM>(G, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
+            lds_update<MM>(*Pm, lkey, lsum, lcnt, lmin, lmax, ne, run_key,
                            run_sum, run_cnt, run_min, run_max);
             have_run = false;
         }
         __syncthreads();
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             if (lkey[i] == KEY_EMPTY) continue;
-            agg_update(G.P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
+            agg_update(*Pm, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
                        MM ? ordered_f64(lmin[i]) : 0.0,
                        MM ? ordered_f64(lmax[i]) : 0.0);
         }
@@ -470,7 +484,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp) {
     for (int off = 32; off > 0; off >>= 1)
         my_matched += __shfl_down(my_matched, off, 64);
     if ((threadIdx.x & 63) == 0 && my_matched)
-        atomicAdd(P_hot.matched, my_matched);
+        atomicAdd(matched_ptr, my_matched);
 }
 
 // ---------------------------------------------------------------------------
@@ -816,7 +830,7 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
         uint32_t ne = (uint32_t)strtoul(nee, nullptr, 10);
         if (ne >= 1024 && ne <= 8192 && !(ne & (ne - 1))) G.ne = ne;
     }
-    size_t lds = (size_t)G.ne * (minmax ? 36 : 20) +
+    size_t lds = 256 + (size_t)G.ne * (minmax ? 36 : 20) +
                  (size_t)G.gang_size * sizeof(RgDesc);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
     hipError_t e = hipMemcpyAsync(d_params, h_params, sizeof(GangParams),
