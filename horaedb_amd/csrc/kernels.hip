@@ -323,21 +323,27 @@ __device__ __forceinline__ void lds_update(const AggParams& P, uint64_t* lkey,
 
 template <bool MM>
 __global__ void __launch_bounds__(1024)
-k_scan_agg_gang(const GangParams* __restrict__ gp) {
+k_scan_agg_gang(const GangParams* __restrict__ gp,
+                const uint8_t* __restrict__ blob_arg,
+                const uint8_t* __restrict__ dec_arg,
+                const RgDesc* __restrict__ rgs_arg) {
+    // blob/dec/rgs come as KERNEL ARGUMENTS: pointers loaded from memory are
+    // generic (flat_load — slow); argument pointers keep the global address
+    // space (global_load).
     const GangParams& G = *gp;
     // Param handling: ~12 hot scalars live in SGPRs; everything the cold
     // paths need (table pointers, cluster arrays, counters) is mirrored
     // into LDS once per block — neither SGPR spills (by-value struct) nor
     // per-iteration s_load stalls (pointer chasing).
-    const uint8_t* const blob = G.P.blob;
-    const uint8_t* const dec = G.P.dec;
+    const uint8_t* const blob = blob_arg;
+    const uint8_t* const dec = dec_arg;
     const int64_t ts_lo = G.P.ts_lo, ts_hi = G.P.ts_hi;
     const int32_t skip = G.P.skip;
     const int32_t use_sset = G.P.use_sset;
     const uint32_t gang_size = G.gang_size;
     const uint32_t n_gangs = G.n_gangs;
     const uint32_t n_rgs_all = G.P.n_rgs;
-    const RgDesc* const rgs_all = G.P.rgs;
+    const RgDesc* const rgs_all = rgs_arg;
     const SstDev* const ssts_all = G.P.ssts;
     unsigned long long* const matched_ptr = G.P.matched;
     // Transposed gang walk: a thread owns TWO row positions of the aligned
@@ -837,9 +843,11 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
                                   hipMemcpyHostToDevice, s);
     if (e != hipSuccess) return e;
     if (minmax)
-        hipLaunchKernelGGL(k_scan_agg_gang<true>, dim3(grid), dim3(1024), lds, s, d_params);
+        hipLaunchKernelGGL(k_scan_agg_gang<true>, dim3(grid), dim3(1024), lds,
+                           s, d_params, p.blob, p.dec, p.rgs);
     else
-        hipLaunchKernelGGL(k_scan_agg_gang<false>, dim3(grid), dim3(1024), lds, s, d_params);
+        hipLaunchKernelGGL(k_scan_agg_gang<false>, dim3(grid), dim3(1024), lds,
+                           s, d_params, p.blob, p.dec, p.rgs);
     return hipGetLastError();
 }
 
