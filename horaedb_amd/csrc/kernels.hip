@@ -406,70 +406,102 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
         uint32_t run_cnt = 0;
         for (uint32_t roff = 0; roff < max_n; roff += blockDim.x) {
             const uint32_t r = roff + threadIdx.x;
-#pragma unroll 2
-            for (uint32_t u = 0; u < nu; u++) {
+            // two-stage software pipeline over the units: issue unit u+1's
+            // column loads before processing unit u, so each iteration's
+            // dependent-load stall overlaps the next iteration's fetches
+            // (the compiler alone serializes them behind per-iteration
+            // waits — measured ~730 cy/iteration without this).
+            uint64_t sA = KEY_EMPTY, s1A = KEY_EMPTY;
+            int64_t tA = 0, t1A = 0;
+            double vA = 0;
+            uint32_t nA = 0, sstA = 0;
+            int32_t nxA = -1;
+            auto issue = [&](uint32_t u, uint64_t& s_, uint64_t& s1_,
+                             int64_t& t_, int64_t& t1_, double& v_,
+                             uint32_t& n_, uint32_t& sst_, int32_t& nx_) {
                 const RgDesc rg = ldesc[u];
-                const uint32_t n = rg.n_rows;
-                if (__all(r >= n)) continue;
+                n_ = rg.n_rows;
+                sst_ = rg.sst_id;
+                nx_ = rg.next_rg;
+                const bool inb = r < n_;
+                const bool inb1 = r + 1 < n_;
                 const uint64_t* S =
                     (const uint64_t*)hx_ptr(blob, dec, rg.series_off);
                 const int64_t* T =
                     (const int64_t*)hx_ptr(blob, dec, rg.ts_off);
                 const double* V =
                     (const double*)hx_ptr(blob, dec, rg.val_off);
-                const bool inb = r < n;
-                // issue all three column loads together (one wait, not three
-                // serialized dependent waits per unit iteration)
-                const uint64_t sv = inb ? S[r] : KEY_EMPTY;
-                const int64_t tv = inb ? T[r] : 0;
-                const double v = inb ? V[r] : 0.0;
-                if (!inb) continue;
-                bool alive = (tv >= ts_lo) & (tv < ts_hi);
-                if (alive && use_sset) alive = sset_has(*Pm, sv);
-                if (alive && skip < 2) {
-                    bool dup = false;
-                    if (r + 1 < n) {
-                        // successor is the adjacent element (same cache line
-                        // as this lane group's loads; no cross-lane sync)
-                        dup = (S[r + 1] == sv) & (T[r + 1] == tv);
-                    } else if (rg.next_rg >= 0) {
-                        const RgDesc nx = rgs_all[rg.next_rg];
-                        uint64_t s2 = *(const uint64_t*)hx_ptr(blob, dec,
-                                                               nx.series_off);
-                        int64_t t2 = *(const int64_t*)hx_ptr(blob, dec,
-                                                             nx.ts_off);
-                        dup = (s2 == sv) & (t2 == tv);
+                s_ = inb ? S[r] : KEY_EMPTY;
+                t_ = inb ? T[r] : 0;
+                v_ = inb ? V[r] : 0.0;
+                s1_ = inb1 ? S[r + 1] : KEY_EMPTY;
+                t1_ = inb1 ? T[r + 1] : 0;
+            };
+            if (nu > 0)
+                issue(0, sA, s1A, tA, t1A, vA, nA, sstA, nxA);
+#pragma unroll 1
+            for (uint32_t u = 0; u < nu; u++) {
+                uint64_t sB = KEY_EMPTY, s1B = KEY_EMPTY;
+                int64_t tB = 0, t1B = 0;
+                double vB = 0;
+                uint32_t nB = 0, sstB = 0;
+                int32_t nxB = -1;
+                if (u + 1 < nu)
+                    issue(u + 1, sB, s1B, tB, t1B, vB, nB, sstB, nxB);
+                const uint64_t sv = sA;
+                const int64_t tv = tA;
+                if (r < nA) {
+                    bool alive = (tv >= ts_lo) & (tv < ts_hi);
+                    if (alive && use_sset) alive = sset_has(*Pm, sv);
+                    if (alive && skip < 2) {
+                        bool dup = false;
+                        if (r + 1 < nA) {
+                            dup = (s1A == sv) & (t1A == tv);
+                        } else if (nxA >= 0) {
+                            const RgDesc nx = rgs_all[nxA];
+                            uint64_t s2 = *(const uint64_t*)hx_ptr(
+                                blob, dec, nx.series_off);
+                            int64_t t2 = *(const int64_t*)hx_ptr(blob, dec,
+                                                                 nx.ts_off);
+                            dup = (s2 == sv) & (t2 == tv);
+                        }
+                        if (!dup) {
+                            const SstDev sst = ssts_all[sstA];
+                            if (sst.cluster >= 0)
+                                dup = shadowed(*Pm, sst, sv, tv);
+                        }
+                        alive = !dup;
                     }
-                    if (!dup) {
-                        const SstDev sst = ssts_all[rg.sst_id];
-                        if (sst.cluster >= 0) dup = shadowed(*Pm, sst, sv, tv);
+                    if (alive) {
+                        const double v = vA;
+                        my_matched++;
+                        if (skip == 1 || skip == 2) {
+                        } else if (have_run && run_key == sv) {
+                            run_sum += v;
+                            run_cnt++;
+                            if (MM) {
+                                run_min = fmin(run_min, v);
+                                run_max = fmax(run_max, v);
+                            }
+                        } else {
+                            if (have_run) {
+                                if (skip == 3) my_flushes++;  // bisect
+                                else lds_update<MM>(*Pm, lkey, lsum, lcnt,
+                                                    lmin, lmax, ne, run_key,
+                                                    run_sum, run_cnt, run_min,
+                                                    run_max);
+                            }
+                            have_run = true;
+                            run_key = sv;
+                            run_sum = v;
+                            run_cnt = 1;
+                            run_min = v;
+                            run_max = v;
+                        }
                     }
-                    alive = !dup;
                 }
-                if (!alive) continue;
-                my_matched++;
-                if (skip == 1 || skip == 2) continue;
-                if (have_run && run_key == sv) {
-                    run_sum += v;
-                    run_cnt++;
-                    if (MM) {
-                        run_min = fmin(run_min, v);
-                        run_max = fmax(run_max, v);
-                    }
-                } else {
-                    if (have_run) {
-                        if (skip == 3) my_flushes++;   // bisect: count only
-                        else lds_update<MM>(*Pm, lkey, lsum, lcnt, lmin, lmax,
-                                            ne, run_key, run_sum, run_cnt,
-                                            run_min, run_max);
-                    }
-                    have_run = true;
-                    run_key = sv;
-                    run_sum = v;
-                    run_cnt = 1;
-                    run_min = v;
-                    run_max = v;
-                }
+                sA = sB; s1A = s1B; tA = tB; t1A = t1B; vA = vB;
+                nA = nB; sstA = sstB; nxA = nxB;
             }
         }
         if (have_run) {
