@@ -1083,7 +1083,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                 s, A, gang, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0,
                 (hx::GangParams*)plan.h_params,
                 (hx::GangParams*)plan.d_params);
-            if (ge2 != hipSuccess) use_gang = false;  // e.g. LDS size rejected
+            if (ge2 != hipSuccess)
+                return fail(HX_ERR_HIP,
+                            std::string("gang kernel launch failed: ") +
+                                hipGetErrorString(ge2));
         }
         if (!use_gang) HIP_TRY(hx::launch_scan_agg(s, A, 0));
         HIP_TRY(hipEventRecord(e1, s));

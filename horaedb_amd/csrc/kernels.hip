@@ -871,6 +871,17 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     size_t lds = 256 + (size_t)G.ne * (minmax ? 36 : 20) +
                  (size_t)G.gang_size * sizeof(RgDesc);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
+    // >64 KiB dynamic LDS needs an explicit opt-in per kernel
+    static bool lds_opted[2] = {false, false};
+    if (!lds_opted[minmax ? 1 : 0]) {
+        const void* f = minmax
+            ? reinterpret_cast<const void*>(&k_scan_agg_gang<true>)
+            : reinterpret_cast<const void*>(&k_scan_agg_gang<false>);
+        hipError_t ae = hipFuncSetAttribute(
+            f, hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        if (ae != hipSuccess) return ae;
+        lds_opted[minmax ? 1 : 0] = true;
+    }
     hipError_t e = hipMemcpyAsync(d_params, h_params, sizeof(GangParams),
                                   hipMemcpyHostToDevice, s);
     if (e != hipSuccess) return e;
