@@ -1050,7 +1050,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipEventCreate(&e0));
         HIP_TRY(hipEventCreate(&e1));
         HIP_TRY(hipEventRecord(e0, s));
-        bool use_gang = !bucket && !getenv("HX_NO_GANG");
+        // the LDS-gang variant measured slower than the wave kernel at every
+        // tested config (see profiles/README r01 notes); opt-in for further
+        // experiments
+        bool use_gang = !bucket && getenv("HX_GANG_ON") != nullptr;
         if (use_gang) {
             // gang = number of SSTs in the dominant size class: one gang
             // then covers EXACTLY one aligned series window (the transposed
