@@ -1,0 +1,99 @@
+# Multi-rank (config 4) semantics on CPU: two gloo ranks own disjoint SST
+# shards (per-rank seed => disjoint series spaces); the sharded result set is
+# the union, verified against a single oracle pass over both shards. This
+# covers bench.py's distributed path (barrier + MAX time + SUM rows + digest
+# exchange) without a GPU. DESIGN.md §6.
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _rank_main(rank, world, tmpdir, q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = os.environ.get("HX_TEST_PORT", "29517")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import oracle
+    from oracle.scan import AGG_SUM, AGG_COUNT
+    from tools.gen_ssts import gen_dataset, middle_range
+
+    shard_dir = os.path.join(tmpdir, f"shard{rank}")
+    m = gen_dataset(shard_dir, n_rows=20_000, n_series=200, n_ssts=4,
+                    seed=100 + rank)
+    ts_range = middle_range(m)
+    ssts = [oracle.read_sst(s["path"]) for s in m["ssts"]]
+    res = oracle.scan_agg(ssts, ts_range, ops=AGG_SUM | AGG_COUNT)
+
+    # the partitioned result set: ranks exchange O(1) digests (bench.py's
+    # timed-region exchange), full tables stay shard-local
+    dist.barrier()
+    digest = torch.tensor([float(len(res["series_id"])),
+                           float(res["count"].sum()),
+                           float(res["sum"].sum())], dtype=torch.float64)
+    dist.all_reduce(digest, op=dist.ReduceOp.SUM)
+    rows = torch.tensor([float(m["n_rows"])], dtype=torch.float64)
+    dist.all_reduce(rows, op=dist.ReduceOp.SUM)
+    dist.barrier()
+
+    q.put((rank, {
+        "digest": digest.tolist(),
+        "rows_total": rows.item(),
+        "series": res["series_id"].tolist(),
+        "sum": res["sum"].tolist(),
+        "count": res["count"].tolist(),
+        "shard_dir": shard_dir,
+        "ts_range": list(ts_range),
+    }))
+    dist.destroy_process_group()
+
+
+def test_two_rank_sharded_aggregate(tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main, args=(r, 2, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(2):
+        rank, data = q.get(timeout=120)
+        outs[rank] = data
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    # both ranks agree on the reduced digest and total rows
+    assert outs[0]["digest"] == outs[1]["digest"]
+    assert outs[0]["rows_total"] == 40_000
+
+    # union of shard tables == oracle over all SSTs of both shards
+    import oracle
+    from oracle.scan import AGG_SUM, AGG_COUNT
+    all_ssts = []
+    for r in (0, 1):
+        ddir = os.path.join(outs[r]["shard_dir"], "data")
+        for f in sorted(os.listdir(ddir)):
+            if f.endswith(".sst"):
+                all_ssts.append(oracle.read_sst(os.path.join(ddir, f)))
+    combined = oracle.scan_agg(all_ssts, tuple(outs[0]["ts_range"]),
+                               ops=AGG_SUM | AGG_COUNT)
+    series_union = sorted(outs[0]["series"] + outs[1]["series"])
+    assert series_union == combined["series_id"].tolist(), \
+        "shards must partition the series space (disjoint per-rank seeds)"
+    merged = {}
+    for r in (0, 1):
+        for s, sm, c in zip(outs[r]["series"], outs[r]["sum"],
+                            outs[r]["count"]):
+            merged[s] = (sm, c)
+    np.testing.assert_allclose(
+        [merged[s][0] for s in combined["series_id"].tolist()],
+        combined["sum"], rtol=1e-12)
+    assert [merged[s][1] for s in combined["series_id"].tolist()] == \
+        combined["count"].tolist()
