@@ -241,6 +241,7 @@ struct DevPlan {
     std::vector<hx::ClusterDev> clusters;
     std::vector<int32_t> cluster_members;
     std::vector<hx::DeltaPageDesc> delta_pages;
+    std::vector<hx::SnappyPageDesc> snappy_pages;
     std::vector<hx::CopyDesc> copies;
     size_t blob_bytes = 0;
     size_t dec_bytes = 0;
@@ -258,6 +259,7 @@ struct DevPlan {
     hx::ClusterDev* d_clusters = nullptr;
     int32_t* d_members = nullptr;
     hx::DeltaPageDesc* d_delta = nullptr;
+    hx::SnappyPageDesc* d_snappy = nullptr;
     hx::CopyDesc* d_copies = nullptr;
 
     // aggregate table (lazily sized)
@@ -477,10 +479,11 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 err_flag = 1;
                 break;
             }
-            if (j.codec != hx::CODEC_UNCOMPRESSED) {
+            if (j.codec != hx::CODEC_UNCOMPRESSED &&
+                j.codec != hx::CODEC_SNAPPY) {
                 std::lock_guard<std::mutex> g(mu);
                 err_msg = j.ss->cat->path + ": codec unsupported in round 1 "
-                          "(uncompressed only; Snappy is next)";
+                          "(uncompressed and Snappy only)";
                 err_flag = 1;
                 break;
             }
@@ -489,19 +492,35 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
             const uint8_t* src = tmp.data() + in_chunk + dp->def_level_bytes;
             std::memcpy(plan.h_blob + j.dst_off, src, payload);
 
+            uint64_t data_off = j.dst_off;  // raw page bytes (post-codec)
+            size_t raw_size = payload;
+            if (j.codec == hx::CODEC_SNAPPY) {
+                std::lock_guard<std::mutex> g(mu);
+                hx::SnappyPageDesc sp{};
+                sp.src_off = j.dst_off;
+                sp.comp_len = (uint32_t)payload;
+                sp.uncomp_len = (uint32_t)(size_t(dp->uncompressed_size) -
+                                           size_t(dp->def_level_bytes));
+                sp.dst_off = hx::OFF_DEC | dec_off;
+                dec_off = align64(dec_off + sp.uncomp_len);
+                plan.snappy_pages.push_back(sp);
+                data_off = sp.dst_off;
+                raw_size = sp.uncomp_len;
+            }
+
             if (dp->encoding == hx::ENC_PLAIN) {
-                if (payload != size_t(j.num_values) * 8) {
+                if (raw_size != size_t(j.num_values) * 8) {
                     std::lock_guard<std::mutex> g(mu);
                     err_msg = j.ss->cat->path + ": PLAIN payload size mismatch";
                     err_flag = 1;
                     break;
                 }
-                j.final_off = j.dst_off;
+                j.final_off = data_off;
             } else if (dp->encoding == hx::ENC_DELTA_BINARY_PACKED && j.col != 2) {
                 std::lock_guard<std::mutex> g(mu);
                 hx::DeltaPageDesc dd{};
-                dd.src_off = j.dst_off;
-                dd.src_len = (uint32_t)payload;
+                dd.src_off = data_off;
+                dd.src_len = (uint32_t)raw_size;
                 dd.n_values = (uint32_t)j.num_values;
                 if (j.num_values > 8192) {
                     err_msg = j.ss->cat->path + ": delta page > 8192 values";
@@ -657,6 +676,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     HIP_TRY(upload(plan.d_clusters, plan.clusters));
     HIP_TRY(upload(plan.d_members, plan.cluster_members));
     HIP_TRY(upload(plan.d_delta, plan.delta_pages));
+    HIP_TRY(upload(plan.d_snappy, plan.snappy_pages));
     HIP_TRY(upload(plan.d_copies, plan.copies));
     HIP_TRY(hipMalloc((void**)&plan.d_counters, 4 * sizeof(unsigned long long)));
     HIP_TRY(hipStreamSynchronize(plan.stream));
@@ -849,6 +869,7 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
         for (void* p : {(void*)plan.d_blob, (void*)plan.d_dec, (void*)plan.d_rgs,
                         (void*)plan.d_ssts, (void*)plan.d_clusters,
                         (void*)plan.d_members, (void*)plan.d_delta,
+                        (void*)plan.d_snappy,
                         (void*)plan.d_copies, (void*)plan.t_series,
                         (void*)plan.t_bucket, (void*)plan.t_state,
                         (void*)plan.t_sum, (void*)plan.t_cnt, (void*)plan.t_min,
@@ -964,6 +985,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipEventCreate(&d1));
         HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
         HIP_TRY(hipEventRecord(d0, s));
+        if (!plan.snappy_pages.empty())
+            HIP_TRY(hx::launch_snappy(s, plan.d_blob, plan.d_dec, plan.d_snappy,
+                                      (uint32_t)plan.snappy_pages.size(),
+                                      plan.d_counters + 1));
         if (!plan.delta_pages.empty())
             HIP_TRY(hx::launch_decode_delta(s, plan.d_blob, plan.d_dec,
                                             plan.d_delta,
@@ -984,8 +1009,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         hipEventDestroy(d0);
         hipEventDestroy(d1);
         if (decode_err)
-            return fail(HX_ERR_FORMAT, "delta page decode failed (malformed "
-                                       "or >8192-value page)");
+            return fail(HX_ERR_FORMAT, "page decode failed (malformed snappy "
+                                       "stream or delta page)");
         plan.decoded = true;
     }
 

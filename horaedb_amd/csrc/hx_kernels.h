@@ -19,6 +19,9 @@ struct CompactOut {
 hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                                const DeltaPageDesc* pages, uint32_t n_pages,
                                unsigned long long* err_flag);
+hipError_t launch_snappy(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                         const SnappyPageDesc* pages, uint32_t n_pages,
+                         unsigned long long* err_flag);
 hipError_t launch_copy_u64(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                            const CopyDesc* descs, uint32_t n_descs);
 hipError_t launch_scan_agg(hipStream_t s, const AggParams& p, uint32_t grid);

@@ -625,6 +625,106 @@ k_copy_u64(const uint8_t* blob, const uint8_t* dec_in, uint8_t* dec,
 }
 
 // ---------------------------------------------------------------------------
+// Snappy raw-block decompression (the reference's default page codec,
+// config.rs:120-133). One WAVE per page: the tag stream is inherently
+// sequential, so every lane parses the (wave-uniform) element headers and
+// the 64 lanes copy the element's bytes cooperatively. Overlapping copies
+// (offset < length) use the periodic-pattern equivalence of the sequential
+// byte copy. Parallelism comes from the page count (~one 64KB page per
+// 8192-row column chunk).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ uint32_t snappy_varint(const uint8_t* p,
+                                                  uint32_t& pos, uint32_t len,
+                                                  int* err) {
+    uint32_t r = 0;
+    int s = 0;
+    for (;;) {
+        if (pos >= len || s > 28) { *err = 1; return 0; }
+        uint8_t b = p[pos++];
+        r |= (uint32_t)(b & 0x7f) << s;
+        if (!(b & 0x80)) return r;
+        s += 7;
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
+                    const SnappyPageDesc* __restrict__ pages, uint32_t n_pages,
+                    unsigned long long* err_flag) {
+    const uint32_t lane = threadIdx.x & 63;
+    const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+    for (uint32_t pg = wave; pg < n_pages; pg += n_waves) {
+        const SnappyPageDesc pd = pages[pg];
+        const uint8_t* src = blob + (pd.src_off & OFF_MASK);
+        uint8_t* dst = dec + (pd.dst_off & OFF_MASK);
+        const uint32_t clen = pd.comp_len;
+        int err = 0;
+        uint32_t pos = 0;
+        const uint32_t ulen = snappy_varint(src, pos, clen, &err);
+        if (err || ulen != pd.uncomp_len) {
+            if (lane == 0) atomicAdd(err_flag, 1ull);
+            continue;
+        }
+        uint32_t d = 0;
+        while (pos < clen && d < ulen && !err) {
+            const uint8_t tag = src[pos++];
+            const uint32_t kind = tag & 3u;
+            if (kind == 0) {  // literal
+                uint32_t len = (tag >> 2) + 1;
+                if (len > 60) {
+                    const uint32_t nb = len - 60;
+                    if (pos + nb > clen) { err = 1; break; }
+                    len = 0;
+                    for (uint32_t i = 0; i < nb; i++)
+                        len |= (uint32_t)src[pos + i] << (8 * i);
+                    len += 1;
+                    pos += nb;
+                }
+                if (pos + len > clen || d + len > ulen) { err = 1; break; }
+                for (uint32_t i = lane; i < len; i += 64)
+                    dst[d + i] = src[pos + i];
+                pos += len;
+                d += len;
+            } else {
+                uint32_t len, off;
+                if (kind == 1) {
+                    len = ((tag >> 2) & 0x7u) + 4;
+                    if (pos >= clen) { err = 1; break; }
+                    off = ((uint32_t)(tag >> 5) << 8) | src[pos];
+                    pos += 1;
+                } else if (kind == 2) {
+                    len = (tag >> 2) + 1;
+                    if (pos + 2 > clen) { err = 1; break; }
+                    off = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
+                    pos += 2;
+                } else {
+                    len = (tag >> 2) + 1;
+                    if (pos + 4 > clen) { err = 1; break; }
+                    off = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8) |
+                          ((uint32_t)src[pos + 2] << 16) |
+                          ((uint32_t)src[pos + 3] << 24);
+                    pos += 4;
+                }
+                if (off == 0 || off > d || d + len > ulen) { err = 1; break; }
+                // drain our own outstanding stores so lanes can read bytes
+                // written by other lanes of this wave (L2-coherent)
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                if (off >= len) {
+                    for (uint32_t i = lane; i < len; i += 64)
+                        dst[d + i] = dst[d - off + i];
+                } else {
+                    for (uint32_t i = lane; i < len; i += 64)
+                        dst[d + i] = dst[d - off + (i % off)];
+                }
+                d += len;
+            }
+        }
+        if ((err || d != ulen) && lane == 0) atomicAdd(err_flag, 1ull);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // DELTA_BINARY_PACKED i64 decode (parquet-format Encodings.md; the encoding
 // the reference's config enables for ts, config.rs:54-75). One workgroup per
 // page: lane-serial header walk, parallel miniblock bit-unpack, hierarchical
@@ -834,6 +934,18 @@ hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                                unsigned long long* err_flag) {
     uint32_t grid = n_pages > 4096 ? 4096 : n_pages;
     hipLaunchKernelGGL(k_decode_delta_i64, dim3(grid), dim3(256), 0, s,
+                       blob, dec, pages, n_pages, err_flag);
+    return hipGetLastError();
+}
+
+hipError_t launch_snappy(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                         const SnappyPageDesc* pages, uint32_t n_pages,
+                         unsigned long long* err_flag) {
+    uint32_t waves_needed = n_pages;
+    uint32_t blocks = (waves_needed + 3) / 4;  // 4 waves per 256-thread block
+    if (blocks > 8192) blocks = 8192;
+    if (blocks == 0) blocks = 1;
+    hipLaunchKernelGGL(k_snappy_decompress, dim3(blocks), dim3(256), 0, s,
                        blob, dec, pages, n_pages, err_flag);
     return hipGetLastError();
 }

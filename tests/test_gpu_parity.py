@@ -60,6 +60,14 @@ def ds_plain(tmp_path_factory):
 
 
 @pytest.fixture(scope="module")
+def ds_snappy(tmp_path_factory):
+    out = str(tmp_path_factory.mktemp("snappy"))
+    m = gen_dataset(out, n_rows=200_000, n_series=1_000, n_ssts=4, seed=44,
+                    compression="snappy")
+    return out, m
+
+
+@pytest.fixture(scope="module")
 def ds_delta(tmp_path_factory):
     out = str(tmp_path_factory.mktemp("delta"))
     m = gen_dataset(out, n_rows=200_000, n_series=1_000, n_ssts=4, seed=43,
@@ -108,6 +116,20 @@ def test_delta_encoded_timestamps(ds_delta):
     out, m = ds_delta
     check_parity(out, middle_range(m), ops=OPS_ALL)
     check_parity(out, (0, 2**62), ops=AGG_SUM | AGG_COUNT)
+
+
+def test_snappy_pages(ds_snappy):
+    # the reference's DEFAULT codec (config.rs:120-133): GPU decompress
+    out, m = ds_snappy
+    check_parity(out, middle_range(m), ops=OPS_ALL)
+    check_parity(out, (0, 2**62), ops=AGG_SUM | AGG_COUNT)
+
+
+def test_snappy_delta_combined(tmp_path_factory):
+    out = str(tmp_path_factory.mktemp("snapdelta"))
+    m = gen_dataset(out, n_rows=100_000, n_series=500, n_ssts=2, seed=45,
+                    compression="snappy", ts_encoding="DELTA_BINARY_PACKED")
+    check_parity(out, middle_range(m), ops=AGG_SUM | AGG_COUNT)
 
 
 # ---------------------------------------------------------------------------
