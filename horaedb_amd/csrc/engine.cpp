@@ -942,6 +942,90 @@ uint32_t kernel_ops(uint32_t ops) {
     return k;
 }
 
+hx_status ensure_decoded(DevPlan& plan, hipStream_t s) {
+    if (plan.decoded) return HX_OK;
+    hipEvent_t d0, d1;
+    HIP_TRY(hipEventCreate(&d0));
+    HIP_TRY(hipEventCreate(&d1));
+    HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
+    HIP_TRY(hipEventRecord(d0, s));
+    if (!plan.snappy_pages.empty())
+        HIP_TRY(hx::launch_snappy(s, plan.d_blob, plan.d_dec, plan.d_snappy,
+                                  (uint32_t)plan.snappy_pages.size(),
+                                  plan.d_counters + 1));
+    if (!plan.delta_pages.empty())
+        HIP_TRY(hx::launch_decode_delta(s, plan.d_blob, plan.d_dec,
+                                        plan.d_delta,
+                                        (uint32_t)plan.delta_pages.size(),
+                                        plan.d_counters + 1));
+    if (!plan.copies.empty())
+        HIP_TRY(hx::launch_copy_u64(s, plan.d_blob, plan.d_dec, plan.d_copies,
+                                    (uint32_t)plan.copies.size()));
+    HIP_TRY(hipEventRecord(d1, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    float ms = 0;
+    HIP_TRY(hipEventElapsedTime(&ms, d0, d1));
+    plan.decode_ms = ms;
+    unsigned long long decode_err = 0;
+    HIP_TRY(hipMemcpy(&decode_err, plan.d_counters + 1, 8,
+                      hipMemcpyDeviceToHost));
+    hipEventDestroy(d0);
+    hipEventDestroy(d1);
+    if (decode_err)
+        return fail(HX_ERR_FORMAT, "page decode failed (malformed snappy "
+                                   "stream or delta page)");
+    plan.decoded = true;
+    return HX_OK;
+}
+
+hx_status ensure_sset(hx_prepared* P, DevPlan& plan) {
+    if (P->sset_keys.empty() || plan.d_sset) return HX_OK;
+    std::vector<uint64_t> keys(P->sset_keys);
+    std::sort(keys.begin(), keys.end());
+    keys.erase(std::unique(keys.begin(), keys.end()), keys.end());
+    uint64_t empty = ~0ull;
+    while (std::binary_search(keys.begin(), keys.end(), empty)) empty--;
+    uint32_t cap = next_pow2_u32(keys.size() * 2 + 16);
+    std::vector<uint64_t> table(cap, empty);
+    auto mix = [](uint64_t x) {
+        x += 0x9E3779B97F4A7C15ull;
+        x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+        x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+        return x ^ (x >> 31);
+    };
+    for (uint64_t kk : keys) {
+        uint32_t i = (uint32_t)mix(kk) & (cap - 1);
+        while (table[i] != empty) i = (i + 1) & (cap - 1);
+        table[i] = kk;
+    }
+    HIP_TRY(hipMalloc((void**)&plan.d_sset, size_t(cap) * 8));
+    HIP_TRY(hipMemcpy(plan.d_sset, table.data(), size_t(cap) * 8,
+                      hipMemcpyHostToDevice));
+    plan.sset_mask = cap - 1;
+    plan.sset_empty = empty;
+    return HX_OK;
+}
+
+// builds the filter/dedup AggParams shared by the aggregate and streaming
+// kernels (table fields zeroed for streaming use)
+static hx::AggParams base_params(hx_prepared* P, DevPlan& plan) {
+    hx::AggParams A{};
+    A.rgs = plan.d_rgs;
+    A.n_rgs = (uint32_t)plan.rgs.size();
+    A.ssts = plan.d_ssts;
+    A.clusters = plan.d_clusters;
+    A.cluster_members = plan.d_members;
+    A.blob = plan.d_blob;
+    A.dec = plan.d_dec;
+    A.ts_lo = P->spec.range.start;
+    A.ts_hi = P->spec.range.end;
+    A.sset = plan.d_sset;
+    A.sset_mask = plan.sset_mask;
+    A.sset_empty = plan.sset_empty;
+    A.use_sset = plan.d_sset ? 1 : 0;
+    return A;
+}
+
 hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                     HostTable& out, double* agg_kernel_ms,
                     unsigned long long* matched_out) {
@@ -950,69 +1034,11 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     const uint32_t ops = kernel_ops(agg->ops);
     const bool bucket = agg->bucket_ms > 0;
 
-    // series-set predicate upload (once per prepared)
-    if (!P->sset_keys.empty() && !plan.d_sset) {
-        // open hash set; sentinel = a value not in the set
-        std::vector<uint64_t> keys(P->sset_keys);
-        std::sort(keys.begin(), keys.end());
-        keys.erase(std::unique(keys.begin(), keys.end()), keys.end());
-        uint64_t empty = ~0ull;
-        while (std::binary_search(keys.begin(), keys.end(), empty)) empty--;
-        uint32_t cap = next_pow2_u32(keys.size() * 2 + 16);
-        std::vector<uint64_t> table(cap, empty);
-        auto mix = [](uint64_t x) {
-            x += 0x9E3779B97F4A7C15ull;
-            x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
-            x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
-            return x ^ (x >> 31);
-        };
-        for (uint64_t k : keys) {
-            uint32_t i = (uint32_t)mix(k) & (cap - 1);
-            while (table[i] != empty) i = (i + 1) & (cap - 1);
-            table[i] = k;
-        }
-        HIP_TRY(hipMalloc((void**)&plan.d_sset, size_t(cap) * 8));
-        HIP_TRY(hipMemcpy(plan.d_sset, table.data(), size_t(cap) * 8,
-                          hipMemcpyHostToDevice));
-        plan.sset_mask = cap - 1;
-        plan.sset_empty = empty;
-    }
+    hx_status sst_ = ensure_sset(P, plan);
+    if (sst_ != HX_OK) return sst_;
 
-    // decode passes (persist across exec calls)
-    if (!plan.decoded) {
-        hipEvent_t d0, d1;
-        HIP_TRY(hipEventCreate(&d0));
-        HIP_TRY(hipEventCreate(&d1));
-        HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
-        HIP_TRY(hipEventRecord(d0, s));
-        if (!plan.snappy_pages.empty())
-            HIP_TRY(hx::launch_snappy(s, plan.d_blob, plan.d_dec, plan.d_snappy,
-                                      (uint32_t)plan.snappy_pages.size(),
-                                      plan.d_counters + 1));
-        if (!plan.delta_pages.empty())
-            HIP_TRY(hx::launch_decode_delta(s, plan.d_blob, plan.d_dec,
-                                            plan.d_delta,
-                                            (uint32_t)plan.delta_pages.size(),
-                                            plan.d_counters + 1));
-        if (!plan.copies.empty())
-            HIP_TRY(hx::launch_copy_u64(s, plan.d_blob, plan.d_dec,
-                                        plan.d_copies,
-                                        (uint32_t)plan.copies.size()));
-        HIP_TRY(hipEventRecord(d1, s));
-        HIP_TRY(hipStreamSynchronize(s));
-        float ms = 0;
-        HIP_TRY(hipEventElapsedTime(&ms, d0, d1));
-        plan.decode_ms = ms;
-        unsigned long long decode_err = 0;
-        HIP_TRY(hipMemcpy(&decode_err, plan.d_counters + 1, 8,
-                          hipMemcpyDeviceToHost));
-        hipEventDestroy(d0);
-        hipEventDestroy(d1);
-        if (decode_err)
-            return fail(HX_ERR_FORMAT, "page decode failed (malformed snappy "
-                                       "stream or delta page)");
-        plan.decoded = true;
-    }
+    hx_status dst_ = ensure_decoded(plan, s);
+    if (dst_ != HX_OK) return dst_;
 
     // one-CAS key claim: series-only grouping with a stats-proven sentinel
     const int32_t key_claim = (!bucket && P->key_claim_safe) ? 1 : 0;
@@ -1395,9 +1421,125 @@ extern "C" hx_status hx_scan_agg(hx_handle* h, const hx_scan_spec* spec,
 extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
                              const hx_device_set* devs, hx_batch_cb cb,
                              void* ctx) {
-    (void)h; (void)spec; (void)devs; (void)cb; (void)ctx;
-    return fail(HX_ERR_UNSUPPORTED,
-                "hx_scan streaming parity mode: implemented later this round");
+    // Streaming parity mode: the merged, deduplicated
+    // row stream of ColumnarStorage::scan (storage.rs:335-370): segments in
+    // ascending time order, rows sorted by (series_id, timestamp) within
+    // each segment (the per-segment SortPreservingMerge + MergeExec output,
+    // read.rs:429-494). GPU: filter/dedup append + radix sorts; single
+    // device.
+    if (!h || !spec || !cb) return fail(HX_ERR_INVALID, "null argument");
+    int32_t dev0 = (devs && devs->device_ids && devs->n_devices > 0)
+                       ? devs->device_ids[0] : 0;
+    hx_device_set one{&dev0, 1};
+    hx_prepared* P = nullptr;
+    hx_status st = hx_prepare(h, spec, &one, &P);
+    if (st != HX_OK) return st;
+    std::unique_ptr<hx_prepared, void (*)(hx_prepared*)> guard(
+        P, hx_prepared_free);
+    DevPlan& plan = P->plans[0];
+    HIP_TRY(hipSetDevice(plan.device));
+    hipStream_t s = plan.stream;
+    st = ensure_decoded(plan, s);
+    if (st != HX_OK) return st;
+    st = ensure_sset(P, plan);
+    if (st != HX_OK) return st;
+
+    const uint64_t cap = (uint64_t)plan.rows_scanned;
+    if (cap == 0) return HX_OK;
+    // device buffers: 3 columns + sort keys/scratch + 3 perms
+    size_t need = cap * 8 * 5 + cap * 4 * 3 + 64;
+    st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
+    if (st != HX_OK) return st;
+    uint8_t* base = (uint8_t*)plan.d_scratch;
+    auto carve8 = [&](size_t count) {
+        uint8_t* p = base;
+        base += (count * 8 + 7) & ~size_t(7);
+        return p;
+    };
+    uint64_t* d_series = (uint64_t*)carve8(cap);
+    long long* d_ts = (long long*)carve8(cap);
+    double* d_val = (double*)carve8(cap);
+    uint64_t* d_keys = (uint64_t*)carve8(cap);
+    uint64_t* d_keys_out = (uint64_t*)carve8(cap);
+    unsigned long long* d_cursor = (unsigned long long*)carve8(1);
+    uint32_t* perm_a = (uint32_t*)base; base += cap * 4;
+    uint32_t* perm_b = (uint32_t*)base; base += cap * 4;
+    uint32_t* perm_c = (uint32_t*)base; base += cap * 4;
+
+    HIP_TRY(hipMemsetAsync(d_cursor, 0, 8, s));
+    hx::AggParams A = base_params(P, plan);
+    HIP_TRY(hx::launch_scan_rows(s, A, 0, A.n_rgs, d_series, d_ts, d_val,
+                                 d_cursor, cap));
+    unsigned long long n64 = 0;
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(&n64, d_cursor, 8, hipMemcpyDeviceToHost));
+    if (n64 > cap)
+        return fail(HX_ERR_HIP, "scan row buffer overflow");
+    const uint32_t n = (uint32_t)n64;
+    if (n == 0) return HX_OK;
+
+    // LSD-stable ordering: ts, then series, then time segment
+    HIP_TRY(hx::launch_iota(s, perm_a, n));
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_ts, perm_a,
+                                  (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_a, perm_b, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_series,
+                                  perm_b, (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_b, perm_c, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    // segment keys from the (permuted) ts values
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_ts, perm_c,
+                                  (unsigned long long*)d_keys_out, n));
+    HIP_TRY(hx::launch_seg_keys(s, (const long long*)d_keys_out, h->segment_ms,
+                                (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_c, perm_a, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    const uint32_t* perm = perm_a;
+
+    // gather the three columns by the final permutation, one D2H
+    const unsigned long long* srcs[3] = {
+        (const unsigned long long*)d_series, (const unsigned long long*)d_ts,
+        (const unsigned long long*)d_val};
+    // reuse keys buffers as gather dst (need 3n; keys has 2n) — allocate
+    std::vector<uint64_t> host(3 * size_t(n));
+    unsigned long long* d_dst = nullptr;
+    HIP_TRY(hipMalloc((void**)&d_dst, 3 * size_t(n) * 8));
+    HIP_TRY(hx::launch_gather_multi(s, srcs, 3, perm, d_dst, n));
+    HIP_TRY(hipMemcpyAsync(host.data(), d_dst, 3 * size_t(n) * 8,
+                           hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    hipFree(d_dst);
+
+    // projection over user columns (storage.rs:65-70; builtins stripped as
+    // MergeStream does, read.rs:330-343)
+    std::vector<int32_t> proj;
+    if (spec->projection && spec->n_projection) {
+        for (size_t i = 0; i < spec->n_projection; i++) {
+            int32_t c = spec->projection[i];
+            if (c < 0 || c > 2)
+                return fail(HX_ERR_INVALID, "projection index out of range");
+            proj.push_back(c);
+        }
+    } else {
+        proj = {0, 1, 2};
+    }
+    static const int32_t kTypes[3] = {0, 1, 2};  // u64, i64, f64
+    const uint64_t* cols_base[3] = {host.data(), host.data() + n,
+                                    host.data() + 2 * size_t(n)};
+    const uint32_t BATCH = 65536;
+    std::vector<const void*> colptrs(proj.size());
+    std::vector<int32_t> coltypes(proj.size());
+    for (size_t i = 0; i < proj.size(); i++) coltypes[i] = kTypes[proj[i]];
+    for (uint32_t off = 0; off < n; off += BATCH) {
+        uint32_t len = std::min(BATCH, n - off);
+        for (size_t i = 0; i < proj.size(); i++)
+            colptrs[i] = (const void*)(cols_base[proj[i]] + off);
+        hx_col_batch b{len, proj.size(), colptrs.data(), coltypes.data()};
+        if (cb(ctx, &b)) break;
+    }
+    return HX_OK;
 }
 
 // introspection used by CPU tests (no GPU needed): per-SST catalog entries

@@ -29,6 +29,11 @@ struct GangParams;  // kernels.hip internal; caller provides param buffers
 hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
                                 uint32_t gang_size, bool minmax,
                                 GangParams* h_params, GangParams* d_params);
+hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
+                            uint32_t rg_first, uint32_t rg_last,
+                            uint64_t* out_series, long long* out_ts,
+                            double* out_value, unsigned long long* cursor,
+                            unsigned long long cap);
 hipError_t launch_gather_multi(hipStream_t s,
                                const unsigned long long* const* srcs,
                                uint32_t n_arrays, const uint32_t* perm,
@@ -42,6 +47,8 @@ hipError_t launch_gather_u64(hipStream_t s, const unsigned long long* in,
 hipError_t launch_avg(hipStream_t s, const double* sum,
                       const unsigned long long* cnt, double* avg, uint32_t n);
 hipError_t launch_iota(hipStream_t s, uint32_t* out, uint32_t n);
+hipError_t launch_seg_keys(hipStream_t s, const long long* ts, long long seg_ms,
+                           unsigned long long* keys, uint32_t n);
 hipError_t launch_xor_sign(hipStream_t s, unsigned long long* buf, uint32_t n);
 
 // rocPRIM stable LSD radix sort: sorts values (u32 perm) by u64 keys.

@@ -526,6 +526,46 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
 }
 
 // ---------------------------------------------------------------------------
+// Streaming parity mode (hx_scan): emit the filtered, deduplicated rows
+// themselves. Appends survivors (series, ts, value) unordered; the host then
+// radix-sorts by (series, ts) — MergeStream's PK order (equal PKs cannot
+// survive dedup, so no stability requirement remains).
+// ---------------------------------------------------------------------------
+struct ScanRowsParams {
+    AggParams P;              // filter/dedup context (table unused)
+    uint32_t rg_first, rg_last;   // process rgs [first, last) (one segment)
+    uint64_t* out_series;
+    long long* out_ts;
+    double* out_value;
+    unsigned long long* cursor;
+    unsigned long long cap;
+};
+
+extern "C" __global__ void __launch_bounds__(256)
+k_scan_rows(ScanRowsParams R) {
+    const AggParams& P = R.P;
+    for (uint32_t rgi = R.rg_first + blockIdx.x; rgi < R.rg_last;
+         rgi += gridDim.x) {
+        const RgDesc rg = P.rgs[rgi];
+        const uint64_t* S = (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+        const int64_t* T = (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+        const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+        const SstDev sst = P.ssts[rg.sst_id];
+        const uint32_t n = rg.n_rows;
+        for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
+            const int64_t t = T[r];
+            const uint64_t s = S[r];
+            if (!row_alive(P, rg, sst, S, T, r, n, s, t)) continue;
+            unsigned long long j = atomicAdd(R.cursor, 1ull);
+            if (j >= R.cap) continue;  // host re-runs with a larger buffer
+            R.out_series[j] = s;
+            R.out_ts[j] = t;
+            R.out_value[j] = V[r];
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Result compaction: live slots -> dense arrays (unsorted; host sorts with
 // rocPRIM then gathers).
 // ---------------------------------------------------------------------------
@@ -918,6 +958,18 @@ k_decode_delta_i64(const uint8_t* blob, uint8_t* dec,
 namespace hx {
 
 extern "C" __global__ void __launch_bounds__(256)
+k_seg_keys(const long long* ts, long long seg_ms, unsigned long long* keys,
+           uint32_t n) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += blockDim.x * gridDim.x) {
+        long long t = ts[i];
+        long long q = t / seg_ms;
+        if ((t % seg_ms) != 0 && t < 0) q--;
+        keys[i] = (unsigned long long)q ^ 0x8000000000000000ull;
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
 k_iota(uint32_t* out, uint32_t n) {
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += blockDim.x * gridDim.x)
@@ -1006,6 +1058,26 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     return hipGetLastError();
 }
 
+hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
+                            uint32_t rg_first, uint32_t rg_last,
+                            uint64_t* out_series, long long* out_ts,
+                            double* out_value, unsigned long long* cursor,
+                            unsigned long long cap) {
+    ScanRowsParams R;
+    R.P = p;
+    R.rg_first = rg_first;
+    R.rg_last = rg_last;
+    R.out_series = out_series;
+    R.out_ts = out_ts;
+    R.out_value = out_value;
+    R.cursor = cursor;
+    R.cap = cap;
+    uint32_t n = rg_last - rg_first;
+    hipLaunchKernelGGL(k_scan_rows, dim3(n > 4096 ? 4096 : (n ? n : 1)),
+                       dim3(256), 0, s, R);
+    return hipGetLastError();
+}
+
 hipError_t launch_gather_multi(hipStream_t s,
                                const unsigned long long* const* srcs,
                                uint32_t n_arrays, const uint32_t* perm,
@@ -1064,6 +1136,13 @@ k_xor_sign(unsigned long long* buf, uint32_t n) {
 
 hipError_t launch_xor_sign(hipStream_t s, unsigned long long* buf, uint32_t n) {
     hipLaunchKernelGGL(k_xor_sign, dim3(grid_for(n, 256)), dim3(256), 0, s, buf, n);
+    return hipGetLastError();
+}
+
+hipError_t launch_seg_keys(hipStream_t s, const long long* ts, long long seg_ms,
+                           unsigned long long* keys, uint32_t n) {
+    hipLaunchKernelGGL(k_seg_keys, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       ts, seg_ms, keys, n);
     return hipGetLastError();
 }
 

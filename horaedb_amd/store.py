@@ -61,6 +61,12 @@ class _ResultTable(C.Structure):
                 ("avg", C.POINTER(C.c_double))]
 
 
+class _ColBatch(C.Structure):
+    _fields_ = [("n_rows", C.c_size_t), ("n_cols", C.c_size_t),
+                ("cols", C.POINTER(C.c_void_p)),
+                ("col_types", C.POINTER(C.c_int32))]
+
+
 class _ExecStats(C.Structure):
     _fields_ = [("exec_ms", C.c_double), ("agg_kernel_ms", C.c_double),
                 ("decode_kernel_ms", C.c_double), ("rows_scanned", C.c_int64),
@@ -90,6 +96,8 @@ def _load():
                                 C.POINTER(_AggSpec), C.POINTER(_DeviceSet),
                                 C.POINTER(C.POINTER(_ResultTable))]
     lib.hx_get_stats.argtypes = [C.c_void_p, C.POINTER(_ExecStats)]
+    lib.hx_scan.argtypes = [C.c_void_p, C.POINTER(_ScanSpec),
+                            C.POINTER(_DeviceSet), C.c_void_p, C.c_void_p]
     lib.hx_catalog_size.argtypes = [C.c_void_p, C.POINTER(C.c_size_t)]
     lib.hx_catalog_entry.argtypes = [C.c_void_p, C.c_size_t,
                                      C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
@@ -251,3 +259,44 @@ class Store:
         """One-shot scan+aggregate (prepare + exec + release)."""
         with self.prepare(ts_range, series_in, devices) as p:
             return p.exec_agg(ops=ops, bucket_ms=bucket_ms)
+
+    def scan(self, ts_range, series_in=None, projection=None, devices=None):
+        """Streaming parity mode (hx_scan): the merged, deduplicated row
+        stream itself — ColumnarStorage::scan semantics (storage.rs:335-370).
+        Returns dict of concatenated column arrays in stream order."""
+        spec = self._spec(ts_range, series_in)
+        if projection is not None:
+            parr = (C.c_int32 * len(projection))(*projection)
+            spec.projection = parr
+            spec.n_projection = len(projection)
+            self._keepalive.append(parr)
+        ds = self._devset(devices)
+        chunks = []
+
+        _BATCH = C.CFUNCTYPE(C.c_int32, C.c_void_p, C.POINTER(_ColBatch))
+
+        def on_batch(ctx, bp):
+            b = bp.contents
+            cols = []
+            for i in range(b.n_cols):
+                t = b.col_types[i]
+                dt = {0: np.uint64, 1: np.int64, 2: np.float64}[t]
+                ptr = C.cast(b.cols[i], C.POINTER(C.c_uint64))
+                arr = np.ctypeslib.as_array(ptr, shape=(b.n_rows,)).copy()
+                cols.append(arr.view(dt))
+            chunks.append(cols)
+            return 0
+
+        cb = _BATCH(on_batch)
+        _check(_lib.hx_scan(self._h, C.byref(spec),
+                            C.byref(ds) if ds else None,
+                            C.cast(cb, C.c_void_p), None))
+        ncols = len(projection) if projection is not None else 3
+        names = ["series_id", "timestamp", "value"]
+        sel = projection if projection is not None else [0, 1, 2]
+        out = {}
+        for i in range(ncols):
+            parts = [c[i] for c in chunks]
+            out[names[sel[i]]] = (np.concatenate(parts) if parts else
+                                  np.empty(0))
+        return out

@@ -26,4 +26,5 @@ from .scan import (  # noqa: F401
     read_sst,
     truncate_by,
     fill_required_projections,
+    scan_rows,
 )

@@ -280,3 +280,30 @@ def fill_required_projections(projection, num_primary_keys, seq_idx):
     if seq_idx not in proj:
         proj.append(seq_idx)
     return proj
+
+
+def scan_rows(ssts, ts_range, series_set=None, segment_ms=12 * 3600 * 1000):
+    """The merged, deduplicated row STREAM (hx_scan parity): segments in
+    ascending time order (union of per-segment plans, storage.rs:343-368),
+    rows sorted by (series_id, timestamp) within each segment."""
+    lo, hi = ts_range
+    sset = None
+    if series_set is not None:
+        sset = np.asarray(sorted(set(int(x) for x in series_set)), np.uint64)
+
+    def pred(cols):
+        m = (cols[1] >= lo) & (cols[1] < hi)
+        if sset is not None:
+            m &= np.isin(cols[0], sset)
+        return m
+
+    merged = merge_scan(ssts, num_primary_keys=2, merge_op=MERGE_LAST,
+                        predicate=pred)
+    if not merged or len(merged[0]) == 0:
+        return {"series_id": np.empty(0, np.uint64),
+                "timestamp": np.empty(0, np.int64),
+                "value": np.empty(0, np.float64)}
+    s, t, v = merged[0], merged[1], merged[2]
+    seg = np.floor_divide(t, np.int64(segment_ms))
+    order = np.lexsort((t, s, seg))
+    return {"series_id": s[order], "timestamp": t[order], "value": v[order]}

@@ -207,3 +207,57 @@ def test_multi_device_if_available(ds_plain):
     exp = oracle.scan_agg(ssts, middle_range(m))
     assert res["series_id"].tolist() == exp["series_id"].tolist()
     np.testing.assert_allclose(res["sum"], exp["sum"], rtol=1e-9)
+
+
+# ---------------------------------------------------------------------------
+# streaming parity mode (hx_scan): row-level merged output
+# ---------------------------------------------------------------------------
+
+def check_scan_rows(store_dir, ts_range, series_in=None, segment_ms=0):
+    from horaedb_amd import Store
+    kw = {"segment_duration_ms": segment_ms} if segment_ms else {}
+    with Store(store_dir, **kw) as st:
+        res = st.scan(ts_range, series_in=series_in, devices=[0])
+    ddir = os.path.join(store_dir, "data")
+    paths = sorted((os.path.join(ddir, f) for f in os.listdir(ddir)
+                    if f.endswith(".sst")),
+                   key=lambda p: int(os.path.basename(p).split(".")[0]))
+    ssts = [oracle.read_sst(p) for p in paths]
+    exp = oracle.scan_rows(ssts, ts_range, series_set=series_in,
+                           segment_ms=segment_ms or 12 * 3600 * 1000)
+    assert res["series_id"].tolist() == exp["series_id"].tolist()
+    assert res["timestamp"].tolist() == exp["timestamp"].tolist()
+    np.testing.assert_array_equal(res["value"], exp["value"])
+    return res
+
+
+def test_scan_rows_basic(ds_plain):
+    out, m = ds_plain
+    check_scan_rows(out, middle_range(m))
+
+
+def test_scan_rows_dedup(tmp_path):
+    store = str(tmp_path)
+    gen_sst_from_arrays(store, 1, [5, 5, 7], [100, 200, 100], [1.0, 2.0, 3.0])
+    gen_sst_from_arrays(store, 2, [5, 7], [200, 100], [20.0, 30.0])
+    res = check_scan_rows(store, (0, 1000))
+    assert res["value"].tolist() == [1.0, 20.0, 30.0]
+
+
+def test_scan_rows_segments(tmp_path):
+    # two time segments: output is PER-SEGMENT sorted (segment-major), not
+    # globally PK-sorted (the reference's per-segment union order)
+    store = str(tmp_path)
+    gen_sst_from_arrays(store, 1, [9, 5], [100, 150], [1.0, 2.0])
+    gen_sst_from_arrays(store, 2, [5, 9], [1100, 1150], [3.0, 4.0])
+    res = check_scan_rows(store, (0, 5000), segment_ms=1000)
+    assert res["series_id"].tolist() == [5, 9, 5, 9]
+    assert res["value"].tolist() == [2.0, 1.0, 3.0, 4.0]
+
+
+def test_scan_rows_projection(ds_plain):
+    from horaedb_amd import Store
+    out, m = ds_plain
+    with Store(out) as st:
+        res = st.scan(middle_range(m), projection=[2, 0], devices=[0])
+    assert set(res.keys()) == {"value", "series_id"}
