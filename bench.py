@@ -190,6 +190,16 @@ def main():
         achieved = algo_bytes / (k_ms * 1e-3)
         peak = 8.0e12
         traffic_env = os.environ.get("HX_ROOFLINE_TRAFFIC")
+        if not traffic_env:
+            # PMC-measured HBM traffic per scanned row (rocprofv3 --pmc, see
+            # profiles/roofline_traffic.json + profiles/README.md)
+            tj = os.path.join(REPO, "profiles", "roofline_traffic.json")
+            if os.path.exists(tj):
+                with open(tj) as f:
+                    tdata = json.load(f).get("config2_1b_rows_ts_range_sum_count")
+                if tdata:
+                    traffic_env = str(tdata["traffic_bytes_per_scanned_row"] *
+                                      my_rows)
         roofline = {
             "bound": "hbm",
             "achieved": achieved / 1e9,
