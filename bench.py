@@ -99,6 +99,8 @@ def main():
     p.add_argument("--ts-encoding", default="PLAIN")
     p.add_argument("--range-frac", type=float, default=0.5)
     p.add_argument("--bucket-ms", type=int, default=0)
+    p.add_argument("--selectivity", type=float, default=0,
+                   help="config 3: series-set predicate keeping this fraction")
     p.add_argument("--ops", default="sum,count")
     p.add_argument("--data-dir", default="/tmp/hx_bench_data")
     p.add_argument("--no-cpu-baseline", action="store_true")
@@ -129,13 +131,21 @@ def main():
     from tools.gen_ssts import middle_range
     store_dir, m = get_dataset(args, rank)
     ts_range = middle_range(m, args.range_frac)
+    series_in = None
+    if args.selectivity > 0:
+        ids = np.load(os.path.join(store_dir, "series_ids.npy"))
+        k = max(1, int(len(ids) * args.selectivity))
+        rng = np.random.default_rng(args.seed)
+        series_in = rng.choice(ids, size=k, replace=False).tolist()
+        log(f"rank{rank}: series-set predicate with {k} ids "
+            f"({args.selectivity:.2%})")
 
     device = local_rank
     torch.cuda.set_device(device)
 
     store = Store(store_dir)
     t0 = time.time()
-    prep = store.prepare(ts_range, devices=[device])
+    prep = store.prepare(ts_range, series_in=series_in, devices=[device])
     log(f"rank{rank}: staged in {time.time() - t0:.1f}s")
 
     def step(copy=False):
