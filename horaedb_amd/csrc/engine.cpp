@@ -271,6 +271,8 @@ struct DevPlan {
     unsigned long long* t_cnt = nullptr;
     unsigned long long* t_min = nullptr;
     unsigned long long* t_max = nullptr;
+    uint8_t* t_slab = nullptr;      // AoS key-claim table
+    uint32_t slab_stride = 0;
     unsigned long long* d_counters = nullptr;  // fill, overflow, matched, n_out
     uint64_t* d_sset = nullptr;
     size_t sset_cap = 0;
@@ -880,7 +882,8 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
                         (void*)plan.d_copies, (void*)plan.t_series,
                         (void*)plan.t_bucket, (void*)plan.t_state,
                         (void*)plan.t_sum, (void*)plan.t_cnt, (void*)plan.t_min,
-                        (void*)plan.t_max, (void*)plan.d_counters,
+                        (void*)plan.t_max, (void*)plan.t_slab,
+                        (void*)plan.d_counters,
                         (void*)plan.d_sset})
             if (p) hipFree(p);
         if (plan.stream) hipStreamDestroy(plan.stream);
@@ -920,14 +923,25 @@ uint32_t next_pow2_u32(uint64_t x) {
     return p;
 }
 
-hx_status alloc_table(DevPlan& plan, uint32_t slots, uint32_t ops, bool bucket) {
-    if (plan.slots == slots && plan.t_series) return HX_OK;
+hx_status alloc_table(DevPlan& plan, uint32_t slots, uint32_t ops, bool bucket,
+                      bool key_claim) {
+    uint32_t stride = key_claim
+                          ? ((ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 64u : 32u)
+                          : 0u;
+    if (plan.slots == slots && (key_claim ? (plan.t_slab && plan.slab_stride == stride)
+                                          : plan.t_series != nullptr))
+        return HX_OK;
     for (void** p : {(void**)&plan.t_series, (void**)&plan.t_bucket,
                      (void**)&plan.t_state, (void**)&plan.t_sum,
                      (void**)&plan.t_cnt, (void**)&plan.t_min,
-                     (void**)&plan.t_max})
+                     (void**)&plan.t_max, (void**)&plan.t_slab})
         if (*p) { hipFree(*p); *p = nullptr; }
     plan.slots = slots;
+    plan.slab_stride = stride;
+    if (key_claim) {
+        HIP_TRY(hipMalloc((void**)&plan.t_slab, size_t(slots) * stride));
+        return HX_OK;
+    }
     HIP_TRY(hipMalloc((void**)&plan.t_series, size_t(slots) * 8));
     HIP_TRY(hipMalloc((void**)&plan.t_state, size_t(slots) * 4));
     if (bucket) HIP_TRY(hipMalloc((void**)&plan.t_bucket, size_t(slots) * 8));
@@ -1062,17 +1076,19 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
 
     unsigned long long counters[4];
     for (int attempt = 0; attempt < 4; attempt++) {
-        hx_status st = alloc_table(plan, slots, ops, bucket);
+        hx_status st = alloc_table(plan, slots, ops, bucket, key_claim);
         if (st != HX_OK) return st;
         // reset table + counters (part of the step)
-        if (key_claim)
-            HIP_TRY(hipMemsetAsync(plan.t_series, 0xFF, size_t(slots) * 8, s));
-        else
+        if (key_claim) {
+            HIP_TRY(hx::launch_init_slab(s, plan.t_slab, slots,
+                                         plan.slab_stride));
+        } else {
             HIP_TRY(hipMemsetAsync(plan.t_state, 0, size_t(slots) * 4, s));
-        if (plan.t_sum) HIP_TRY(hipMemsetAsync(plan.t_sum, 0, size_t(slots) * 8, s));
-        if (plan.t_cnt) HIP_TRY(hipMemsetAsync(plan.t_cnt, 0, size_t(slots) * 8, s));
-        if (plan.t_min) HIP_TRY(hipMemsetAsync(plan.t_min, 0xFF, size_t(slots) * 8, s));
-        if (plan.t_max) HIP_TRY(hipMemsetAsync(plan.t_max, 0, size_t(slots) * 8, s));
+            if (plan.t_sum) HIP_TRY(hipMemsetAsync(plan.t_sum, 0, size_t(slots) * 8, s));
+            if (plan.t_cnt) HIP_TRY(hipMemsetAsync(plan.t_cnt, 0, size_t(slots) * 8, s));
+            if (plan.t_min) HIP_TRY(hipMemsetAsync(plan.t_min, 0xFF, size_t(slots) * 8, s));
+            if (plan.t_max) HIP_TRY(hipMemsetAsync(plan.t_max, 0, size_t(slots) * 8, s));
+        }
         HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
 
         hx::AggParams A{};
@@ -1094,7 +1110,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.key_claim = key_claim;
         A.skip = getenv("HX_SKIP") ? atoi(getenv("HX_SKIP")) : 0;
         A.table = {plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
-                   plan.t_cnt, plan.t_min, plan.t_max, slots - 1};
+                   plan.t_cnt, plan.t_min, plan.t_max, slots - 1,
+                   plan.t_slab, plan.slab_stride};
         A.fill = plan.d_counters + 0;
         A.overflow = plan.d_counters + 1;
         A.matched = plan.d_counters + 2;
@@ -1169,10 +1186,13 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     out.n = fill;
     if (fill == 0) return HX_OK;
     const uint32_t n = (uint32_t)fill;
-    const bool has_sum = plan.t_sum != nullptr;
-    const bool has_cnt = plan.t_cnt != nullptr;
-    const bool has_min = plan.t_min != nullptr;
-    const bool has_max = plan.t_max != nullptr;
+    const bool aos = plan.t_slab != nullptr;
+    const bool has_sum = aos ? (ops & (HX_AGG_SUM | HX_AGG_AVG)) != 0
+                             : plan.t_sum != nullptr;
+    const bool has_cnt = aos ? (ops & (HX_AGG_COUNT | HX_AGG_AVG)) != 0
+                             : plan.t_cnt != nullptr;
+    const bool has_min = aos ? (ops & HX_AGG_MIN) != 0 : plan.t_min != nullptr;
+    const bool has_max = aos ? (ops & HX_AGG_MAX) != 0 : plan.t_max != nullptr;
     const bool has_avg = (agg->ops & HX_AGG_AVG) != 0;
     const uint32_t n_core = 1 + (bucket ? 1 : 0) + has_sum + has_cnt +
                             has_min + has_max;
@@ -1206,7 +1226,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     HIP_TRY(hipMemsetAsync(d_nout, 0, 8, s));
     hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
     hx::AggTable T{plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
-                   plan.t_cnt, plan.t_min, plan.t_max, plan.slots - 1};
+                   plan.t_cnt, plan.t_min, plan.t_max, plan.slots - 1,
+                   plan.t_slab, plan.slab_stride};
     HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,
                                bucket ? agg->bucket_ms : 0, co));
 

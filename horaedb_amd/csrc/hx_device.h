@@ -58,6 +58,10 @@ struct AggTable {
     unsigned long long* vmin;  // ordered-u64 mapped f64 (bit-exact min/max)
     unsigned long long* vmax;
     uint32_t  mask;            // slots-1 (power of two)
+    // key-claim AoS mode: slot i = slab[i*stride .. ) = {key, sum, cnt
+    // [, min, max]} — probe + update touch ONE cache line instead of three
+    uint8_t*  slab;            // non-null => AoS mode (series-only grouping)
+    uint32_t  stride;          // 32 (sum/cnt) or 64 (with min/max)
 };
 
 struct AggParams {

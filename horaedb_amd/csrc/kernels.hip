@@ -95,22 +95,24 @@ __device__ __forceinline__ void table_add(const AggParams& P, uint32_t i,
     if (P.ops & HXK_MAX) atomicMax(&P.table.vmax[i], f64_ordered(mx));
 }
 
-// Fast claim path: series_id IS the slot word, claimed by one CAS against
-// KEY_EMPTY (host proves via column statistics that no series == KEY_EMPTY;
-// series-only grouping). One L3 load per probe, no state word.
+// Fast claim path (AoS slab): slot i = {key, sum, cnt[, min, max]} in one
+// cache line. key claimed by one CAS against KEY_EMPTY (host proves via
+// column statistics that no series == KEY_EMPTY; series-only grouping);
+// probe is a plain cached load (a slot transitions KEY_EMPTY -> key exactly
+// once per exec, so a stale read can only be KEY_EMPTY, which the CAS
+// corrects).
 __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s,
                                                   double vsum,
                                                   unsigned long long cnt,
                                                   double mn, double mx) {
+    const uint32_t stride = P.table.stride;
     uint32_t i = (uint32_t)mix64(s) & P.table.mask;
     for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
-        // plain (cached) probe: a slot transitions KEY_EMPTY -> key exactly
-        // once per exec, so a stale read can only be KEY_EMPTY — which falls
-        // through to the CAS (coherent) and learns the truth.
-        uint64_t k = P.table.series[i];
+        uint8_t* slot = P.table.slab + (size_t)i * stride;
+        uint64_t k = *(uint64_t*)slot;
         if (k == KEY_EMPTY) {
             uint64_t expected = KEY_EMPTY;
-            if (__hip_atomic_compare_exchange_strong(&P.table.series[i],
+            if (__hip_atomic_compare_exchange_strong((uint64_t*)slot,
                     &expected, s, RLX, RLX, AGT)) {
                 __hip_atomic_fetch_add(P.fill, 1ull, RLX, AGT);
                 k = s;
@@ -119,12 +121,35 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
             }
         }
         if (k == s) {
-            table_add(P, i, vsum, cnt, mn, mx);
+            if (P.ops & (HXK_SUM | HXK_AVG))
+                atomicAdd((double*)(slot + 8), vsum);
+            if (P.ops & (HXK_COUNT | HXK_AVG))
+                atomicAdd((unsigned long long*)(slot + 16), cnt);
+            if (P.ops & HXK_MIN)
+                atomicMin((unsigned long long*)(slot + 24), f64_ordered(mn));
+            if (P.ops & HXK_MAX)
+                atomicMax((unsigned long long*)(slot + 32), f64_ordered(mx));
             return;
         }
         i = (i + 1) & P.table.mask;
     }
     __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+}
+
+// slab init: key=KEY_EMPTY, sum=0, cnt=0, min=~0 (ordered), max=0
+extern "C" __global__ void __launch_bounds__(256)
+k_init_slab(uint8_t* slab, uint32_t n_slots, uint32_t stride) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n_slots;
+         i += blockDim.x * gridDim.x) {
+        uint8_t* slot = slab + (size_t)i * stride;
+        ((uint64_t*)slot)[0] = KEY_EMPTY;
+        ((uint64_t*)slot)[1] = 0;   // sum
+        ((uint64_t*)slot)[2] = 0;   // cnt
+        if (stride >= 64) {
+            ((uint64_t*)slot)[3] = ~0ull;  // min (ordered)
+            ((uint64_t*)slot)[4] = 0;      // max (ordered)
+        }
+    }
 }
 
 // General path (16-byte keys, e.g. (series,bucket)): claim via a CAS on a
@@ -590,8 +615,10 @@ k_compact(CompactParams C) {
     const uint32_t stride = blockDim.x * gridDim.x;
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
          __any(i < C.n_slots); i += stride) {
+        const uint8_t* slot =
+            C.key_claim ? C.table.slab + (size_t)i * C.table.stride : nullptr;
         const bool live = (i < C.n_slots) &&
-                          (C.key_claim ? (C.table.series[i] != KEY_EMPTY)
+                          (C.key_claim ? (*(const uint64_t*)slot != KEY_EMPTY)
                                        : (C.table.state[i] == 2u));
         const unsigned long long mask = __ballot(live);
         if (!mask) continue;
@@ -604,12 +631,25 @@ k_compact(CompactParams C) {
         if (live) {
             const unsigned long long j =
                 wave_base + __popcll(mask & ((1ull << lane) - 1ull));
-            C.out_series[j] = C.table.series[i];
-            if (C.bucket_ms) C.out_bucket[j] = C.table.bucket[i];
-            if (C.out_sum) C.out_sum[j] = C.table.sum[i];
-            if (C.out_cnt) C.out_cnt[j] = C.table.cnt[i];
-            if (C.out_min) C.out_min[j] = ordered_f64(C.table.vmin[i]);
-            if (C.out_max) C.out_max[j] = ordered_f64(C.table.vmax[i]);
+            if (C.key_claim) {
+                C.out_series[j] = *(const uint64_t*)slot;
+                if (C.out_sum) C.out_sum[j] = *(const double*)(slot + 8);
+                if (C.out_cnt)
+                    C.out_cnt[j] = *(const unsigned long long*)(slot + 16);
+                if (C.out_min)
+                    C.out_min[j] = ordered_f64(
+                        *(const unsigned long long*)(slot + 24));
+                if (C.out_max)
+                    C.out_max[j] = ordered_f64(
+                        *(const unsigned long long*)(slot + 32));
+            } else {
+                C.out_series[j] = C.table.series[i];
+                if (C.bucket_ms) C.out_bucket[j] = C.table.bucket[i];
+                if (C.out_sum) C.out_sum[j] = C.table.sum[i];
+                if (C.out_cnt) C.out_cnt[j] = C.table.cnt[i];
+                if (C.out_min) C.out_min[j] = ordered_f64(C.table.vmin[i]);
+                if (C.out_max) C.out_max[j] = ordered_f64(C.table.vmax[i]);
+            }
         }
     }
 }
@@ -1143,6 +1183,13 @@ hipError_t launch_seg_keys(hipStream_t s, const long long* ts, long long seg_ms,
                            unsigned long long* keys, uint32_t n) {
     hipLaunchKernelGGL(k_seg_keys, dim3(grid_for(n, 256)), dim3(256), 0, s,
                        ts, seg_ms, keys, n);
+    return hipGetLastError();
+}
+
+hipError_t launch_init_slab(hipStream_t s, uint8_t* slab, uint32_t n_slots,
+                            uint32_t stride) {
+    hipLaunchKernelGGL(k_init_slab, dim3(grid_for(n_slots, 256)), dim3(256),
+                       0, s, slab, n_slots, stride);
     return hipGetLastError();
 }
 
