@@ -1062,7 +1062,9 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     if (dst_ != HX_OK) return dst_;
 
     // one-CAS key claim: series-only grouping with a stats-proven sentinel
-    const int32_t key_claim = (!bucket && P->key_claim_safe) ? 1 : 0;
+    // (HX_FORCE_STATE=1 forces the generic state-word path for A/B runs)
+    const int32_t key_claim =
+        (!bucket && P->key_claim_safe && !getenv("HX_FORCE_STATE")) ? 1 : 0;
 
     // table size heuristic; grows on overflow
     uint32_t slots = plan.slots;
