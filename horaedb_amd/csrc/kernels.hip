@@ -236,6 +236,79 @@ __device__ __forceinline__ bool row_alive(const AggParams& P, const RgDesc& rg,
 // filter + MergeExec dedup + hash group-by aggregate. One workgroup per row
 // group, grid-stride; each thread strides rows (coalesced 8B column loads).
 // ---------------------------------------------------------------------------
+// Per-window worker for k_scan_agg: 64 consecutive rows on one wave —
+// loads, filter, dedup, wave-segmented pre-reduction. Returns whether this
+// lane must issue a table update (run head with survivors) via out params.
+struct WinResult {
+    uint64_t s;
+    int64_t b;
+    double vv, mn, mx;
+    unsigned long long c;
+    bool head;
+};
+
+__device__ __forceinline__ void scan_window(const AggParams& P,
+                                            const RgDesc& rg,
+                                            const SstDev& sst,
+                                            const uint64_t* S,
+                                            const int64_t* T, const double* V,
+                                            uint32_t base, uint32_t n,
+                                            int lane,
+                                            unsigned long long& my_matched,
+                                            WinResult& W) {
+    const uint32_t r = base + threadIdx.x;
+    const bool inb = r < n;
+    uint64_t s = KEY_EMPTY;
+    int64_t t = 0;
+    double v = 0.0;
+    bool alive = false;
+    if (inb) {
+        t = T[r];
+        s = S[r];
+        alive = row_alive(P, rg, sst, S, T, r, n, s, t);
+        if (alive) v = V[r];
+    }
+    int64_t b = (inb && P.bucket_ms) ? floordiv(t, P.bucket_ms) : 0;
+    unsigned long long c = alive ? 1ull : 0ull;
+    double vv = alive ? v : 0.0;
+    double mn = alive ? v : HUGE_VAL;
+    double mx = alive ? v : -HUGE_VAL;
+    my_matched += c;
+    const uint64_t sp = __shfl_up(s, 1, 64);
+    const int64_t bp = __shfl_up((long long)b, 1, 64);
+    const bool head = (lane == 0) || sp != s || bp != b;
+    bool done = false;
+    for (int d = 1; d < 64; d++) {
+        const uint64_t s2 = __shfl_down(s, d, 64);
+        const long long b2 = __shfl_down((long long)b, d, 64);
+        const double v2 = __shfl_down(vv, d, 64);
+        const unsigned long long c2 = __shfl_down(c, d, 64);
+        const double mn2 = __shfl_down(mn, d, 64);
+        const double mx2 = __shfl_down(mx, d, 64);
+        done = done || (lane + d >= 64) || s2 != s || b2 != b;
+        if (head && !done) {
+            vv += v2;
+            c += c2;
+            mn = fmin(mn, mn2);
+            mx = fmax(mx, mx2);
+        }
+        if (__all(done)) break;
+    }
+    W.s = s;
+    W.b = b;
+    W.vv = vv;
+    W.mn = mn;
+    W.mx = mx;
+    W.c = c;
+    W.head = head;
+}
+
+// The headline fused kernel: decode(PLAIN in place) + ts-range/series-set
+// filter + MergeExec dedup + hash group-by aggregate. One workgroup per
+// row-group slice, grid-stride; TWO 64-row windows per iteration so two
+// dependent table-update chains (L3 probe + atomics) overlap — the kernel
+// is memory-latency-bound (SQ_WAIT_ANY 76%) at full occupancy, so the
+// lever is per-wave memory-level parallelism.
 extern "C" __global__ void __launch_bounds__(256)
 k_scan_agg(AggParams P) {
     unsigned long long my_matched = 0;
@@ -247,53 +320,26 @@ k_scan_agg(AggParams P) {
         const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
         const SstDev sst = P.ssts[rg.sst_id];
         const uint32_t n = rg.n_rows;
-        for (uint32_t base = 0; base < n; base += blockDim.x) {
-            const uint32_t r = base + threadIdx.x;
-            const bool inb = r < n;
-            uint64_t s = KEY_EMPTY;
-            int64_t t = 0;
-            double v = 0.0;
-            bool alive = false;
-            if (inb) {
-                t = T[r];
-                s = S[r];
-                alive = row_alive(P, rg, sst, S, T, r, n, s, t);
-                if (alive) v = V[r];
+        for (uint32_t base = 0; base < n; base += blockDim.x * 2) {
+            WinResult A, B;
+            B.c = 0;
+            B.head = false;
+            scan_window(P, rg, sst, S, T, V, base, n, lane, my_matched, A);
+            const uint32_t base2 = base + blockDim.x;
+            if (base2 < n)
+                scan_window(P, rg, sst, S, T, V, base2, n, lane, my_matched, B);
+            const bool upA = A.head && A.c > 0;
+            const bool upB = B.head && B.c > 0;
+            if (upA && upB) {  // two independent chains: overlap them
+                agg_update(P, A.s, A.b, A.vv, A.c, A.mn, A.mx);
+                agg_update(P, B.s, B.b, B.vv, B.c, B.mn, B.mx);
+            } else if (upA) {
+                agg_update(P, A.s, A.b, A.vv, A.c, A.mn, A.mx);
+            } else if (upB) {
+                agg_update(P, B.s, B.b, B.vv, B.c, B.mn, B.mx);
             }
-            int64_t b = (inb && P.bucket_ms) ? floordiv(t, P.bucket_ms) : 0;
-            // Wave-level segmented pre-reduction over the sorted rows: lanes
-            // hold 64 consecutive rows; equal (series,bucket) runs are
-            // contiguous (SSTs are PK-sorted), so each run head accumulates
-            // its run and issues ONE table update (guideline 12).
-            unsigned long long c = alive ? 1ull : 0ull;
-            double vv = alive ? v : 0.0;
-            double mn = alive ? v : HUGE_VAL;
-            double mx = alive ? v : -HUGE_VAL;
-            my_matched += c;
-            const uint64_t sp = __shfl_up(s, 1, 64);
-            const int64_t bp = __shfl_up((long long)b, 1, 64);
-            const bool head = (lane == 0) || sp != s || bp != b;
-            bool done = false;
-            for (int d = 1; d < 64; d++) {
-                const uint64_t s2 = __shfl_down(s, d, 64);
-                const long long b2 = __shfl_down((long long)b, d, 64);
-                const double v2 = __shfl_down(vv, d, 64);
-                const unsigned long long c2 = __shfl_down(c, d, 64);
-                const double mn2 = __shfl_down(mn, d, 64);
-                const double mx2 = __shfl_down(mx, d, 64);
-                done = done || (lane + d >= 64) || s2 != s || b2 != b;
-                if (head && !done) {
-                    vv += v2;
-                    c += c2;
-                    mn = fmin(mn, mn2);
-                    mx = fmax(mx, mx2);
-                }
-                if (__all(done)) break;
-            }
-            if (head && c > 0) agg_update(P, s, b, vv, c, mn, mx);
         }
     }
-    // one atomic per wave for the matched counter (guideline 12)
     for (int off = 32; off > 0; off >>= 1)
         my_matched += __shfl_down(my_matched, off, 64);
     if ((threadIdx.x & 63) == 0 && my_matched)
