@@ -44,7 +44,8 @@ def get_dataset(args, rank):
             log(f"rank{rank}: reusing dataset {out}")
             return out, m
     t0 = time.time()
-    workers = min(16, os.cpu_count() or 1)
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    workers = min(16, max(2, (os.cpu_count() or 8) // max(1, world)))
     m = gen_dataset(out, args.rows, args.series, args.ssts,
                     seed=args.seed + rank, compression=args.compression,
                     ts_encoding=args.ts_encoding, workers=workers)

@@ -1133,8 +1133,6 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
             uint32_t gang = 1;
             {
                 std::map<int64_t, int> class_count;
-                for (const auto& sd : plan.ssts) (void)sd;
-                std::map<int64_t, int64_t> class_rows;
                 std::vector<int64_t> per_sst(plan.ssts.size(), 0);
                 for (const auto& rd : plan.rgs) {
                     int64_t end = rd.row_base + rd.n_rows;
