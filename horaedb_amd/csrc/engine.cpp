@@ -242,6 +242,7 @@ struct DevPlan {
     std::vector<int32_t> cluster_members;
     std::vector<hx::DeltaPageDesc> delta_pages;
     std::vector<hx::SnappyPageDesc> snappy_pages;
+    std::vector<hx::RleDictPageDesc> rledict_pages;
     std::vector<hx::CopyDesc> copies;
     size_t blob_bytes = 0;
     size_t dec_bytes = 0;
@@ -260,6 +261,7 @@ struct DevPlan {
     int32_t* d_members = nullptr;
     hx::DeltaPageDesc* d_delta = nullptr;
     hx::SnappyPageDesc* d_snappy = nullptr;
+    hx::RleDictPageDesc* d_rledict = nullptr;
     hx::CopyDesc* d_copies = nullptr;
 
     // aggregate table (lazily sized)
@@ -459,18 +461,17 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 err_flag = 1;
                 break;
             }
-            // round-1 layout contract: exactly one v1/v2 data page per chunk,
-            // no dictionary page, uncompressed (Snappy next)
+            // layout contract: exactly one v1/v2 data page per chunk, plus an
+            // optional dictionary page (RLE_DICTIONARY chunks)
             const hx::PageDesc* dp = nullptr;
+            const hx::PageDesc* dictp = nullptr;
             int n_data = 0;
             for (const auto& p : pages) {
                 if (p.page_type == 0 || p.page_type == 3) {
                     dp = &p;
                     n_data++;
                 } else if (p.page_type == 2) {
-                    std::lock_guard<std::mutex> g(mu);
-                    err_msg = j.ss->cat->path + ": dictionary pages unsupported (round 1)";
-                    err_flag = 1;
+                    dictp = &p;
                 }
             }
             if (err_flag.load()) break;
@@ -533,6 +534,52 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 dec_off = align64(dec_off + size_t(j.num_values) * 8);
                 plan.delta_pages.push_back(dd);
                 j.final_off = dd.dst_off;
+            } else if ((dp->encoding == hx::ENC_RLE_DICTIONARY ||
+                        dp->encoding == hx::ENC_PLAIN_DICTIONARY) && dictp) {
+                // dict payload was copied at dst_off together with the data
+                // payload (the whole-chunk staging copy keeps page payloads
+                // at their in-chunk offsets) — recompute both offsets
+                std::lock_guard<std::mutex> g(mu);
+                hx::RleDictPageDesc rd{};
+                size_t dict_in_chunk = size_t(dictp->payload_off - j.chunk_start);
+                size_t dict_payload = size_t(dictp->compressed_size);
+                // re-copy the dictionary payload right after the data payload
+                size_t dict_dst = j.dst_off + ((payload + 15) & ~size_t(15));
+                if (dict_dst + dict_payload >
+                    j.dst_off + size_t(j.comp_size) + 64) {
+                    err_msg = j.ss->cat->path + ": dictionary staging overflow";
+                    err_flag = 1;
+                    break;
+                }
+                std::memcpy(plan.h_blob + dict_dst, tmp.data() + dict_in_chunk,
+                            dict_payload);
+                uint64_t dict_data_off = dict_dst;
+                uint32_t dict_raw = (uint32_t)dict_payload;
+                if (j.codec == hx::CODEC_SNAPPY) {
+                    hx::SnappyPageDesc sp{};
+                    sp.src_off = dict_dst;
+                    sp.comp_len = (uint32_t)dict_payload;
+                    sp.uncomp_len = (uint32_t)dictp->uncompressed_size;
+                    sp.dst_off = hx::OFF_DEC | dec_off;
+                    dec_off = align64(dec_off + sp.uncomp_len);
+                    plan.snappy_pages.push_back(sp);
+                    dict_data_off = sp.dst_off;
+                    dict_raw = sp.uncomp_len;
+                }
+                if (dict_raw != (uint32_t)dictp->num_values * 8) {
+                    err_msg = j.ss->cat->path + ": dictionary size mismatch";
+                    err_flag = 1;
+                    break;
+                }
+                rd.dict_off = dict_data_off;
+                rd.dict_n = (uint32_t)dictp->num_values;
+                rd.idx_off = data_off;
+                rd.idx_len = (uint32_t)raw_size;
+                rd.n_values = (uint32_t)j.num_values;
+                rd.dst_off = hx::OFF_DEC | dec_off;
+                dec_off = align64(dec_off + size_t(j.num_values) * 8);
+                plan.rledict_pages.push_back(rd);
+                j.final_off = rd.dst_off;
             } else {
                 std::lock_guard<std::mutex> g(mu);
                 err_msg = j.ss->cat->path + ": encoding " +
@@ -659,12 +706,12 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     HIP_TRY(hipSetDevice(plan.device));
     if (!plan.stream) HIP_TRY(hipStreamCreate(&plan.stream));
     if (plan.blob_bytes) {
-        HIP_TRY(hipMalloc((void**)&plan.d_blob, plan.blob_bytes));
+        HIP_TRY(hipMalloc((void**)&plan.d_blob, plan.blob_bytes + 64));
         HIP_TRY(hipMemcpyAsync(plan.d_blob, plan.h_blob, plan.blob_bytes,
                                hipMemcpyHostToDevice, plan.stream));
     }
     if (plan.dec_bytes)
-        HIP_TRY(hipMalloc((void**)&plan.d_dec, plan.dec_bytes));
+        HIP_TRY(hipMalloc((void**)&plan.d_dec, plan.dec_bytes + 64));
     auto upload = [&](auto*& dptr, const auto& vec) -> hipError_t {
         using T = std::remove_reference_t<decltype(vec[0])>;
         if (vec.empty()) { dptr = nullptr; return hipSuccess; }
@@ -679,6 +726,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     HIP_TRY(upload(plan.d_members, plan.cluster_members));
     HIP_TRY(upload(plan.d_delta, plan.delta_pages));
     HIP_TRY(upload(plan.d_snappy, plan.snappy_pages));
+    HIP_TRY(upload(plan.d_rledict, plan.rledict_pages));
     HIP_TRY(upload(plan.d_copies, plan.copies));
     HIP_TRY(hipMalloc((void**)&plan.d_counters, 4 * sizeof(unsigned long long)));
     HIP_TRY(hipStreamSynchronize(plan.stream));
@@ -878,7 +926,7 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
         for (void* p : {(void*)plan.d_blob, (void*)plan.d_dec, (void*)plan.d_rgs,
                         (void*)plan.d_ssts, (void*)plan.d_clusters,
                         (void*)plan.d_members, (void*)plan.d_delta,
-                        (void*)plan.d_snappy,
+                        (void*)plan.d_snappy, (void*)plan.d_rledict,
                         (void*)plan.d_copies, (void*)plan.t_series,
                         (void*)plan.t_bucket, (void*)plan.t_state,
                         (void*)plan.t_sum, (void*)plan.t_cnt, (void*)plan.t_min,
@@ -974,6 +1022,10 @@ hx_status ensure_decoded(DevPlan& plan, hipStream_t s) {
         HIP_TRY(hx::launch_snappy(s, plan.d_blob, plan.d_dec, plan.d_snappy,
                                   (uint32_t)plan.snappy_pages.size(),
                                   plan.d_counters + 1));
+    if (!plan.rledict_pages.empty())
+        HIP_TRY(hx::launch_rledict(s, plan.d_blob, plan.d_dec, plan.d_rledict,
+                                   (uint32_t)plan.rledict_pages.size(),
+                                   plan.d_counters + 1));
     if (!plan.delta_pages.empty())
         HIP_TRY(hx::launch_decode_delta(s, plan.d_blob, plan.d_dec,
                                         plan.d_delta,

@@ -96,6 +96,19 @@ struct DeltaPageDesc {
     uint32_t src_len;
 };
 
+// RLE_DICTIONARY decode unit: one data page + its dictionary page ->
+// dense 8-byte values at dst_off (dec). Indices are the Parquet RLE/
+// bit-packed hybrid; dictionary is PLAIN 8-byte values.
+struct RleDictPageDesc {
+    uint64_t dict_off;    // PLAIN dictionary payload (bit63: dec blob)
+    uint64_t idx_off;     // data page payload (starts with bit-width byte)
+    uint64_t dst_off;     // dec blob
+    uint32_t dict_n;
+    uint32_t idx_len;
+    uint32_t n_values;
+    uint32_t _pad;
+};
+
 // Snappy decompress unit: one page.
 struct SnappyPageDesc {
     uint64_t src_off;     // blob

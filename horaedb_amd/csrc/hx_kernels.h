@@ -19,6 +19,9 @@ struct CompactOut {
 hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                                const DeltaPageDesc* pages, uint32_t n_pages,
                                unsigned long long* err_flag);
+hipError_t launch_rledict(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                          const RleDictPageDesc* pages, uint32_t n_pages,
+                          unsigned long long* err_flag);
 hipError_t launch_snappy(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                          const SnappyPageDesc* pages, uint32_t n_pages,
                          unsigned long long* err_flag);

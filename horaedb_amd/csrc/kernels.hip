@@ -597,6 +597,120 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
 }
 
 // ---------------------------------------------------------------------------
+// RLE_DICTIONARY decode (parquet-format Encodings.md "RLE/bit-packed
+// hybrid"; the encoding config.rs:54-75 enables with dictionaries on).
+// One workgroup per page: lane-serial run-header walk into LDS (varints are
+// sequential), then threads expand runs in parallel through the dictionary.
+// ---------------------------------------------------------------------------
+#define RLED_MAX_RUNS 8192
+
+extern "C" __global__ void __launch_bounds__(256)
+k_decode_rledict(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
+                 const RleDictPageDesc* __restrict__ pages, uint32_t n_pages,
+                 unsigned long long* err_flag) {
+    struct Run { uint32_t out_pos, count, byte_off, rle_val_or_flag; };
+    __shared__ Run runs[RLED_MAX_RUNS];
+    __shared__ int n_runs;
+    __shared__ int hdr_err;
+    __shared__ uint32_t bw_sh;
+    for (uint32_t pg = blockIdx.x; pg < n_pages; pg += gridDim.x) {
+        const RleDictPageDesc pd = pages[pg];
+        const uint8_t* idx = ((pd.idx_off & OFF_DEC) ? dec : blob) +
+                             (pd.idx_off & OFF_MASK);
+        const uint64_t* dict = (const uint64_t*)(((pd.dict_off & OFF_DEC)
+                                                      ? dec : blob) +
+                                                 (pd.dict_off & OFF_MASK));
+        uint64_t* dst = (uint64_t*)(dec + (pd.dst_off & OFF_MASK));
+        if (threadIdx.x == 0) {
+            hdr_err = 0;
+            n_runs = 0;
+            const uint32_t len = pd.idx_len;
+            uint32_t pos = 0;
+            uint32_t bw = (len > 0) ? idx[pos++] : 0xFF;
+            bw_sh = bw;
+            if (bw > 32) hdr_err = 1;
+            uint32_t out = 0;
+            while (!hdr_err && out < pd.n_values) {
+                // varint header
+                uint64_t hdrv = 0;
+                int sh = 0;
+                for (;;) {
+                    if (pos >= len || sh > 28) { hdr_err = 2; break; }
+                    uint8_t bb = idx[pos++];
+                    hdrv |= (uint64_t)(bb & 0x7f) << sh;
+                    if (!(bb & 0x80)) break;
+                    sh += 7;
+                }
+                if (hdr_err) break;
+                if (n_runs >= RLED_MAX_RUNS) { hdr_err = 3; break; }
+                if (hdrv & 1) {  // bit-packed: (hdr>>1) groups of 8
+                    uint32_t cnt = (uint32_t)(hdrv >> 1) * 8;
+                    uint32_t nbytes = (uint32_t)(hdrv >> 1) * bw;
+                    if (pos + nbytes > len) { hdr_err = 4; break; }
+                    if (cnt > pd.n_values - out) cnt = pd.n_values - out;
+                    runs[n_runs++] = {out, cnt, pos, 0x80000000u};
+                    pos += nbytes;
+                    out += cnt;
+                } else {          // RLE run
+                    uint32_t cnt = (uint32_t)(hdrv >> 1);
+                    uint32_t vbytes = (bw + 7) / 8;
+                    if (pos + vbytes > len) { hdr_err = 5; break; }
+                    uint32_t v = 0;
+                    for (uint32_t i = 0; i < vbytes; i++)
+                        v |= (uint32_t)idx[pos + i] << (8 * i);
+                    pos += vbytes;
+                    if (cnt > pd.n_values - out) cnt = pd.n_values - out;
+                    runs[n_runs++] = {out, cnt, 0, v};
+                    out += cnt;
+                }
+            }
+            if (!hdr_err && out < pd.n_values) hdr_err = 6;
+        }
+        __syncthreads();
+        if (hdr_err) {
+            if (threadIdx.x == 0) atomicAdd(err_flag, 1ull);
+            __syncthreads();
+            continue;
+        }
+        const uint32_t bw = bw_sh;
+        const int nr = n_runs;
+        for (int rI = 0; rI < nr; rI++) {
+            const Run run = runs[rI];
+            if (run.rle_val_or_flag != 0x80000000u) {  // RLE: broadcast
+                const uint32_t v = run.rle_val_or_flag;
+                if (v >= pd.dict_n) {
+                    if (threadIdx.x == 0) atomicAdd(err_flag, 1ull);
+                    break;
+                }
+                const uint64_t dv = dict[v];
+                for (uint32_t i = threadIdx.x; i < run.count; i += blockDim.x)
+                    dst[run.out_pos + i] = dv;
+            } else {  // bit-packed: value i at bit i*bw
+                bool bad = false;
+                for (uint32_t i = threadIdx.x; i < run.count; i += blockDim.x) {
+                    const uint64_t bitpos = (uint64_t)i * bw;
+                    const uint8_t* base = idx + run.byte_off + (bitpos >> 3);
+                    const int shift = (int)(bitpos & 7);
+                    uint64_t window = 0;
+                    for (int b2 = 0; b2 < 5; b2++)
+                        window |= (uint64_t)base[b2] << (8 * b2);
+                    uint32_t v = (uint32_t)((window >> shift) &
+                                            ((bw < 32) ? ((1u << bw) - 1u)
+                                                       : 0xFFFFFFFFu));
+                    if (v >= pd.dict_n) { bad = true; break; }
+                    dst[run.out_pos + i] = dict[v];
+                }
+                if (bad) {
+                    atomicAdd(err_flag, 1ull);
+                    break;
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Streaming parity mode (hx_scan): emit the filtered, deduplicated rows
 // themselves. Appends survivors (series, ts, value) unordered; the host then
 // radix-sorts by (series, ts) — MergeStream's PK order (equal PKs cannot
@@ -1072,6 +1186,15 @@ hipError_t launch_decode_delta(hipStream_t s, const uint8_t* blob, uint8_t* dec,
                                unsigned long long* err_flag) {
     uint32_t grid = n_pages > 4096 ? 4096 : n_pages;
     hipLaunchKernelGGL(k_decode_delta_i64, dim3(grid), dim3(256), 0, s,
+                       blob, dec, pages, n_pages, err_flag);
+    return hipGetLastError();
+}
+
+hipError_t launch_rledict(hipStream_t s, const uint8_t* blob, uint8_t* dec,
+                          const RleDictPageDesc* pages, uint32_t n_pages,
+                          unsigned long long* err_flag) {
+    uint32_t grid = n_pages > 4096 ? 4096 : (n_pages ? n_pages : 1);
+    hipLaunchKernelGGL(k_decode_rledict, dim3(grid), dim3(256), 0, s,
                        blob, dec, pages, n_pages, err_flag);
     return hipGetLastError();
 }

@@ -125,6 +125,32 @@ def test_snappy_pages(ds_snappy):
     check_parity(out, (0, 2**62), ops=AGG_SUM | AGG_COUNT)
 
 
+def test_rle_dictionary(tmp_path):
+    # RLE_DICTIONARY chunks (dict page + RLE/bit-packed hybrid indices)
+    store = str(tmp_path)
+    rng = np.random.default_rng(7)
+    n = 60_000
+    series = np.sort(rng.integers(0, 300, n).astype(np.uint64))
+    ts = np.arange(n, dtype=np.int64) * 10
+    vals = rng.random(n)
+    gen_sst_from_arrays(store, 1, series, ts, vals,
+                        dict_columns=("series_id", "value"))
+    check_parity(store, (0, 10**9), ops=OPS_ALL)
+    check_parity(store, (100_000, 400_000), ops=AGG_SUM | AGG_COUNT)
+
+
+def test_rle_dictionary_snappy(tmp_path):
+    store = str(tmp_path)
+    rng = np.random.default_rng(8)
+    n = 30_000
+    series = np.sort(rng.integers(0, 100, n).astype(np.uint64))
+    ts = np.arange(n, dtype=np.int64) * 10
+    vals = rng.random(n)
+    gen_sst_from_arrays(store, 1, series, ts, vals, compression="snappy",
+                        dict_columns=("series_id",))
+    check_parity(store, (0, 10**9), ops=AGG_SUM | AGG_COUNT)
+
+
 def test_snappy_delta_combined(tmp_path_factory):
     out = str(tmp_path_factory.mktemp("snapdelta"))
     m = gen_dataset(out, n_rows=100_000, n_series=500, n_ssts=2, seed=45,

@@ -69,7 +69,7 @@ def make_series_ids(n_series, seed):
 
 
 def write_sst(path, series, ts, value, seq, row_group=8192,
-              compression="none", ts_encoding="PLAIN"):
+              compression="none", ts_encoding="PLAIN", dict_columns=()):
     """Write one SST from explicit row arrays (must be (series,ts)-sorted)."""
     import pyarrow as pa
     import pyarrow.parquet as pq
@@ -81,13 +81,25 @@ def write_sst(path, series, ts, value, seq, row_group=8192,
         "__seq__": np.full(n, seq, dtype=np.uint64),
         "__reserved__": np.zeros(n, dtype=np.uint64),
     }, schema=_pa_schema())
-    enc = {c: "PLAIN" for c in SCHEMA_COLS}
-    enc["timestamp"] = ts_encoding
+    kw = {}
+    if dict_columns:
+        # RLE_DICTIONARY chunks (config.rs:54-75 with dictionaries on);
+        # large limits keep one dictionary per chunk (no PLAIN fallback)
+        kw["use_dictionary"] = list(dict_columns)
+        kw["dictionary_pagesize_limit"] = 1 << 30
+        enc = {c: "PLAIN" for c in SCHEMA_COLS if c not in dict_columns}
+        enc["timestamp"] = ts_encoding if "timestamp" not in dict_columns \
+            else None
+        enc = {c: v for c, v in enc.items() if v}
+    else:
+        kw["use_dictionary"] = False
+        enc = {c: "PLAIN" for c in SCHEMA_COLS}
+        enc["timestamp"] = ts_encoding
     pq.write_table(
-        tbl, path, row_group_size=row_group, use_dictionary=False,
+        tbl, path, row_group_size=row_group,
         compression="NONE" if compression == "none" else compression.upper(),
         data_page_version="1.0", column_encoding=enc,
-        write_statistics=True)
+        write_statistics=True, **kw)
     return n
 
 
