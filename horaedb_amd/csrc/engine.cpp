@@ -5,6 +5,7 @@
 // §3-§5). Compute is GPU-only; no CPU fallback exists here.
 #include "../../include/horaedb_hx.h"
 #include "parquet_meta.h"
+#include "parquet_writer.h"
 #include "hx_device.h"
 
 #include <hip/hip_runtime.h>
@@ -1619,6 +1620,164 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
         hx_col_batch b{len, proj.size(), colptrs.data(), coltypes.data()};
         if (cb(ctx, &b)) break;
     }
+    return HX_OK;
+}
+
+extern "C" hx_status hx_write_sst(const char* path, const uint64_t* series,
+                                  const int64_t* ts, const double* value,
+                                  uint64_t seq, int64_t n_rows,
+                                  int64_t row_group) {
+    if (!path || !series || !ts || !value || n_rows <= 0 || row_group <= 0)
+        return fail(HX_ERR_INVALID, "bad argument");
+    std::string err = hx::write_metric_sst(path, series, ts, value, seq,
+                                           n_rows, row_group);
+    if (!err.empty()) return fail(HX_ERR_IO, err);
+    return HX_OK;
+}
+
+extern "C" hx_status hx_compact(hx_handle* h, hx_time_range range,
+                                const hx_device_set* devs,
+                                uint64_t* out_new_seq) {
+    // Compaction (Executor::do_compaction,
+    // executor.rs:155-222): re-runs the scan (GPU decode+dedup) over the
+    // ts-overlap CLOSURE of the files overlapping `range` and rewrites them
+    // as ONE new SST (native Parquet writer). The closure guarantees no
+    // remaining file shares primary keys with the inputs, so the output may
+    // carry one constant __seq__ (the freshly allocated file id) without
+    // changing any future merge outcome — the reference instead preserves
+    // per-row seqs (keep_builtin), which only matters when compacting a
+    // PARTIAL overlap set (picker.rs smallest-first can do that; round 2).
+    if (!h || !out_new_seq) return fail(HX_ERR_INVALID, "null argument");
+    *out_new_seq = 0;
+    // ts-overlap closure over the catalog
+    std::vector<char> in_set(h->ssts.size(), 0);
+    bool grew = true;
+    for (size_t i = 0; i < h->ssts.size(); i++)
+        if (overlaps(h->ssts[i], range)) in_set[i] = 1;
+    while (grew) {
+        grew = false;
+        for (size_t i = 0; i < h->ssts.size(); i++) {
+            if (in_set[i]) continue;
+            for (size_t j = 0; j < h->ssts.size(); j++) {
+                if (!in_set[j]) continue;
+                if (h->ssts[i].ts_min <= h->ssts[j].ts_max &&
+                    h->ssts[j].ts_min <= h->ssts[i].ts_max) {
+                    in_set[i] = 1;
+                    grew = true;
+                    break;
+                }
+            }
+        }
+    }
+    std::vector<hx_sst_desc> inputs;
+    uint64_t max_seq = 0;
+    for (size_t i = 0; i < h->ssts.size(); i++) {
+        if (!in_set[i]) continue;
+        inputs.push_back({h->ssts[i].path.c_str(), h->ssts[i].seq});
+    }
+    for (const auto& s : h->ssts) max_seq = std::max(max_seq, s.seq);
+    if (inputs.size() < 1) return HX_OK;
+
+    hx_scan_spec spec{};
+    spec.range = {INT64_MIN, INT64_MAX};   // whole files, like the executor
+    spec.ssts = inputs.data();
+    spec.n_ssts = inputs.size();
+    int32_t dev0 = (devs && devs->device_ids && devs->n_devices > 0)
+                       ? devs->device_ids[0] : 0;
+    hx_device_set one{&dev0, 1};
+    hx_prepared* P = nullptr;
+    hx_status st = hx_prepare(h, &spec, &one, &P);
+    if (st != HX_OK) return st;
+    std::unique_ptr<hx_prepared, void (*)(hx_prepared*)> guard(
+        P, hx_prepared_free);
+    DevPlan& plan = P->plans[0];
+    HIP_TRY(hipSetDevice(plan.device));
+    hipStream_t s = plan.stream;
+    st = ensure_decoded(plan, s);
+    if (st != HX_OK) return st;
+
+    const uint64_t cap = (uint64_t)plan.rows_scanned;
+    if (cap == 0) return HX_OK;
+    size_t need = cap * 8 * 5 + cap * 4 * 3 + 64;
+    st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
+    if (st != HX_OK) return st;
+    uint8_t* base = (uint8_t*)plan.d_scratch;
+    auto carve8 = [&](size_t count) {
+        uint8_t* p = base;
+        base += (count * 8 + 7) & ~size_t(7);
+        return p;
+    };
+    uint64_t* d_series = (uint64_t*)carve8(cap);
+    long long* d_ts = (long long*)carve8(cap);
+    double* d_val = (double*)carve8(cap);
+    uint64_t* d_keys = (uint64_t*)carve8(cap);
+    uint64_t* d_keys_out = (uint64_t*)carve8(cap);
+    unsigned long long* d_cursor = (unsigned long long*)carve8(1);
+    uint32_t* perm_a = (uint32_t*)base; base += cap * 4;
+    uint32_t* perm_b = (uint32_t*)base; base += cap * 4;
+    uint32_t* perm_c = (uint32_t*)base; base += cap * 4;
+
+    HIP_TRY(hipMemsetAsync(d_cursor, 0, 8, s));
+    hx::AggParams A = base_params(P, plan);
+    HIP_TRY(hx::launch_scan_rows(s, A, 0, A.n_rgs, d_series, d_ts, d_val,
+                                 d_cursor, cap));
+    unsigned long long n64 = 0;
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(&n64, d_cursor, 8, hipMemcpyDeviceToHost));
+    if (n64 > cap) return fail(HX_ERR_HIP, "compact row buffer overflow");
+    const uint32_t n = (uint32_t)n64;
+    if (n == 0) return HX_OK;
+
+    HIP_TRY(hx::launch_iota(s, perm_a, n));
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_ts, perm_a,
+                                  (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_a, perm_b, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_series,
+                                  perm_b, (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_b, perm_c, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    const uint32_t* perm = perm_c;
+
+    std::vector<uint64_t> host(3 * size_t(n));
+    unsigned long long* d_dst = nullptr;
+    HIP_TRY(hipMalloc((void**)&d_dst, 3 * size_t(n) * 8));
+    const unsigned long long* srcs[3] = {
+        (const unsigned long long*)d_series, (const unsigned long long*)d_ts,
+        (const unsigned long long*)d_val};
+    HIP_TRY(hx::launch_gather_multi(s, srcs, 3, perm, d_dst, n));
+    HIP_TRY(hipMemcpyAsync(host.data(), d_dst, 3 * size_t(n) * 8,
+                           hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    hipFree(d_dst);
+
+    const uint64_t new_seq = max_seq + 1;  // fresh file id (sst.rs:39-46)
+    std::string out_path = h->store + "/data/" + std::to_string(new_seq) +
+                           ".sst";
+    std::string werr = hx::write_metric_sst(
+        out_path, host.data(), (const int64_t*)(host.data() + n),
+        (const double*)(host.data() + 2 * size_t(n)), new_seq, n, 8192);
+    if (!werr.empty()) return fail(HX_ERR_IO, werr);
+
+    // catalog update: add new file, drop + unlink inputs (add-before-delete,
+    // executor.rs:206-220)
+    CatSst fresh;
+    st = read_file_meta(out_path, new_seq, fresh);
+    if (st != HX_OK) return st;
+    std::vector<CatSst> kept;
+    for (size_t i = 0; i < h->ssts.size(); i++) {
+        if (in_set[i]) {
+            unlink(h->ssts[i].path.c_str());  // best-effort, like the reference
+            continue;
+        }
+        kept.push_back(std::move(h->ssts[i]));
+    }
+    kept.push_back(std::move(fresh));
+    std::sort(kept.begin(), kept.end(),
+              [](const CatSst& a, const CatSst& b) { return a.seq < b.seq; });
+    h->ssts = std::move(kept);
+    *out_new_seq = new_seq;
     return HX_OK;
 }
 

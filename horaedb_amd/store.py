@@ -98,6 +98,11 @@ def _load():
     lib.hx_get_stats.argtypes = [C.c_void_p, C.POINTER(_ExecStats)]
     lib.hx_scan.argtypes = [C.c_void_p, C.POINTER(_ScanSpec),
                             C.POINTER(_DeviceSet), C.c_void_p, C.c_void_p]
+    lib.hx_compact.argtypes = [C.c_void_p, _TimeRange, C.POINTER(_DeviceSet),
+                               C.POINTER(C.c_uint64)]
+    lib.hx_write_sst.argtypes = [C.c_char_p, C.POINTER(C.c_uint64),
+                                 C.POINTER(C.c_int64), C.POINTER(C.c_double),
+                                 C.c_uint64, C.c_int64, C.c_int64]
     lib.hx_catalog_size.argtypes = [C.c_void_p, C.POINTER(C.c_size_t)]
     lib.hx_catalog_entry.argtypes = [C.c_void_p, C.c_size_t,
                                      C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
@@ -260,6 +265,17 @@ class Store:
         with self.prepare(ts_range, series_in, devices) as p:
             return p.exec_agg(ops=ops, bucket_ms=bucket_ms)
 
+    def compact(self, ts_range, devices=None):
+        """ColumnarStorage::compact (storage.rs:76-89): GPU-merge the
+        ts-overlap closure of the range's SSTs into one new SST. Returns the
+        new file's sequence (0 = nothing to compact)."""
+        self._keepalive = []
+        ds = self._devset(devices)
+        out = C.c_uint64()
+        _check(_lib.hx_compact(self._h, _TimeRange(*ts_range),
+                               C.byref(ds) if ds else None, C.byref(out)))
+        return out.value
+
     def scan(self, ts_range, series_in=None, projection=None, devices=None):
         """Streaming parity mode (hx_scan): the merged, deduplicated row
         stream itself — ColumnarStorage::scan semantics (storage.rs:335-370).
@@ -300,3 +316,16 @@ class Store:
             out[names[sel[i]]] = (np.concatenate(parts) if parts else
                                   np.empty(0))
         return out
+
+
+def write_sst_native(path, series, ts, value, seq, row_group=8192):
+    """Native C++ SST writer (the compaction output path), exposed for
+    tests: writes the reference metric-SST layout without pyarrow."""
+    series = np.ascontiguousarray(series, dtype=np.uint64)
+    ts = np.ascontiguousarray(ts, dtype=np.int64)
+    value = np.ascontiguousarray(value, dtype=np.float64)
+    n = len(series)
+    _check(_lib.hx_write_sst(
+        path.encode(), series.ctypes.data_as(C.POINTER(C.c_uint64)),
+        ts.ctypes.data_as(C.POINTER(C.c_int64)),
+        value.ctypes.data_as(C.POINTER(C.c_double)), seq, n, row_group))

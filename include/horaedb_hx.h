@@ -168,6 +168,21 @@ hx_status hx_scan_agg(hx_handle*, const hx_scan_spec*, const hx_agg_spec*,
 hx_status hx_scan(hx_handle*, const hx_scan_spec*, const hx_device_set*,
                   hx_batch_cb cb, void* ctx);
 
+/* Compaction (ColumnarStorage::compact, storage.rs:76-89; executor
+ * semantics executor.rs:155-222): GPU-merges the ts-overlap closure of the
+ * SSTs overlapping `range` into ONE new SST (fresh file id = old max + 1),
+ * updates the catalog, unlinks the inputs (add-before-delete). The closure
+ * guarantees no remaining file shares primary keys with the inputs, so the
+ * output carries one constant __seq__ without changing future merges. */
+hx_status hx_compact(hx_handle*, hx_time_range range, const hx_device_set*,
+                     uint64_t* out_new_seq);
+
+/* Native SST writer (the compaction output path; PLAIN uncompressed,
+ * row-group 8192 contract — storage.rs:193-298). Exposed for tests/ingest. */
+hx_status hx_write_sst(const char* path, const uint64_t* series,
+                       const int64_t* ts, const double* value, uint64_t seq,
+                       int64_t n_rows, int64_t row_group);
+
 /* ---- introspection for bench/tests ------------------------------------ */
 typedef struct {
     double  exec_ms;          /* wall of last hx_exec_agg (HIP events)      */

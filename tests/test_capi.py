@@ -104,3 +104,41 @@ def test_open_rejects_non_parquet(tmp_path):
     with pytest.raises(HxError) as ei:
         Store(str(tmp_path))
     assert ei.value.code == 2  # FORMAT
+
+
+def test_native_sst_writer_roundtrip(tmp_path):
+    # the compaction output writer (csrc/parquet_writer.cpp): our footer +
+    # pages must be readable by pyarrow AND by our own catalog reader
+    import pyarrow.parquet as pq
+    from horaedb_amd.store import write_sst_native
+
+    n = 20_000
+    rng = np.random.default_rng(3)
+    series = np.sort(rng.integers(0, 500, n).astype(np.uint64))
+    ts = np.arange(n, dtype=np.int64) * 7 + 1000
+    val = rng.random(n)
+    ddir = tmp_path / "data"
+    ddir.mkdir()
+    path = str(ddir / "9.sst")
+    write_sst_native(path, series, ts, val, seq=9)
+
+    t = pq.read_table(path)
+    assert t.num_rows == n
+    assert t.schema.names == ["series_id", "timestamp", "value", "__seq__",
+                              "__reserved__"]
+    np.testing.assert_array_equal(t.column("series_id").to_numpy(), series)
+    np.testing.assert_array_equal(t.column("timestamp").to_numpy(), ts)
+    np.testing.assert_array_equal(t.column("value").to_numpy(), val)
+    assert t.column("__seq__").to_numpy().tolist() == [9] * n
+    pf = pq.ParquetFile(path)
+    assert pf.metadata.num_row_groups == (n + 8191) // 8192
+    c = pf.metadata.row_group(0).column(1)
+    assert c.statistics.min == 1000
+
+    with Store(str(tmp_path)) as st:
+        cat = st.catalog()
+        assert len(cat) == 1
+        assert cat[0]["seq"] == 9
+        assert cat[0]["n_rows"] == n
+        assert cat[0]["ts_min"] == 1000
+        assert cat[0]["ts_max"] == int(ts.max())

@@ -287,3 +287,80 @@ def test_scan_rows_projection(ds_plain):
     with Store(out) as st:
         res = st.scan(middle_range(m), projection=[2, 0], devices=[0])
     assert set(res.keys()) == {"value", "series_id"}
+
+
+# ---------------------------------------------------------------------------
+# compaction: GPU merge-dedup -> one new SST
+# ---------------------------------------------------------------------------
+
+def test_compact_overlapping_ssts(tmp_path):
+    from horaedb_amd import Store
+    store = str(tmp_path)
+    gen_sst_from_arrays(store, 1, [5, 5, 7], [100, 200, 100], [1.0, 2.0, 3.0])
+    gen_sst_from_arrays(store, 2, [5, 7, 8], [200, 100, 50],
+                        [20.0, 30.0, 40.0])
+    # oracle BEFORE compaction (the inputs define the expected result)
+    ddir = os.path.join(store, "data")
+    ssts = [oracle.read_sst(os.path.join(ddir, f"{q}.sst")) for q in (1, 2)]
+    exp = oracle.scan_agg(ssts, (0, 10**6), ops=AGG_SUM | AGG_COUNT)
+    exp_rows = oracle.scan_rows(ssts, (0, 10**6))
+
+    with Store(store) as st:
+        new_seq = st.compact((0, 10**6), devices=[0])
+        assert new_seq == 3
+        cat = st.catalog()
+        assert [c["seq"] for c in cat] == [3]
+        assert cat[0]["n_rows"] == 4  # 5 rows deduped to 4
+        # post-compaction scans reproduce the pre-compaction results
+        res = st.scan_agg((0, 10**6), ops=AGG_SUM | AGG_COUNT, devices=[0])
+        assert res["series_id"].tolist() == exp["series_id"].tolist()
+        np.testing.assert_allclose(res["sum"], exp["sum"], rtol=1e-9)
+        assert res["count"].tolist() == exp["count"].tolist()
+        rows = st.scan((0, 10**6), devices=[0])
+        assert rows["series_id"].tolist() == exp_rows["series_id"].tolist()
+        np.testing.assert_array_equal(rows["value"], exp_rows["value"])
+    assert sorted(os.listdir(ddir)) == ["3.sst"]
+    # the output is standard parquet (readable by pyarrow)
+    import pyarrow.parquet as pq
+    t = pq.read_table(os.path.join(ddir, "3.sst"))
+    assert t.num_rows == 4
+
+
+def test_compact_larger_randomized(tmp_path):
+    from horaedb_amd import Store
+    store = str(tmp_path)
+    rng = np.random.default_rng(11)
+    for seq in (1, 2, 3):
+        n = 30_000
+        series = rng.integers(0, 400, n).astype(np.uint64)
+        ts = rng.integers(0, 5000, n).astype(np.int64) * 10
+        gen_sst_from_arrays(store, seq, series, ts, rng.random(n))
+    ddir = os.path.join(store, "data")
+    ssts = [oracle.read_sst(os.path.join(ddir, f"{q}.sst"))
+            for q in (1, 2, 3)]
+    exp = oracle.scan_agg(ssts, (0, 10**9), ops=OPS_ALL)
+    with Store(store) as st:
+        new_seq = st.compact((0, 10**9), devices=[0])
+        assert new_seq == 4
+        res = st.scan_agg((0, 10**9), ops=OPS_ALL, devices=[0])
+    assert res["series_id"].tolist() == exp["series_id"].tolist()
+    assert res["count"].tolist() == exp["count"].tolist()
+    np.testing.assert_array_equal(res["vmin"], exp["vmin"])
+    np.testing.assert_array_equal(res["vmax"], exp["vmax"])
+    np.testing.assert_allclose(res["sum"], exp["sum"], rtol=1e-9)
+
+
+def test_compact_leaves_disjoint_files(tmp_path):
+    from horaedb_amd import Store
+    store = str(tmp_path)
+    gen_sst_from_arrays(store, 1, [5], [100], [1.0])
+    gen_sst_from_arrays(store, 2, [5], [150], [2.0])
+    gen_sst_from_arrays(store, 3, [5], [10**7], [3.0])  # far away in time
+    with Store(store) as st:
+        new_seq = st.compact((0, 1000), devices=[0])
+        assert new_seq == 4
+        cat = st.catalog()
+        assert sorted(c["seq"] for c in cat) == [3, 4]
+        res = st.scan_agg((0, 2 * 10**7), ops=AGG_SUM | AGG_COUNT, devices=[0])
+        assert res["count"].tolist() == [3]
+        np.testing.assert_allclose(res["sum"], [6.0])
