@@ -236,10 +236,15 @@ std::string write_metric_sst(const std::string& path, const uint64_t* series,
         w.i64f(l2, 3, rg.num_rows);  // num_rows
         w.stop();
     }
-    {
-        int16_t l2 = last;
-        w.str(l2, 6, "horaedb-amd hx_compact writer");
-        last = l2;
+    w.str(last, 6, "horaedb-amd hx_compact writer");
+    // column_orders: TYPE_ORDER for every leaf — required for readers to
+    // trust min_value/max_value (parquet.thrift ColumnOrder union)
+    w.list_header(last, 7, 12, 5);
+    for (int c = 0; c < 5; c++) {
+        int16_t l2 = 0;
+        w.field(l2, 1, 12);  // TypeDefinedOrder (empty struct)
+        w.stop();
+        w.stop();
     }
     w.stop();
 
