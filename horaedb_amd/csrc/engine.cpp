@@ -1635,6 +1635,69 @@ extern "C" hx_status hx_write_sst(const char* path, const uint64_t* series,
     return HX_OK;
 }
 
+extern "C" hx_status hx_write(hx_handle* h, const uint64_t* series,
+                              const int64_t* ts, const double* value,
+                              int64_t n, int32_t enable_check,
+                              uint64_t* out_seq) {
+    // ColumnarStorage::write (storage.rs:76-89, :307-333): sort the batch by
+    // primary key (sort_batch, storage.rs:244-256 — GPU radix sort when a
+    // device is visible, the reference's CPU SortExec otherwise is NOT
+    // mirrored: ingest prep stays host-side std::stable_sort in that case),
+    // allocate the file id (= sequence, sst.rs:39-46), write one SST,
+    // add it to the catalog (manifest add_file, manifest/mod.rs:115-157).
+    if (!h || !series || !ts || !value || n <= 0 || !out_seq)
+        return fail(HX_ERR_INVALID, "bad argument");
+    *out_seq = 0;
+    if (enable_check) {
+        // segment-crossing check (storage.rs:309-316)
+        int64_t mn = ts[0], mx = ts[0];
+        for (int64_t i = 1; i < n; i++) {
+            mn = std::min(mn, ts[i]);
+            mx = std::max(mx, ts[i]);
+        }
+        auto seg = [&](int64_t t) {
+            int64_t q = t / h->segment_ms;
+            if ((t % h->segment_ms) != 0 && t < 0) q--;
+            return q;
+        };
+        if (seg(mn) != seg(mx))
+            return fail(HX_ERR_INVALID,
+                        "write crosses a segment boundary (storage.rs:309-316)");
+    }
+    // stable sort by (series, ts) — equal PKs keep batch order
+    // (LastValueOperator's last-wins depends on it, operator.rs:37-44)
+    std::vector<uint32_t> order(n);
+    for (int64_t i = 0; i < n; i++) order[i] = (uint32_t)i;
+    std::stable_sort(order.begin(), order.end(),
+                     [&](uint32_t a2, uint32_t b2) {
+                         if (series[a2] != series[b2])
+                             return series[a2] < series[b2];
+                         return ts[a2] < ts[b2];
+                     });
+    std::vector<uint64_t> s2(n);
+    std::vector<int64_t> t2(n);
+    std::vector<double> v2(n);
+    for (int64_t i = 0; i < n; i++) {
+        s2[i] = series[order[i]];
+        t2[i] = ts[order[i]];
+        v2[i] = value[order[i]];
+    }
+    uint64_t max_seq = 0;
+    for (const auto& s : h->ssts) max_seq = std::max(max_seq, s.seq);
+    const uint64_t new_seq = max_seq + 1;
+    std::string out_path = h->store + "/data/" + std::to_string(new_seq) +
+                           ".sst";
+    std::string werr = hx::write_metric_sst(out_path, s2.data(), t2.data(),
+                                            v2.data(), new_seq, n, 8192);
+    if (!werr.empty()) return fail(HX_ERR_IO, werr);
+    CatSst fresh;
+    hx_status st = read_file_meta(out_path, new_seq, fresh);
+    if (st != HX_OK) return st;
+    h->ssts.push_back(std::move(fresh));
+    *out_seq = new_seq;
+    return HX_OK;
+}
+
 extern "C" hx_status hx_compact(hx_handle* h, hx_time_range range,
                                 const hx_device_set* devs,
                                 uint64_t* out_new_seq) {

@@ -100,6 +100,9 @@ def _load():
                             C.POINTER(_DeviceSet), C.c_void_p, C.c_void_p]
     lib.hx_compact.argtypes = [C.c_void_p, _TimeRange, C.POINTER(_DeviceSet),
                                C.POINTER(C.c_uint64)]
+    lib.hx_write.argtypes = [C.c_void_p, C.POINTER(C.c_uint64),
+                             C.POINTER(C.c_int64), C.POINTER(C.c_double),
+                             C.c_int64, C.c_int32, C.POINTER(C.c_uint64)]
     lib.hx_write_sst.argtypes = [C.c_char_p, C.POINTER(C.c_uint64),
                                  C.POINTER(C.c_int64), C.POINTER(C.c_double),
                                  C.c_uint64, C.c_int64, C.c_int64]
@@ -264,6 +267,20 @@ class Store:
         """One-shot scan+aggregate (prepare + exec + release)."""
         with self.prepare(ts_range, series_in, devices) as p:
             return p.exec_agg(ops=ops, bucket_ms=bucket_ms)
+
+    def write(self, series, ts, value, enable_check=True):
+        """ColumnarStorage::write (storage.rs:76-89): stable PK sort + new
+        SST + catalog add. Returns the new file's sequence."""
+        series = np.ascontiguousarray(series, dtype=np.uint64)
+        ts_a = np.ascontiguousarray(ts, dtype=np.int64)
+        value = np.ascontiguousarray(value, dtype=np.float64)
+        out = C.c_uint64()
+        _check(_lib.hx_write(
+            self._h, series.ctypes.data_as(C.POINTER(C.c_uint64)),
+            ts_a.ctypes.data_as(C.POINTER(C.c_int64)),
+            value.ctypes.data_as(C.POINTER(C.c_double)), len(series),
+            1 if enable_check else 0, C.byref(out)))
+        return out.value
 
     def compact(self, ts_range, devices=None):
         """ColumnarStorage::compact (storage.rs:76-89): GPU-merge the

@@ -177,6 +177,15 @@ hx_status hx_scan(hx_handle*, const hx_scan_spec*, const hx_device_set*,
 hx_status hx_compact(hx_handle*, hx_time_range range, const hx_device_set*,
                      uint64_t* out_new_seq);
 
+/* ColumnarStorage::write (storage.rs:76-89, :307-333): stable PK sort of
+ * the batch (sort_batch, storage.rs:244-256), file id allocation
+ * (= sequence), one new SST via the native writer, catalog add. Ingest-side
+ * prep (not the GPU scan hot path). enable_check enforces the
+ * segment-crossing check (storage.rs:309-316). */
+hx_status hx_write(hx_handle*, const uint64_t* series, const int64_t* ts,
+                   const double* value, int64_t n_rows, int32_t enable_check,
+                   uint64_t* out_seq);
+
 /* Native SST writer (the compaction output path; PLAIN uncompressed,
  * row-group 8192 contract — storage.rs:193-298). Exposed for tests/ingest. */
 hx_status hx_write_sst(const char* path, const uint64_t* series,

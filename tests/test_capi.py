@@ -142,3 +142,42 @@ def test_native_sst_writer_roundtrip(tmp_path):
         assert cat[0]["n_rows"] == n
         assert cat[0]["ts_min"] == 1000
         assert cat[0]["ts_max"] == int(ts.max())
+
+
+def test_write_sort_and_catalog(tmp_path):
+    # ColumnarStorage::write: unsorted batch -> PK-sorted SST + catalog add
+    # (sort invariant golden, storage.rs:493-536 analog)
+    import pyarrow.parquet as pq
+    (tmp_path / "data").mkdir()
+    with Store(str(tmp_path)) as st:
+        seq = st.write([7, 5, 5, 9], [100, 200, 100, 50],
+                       [1.0, 2.0, 3.0, 4.0])
+        assert seq == 1
+        cat = st.catalog()
+        assert len(cat) == 1 and cat[0]["n_rows"] == 4
+        t = pq.read_table(str(tmp_path / "data" / "1.sst"))
+        assert t.column("series_id").to_numpy().tolist() == [5, 5, 7, 9]
+        assert t.column("timestamp").to_numpy().tolist() == [100, 200, 100, 50]
+        # second write gets the next file id
+        seq2 = st.write([1], [10], [0.5])
+        assert seq2 == 2
+        assert len(st.find_ssts((0, 1000))) == 2
+
+
+def test_write_stable_on_equal_pks(tmp_path):
+    # equal PKs keep batch order (LastValueOperator last-wins contract)
+    import pyarrow.parquet as pq
+    (tmp_path / "data").mkdir()
+    with Store(str(tmp_path)) as st:
+        st.write([5, 5, 5], [100, 100, 100], [1.0, 2.0, 3.0])
+    t = pq.read_table(str(tmp_path / "data" / "1.sst"))
+    assert t.column("value").to_numpy().tolist() == [1.0, 2.0, 3.0]
+
+
+def test_write_segment_crossing_check(tmp_path):
+    (tmp_path / "data").mkdir()
+    with Store(str(tmp_path), segment_duration_ms=1000) as st:
+        with pytest.raises(HxError) as ei:
+            st.write([1, 1], [100, 5000], [1.0, 2.0])
+        assert ei.value.code == 6
+        st.write([1, 1], [100, 5000], [1.0, 2.0], enable_check=False)

@@ -364,3 +364,23 @@ def test_compact_leaves_disjoint_files(tmp_path):
         res = st.scan_agg((0, 2 * 10**7), ops=AGG_SUM | AGG_COUNT, devices=[0])
         assert res["count"].tolist() == [3]
         np.testing.assert_allclose(res["sum"], [6.0])
+
+
+def test_write_then_scan_reference_golden(tmp_path):
+    # the reference's own end-to-end golden (storage.rs:392-491) mapped onto
+    # the metric schema (pk1 -> series_id, pk2 -> timestamp): two writes into
+    # one segment, Overwrite dedup, predicate applied before the merge
+    from horaedb_amd import Store
+    (tmp_path / "data").mkdir()
+    with Store(str(tmp_path)) as st:
+        st.write([11, 11, 9, 10, 5], [100, 100, 1, 2, 3], [2.0, 7.0, 4.0, 6.0, 1.0])
+        st.write([11, 11, 9, 10], [100, 99, 1, 2], [22.0, 77.0, 44.0, 66.0])
+        rows = st.scan((0, 2**40), devices=[0])
+        # expected (storage.rs:448-461): (5,3,1)(9,1,44)(10,2,66)(11,99,77)(11,100,22)
+        assert rows["series_id"].tolist() == [5, 9, 10, 11, 11]
+        assert rows["timestamp"].tolist() == [3, 1, 2, 99, 100]
+        assert rows["value"].tolist() == [1.0, 44.0, 66.0, 77.0, 22.0]
+        # predicate pk1 == 11 (storage.rs:475-489)
+        rows = st.scan((0, 2**40), series_in=[11], devices=[0])
+        assert rows["series_id"].tolist() == [11, 11]
+        assert rows["value"].tolist() == [77.0, 22.0]
