@@ -737,15 +737,31 @@ k_scan_rows(ScanRowsParams R) {
         const double* V = (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
         const SstDev sst = P.ssts[rg.sst_id];
         const uint32_t n = rg.n_rows;
-        for (uint32_t r = threadIdx.x; r < n; r += blockDim.x) {
-            const int64_t t = T[r];
-            const uint64_t s = S[r];
-            if (!row_alive(P, rg, sst, S, T, r, n, s, t)) continue;
-            unsigned long long j = atomicAdd(R.cursor, 1ull);
-            if (j >= R.cap) continue;  // host re-runs with a larger buffer
-            R.out_series[j] = s;
-            R.out_ts[j] = t;
-            R.out_value[j] = V[r];
+        const int lane = threadIdx.x & 63;
+        for (uint32_t base = 0; base < n; base += blockDim.x) {
+            const uint32_t r = base + threadIdx.x;
+            const bool inb = r < n;
+            const int64_t t = inb ? T[r] : 0;
+            const uint64_t s = inb ? S[r] : 0;
+            const bool live = inb && row_alive(P, rg, sst, S, T, r, n, s, t);
+            // one cursor atomic per wave (guideline 12)
+            const unsigned long long mask = __ballot(live);
+            if (!mask) continue;
+            const int leader = __ffsll((unsigned long long)mask) - 1;
+            unsigned long long wave_base = 0;
+            if (lane == leader)
+                wave_base = atomicAdd(R.cursor,
+                                      (unsigned long long)__popcll(mask));
+            wave_base = __shfl(wave_base, leader, 64);
+            if (live) {
+                const unsigned long long j =
+                    wave_base + __popcll(mask & ((1ull << lane) - 1ull));
+                if (j < R.cap) {
+                    R.out_series[j] = s;
+                    R.out_ts[j] = t;
+                    R.out_value[j] = V[r];
+                }
+            }
         }
     }
 }
