@@ -104,9 +104,22 @@ __device__ __forceinline__ void table_add(const AggParams& P, uint32_t i,
 __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s,
                                                   double vsum,
                                                   unsigned long long cnt,
-                                                  double mn, double mx) {
+                                                  double mn, double mx,
+                                                  uint32_t hint_i = 0xFFFFFFFFu,
+                                                  uint64_t hint_k = 0) {
     const uint32_t stride = P.table.stride;
     uint32_t i = (uint32_t)mix64(s) & P.table.mask;
+    if (hint_i == i && hint_k == s) {  // prefetched probe already matched
+        uint8_t* slot = P.table.slab + (size_t)i * stride;
+        if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd((double*)(slot + 8), vsum);
+        if (P.ops & (HXK_COUNT | HXK_AVG))
+            atomicAdd((unsigned long long*)(slot + 16), cnt);
+        if (P.ops & HXK_MIN)
+            atomicMin((unsigned long long*)(slot + 24), f64_ordered(mn));
+        if (P.ops & HXK_MAX)
+            atomicMax((unsigned long long*)(slot + 32), f64_ordered(mx));
+        return;
+    }
     for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
         uint8_t* slot = P.table.slab + (size_t)i * stride;
         uint64_t k = *(uint64_t*)slot;
@@ -159,9 +172,11 @@ k_init_slab(uint8_t* slab, uint32_t n_slots, uint32_t stride) {
 __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
                                            int64_t b, double vsum,
                                            unsigned long long cnt,
-                                           double mn, double mx) {
+                                           double mn, double mx,
+                                           uint32_t hint_i = 0xFFFFFFFFu,
+                                           uint64_t hint_k = 0) {
     if (P.key_claim) {
-        agg_update_keycas(P, s, vsum, cnt, mn, mx);
+        agg_update_keycas(P, s, vsum, cnt, mn, mx, hint_i, hint_k);
         return;
     }
     const bool use_b = P.bucket_ms != 0;
@@ -245,6 +260,8 @@ struct WinResult {
     double vv, mn, mx;
     unsigned long long c;
     bool head;
+    uint32_t hint_i;    // early-probed slot index (key-claim tables)
+    uint64_t hint_k;    // its key value at probe time
 };
 
 __device__ __forceinline__ void scan_window(const AggParams& P,
@@ -274,6 +291,15 @@ __device__ __forceinline__ void scan_window(const AggParams& P,
     double mn = alive ? v : HUGE_VAL;
     double mx = alive ? v : -HUGE_VAL;
     my_matched += c;
+    // issue the table probe NOW: its L3 latency hides under the cross-lane
+    // reduction, and a hit skips the dependent probe load in agg_update
+    W.hint_i = 0xFFFFFFFFu;
+    W.hint_k = 0;
+    if (P.key_claim && alive) {
+        W.hint_i = (uint32_t)mix64(s) & P.table.mask;
+        W.hint_k = *(const uint64_t*)(P.table.slab +
+                                      (size_t)W.hint_i * P.table.stride);
+    }
     const uint64_t sp = __shfl_up(s, 1, 64);
     const int64_t bp = __shfl_up((long long)b, 1, 64);
     const bool head = (lane == 0) || sp != s || bp != b;
@@ -331,12 +357,16 @@ k_scan_agg(AggParams P) {
             const bool upA = A.head && A.c > 0;
             const bool upB = B.head && B.c > 0;
             if (upA && upB) {  // two independent chains: overlap them
-                agg_update(P, A.s, A.b, A.vv, A.c, A.mn, A.mx);
-                agg_update(P, B.s, B.b, B.vv, B.c, B.mn, B.mx);
+                agg_update(P, A.s, A.b, A.vv, A.c, A.mn, A.mx, A.hint_i,
+                           A.hint_k);
+                agg_update(P, B.s, B.b, B.vv, B.c, B.mn, B.mx, B.hint_i,
+                           B.hint_k);
             } else if (upA) {
-                agg_update(P, A.s, A.b, A.vv, A.c, A.mn, A.mx);
+                agg_update(P, A.s, A.b, A.vv, A.c, A.mn, A.mx, A.hint_i,
+                           A.hint_k);
             } else if (upB) {
-                agg_update(P, B.s, B.b, B.vv, B.c, B.mn, B.mx);
+                agg_update(P, B.s, B.b, B.vv, B.c, B.mn, B.mx, B.hint_i,
+                           B.hint_k);
             }
         }
     }
