@@ -275,15 +275,42 @@ __device__ __forceinline__ void scan_window(const AggParams& P,
                                             WinResult& W) {
     const uint32_t r = base + threadIdx.x;
     const bool inb = r < n;
+    const bool has_next = r + 1 < n;
     uint64_t s = KEY_EMPTY;
     int64_t t = 0;
     double v = 0.0;
+    uint64_t s1 = 0;
+    int64_t t1 = 0;
     bool alive = false;
+    // issue ALL of the window's loads together (row, value, successor):
+    // one memory round-trip per window instead of a 3-deep dependent chain
     if (inb) {
         t = T[r];
         s = S[r];
-        alive = row_alive(P, rg, sst, S, T, r, n, s, t);
-        if (alive) v = V[r];
+        v = V[r];
+        if (has_next) {
+            s1 = S[r + 1];
+            t1 = T[r + 1];
+        }
+    }
+    if (inb) {
+        alive = (t >= P.ts_lo) & (t < P.ts_hi);
+        if (alive && P.use_sset) alive = sset_has(P, s);
+        if (alive && P.skip < 2) {
+            bool dup = false;
+            if (has_next) {
+                dup = (s1 == s) & (t1 == t);
+            } else if (rg.next_rg >= 0) {
+                const RgDesc nx = P.rgs[rg.next_rg];
+                uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
+                                                       nx.series_off);
+                int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
+                dup = (s2 == s) & (t2 == t);
+            }
+            if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+            alive = !dup;
+        }
+        if (!alive) v = 0.0;
     }
     int64_t b = (inb && P.bucket_ms) ? floordiv(t, P.bucket_ms) : 0;
     unsigned long long c = alive ? 1ull : 0ull;
