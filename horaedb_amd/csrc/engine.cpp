@@ -1527,6 +1527,9 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
 
     const uint64_t cap = (uint64_t)plan.rows_scanned;
     if (cap == 0) return HX_OK;
+    if (cap > 0xFFFFFFFFull)
+        return fail(HX_ERR_UNSUPPORTED,
+                    "row streams above 2^32 rows per call: split the range");
     // device buffers: 3 columns + sort keys/scratch + 3 perms
     size_t need = cap * 8 * 5 + cap * 4 * 3 + 64;
     st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
@@ -1761,6 +1764,9 @@ extern "C" hx_status hx_compact(hx_handle* h, hx_time_range range,
 
     const uint64_t cap = (uint64_t)plan.rows_scanned;
     if (cap == 0) return HX_OK;
+    if (cap > 0xFFFFFFFFull)
+        return fail(HX_ERR_UNSUPPORTED,
+                    "row streams above 2^32 rows per call: split the range");
     size_t need = cap * 8 * 5 + cap * 4 * 3 + 64;
     st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
     if (st != HX_OK) return st;
