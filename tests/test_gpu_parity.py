@@ -384,3 +384,38 @@ def test_write_then_scan_reference_golden(tmp_path):
         rows = st.scan((0, 2**40), series_in=[11], devices=[0])
         assert rows["series_id"].tolist() == [11, 11]
         assert rows["value"].tolist() == [77.0, 22.0]
+
+
+# ---------------------------------------------------------------------------
+# randomized property parity: arbitrary small stores
+# ---------------------------------------------------------------------------
+
+def test_randomized_stores_parity(tmp_path_factory):
+    rng = np.random.default_rng(12345)
+    for trial in range(6):
+        store = str(tmp_path_factory.mktemp(f"rand{trial}"))
+        n_ssts = int(rng.integers(1, 4))
+        for seq in range(1, n_ssts + 1):
+            n = int(rng.integers(1, 20_000))
+            series = rng.integers(0, int(rng.integers(1, 1000)), n).astype(np.uint64)
+            ts = rng.integers(0, int(rng.integers(10, 100_000)), n).astype(np.int64)
+            gen_sst_from_arrays(store, seq, series, ts, rng.random(n))
+        lo = int(rng.integers(0, 50_000))
+        hi = lo + int(rng.integers(1, 100_000))
+        check_parity(store, (lo, hi), ops=OPS_ALL)
+        check_parity(store, (lo, hi), ops=AGG_SUM | AGG_COUNT,
+                     bucket_ms=int(rng.integers(1, 5000)))
+
+
+def test_delta_wide_bitwidths(tmp_path):
+    # delta pages with huge jumps (wide miniblock bit widths incl. >32)
+    store = str(tmp_path)
+    rng = np.random.default_rng(77)
+    n = 30_000
+    series = np.sort(rng.integers(0, 100, n).astype(np.uint64))
+    ts = np.cumsum(rng.integers(-2**40, 2**40, n)).astype(np.int64)
+    order = np.lexsort((ts, series))
+    gen_sst_from_arrays(store, 1, series[order], ts[order], rng.random(n),
+                        sort=False, ts_encoding="DELTA_BINARY_PACKED")
+    lo, hi = int(ts.min()), int(ts.max()) + 1
+    check_parity(store, (lo, hi), ops=AGG_SUM | AGG_COUNT)
