@@ -152,7 +152,7 @@ class Prepared:
             _lib.hx_result_free(out)
             return {"n_groups": n}
         res = {"series_id": _np(t.series_id, n, np.uint64)}
-        if t.bucket:
+        if bucket_ms:
             res["bucket"] = _np(t.bucket, n, np.int64)
         if ops & AGG_SUM:
             res["sum"] = _np(t.sum, n, np.float64)
