@@ -179,10 +179,13 @@ def main():
     # max over ranks of elapsed; sum over ranks of rows
     rows_scanned = prep.stats()["rows_scanned"]
     if dist:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        dev = torch.device(f"cuda:{device}") if torch.cuda.is_available() \
+            else torch.device("cpu")  # nccl needs device tensors
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
-        r = torch.tensor([float(rows_scanned)], dtype=torch.float64)
+        r = torch.tensor([float(rows_scanned)], dtype=torch.float64,
+                         device=dev)
         dist.all_reduce(r, op=dist.ReduceOp.SUM)
         rows_scanned = int(r.item())
 
