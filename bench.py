@@ -113,12 +113,15 @@ def main():
     n_gpus = max(args.gpus, world)
 
     dist = None
+    backend = None
     if world > 1:
         import torch
         import torch.distributed as torch_dist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        torch_dist.init_process_group(backend="nccl" if
-                                      torch.cuda.is_available() else "gloo")
+        backend = os.environ.get(
+            "HX_BENCH_BACKEND",
+            "nccl" if torch.cuda.is_available() else "gloo")
+        torch_dist.init_process_group(backend=backend)
         dist = torch_dist
 
     import torch
@@ -141,7 +144,7 @@ def main():
         log(f"rank{rank}: series-set predicate with {k} ids "
             f"({args.selectivity:.2%})")
 
-    device = local_rank
+    device = int(os.environ.get("HX_DEV_OVERRIDE", local_rank))
     torch.cuda.set_device(device)
 
     store = Store(store_dir)
@@ -179,7 +182,7 @@ def main():
     # max over ranks of elapsed; sum over ranks of rows
     rows_scanned = prep.stats()["rows_scanned"]
     if dist:
-        dev = torch.device(f"cuda:{device}") if torch.cuda.is_available() \
+        dev = torch.device(f"cuda:{device}") if backend == "nccl" \
             else torch.device("cpu")  # nccl needs device tensors
         t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
