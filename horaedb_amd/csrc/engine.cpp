@@ -276,6 +276,8 @@ struct DevPlan {
     unsigned long long* t_max = nullptr;
     uint8_t* t_slab = nullptr;      // AoS key-claim table
     uint32_t slab_stride = 0;
+    uint8_t* t_rep = nullptr;       // per-XCD accumulator replicas (opt-in)
+    uint32_t rep_stride = 0;
     unsigned long long* d_counters = nullptr;  // fill, overflow, matched, n_out
     uint64_t* d_sset = nullptr;
     size_t sset_cap = 0;
@@ -932,6 +934,7 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
                         (void*)plan.t_bucket, (void*)plan.t_state,
                         (void*)plan.t_sum, (void*)plan.t_cnt, (void*)plan.t_min,
                         (void*)plan.t_max, (void*)plan.t_slab,
+                        (void*)plan.t_rep,
                         (void*)plan.d_counters,
                         (void*)plan.d_sset})
             if (p) hipFree(p);
@@ -977,18 +980,27 @@ hx_status alloc_table(DevPlan& plan, uint32_t slots, uint32_t ops, bool bucket,
     uint32_t stride = key_claim
                           ? ((ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 64u : 32u)
                           : 0u;
-    if (plan.slots == slots && (key_claim ? (plan.t_slab && plan.slab_stride == stride)
-                                          : plan.t_series != nullptr))
+    const bool want_rep = key_claim && getenv("HX_XCD_REP") != nullptr;
+    if (plan.slots == slots &&
+        (key_claim ? (plan.t_slab && plan.slab_stride == stride &&
+                      (plan.t_rep != nullptr) == want_rep)
+                   : plan.t_series != nullptr))
         return HX_OK;
     for (void** p : {(void**)&plan.t_series, (void**)&plan.t_bucket,
                      (void**)&plan.t_state, (void**)&plan.t_sum,
                      (void**)&plan.t_cnt, (void**)&plan.t_min,
-                     (void**)&plan.t_max, (void**)&plan.t_slab})
+                     (void**)&plan.t_max, (void**)&plan.t_slab,
+                     (void**)&plan.t_rep})
         if (*p) { hipFree(*p); *p = nullptr; }
     plan.slots = slots;
     plan.slab_stride = stride;
     if (key_claim) {
         HIP_TRY(hipMalloc((void**)&plan.t_slab, size_t(slots) * stride));
+        if (getenv("HX_XCD_REP")) {
+            plan.rep_stride = (ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 32u : 16u;
+            HIP_TRY(hipMalloc((void**)&plan.t_rep,
+                              8ull * slots * plan.rep_stride));
+        }
         return HX_OK;
     }
     HIP_TRY(hipMalloc((void**)&plan.t_series, size_t(slots) * 8));
@@ -1137,6 +1149,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         if (key_claim) {
             HIP_TRY(hx::launch_init_slab(s, plan.t_slab, slots,
                                          plan.slab_stride));
+            if (plan.t_rep)
+                HIP_TRY(hx::launch_init_rep(
+                    s, plan.t_rep, 8ull * slots, plan.rep_stride,
+                    (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
         } else {
             HIP_TRY(hipMemsetAsync(plan.t_state, 0, size_t(slots) * 4, s));
             if (plan.t_sum) HIP_TRY(hipMemsetAsync(plan.t_sum, 0, size_t(slots) * 8, s));
@@ -1166,7 +1182,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.skip = getenv("HX_SKIP") ? atoi(getenv("HX_SKIP")) : 0;
         A.table = {plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
                    plan.t_cnt, plan.t_min, plan.t_max, slots - 1,
-                   plan.t_slab, plan.slab_stride};
+                   plan.t_slab, plan.slab_stride, plan.t_rep,
+                   plan.rep_stride};
         A.fill = plan.d_counters + 0;
         A.overflow = plan.d_counters + 1;
         A.matched = plan.d_counters + 2;
@@ -1280,7 +1297,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
     hx::AggTable T{plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
                    plan.t_cnt, plan.t_min, plan.t_max, plan.slots - 1,
-                   plan.t_slab, plan.slab_stride};
+                   plan.t_slab, plan.slab_stride, plan.t_rep,
+                   plan.rep_stride};
     HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,
                                bucket ? agg->bucket_ms : 0, co));
 

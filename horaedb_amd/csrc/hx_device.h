@@ -62,6 +62,11 @@ struct AggTable {
     // [, min, max]} — probe + update touch ONE cache line instead of three
     uint8_t*  slab;            // non-null => AoS mode (series-only grouping)
     uint32_t  stride;          // 32 (sum/cnt) or 64 (with min/max)
+    // optional per-XCD accumulator replicas (HX_XCD_REP): keys live in the
+    // shared slab; each XCD's blocks accumulate into their own replica to
+    // avoid cross-L2 line ping-pong; compact merges the replicas.
+    uint8_t*  rep;             // 8 replicas x slots x rep_stride, or null
+    uint32_t  rep_stride;      // 16 (sum,cnt) or 32 (+min,max)
 };
 
 struct AggParams {

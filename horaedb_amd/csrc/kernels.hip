@@ -95,6 +95,35 @@ __device__ __forceinline__ void table_add(const AggParams& P, uint32_t i,
     if (P.ops & HXK_MAX) atomicMax(&P.table.vmax[i], f64_ordered(mx));
 }
 
+// accumulate into slot i: the shared slab, or this XCD's replica when
+// per-XCD replication is on (the replica's lines stay exclusive to one L2
+// — no cross-XCD ping-pong on the hot accumulator lines)
+__device__ __forceinline__ void keycas_add(const AggParams& P, uint32_t i,
+                                           double vsum,
+                                           unsigned long long cnt, double mn,
+                                           double mx) {
+    uint8_t* slot;
+    uint32_t off8, off16, off24, off32;
+    if (P.table.rep) {
+        uint32_t xcc;
+        asm("s_getreg_b32 %0, hwreg(HW_REG_XCC_ID)" : "=s"(xcc));
+        xcc &= 7u;
+        slot = P.table.rep +
+               ((size_t)xcc * (P.table.mask + 1ull) + i) * P.table.rep_stride;
+        off8 = 0; off16 = 8; off24 = 16; off32 = 24;
+    } else {
+        slot = P.table.slab + (size_t)i * P.table.stride;
+        off8 = 8; off16 = 16; off24 = 24; off32 = 32;
+    }
+    if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd((double*)(slot + off8), vsum);
+    if (P.ops & (HXK_COUNT | HXK_AVG))
+        atomicAdd((unsigned long long*)(slot + off16), cnt);
+    if (P.ops & HXK_MIN)
+        atomicMin((unsigned long long*)(slot + off24), f64_ordered(mn));
+    if (P.ops & HXK_MAX)
+        atomicMax((unsigned long long*)(slot + off32), f64_ordered(mx));
+}
+
 // Fast claim path (AoS slab): slot i = {key, sum, cnt[, min, max]} in one
 // cache line. key claimed by one CAS against KEY_EMPTY (host proves via
 // column statistics that no series == KEY_EMPTY; series-only grouping);
@@ -110,14 +139,7 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
     const uint32_t stride = P.table.stride;
     uint32_t i = (uint32_t)mix64(s) & P.table.mask;
     if (hint_i == i && hint_k == s) {  // prefetched probe already matched
-        uint8_t* slot = P.table.slab + (size_t)i * stride;
-        if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd((double*)(slot + 8), vsum);
-        if (P.ops & (HXK_COUNT | HXK_AVG))
-            atomicAdd((unsigned long long*)(slot + 16), cnt);
-        if (P.ops & HXK_MIN)
-            atomicMin((unsigned long long*)(slot + 24), f64_ordered(mn));
-        if (P.ops & HXK_MAX)
-            atomicMax((unsigned long long*)(slot + 32), f64_ordered(mx));
+        keycas_add(P, i, vsum, cnt, mn, mx);
         return;
     }
     for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
@@ -134,19 +156,26 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
             }
         }
         if (k == s) {
-            if (P.ops & (HXK_SUM | HXK_AVG))
-                atomicAdd((double*)(slot + 8), vsum);
-            if (P.ops & (HXK_COUNT | HXK_AVG))
-                atomicAdd((unsigned long long*)(slot + 16), cnt);
-            if (P.ops & HXK_MIN)
-                atomicMin((unsigned long long*)(slot + 24), f64_ordered(mn));
-            if (P.ops & HXK_MAX)
-                atomicMax((unsigned long long*)(slot + 32), f64_ordered(mx));
+            keycas_add(P, i, vsum, cnt, mn, mx);
             return;
         }
         i = (i + 1) & P.table.mask;
     }
     __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_init_rep(uint8_t* rep, size_t total_slots, uint32_t stride, uint32_t mm) {
+    for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total_slots;
+         i += (size_t)blockDim.x * gridDim.x) {
+        uint8_t* slot = rep + i * stride;
+        ((uint64_t*)slot)[0] = 0;   // sum
+        ((uint64_t*)slot)[1] = 0;   // cnt
+        if (mm) {
+            ((uint64_t*)slot)[2] = ~0ull;
+            ((uint64_t*)slot)[3] = 0;
+        }
+    }
 }
 
 // slab init: key=KEY_EMPTY, sum=0, cnt=0, min=~0 (ordered), max=0
@@ -864,7 +893,29 @@ k_compact(CompactParams C) {
         if (live) {
             const unsigned long long j =
                 wave_base + __popcll(mask & ((1ull << lane) - 1ull));
-            if (C.key_claim) {
+            if (C.key_claim && C.table.rep) {
+                C.out_series[j] = *(const uint64_t*)slot;
+                double sum = 0, mnv = HUGE_VAL, mxv = -HUGE_VAL;
+                unsigned long long cntv = 0;
+                for (uint32_t x = 0; x < 8; x++) {
+                    const uint8_t* r2 =
+                        C.table.rep +
+                        ((size_t)x * (C.table.mask + 1ull) + i) *
+                            C.table.rep_stride;
+                    sum += *(const double*)r2;
+                    cntv += *(const unsigned long long*)(r2 + 8);
+                    if (C.out_min)
+                        mnv = fmin(mnv, ordered_f64(
+                                            *(const unsigned long long*)(r2 + 16)));
+                    if (C.out_max)
+                        mxv = fmax(mxv, ordered_f64(
+                                            *(const unsigned long long*)(r2 + 24)));
+                }
+                if (C.out_sum) C.out_sum[j] = sum;
+                if (C.out_cnt) C.out_cnt[j] = cntv;
+                if (C.out_min) C.out_min[j] = mnv;
+                if (C.out_max) C.out_max[j] = mxv;
+            } else if (C.key_claim) {
                 C.out_series[j] = *(const uint64_t*)slot;
                 if (C.out_sum) C.out_sum[j] = *(const double*)(slot + 8);
                 if (C.out_cnt)
@@ -1432,6 +1483,16 @@ hipError_t launch_init_slab(hipStream_t s, uint8_t* slab, uint32_t n_slots,
                             uint32_t stride) {
     hipLaunchKernelGGL(k_init_slab, dim3(grid_for(n_slots, 256)), dim3(256),
                        0, s, slab, n_slots, stride);
+    return hipGetLastError();
+}
+
+hipError_t launch_init_rep(hipStream_t s, uint8_t* rep, size_t total_slots,
+                           uint32_t stride, bool mm) {
+    hipLaunchKernelGGL(k_init_rep,
+                       dim3(grid_for((uint32_t)std::min<size_t>(
+                                         total_slots, 0x7FFFFFFF), 256)),
+                       dim3(256), 0, s, rep, total_slots, stride,
+                       mm ? 1u : 0u);
     return hipGetLastError();
 }
 
