@@ -446,6 +446,12 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 err_flag = 1;
                 break;
             }
+            if (j.comp_size <= 0 || j.comp_size > (int64_t)1 << 31) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": implausible chunk size";
+                err_flag = 1;
+                break;
+            }
             tmp.resize(j.comp_size);
             if (pread(last_fd, tmp.data(), j.comp_size, j.chunk_start) !=
                 (ssize_t)j.comp_size) {

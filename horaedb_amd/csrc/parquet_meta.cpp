@@ -207,6 +207,8 @@ std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
                 default: r.skip(t);
             }
         }
+        if (comp < 0 || uncomp < 0)
+            throw std::runtime_error("negative page size");
         size_t hdr_len = r.offset(buf + pos);
         pd.payload_off = base_off + int64_t(pos + hdr_len);
         pd.compressed_size = comp;

@@ -10,8 +10,12 @@
  *
  * Conventions: all structs are POD; caller owns all inputs; callee allocates
  * outputs, freed by the matching hx_*_free; status codes + hx_last_error()
- * (thread-local string). A handle is usable from many threads; each call set
- * uses its own HIP stream set per device. All compute is GPU-resident — if no
+ * (thread-local string). Concurrency: scan-side calls (hx_prepare /
+ * hx_scan_agg / hx_scan) may run from many threads of one handle; a given
+ * hx_prepared serves ONE call at a time; catalog mutations (hx_write,
+ * hx_compact) require external serialization against all other calls, and
+ * hx_find_ssts results are invalidated by the next hx_find_ssts on the
+ * same handle. All compute is GPU-resident — if no
  * MI355X/HIP runtime is available every scan entry returns HX_ERR_NO_GPU
  * (there is no CPU fallback in this library).
  */
