@@ -320,7 +320,14 @@ class Store:
             chunks.append(cols)
             return 0
 
-        cb = _BATCH(on_batch)
+        def on_batch_safe(ctx, bp):
+            # exceptions must not cross the C callback boundary
+            try:
+                return on_batch(ctx, bp)
+            except Exception:
+                return 1  # stop the stream
+
+        cb = _BATCH(on_batch_safe)
         _check(_lib.hx_scan(self._h, C.byref(spec),
                             C.byref(ds) if ds else None,
                             C.cast(cb, C.c_void_p), None))
