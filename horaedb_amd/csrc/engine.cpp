@@ -218,6 +218,23 @@ static bool overlaps(const CatSst& s, hx_time_range r) {
     return s.ts_min < r.end && s.ts_max >= r.start;
 }
 
+extern "C" hx_status hx_schema(hx_handle* h, const hx_col_desc** out,
+                               size_t* n_out, size_t* n_primary_keys) {
+    // the metric-engine schema contract (types.rs:150-240): 2 PKs, one f64
+    // value, builtin __seq__/__reserved__ appended
+    static const hx_col_desc kSchema[5] = {
+        {"series_id", 0, 1, 0},   {"timestamp", 1, 1, 0},
+        {"value", 2, 0, 0},       {"__seq__", 0, 0, 1},
+        {"__reserved__", 0, 0, 1},
+    };
+    if (!h || !out || !n_out || !n_primary_keys)
+        return fail(HX_ERR_INVALID, "null argument");
+    *out = kSchema;
+    *n_out = 5;
+    *n_primary_keys = 2;
+    return HX_OK;
+}
+
 extern "C" hx_status hx_find_ssts(hx_handle* h, hx_time_range range,
                                   const hx_sst_desc** out, size_t* n_out) {
     if (!h || !out || !n_out) return fail(HX_ERR_INVALID, "null argument");

@@ -144,6 +144,18 @@ hx_status hx_open(const char* store_path, int64_t segment_duration_ms,
                   hx_handle** out);
 void      hx_close(hx_handle*);
 
+/* ColumnarStorage::schema (storage.rs:76-89; StorageSchema contract
+ * types.rs:150-240): the store's column layout — user columns (primary keys
+ * first) plus the appended builtin __seq__/__reserved__ UInt64 columns. */
+typedef struct {
+    const char* name;
+    int32_t     col_type;     /* 0=u64, 1=i64(ts ms), 2=f64 */
+    int32_t     is_primary_key;
+    int32_t     is_builtin;
+} hx_col_desc;
+hx_status hx_schema(hx_handle*, const hx_col_desc** out, size_t* n_out,
+                    size_t* n_primary_keys);
+
 /* Catalog introspection (Manifest::find_ssts, manifest/mod.rs:165-172). */
 hx_status hx_find_ssts(hx_handle*, hx_time_range range,
                        const hx_sst_desc** out, size_t* n_out);
