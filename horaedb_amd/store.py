@@ -106,6 +106,8 @@ def _load():
     lib.hx_write_sst.argtypes = [C.c_char_p, C.POINTER(C.c_uint64),
                                  C.POINTER(C.c_int64), C.POINTER(C.c_double),
                                  C.c_uint64, C.c_int64, C.c_int64]
+    lib.hx_schema.argtypes = [C.c_void_p, C.c_void_p, C.POINTER(C.c_size_t),
+                              C.POINTER(C.c_size_t)]
     lib.hx_catalog_size.argtypes = [C.c_void_p, C.POINTER(C.c_size_t)]
     lib.hx_catalog_entry.argtypes = [C.c_void_p, C.c_size_t,
                                      C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
@@ -203,6 +205,27 @@ class Store:
 
     def __exit__(self, *a):
         self.close()
+
+    def schema(self):
+        """ColumnarStorage::schema (storage.rs:76-89): column layout with
+        PK/builtin markers (types.rs:150-240 contract)."""
+
+        class _Col(C.Structure):
+            _fields_ = [("name", C.c_char_p), ("col_type", C.c_int32),
+                        ("is_primary_key", C.c_int32),
+                        ("is_builtin", C.c_int32)]
+
+        out = C.POINTER(_Col)()
+        n = C.c_size_t()
+        npk = C.c_size_t()
+        _check(_lib.hx_schema(self._h, C.cast(C.byref(out), C.c_void_p),
+                              C.byref(n), C.byref(npk)))
+        tnames = {0: "u64", 1: "i64", 2: "f64"}
+        cols = [{"name": out[i].name.decode(),
+                 "type": tnames[out[i].col_type],
+                 "primary_key": bool(out[i].is_primary_key),
+                 "builtin": bool(out[i].is_builtin)} for i in range(n.value)]
+        return {"columns": cols, "num_primary_keys": npk.value}
 
     def catalog(self):
         n = C.c_size_t()

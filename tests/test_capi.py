@@ -181,3 +181,17 @@ def test_write_segment_crossing_check(tmp_path):
             st.write([1, 1], [100, 5000], [1.0, 2.0])
         assert ei.value.code == 6
         st.write([1, 1], [100, 5000], [1.0, 2.0], enable_check=False)
+
+
+def test_schema_surface(store_dir):
+    # ColumnarStorage::schema: the types.rs:150-240 contract
+    out, m = store_dir
+    with Store(out) as st:
+        sch = st.schema()
+    assert sch["num_primary_keys"] == 2
+    assert [c["name"] for c in sch["columns"]] == \
+        ["series_id", "timestamp", "value", "__seq__", "__reserved__"]
+    assert [c["primary_key"] for c in sch["columns"]] == \
+        [True, True, False, False, False]
+    assert [c["builtin"] for c in sch["columns"]] == \
+        [False, False, False, True, True]
