@@ -54,36 +54,54 @@ def get_dataset(args, rank):
     return out, m
 
 
-def cpu_baseline_leg(store_dir, m, ts_range, budget_s=20.0):
-    """Oracle (numpy/pyarrow CPU restatement) timed on a bounded sample of
-    the same workload — kind 'port' (the reference Rust path cannot be built
-    here: no cargo). Scaled to rows/s of scanned rows."""
+def _cpu_one_sst(args):
+    path, ts_range = args
+    import sys as _s
+    _s.path.insert(0, REPO)
     import oracle
     from oracle.scan import AGG_SUM, AGG_COUNT
+    sst = oracle.read_sst(path)
+    oracle.scan_agg([sst], tuple(ts_range), ops=AGG_SUM | AGG_COUNT)
+    return sst.n_rows
+
+
+def cpu_baseline_leg(store_dir, m, ts_range, budget_s=20.0, cores=1):
+    """Oracle (numpy/pyarrow CPU restatement) timed on a bounded sample of
+    the same workload — kind 'port' (the reference Rust path cannot be built
+    here: no cargo). Scaled to rows/s of scanned rows.
+    cores>1 fans SSTs over a process pool (the all-cores leg BASELINE.md
+    plans; cores = processes actually used)."""
     lo, hi = ts_range
-    rows_done = 0
+    paths = [s["path"] for s in m["ssts"]
+             if s["ts_min"] < hi and s["ts_max"] >= lo]
+    # bound the sample to ~budget_s of single-core work, scaled by cores
+    per_sst_rows = m["n_rows"] // max(1, m["n_ssts"])
+    est_rate = 18e6  # rows/s/core, prior measurements
+    max_ssts = max(1, int(budget_s * est_rate * cores / max(1, per_sst_rows)))
+    sample = paths[:max_ssts]
     t0 = time.time()
-    sample_ssts = 0
-    for s in m["ssts"]:
-        # only SSTs overlapping the range count as scanned
-        if not (s["ts_min"] < hi and s["ts_max"] >= lo):
-            continue
-        sst = oracle.read_sst(s["path"])
-        oracle.scan_agg([sst], ts_range, ops=AGG_SUM | AGG_COUNT)
-        rows_done += sst.n_rows
-        sample_ssts += 1
-        if time.time() - t0 > budget_s:
-            break
+    if cores > 1:
+        import multiprocessing as mp
+        with mp.get_context("spawn").Pool(cores) as pool:
+            counts = pool.map(_cpu_one_sst,
+                              [(p, list(ts_range)) for p in sample])
+        rows_done = sum(counts)
+    else:
+        rows_done = 0
+        for p in sample:
+            rows_done += _cpu_one_sst((p, list(ts_range)))
+            if time.time() - t0 > budget_s:
+                break
     dt = time.time() - t0
     if rows_done == 0 or dt <= 0:
         return None
     return {
         "value": rows_done / dt,
         "unit": "rows/s",
-        "cores": 1,
+        "cores": cores,
         "kind": "port",
-        "sample": f"{sample_ssts} SST(s), {rows_done} rows, {dt:.1f}s "
-                  f"(oracle numpy+pyarrow, single-threaded)",
+        "sample": f"{len(sample)} SST(s), {rows_done} rows, {dt:.1f}s "
+                  f"(oracle numpy+pyarrow, {cores} process(es))",
     }
 
 
@@ -105,6 +123,8 @@ def main():
     p.add_argument("--ops", default="sum,count")
     p.add_argument("--data-dir", default="/tmp/hx_bench_data")
     p.add_argument("--no-cpu-baseline", action="store_true")
+    p.add_argument("--cpu-cores", type=int, default=1,
+                   help="processes for the cpu_baseline leg")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -229,7 +249,8 @@ def main():
         cpu = None
         if not args.no_cpu_baseline and n_gpus == 1:
             log("running cpu_baseline (oracle, bounded sample)...")
-            cpu = cpu_baseline_leg(store_dir, m, ts_range)
+            cpu = cpu_baseline_leg(store_dir, m, ts_range,
+                                   cores=args.cpu_cores)
 
         result = {
             "metric": "scanned rows/sec, 1B-row range+sum (config 2)",
