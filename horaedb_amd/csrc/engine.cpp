@@ -1146,6 +1146,11 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     hx_status sst_ = ensure_sset(P, plan);
     if (sst_ != HX_OK) return sst_;
 
+    // decode results persist across exec calls on one prepared scan; for
+    // benchmarking encoded/compressed workloads HX_REDECODE=1 forces every
+    // step to repeat the decompress/decode stage (no cached decode output
+    // inside the timed region).
+    if (getenv("HX_REDECODE")) plan.decoded = false;
     hx_status dst_ = ensure_decoded(plan, s);
     if (dst_ != HX_OK) return dst_;
 
