@@ -164,6 +164,11 @@ def main():
         log(f"rank{rank}: series-set predicate with {k} ids "
             f"({args.selectivity:.2%})")
 
+    if args.compression != "none" or args.ts_encoding != "PLAIN":
+        # decode/decompress must run in EVERY timed step (no cached decode)
+        os.environ["HX_REDECODE"] = "1"
+        log("workload has a decode stage: HX_REDECODE=1 (per-step decode)")
+
     device = int(os.environ.get("HX_DEV_OVERRIDE", local_rank))
     torch.cuda.set_device(device)
 
