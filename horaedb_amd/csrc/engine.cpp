@@ -1164,9 +1164,11 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     if (!slots) {
         const char* env = getenv("HX_TABLE_SLOTS");
         if (env) slots = next_pow2_u32(strtoull(env, nullptr, 10));
-        else slots = next_pow2_u32(std::max<uint64_t>(1 << 16,
-                                       (uint64_t)plan.rows_scanned / 32));
-        if (slots > (1u << 27)) slots = 1u << 27;
+        else slots = next_pow2_u32(std::max<uint64_t>(
+                 1 << 16,
+                 // bucket keys multiply the group count: start larger
+                 (uint64_t)plan.rows_scanned / (bucket ? 8 : 32)));
+        if (slots > (1u << 28)) slots = 1u << 28;
     }
 
     unsigned long long counters[4];
