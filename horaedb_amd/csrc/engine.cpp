@@ -1166,8 +1166,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         if (env) slots = next_pow2_u32(strtoull(env, nullptr, 10));
         else slots = next_pow2_u32(std::max<uint64_t>(
                  1 << 16,
-                 // bucket keys multiply the group count: start larger
-                 (uint64_t)plan.rows_scanned / (bucket ? 8 : 32)));
+                 // bucket keys multiply the group count: be generous
+                 (uint64_t)plan.rows_scanned / (bucket ? 2 : 32)));
         if (slots > (1u << 28)) slots = 1u << 28;
     }
 
@@ -1214,6 +1214,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                    plan.t_cnt, plan.t_min, plan.t_max, slots - 1,
                    plan.t_slab, plan.slab_stride, plan.t_rep,
                    plan.rep_stride};
+        A.fill_limit = (unsigned long long)(double(slots) * 0.85);
         A.fill = plan.d_counters + 0;
         A.overflow = plan.d_counters + 1;
         A.matched = plan.d_counters + 2;

@@ -88,6 +88,7 @@ struct AggParams {
     int32_t key_claim;             // 1 = one-CAS key-claim mode (no state word)
     int32_t skip;                  // debug bisect: 1 skip table, 2 also skip dedup
     AggTable table;
+    unsigned long long fill_limit; // early-abort when fill exceeds this
     unsigned long long* fill;      // claimed slots
     unsigned long long* overflow;  // !=0 => rerun with a larger table
     unsigned long long* matched;   // rows surviving filter+dedup

@@ -396,6 +396,13 @@ k_scan_agg(AggParams P) {
     unsigned long long my_matched = 0;
     const int lane = threadIdx.x & 63;
     for (uint32_t rgi = blockIdx.x; rgi < P.n_rgs; rgi += gridDim.x) {
+        // early-abort once the table saturates: the host retries with a
+        // larger table, so finishing a doomed pass only burns time
+        if (__hip_atomic_load(P.fill, RLX, AGT) > P.fill_limit) {
+            if (threadIdx.x == 0)
+                __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+            break;
+        }
         const RgDesc rg = P.rgs[rgi];
         const uint64_t* S = (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
         const int64_t* T = (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
