@@ -1000,14 +1000,15 @@ uint32_t next_pow2_u32(uint64_t x) {
 
 hx_status alloc_table(DevPlan& plan, uint32_t slots, uint32_t ops, bool bucket,
                       bool key_claim) {
+    (void)bucket;
+    // AoS slab strides: key-claim {key,sum,cnt[,min,max]} = 32/64;
+    // state mode {state,pad,series,bucket,sum,cnt[,min,max]} = 48/64
     uint32_t stride = key_claim
                           ? ((ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 64u : 32u)
-                          : 0u;
+                          : ((ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 64u : 48u);
     const bool want_rep = key_claim && getenv("HX_XCD_REP") != nullptr;
-    if (plan.slots == slots &&
-        (key_claim ? (plan.t_slab && plan.slab_stride == stride &&
-                      (plan.t_rep != nullptr) == want_rep)
-                   : plan.t_series != nullptr))
+    if (plan.slots == slots && plan.t_slab && plan.slab_stride == stride &&
+        (plan.t_rep != nullptr) == want_rep)
         return HX_OK;
     for (void** p : {(void**)&plan.t_series, (void**)&plan.t_bucket,
                      (void**)&plan.t_state, (void**)&plan.t_sum,
@@ -1017,26 +1018,12 @@ hx_status alloc_table(DevPlan& plan, uint32_t slots, uint32_t ops, bool bucket,
         if (*p) { hipFree(*p); *p = nullptr; }
     plan.slots = slots;
     plan.slab_stride = stride;
-    if (key_claim) {
-        HIP_TRY(hipMalloc((void**)&plan.t_slab, size_t(slots) * stride));
-        if (getenv("HX_XCD_REP")) {
-            plan.rep_stride = (ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 32u : 16u;
-            HIP_TRY(hipMalloc((void**)&plan.t_rep,
-                              8ull * slots * plan.rep_stride));
-        }
-        return HX_OK;
+    HIP_TRY(hipMalloc((void**)&plan.t_slab, size_t(slots) * stride));
+    if (key_claim && getenv("HX_XCD_REP")) {
+        plan.rep_stride = (ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 32u : 16u;
+        HIP_TRY(hipMalloc((void**)&plan.t_rep,
+                          8ull * slots * plan.rep_stride));
     }
-    HIP_TRY(hipMalloc((void**)&plan.t_series, size_t(slots) * 8));
-    HIP_TRY(hipMalloc((void**)&plan.t_state, size_t(slots) * 4));
-    if (bucket) HIP_TRY(hipMalloc((void**)&plan.t_bucket, size_t(slots) * 8));
-    if (ops & (HX_AGG_SUM | HX_AGG_AVG))
-        HIP_TRY(hipMalloc((void**)&plan.t_sum, size_t(slots) * 8));
-    if (ops & (HX_AGG_COUNT | HX_AGG_AVG))
-        HIP_TRY(hipMalloc((void**)&plan.t_cnt, size_t(slots) * 8));
-    if (ops & HX_AGG_MIN)
-        HIP_TRY(hipMalloc((void**)&plan.t_min, size_t(slots) * 8));
-    if (ops & HX_AGG_MAX)
-        HIP_TRY(hipMalloc((void**)&plan.t_max, size_t(slots) * 8));
     return HX_OK;
 }
 
@@ -1184,11 +1171,9 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                     s, plan.t_rep, 8ull * slots, plan.rep_stride,
                     (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
         } else {
-            HIP_TRY(hipMemsetAsync(plan.t_state, 0, size_t(slots) * 4, s));
-            if (plan.t_sum) HIP_TRY(hipMemsetAsync(plan.t_sum, 0, size_t(slots) * 8, s));
-            if (plan.t_cnt) HIP_TRY(hipMemsetAsync(plan.t_cnt, 0, size_t(slots) * 8, s));
-            if (plan.t_min) HIP_TRY(hipMemsetAsync(plan.t_min, 0xFF, size_t(slots) * 8, s));
-            if (plan.t_max) HIP_TRY(hipMemsetAsync(plan.t_max, 0, size_t(slots) * 8, s));
+            HIP_TRY(hx::launch_init_state_slab(
+                s, plan.t_slab, slots, plan.slab_stride,
+                (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
         }
         HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
 
@@ -1210,8 +1195,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.ops = ops;
         A.key_claim = key_claim;
         A.skip = getenv("HX_SKIP") ? atoi(getenv("HX_SKIP")) : 0;
-        A.table = {plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
-                   plan.t_cnt, plan.t_min, plan.t_max, slots - 1,
+        A.table = {nullptr, nullptr, nullptr, nullptr,
+                   nullptr, nullptr, nullptr, slots - 1,
                    plan.t_slab, plan.slab_stride, plan.t_rep,
                    plan.rep_stride};
         A.fill_limit = (unsigned long long)(double(slots) * 0.85);
@@ -1326,8 +1311,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
 
     HIP_TRY(hipMemsetAsync(d_nout, 0, 8, s));
     hx::CompactOut co{c_series, c_bucket, c_sum, c_cnt, c_min, c_max, d_nout};
-    hx::AggTable T{plan.t_series, plan.t_bucket, plan.t_state, plan.t_sum,
-                   plan.t_cnt, plan.t_min, plan.t_max, plan.slots - 1,
+    hx::AggTable T{nullptr, nullptr, nullptr, nullptr,
+                   nullptr, nullptr, nullptr, plan.slots - 1,
                    plan.t_slab, plan.slab_stride, plan.t_rep,
                    plan.rep_stride};
     HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,

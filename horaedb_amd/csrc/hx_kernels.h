@@ -54,6 +54,8 @@ hipError_t launch_init_slab(hipStream_t s, uint8_t* slab, uint32_t n_slots,
                             uint32_t stride);
 hipError_t launch_init_rep(hipStream_t s, uint8_t* rep, size_t total_slots,
                            uint32_t stride, bool mm);
+hipError_t launch_init_state_slab(hipStream_t s, uint8_t* slab,
+                                  uint32_t n_slots, uint32_t stride, bool mm);
 hipError_t launch_seg_keys(hipStream_t s, const long long* ts, long long seg_ms,
                            unsigned long long* keys, uint32_t n);
 hipError_t launch_xor_sign(hipStream_t s, unsigned long long* buf, uint32_t n);

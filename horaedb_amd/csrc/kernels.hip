@@ -86,15 +86,6 @@ __device__ bool shadowed(const AggParams& P, const SstDev& me,
 // sentinel needed); claimer stores key words then releases state=2; readers
 // are gated by the control dependency on state==2 (all table words accessed
 // with agent-scope atomics => L2-coherent, no L1 staleness).
-__device__ __forceinline__ void table_add(const AggParams& P, uint32_t i,
-                                          double vsum, unsigned long long cnt,
-                                          double mn, double mx) {
-    if (P.ops & (HXK_SUM | HXK_AVG)) atomicAdd(&P.table.sum[i], vsum);
-    if (P.ops & (HXK_COUNT | HXK_AVG)) atomicAdd(&P.table.cnt[i], cnt);
-    if (P.ops & HXK_MIN) atomicMin(&P.table.vmin[i], f64_ordered(mn));
-    if (P.ops & HXK_MAX) atomicMax(&P.table.vmax[i], f64_ordered(mx));
-}
-
 // accumulate into slot i: the shared slab, or this XCD's replica when
 // per-XCD replication is on (the replica's lines stay exclusive to one L2
 // — no cross-XCD ping-pong on the hot accumulator lines)
@@ -142,7 +133,8 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
         keycas_add(P, i, vsum, cnt, mn, mx);
         return;
     }
-    for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
+    const uint32_t probe_cap = P.table.mask < 4096u ? P.table.mask : 4096u;
+    for (uint32_t probes = 0; probes <= probe_cap; ++probes) {
         uint8_t* slot = P.table.slab + (size_t)i * stride;
         uint64_t k = *(uint64_t*)slot;
         if (k == KEY_EMPTY) {
@@ -162,6 +154,22 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
         i = (i + 1) & P.table.mask;
     }
     __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_init_state_slab(uint8_t* slab, uint32_t n_slots, uint32_t stride,
+                  uint32_t mm) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n_slots;
+         i += blockDim.x * gridDim.x) {
+        uint8_t* slot = slab + (size_t)i * stride;
+        ((uint64_t*)slot)[0] = 0;   // state + pad
+        ((uint64_t*)slot)[3] = 0;   // sum
+        ((uint64_t*)slot)[4] = 0;   // cnt
+        if (mm) {
+            ((uint64_t*)slot)[5] = ~0ull;
+            ((uint64_t*)slot)[6] = 0;
+        }
+    }
 }
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -208,21 +216,25 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
         agg_update_keycas(P, s, vsum, cnt, mn, mx, hint_i, hint_k);
         return;
     }
-    const bool use_b = P.bucket_ms != 0;
+    // AoS slot: {state u32, pad u32, series u64, bucket i64, sum, cnt
+    // [, min, max]} — probe + claim + update on one/two cache lines
+    const uint32_t stride = P.table.stride;
     uint64_t h = mix64(s ^ ((uint64_t)b * 0xD1B54A32D192ED03ull));
     uint32_t i = (uint32_t)h & P.table.mask;
-    for (uint32_t probes = 0; probes <= P.table.mask; ++probes) {
-        uint32_t st = __hip_atomic_load(&P.table.state[i], RLX, AGT);
+    const uint32_t probe_cap = P.table.mask < 4096u ? P.table.mask : 4096u;
+    for (uint32_t probes = 0; probes <= probe_cap; ++probes) {
+        uint8_t* slot = P.table.slab + (size_t)i * stride;
+        uint32_t* stp = (uint32_t*)slot;
+        uint32_t st = __hip_atomic_load(stp, RLX, AGT);
         if (st == 0) {
             uint32_t expected = 0;
-            if (__hip_atomic_compare_exchange_strong(&P.table.state[i],
-                    &expected, 1u, RLX, RLX, AGT)) {
-                __hip_atomic_store(&P.table.series[i], s, RLX, AGT);
-                if (use_b)
-                    __hip_atomic_store((unsigned long long*)&P.table.bucket[i],
-                                       (unsigned long long)b, RLX, AGT);
+            if (__hip_atomic_compare_exchange_strong(stp, &expected, 1u, RLX,
+                                                     RLX, AGT)) {
+                __hip_atomic_store((uint64_t*)(slot + 8), s, RLX, AGT);
+                __hip_atomic_store((unsigned long long*)(slot + 16),
+                                   (unsigned long long)b, RLX, AGT);
                 __hip_atomic_fetch_add(P.fill, 1ull, RLX, AGT);
-                __hip_atomic_store(&P.table.state[i], 2u, __ATOMIC_RELEASE, AGT);
+                __hip_atomic_store(stp, 2u, __ATOMIC_RELEASE, AGT);
                 st = 2;
             } else {
                 st = expected;
@@ -231,17 +243,24 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
         if (st == 1) {  // claimer is storing the key; bounded spin
             uint32_t spins = 0;
             do {
-                st = __hip_atomic_load(&P.table.state[i], RLX, AGT);
+                st = __hip_atomic_load(stp, RLX, AGT);
                 if (++spins > (1u << 22)) {
                     __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
                     return;
                 }
             } while (st != 2);
         }
-        if (__hip_atomic_load(&P.table.series[i], RLX, AGT) == s &&
-            (!use_b || (int64_t)__hip_atomic_load(
-                           (unsigned long long*)&P.table.bucket[i], RLX, AGT) == b)) {
-            table_add(P, i, vsum, cnt, mn, mx);
+        if (__hip_atomic_load((uint64_t*)(slot + 8), RLX, AGT) == s &&
+            (int64_t)__hip_atomic_load((unsigned long long*)(slot + 16), RLX,
+                                       AGT) == b) {
+            if (P.ops & (HXK_SUM | HXK_AVG))
+                atomicAdd((double*)(slot + 24), vsum);
+            if (P.ops & (HXK_COUNT | HXK_AVG))
+                atomicAdd((unsigned long long*)(slot + 32), cnt);
+            if (P.ops & HXK_MIN)
+                atomicMin((unsigned long long*)(slot + 40), f64_ordered(mn));
+            if (P.ops & HXK_MAX)
+                atomicMax((unsigned long long*)(slot + 48), f64_ordered(mx));
             return;
         }
         i = (i + 1) & P.table.mask;
@@ -884,11 +903,10 @@ k_compact(CompactParams C) {
     const uint32_t stride = blockDim.x * gridDim.x;
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
          __any(i < C.n_slots); i += stride) {
-        const uint8_t* slot =
-            C.key_claim ? C.table.slab + (size_t)i * C.table.stride : nullptr;
+        const uint8_t* slot = C.table.slab + (size_t)i * C.table.stride;
         const bool live = (i < C.n_slots) &&
                           (C.key_claim ? (*(const uint64_t*)slot != KEY_EMPTY)
-                                       : (C.table.state[i] == 2u));
+                                       : (*(const uint32_t*)slot == 2u));
         const unsigned long long mask = __ballot(live);
         if (!mask) continue;
         // one atomic per wave (guideline 12): first live lane reserves
@@ -934,12 +952,18 @@ k_compact(CompactParams C) {
                     C.out_max[j] = ordered_f64(
                         *(const unsigned long long*)(slot + 32));
             } else {
-                C.out_series[j] = C.table.series[i];
-                if (C.bucket_ms) C.out_bucket[j] = C.table.bucket[i];
-                if (C.out_sum) C.out_sum[j] = C.table.sum[i];
-                if (C.out_cnt) C.out_cnt[j] = C.table.cnt[i];
-                if (C.out_min) C.out_min[j] = ordered_f64(C.table.vmin[i]);
-                if (C.out_max) C.out_max[j] = ordered_f64(C.table.vmax[i]);
+                C.out_series[j] = *(const uint64_t*)(slot + 8);
+                if (C.bucket_ms)
+                    C.out_bucket[j] = *(const long long*)(slot + 16);
+                if (C.out_sum) C.out_sum[j] = *(const double*)(slot + 24);
+                if (C.out_cnt)
+                    C.out_cnt[j] = *(const unsigned long long*)(slot + 32);
+                if (C.out_min)
+                    C.out_min[j] = ordered_f64(
+                        *(const unsigned long long*)(slot + 40));
+                if (C.out_max)
+                    C.out_max[j] = ordered_f64(
+                        *(const unsigned long long*)(slot + 48));
             }
         }
     }
@@ -1490,6 +1514,13 @@ hipError_t launch_init_slab(hipStream_t s, uint8_t* slab, uint32_t n_slots,
                             uint32_t stride) {
     hipLaunchKernelGGL(k_init_slab, dim3(grid_for(n_slots, 256)), dim3(256),
                        0, s, slab, n_slots, stride);
+    return hipGetLastError();
+}
+
+hipError_t launch_init_state_slab(hipStream_t s, uint8_t* slab,
+                                  uint32_t n_slots, uint32_t stride, bool mm) {
+    hipLaunchKernelGGL(k_init_state_slab, dim3(grid_for(n_slots, 256)),
+                       dim3(256), 0, s, slab, n_slots, stride, mm ? 1u : 0u);
     return hipGetLastError();
 }
 
