@@ -416,11 +416,18 @@ k_scan_agg(AggParams P) {
     const int lane = threadIdx.x & 63;
     for (uint32_t rgi = blockIdx.x; rgi < P.n_rgs; rgi += gridDim.x) {
         // early-abort once the table saturates: the host retries with a
-        // larger table, so finishing a doomed pass only burns time
-        if (__hip_atomic_load(P.fill, RLX, AGT) > P.fill_limit) {
-            if (threadIdx.x == 0)
-                __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
-            break;
+        // larger table, so finishing a doomed pass only burns time.
+        // One lane per wave polls the hot counter (same-address loads from
+        // every lane would serialize at the coherence point).
+        {
+            unsigned long long f = 0;
+            if (lane == 0) f = __hip_atomic_load(P.fill, RLX, AGT);
+            f = __shfl(f, 0, 64);
+            if (f > P.fill_limit) {
+                if (threadIdx.x == 0)
+                    __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+                break;
+            }
         }
         const RgDesc rg = P.rgs[rgi];
         const uint64_t* S = (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
