@@ -419,8 +419,8 @@ k_scan_agg(AggParams P) {
         // larger table, so finishing a doomed pass only burns time.
         // One lane per wave polls the hot counter (same-address loads from
         // every lane would serialize at the coherence point).
-        {
-            unsigned long long f = 0;
+        if ((rgi & 7u) == (blockIdx.x & 7u)) {  // throttled: hot-word loads
+            unsigned long long f = 0;                // serialize at ~88/us
             if (lane == 0) f = __hip_atomic_load(P.fill, RLX, AGT);
             f = __shfl(f, 0, 64);
             if (f > P.fill_limit) {
