@@ -419,9 +419,13 @@ k_scan_agg(AggParams P) {
         // larger table, so finishing a doomed pass only burns time.
         // One lane per wave polls the hot counter (same-address loads from
         // every lane would serialize at the coherence point).
-        if ((rgi & 7u) == (blockIdx.x & 7u)) {  // throttled: hot-word loads
-            unsigned long long f = 0;                // serialize at ~88/us
-            if (lane == 0) f = __hip_atomic_load(P.fill, RLX, AGT);
+        if ((rgi & 7u) == (blockIdx.x & 7u)) {
+            // best-effort saturation check: a PLAIN (L1-served) load — a
+            // coherent load of this hot word from every wave serializes at
+            // the coherence point (~88/us) and cost 50% kernel time; stale
+            // reads only delay the abort, which the probe cap bounds anyway
+            unsigned long long f = 0;
+            if (lane == 0) f = *(volatile const unsigned long long*)P.fill;
             f = __shfl(f, 0, 64);
             if (f > P.fill_limit) {
                 if (threadIdx.x == 0)
