@@ -295,6 +295,8 @@ struct DevPlan {
     uint32_t slab_stride = 0;
     uint8_t* t_rep = nullptr;       // per-XCD accumulator replicas (opt-in)
     uint32_t rep_stride = 0;
+    uint8_t* t_bstore = nullptr;    // direct-indexed bucket accumulators
+    uint64_t bstore_cap = 0;
     unsigned long long* d_counters = nullptr;  // fill, overflow, matched, n_out
     uint64_t* d_sset = nullptr;
     size_t sset_cap = 0;
@@ -957,7 +959,7 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
                         (void*)plan.t_bucket, (void*)plan.t_state,
                         (void*)plan.t_sum, (void*)plan.t_cnt, (void*)plan.t_min,
                         (void*)plan.t_max, (void*)plan.t_slab,
-                        (void*)plan.t_rep,
+                        (void*)plan.t_rep, (void*)plan.t_bstore,
                         (void*)plan.d_counters,
                         (void*)plan.d_sset})
             if (p) hipFree(p);
@@ -1142,9 +1144,48 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     if (dst_ != HX_OK) return dst_;
 
     // one-CAS key claim: series-only grouping with a stats-proven sentinel
-    // (HX_FORCE_STATE=1 forces the generic state-word path for A/B runs)
+    // (HX_FORCE_STATE=1 forces the generic state-word path for A/B runs).
+    // Bucket queries take the DIRECT-INDEXED path when the bucket range
+    // (known exactly from the scan range clamped to the catalog) fits a
+    // dense per-slot accumulator row: table keyed by series only, bucket
+    // becomes an array index — no (series,bucket) hashing at all.
+    int64_t lo_bucket = 0;
+    uint32_t n_buckets = 0;
+    uint32_t bstride = 0;
+    if (bucket && P->key_claim_safe && !getenv("HX_FORCE_STATE")) {
+        int64_t lo_ts = P->spec.range.start, hi_ts = P->spec.range.end;
+        int64_t cat_lo = INT64_MAX, cat_hi = INT64_MIN;
+        for (const auto& c : P->h->ssts) {
+            cat_lo = std::min(cat_lo, c.ts_min);
+            cat_hi = std::max(cat_hi, c.ts_max);
+        }
+        if (cat_lo <= cat_hi) {
+            lo_ts = std::max(lo_ts, cat_lo);
+            hi_ts = std::min(hi_ts, cat_hi + 1);
+        }
+        if (lo_ts < hi_ts) {
+            auto fdiv = [&](int64_t t) {
+                int64_t q = t / agg->bucket_ms;
+                if ((t % agg->bucket_ms) != 0 && t < 0) q--;
+                return q;
+            };
+            lo_bucket = fdiv(lo_ts);
+            int64_t nb = fdiv(hi_ts - 1) - lo_bucket + 1;
+            bstride = (agg->ops & (HX_AGG_MIN | HX_AGG_MAX)) ? 32u : 16u;
+            // memory cap: fall back to the generic path beyond ~24 GB
+            uint64_t est_slots = std::max<uint64_t>(
+                1 << 16, next_pow2_u32((uint64_t)plan.rows_scanned / 32));
+            if (nb > 0 &&
+                (uint64_t)nb * est_slots * bstride <= (24ull << 30)) {
+                n_buckets = (uint32_t)nb;
+            } else {
+                bstride = 0;
+            }
+        }
+    }
     const int32_t key_claim =
-        (!bucket && P->key_claim_safe && !getenv("HX_FORCE_STATE")) ? 1 : 0;
+        ((!bucket || n_buckets) && P->key_claim_safe &&
+         !getenv("HX_FORCE_STATE")) ? 1 : 0;
 
     // table size heuristic; grows on overflow
     uint32_t slots = plan.slots;
@@ -1153,8 +1194,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         if (env) slots = next_pow2_u32(strtoull(env, nullptr, 10));
         else slots = next_pow2_u32(std::max<uint64_t>(
                  1 << 16,
-                 // bucket keys multiply the group count: be generous
-                 (uint64_t)plan.rows_scanned / (bucket ? 2 : 32)));
+                 // generic (series,bucket) hashing needs generous sizing;
+                 // the direct-indexed path is keyed by series only
+                 (uint64_t)plan.rows_scanned /
+                     ((bucket && !n_buckets) ? 2 : 32)));
         if (slots > (1u << 28)) slots = 1u << 28;
     }
 
@@ -1162,10 +1205,24 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     for (int attempt = 0; attempt < 4; attempt++) {
         hx_status st = alloc_table(plan, slots, ops, bucket, key_claim);
         if (st != HX_OK) return st;
+        if (n_buckets) {
+            uint64_t need_b = (uint64_t)slots * n_buckets * bstride;
+            if (need_b > plan.bstore_cap) {
+                if (plan.t_bstore) hipFree(plan.t_bstore);
+                plan.t_bstore = nullptr;
+                plan.bstore_cap = 0;
+                HIP_TRY(hipMalloc((void**)&plan.t_bstore, need_b));
+                plan.bstore_cap = need_b;
+            }
+        }
         // reset table + counters (part of the step)
         if (key_claim) {
             HIP_TRY(hx::launch_init_slab(s, plan.t_slab, slots,
                                          plan.slab_stride));
+            if (n_buckets)
+                HIP_TRY(hx::launch_init_rep(
+                    s, plan.t_bstore, (size_t)slots * n_buckets, bstride,
+                    (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
             if (plan.t_rep)
                 HIP_TRY(hx::launch_init_rep(
                     s, plan.t_rep, 8ull * slots, plan.rep_stride,
@@ -1192,8 +1249,12 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.sset_empty = plan.sset_empty;
         A.use_sset = plan.d_sset ? 1 : 0;
         A.bucket_ms = bucket ? agg->bucket_ms : 0;
-        A.ops = ops;
+        A.ops = n_buckets ? (ops | HX_AGG_COUNT) : ops;  // cnt = liveness
         A.key_claim = key_claim;
+        A.lo_bucket = lo_bucket;
+        A.n_buckets = n_buckets;
+        A.bstride = bstride;
+        A.bstore = plan.t_bstore;
         A.skip = getenv("HX_SKIP") ? atoi(getenv("HX_SKIP")) : 0;
         A.table = {nullptr, nullptr, nullptr, nullptr,
                    nullptr, nullptr, nullptr, slots - 1,
@@ -1316,7 +1377,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                    plan.t_slab, plan.slab_stride, plan.t_rep,
                    plan.rep_stride};
     HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,
-                               bucket ? agg->bucket_ms : 0, co));
+                               bucket ? agg->bucket_ms : 0, lo_bucket,
+                               n_buckets, bstride, plan.t_bstore, co));
 
     // sort: LSD-stable — by bucket first (if any), then by series
     HIP_TRY(hx::launch_iota(s, perm_a, n));

@@ -88,6 +88,13 @@ struct AggParams {
     int32_t key_claim;             // 1 = one-CAS key-claim mode (no state word)
     int32_t skip;                  // debug bisect: 1 skip table, 2 also skip dedup
     AggTable table;
+    // direct-indexed bucket mode (bucket_ms > 0, key-claim safe, bucket
+    // range known from the scan range): table keyed by series only; each
+    // slot owns a dense [n_buckets] accumulator row in bstore.
+    int64_t lo_bucket;
+    uint32_t n_buckets;            // 0 => generic state-word path
+    uint32_t bstride;              // 16 {sum,cnt} or 32 {+min,max}
+    uint8_t* bstore;
     unsigned long long fill_limit; // early-abort when fill exceeds this
     unsigned long long* fill;      // claimed slots
     unsigned long long* overflow;  // !=0 => rerun with a larger table

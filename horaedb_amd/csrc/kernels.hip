@@ -90,12 +90,16 @@ __device__ bool shadowed(const AggParams& P, const SstDev& me,
 // per-XCD replication is on (the replica's lines stay exclusive to one L2
 // — no cross-XCD ping-pong on the hot accumulator lines)
 __device__ __forceinline__ void keycas_add(const AggParams& P, uint32_t i,
-                                           double vsum,
+                                           int64_t b, double vsum,
                                            unsigned long long cnt, double mn,
                                            double mx) {
     uint8_t* slot;
     uint32_t off8, off16, off24, off32;
-    if (P.table.rep) {
+    if (P.n_buckets) {  // direct-indexed bucket row of this series slot
+        slot = P.bstore + ((size_t)i * P.n_buckets +
+                           (size_t)(b - P.lo_bucket)) * P.bstride;
+        off8 = 0; off16 = 8; off24 = 16; off32 = 24;
+    } else if (P.table.rep) {
         uint32_t xcc;
         asm("s_getreg_b32 %0, hwreg(HW_REG_XCC_ID)" : "=s"(xcc));
         xcc &= 7u;
@@ -122,7 +126,7 @@ __device__ __forceinline__ void keycas_add(const AggParams& P, uint32_t i,
 // once per exec, so a stale read can only be KEY_EMPTY, which the CAS
 // corrects).
 __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s,
-                                                  double vsum,
+                                                  int64_t b, double vsum,
                                                   unsigned long long cnt,
                                                   double mn, double mx,
                                                   uint32_t hint_i = 0xFFFFFFFFu,
@@ -130,7 +134,7 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
     const uint32_t stride = P.table.stride;
     uint32_t i = (uint32_t)mix64(s) & P.table.mask;
     if (hint_i == i && hint_k == s) {  // prefetched probe already matched
-        keycas_add(P, i, vsum, cnt, mn, mx);
+        keycas_add(P, i, b, vsum, cnt, mn, mx);
         return;
     }
     const uint32_t probe_cap = P.table.mask < 4096u ? P.table.mask : 4096u;
@@ -148,7 +152,7 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
             }
         }
         if (k == s) {
-            keycas_add(P, i, vsum, cnt, mn, mx);
+            keycas_add(P, i, b, vsum, cnt, mn, mx);
             return;
         }
         i = (i + 1) & P.table.mask;
@@ -213,7 +217,7 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
                                            uint32_t hint_i = 0xFFFFFFFFu,
                                            uint64_t hint_k = 0) {
     if (P.key_claim) {
-        agg_update_keycas(P, s, vsum, cnt, mn, mx, hint_i, hint_k);
+        agg_update_keycas(P, s, b, vsum, cnt, mn, mx, hint_i, hint_k);
         return;
     }
     // AoS slot: {state u32, pad u32, series u64, bucket i64, sum, cnt
@@ -552,8 +556,8 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const uint32_t ne = G.ne;
     AggParams* Pm = (AggParams*)smem;          // LDS mirror for cold paths
-    RgDesc* ldesc = (RgDesc*)(smem + 256);
-    const size_t desc_bytes = 256 + (size_t)gang_size * sizeof(RgDesc);
+    RgDesc* ldesc = (RgDesc*)(smem + 320);
+    const size_t desc_bytes = 320 + (size_t)gang_size * sizeof(RgDesc);
     uint64_t* lkey = (uint64_t*)(smem + desc_bytes);
     double* lsum = (double*)(smem + desc_bytes + (size_t)ne * 8);
     unsigned long long* lmin =
@@ -563,7 +567,7 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
     unsigned int* lcnt =
         (unsigned int*)(smem + desc_bytes + (size_t)ne * (MM ? 32 : 16));
     {
-        static_assert(sizeof(AggParams) <= 256, "grow the LDS mirror");
+        static_assert(sizeof(AggParams) <= 320, "grow the LDS mirror");
         const uint64_t* src = (const uint64_t*)&gp->P;
         uint64_t* dst = (uint64_t*)Pm;
         for (uint32_t i = threadIdx.x; i < sizeof(AggParams) / 8;
@@ -899,6 +903,11 @@ struct CompactParams {
     uint32_t ops;
     int32_t key_claim;
     int64_t bucket_ms;
+    // direct-indexed bucket mode
+    int64_t lo_bucket;
+    uint32_t n_buckets;
+    uint32_t bstride;
+    const uint8_t* bstore;
     uint64_t* out_series;
     long long* out_bucket;
     double* out_sum;
@@ -912,6 +921,48 @@ extern "C" __global__ void __launch_bounds__(256)
 k_compact(CompactParams C) {
     const int lane = threadIdx.x & 63;
     const uint32_t stride = blockDim.x * gridDim.x;
+    if (C.n_buckets) {  // direct-indexed bucket mode: sweep (slot, bucket)
+        const size_t total = (size_t)C.n_slots * C.n_buckets;
+        for (size_t p = blockIdx.x * blockDim.x + threadIdx.x;
+             __any(p < total); p += stride) {
+            bool live = false;
+            uint64_t key = 0;
+            const uint8_t* brow = nullptr;
+            if (p < total) {
+                const uint32_t slot_i = (uint32_t)(p / C.n_buckets);
+                key = *(const uint64_t*)(C.table.slab +
+                                         (size_t)slot_i * C.table.stride);
+                brow = C.bstore + p * C.bstride;
+                live = key != KEY_EMPTY &&
+                       *(const unsigned long long*)(brow + 8) > 0;
+            }
+            const unsigned long long mask = __ballot(live);
+            if (!mask) continue;
+            const int leader = __ffsll((unsigned long long)mask) - 1;
+            unsigned long long wave_base = 0;
+            if (lane == leader)
+                wave_base =
+                    atomicAdd(C.n_out, (unsigned long long)__popcll(mask));
+            wave_base = __shfl(wave_base, leader, 64);
+            if (live) {
+                const unsigned long long j =
+                    wave_base + __popcll(mask & ((1ull << lane) - 1ull));
+                C.out_series[j] = key;
+                C.out_bucket[j] =
+                    C.lo_bucket + (long long)(p % C.n_buckets);
+                if (C.out_sum) C.out_sum[j] = *(const double*)brow;
+                if (C.out_cnt)
+                    C.out_cnt[j] = *(const unsigned long long*)(brow + 8);
+                if (C.out_min)
+                    C.out_min[j] = ordered_f64(
+                        *(const unsigned long long*)(brow + 16));
+                if (C.out_max)
+                    C.out_max[j] = ordered_f64(
+                        *(const unsigned long long*)(brow + 24));
+            }
+        }
+        return;
+    }
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
          __any(i < C.n_slots); i += stride) {
         const uint8_t* slot = C.table.slab + (size_t)i * C.table.stride;
@@ -1407,7 +1458,7 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
         uint32_t ne = (uint32_t)strtoul(nee, nullptr, 10);
         if (ne >= 1024 && ne <= 8192 && !(ne & (ne - 1))) G.ne = ne;
     }
-    size_t lds = 256 + (size_t)G.ne * (minmax ? 36 : 20) +
+    size_t lds = 320 + (size_t)G.ne * (minmax ? 36 : 20) +
                  (size_t)G.gang_size * sizeof(RgDesc);
     uint32_t grid = G.n_gangs > 4096 ? 4096 : (G.n_gangs ? G.n_gangs : 1);
     // >64 KiB dynamic LDS needs an explicit opt-in per kernel
@@ -1469,6 +1520,8 @@ hipError_t launch_gather_multi(hipStream_t s,
 
 hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
                           uint32_t ops, int32_t key_claim, int64_t bucket_ms,
+                          int64_t lo_bucket, uint32_t n_buckets,
+                          uint32_t bstride, const uint8_t* bstore,
                           const CompactOut& o) {
     CompactParams C;
     C.table = t;
@@ -1476,6 +1529,10 @@ hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
     C.ops = ops;
     C.key_claim = key_claim;
     C.bucket_ms = bucket_ms;
+    C.lo_bucket = lo_bucket;
+    C.n_buckets = n_buckets;
+    C.bstride = bstride;
+    C.bstore = bstore;
     C.out_series = o.series;
     C.out_bucket = o.bucket;
     C.out_sum = o.sum;
