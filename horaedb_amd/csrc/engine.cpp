@@ -1332,7 +1332,15 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     out.release();
     out.n = fill;
     if (fill == 0) return HX_OK;
-    const uint32_t n = (uint32_t)fill;
+    // capacity: series-only => one group per claimed slot (= fill);
+    // direct-indexed buckets => up to fill x n_buckets groups, but never
+    // more than the surviving-row count
+    uint64_t cap64 = fill;
+    if (n_buckets)
+        cap64 = std::min<uint64_t>(fill * (uint64_t)n_buckets, counters[2]);
+    if (cap64 > 0xFFFFFFFFull)
+        return fail(HX_ERR_UNSUPPORTED, "result exceeds 2^32 groups");
+    const uint32_t n = (uint32_t)cap64;
     const bool aos = plan.t_slab != nullptr;
     const bool has_sum = aos ? (ops & (HX_AGG_SUM | HX_AGG_AVG)) != 0
                              : plan.t_sum != nullptr;
@@ -1379,23 +1387,31 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,
                                bucket ? agg->bucket_ms : 0, lo_bucket,
                                n_buckets, bstride, plan.t_bstore, co));
+    unsigned long long n_groups64 = 0;
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(&n_groups64, d_nout, 8, hipMemcpyDeviceToHost));
+    if (n_groups64 > n)
+        return fail(HX_ERR_HIP, "compact overran its capacity bound");
+    const uint32_t ng = (uint32_t)n_groups64;
+    out.n = ng;
+    if (ng == 0) return HX_OK;
 
     // sort: LSD-stable — by bucket first (if any), then by series
-    HIP_TRY(hx::launch_iota(s, perm_a, n));
+    HIP_TRY(hx::launch_iota(s, perm_a, ng));
     const uint32_t* perm_in = perm_a;
     uint32_t* perm_out = perm_b;
     if (bucket) {
         HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_bucket,
-                                      perm_a, (unsigned long long*)keys_tmp, n));
-        HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)keys_tmp, n));
-        HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, keys_out, perm_a, perm_out, n,
+                                      perm_a, (unsigned long long*)keys_tmp, ng));
+        HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)keys_tmp, ng));
+        HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, keys_out, perm_a, perm_out, ng,
                                    &plan.d_sort_temp, &plan.sort_temp_cap));
         perm_in = perm_out;
         perm_out = perm_c;
     }
     HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)c_series, perm_in,
-                                  (unsigned long long*)keys_tmp, n));
-    HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, keys_out, perm_in, perm_out, n,
+                                  (unsigned long long*)keys_tmp, ng));
+    HIP_TRY(hx::sort_pairs_u64(s, keys_tmp, keys_out, perm_in, perm_out, ng,
                                &plan.d_sort_temp, &plan.sort_temp_cap));
     const uint32_t* perm = perm_out;
 
@@ -1413,11 +1429,11 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     if (has_min) srcs[na++] = (const unsigned long long*)c_min;
     size_t off_max = has_max ? na : 0;
     if (has_max) srcs[na++] = (const unsigned long long*)c_max;
-    HIP_TRY(hx::launch_gather_multi(s, srcs, na, perm, d_dst, n));
+    HIP_TRY(hx::launch_gather_multi(s, srcs, na, perm, d_dst, ng));
     if (has_avg)
         HIP_TRY(hx::launch_avg(s, (const double*)(d_dst + size_t(off_sum) * n),
                                d_dst + size_t(off_cnt) * n,
-                               (double*)(d_dst + size_t(n_core) * n), n));
+                               (double*)(d_dst + size_t(n_core) * n), ng));
 
     const size_t bytes = size_t(n) * 8 * n_total;
     if (hipHostMalloc(&out.buf, bytes, hipHostMallocDefault) == hipSuccess) {
