@@ -1431,11 +1431,12 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     if (has_max) srcs[na++] = (const unsigned long long*)c_max;
     HIP_TRY(hx::launch_gather_multi(s, srcs, na, perm, d_dst, ng));
     if (has_avg)
-        HIP_TRY(hx::launch_avg(s, (const double*)(d_dst + size_t(off_sum) * n),
-                               d_dst + size_t(off_cnt) * n,
-                               (double*)(d_dst + size_t(n_core) * n), ng));
+        HIP_TRY(hx::launch_avg(s,
+                               (const double*)(d_dst + size_t(off_sum) * ng),
+                               d_dst + size_t(off_cnt) * ng,
+                               (double*)(d_dst + size_t(n_core) * ng), ng));
 
-    const size_t bytes = size_t(n) * 8 * n_total;
+    const size_t bytes = size_t(ng) * 8 * n_total;
     if (hipHostMalloc(&out.buf, bytes, hipHostMallocDefault) == hipSuccess) {
         out.pinned = true;
     } else {
@@ -1447,12 +1448,12 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     HIP_TRY(hipStreamSynchronize(s));
     uint64_t* hb = (uint64_t*)out.buf;
     out.series = hb;
-    out.bucket = bucket ? (int64_t*)(hb + off_bucket * n) : nullptr;
-    out.sum = has_sum ? (double*)(hb + off_sum * n) : nullptr;
-    out.cnt = has_cnt ? (unsigned long long*)(hb + off_cnt * n) : nullptr;
-    out.vmin = has_min ? (double*)(hb + off_min * n) : nullptr;
-    out.vmax = has_max ? (double*)(hb + off_max * n) : nullptr;
-    out.avg = has_avg ? (double*)(hb + size_t(n_core) * n) : nullptr;
+    out.bucket = bucket ? (int64_t*)(hb + off_bucket * ng) : nullptr;
+    out.sum = has_sum ? (double*)(hb + off_sum * ng) : nullptr;
+    out.cnt = has_cnt ? (unsigned long long*)(hb + off_cnt * ng) : nullptr;
+    out.vmin = has_min ? (double*)(hb + off_min * ng) : nullptr;
+    out.vmax = has_max ? (double*)(hb + off_max * ng) : nullptr;
+    out.avg = has_avg ? (double*)(hb + size_t(n_core) * ng) : nullptr;
     return HX_OK;
 }
 
