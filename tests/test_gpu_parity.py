@@ -419,3 +419,20 @@ def test_delta_wide_bitwidths(tmp_path):
                         sort=False, ts_encoding="DELTA_BINARY_PACKED")
     lo, hi = int(ts.min()), int(ts.max()) + 1
     check_parity(store, (lo, hi), ops=AGG_SUM | AGG_COUNT)
+
+
+def test_negative_timestamps_buckets_and_segments(tmp_path):
+    # floor-division semantics for negative ts (truncate_by types.rs:82-86)
+    store = str(tmp_path)
+    rng = np.random.default_rng(21)
+    n = 20_000
+    series = np.sort(rng.integers(0, 200, n).astype(np.uint64))
+    ts = rng.integers(-100_000, 100_000, n).astype(np.int64)
+    order = np.lexsort((ts, series))
+    gen_sst_from_arrays(store, 1, series[order], ts[order], rng.random(n),
+                        sort=False)
+    check_parity(store, (-200_000, 200_000), ops=AGG_SUM | AGG_COUNT,
+                 bucket_ms=7_000)
+    check_parity(store, (-50_000, 50_000), ops=OPS_ALL)
+    # streaming mode with negative segments
+    check_scan_rows(store, (-200_000, 200_000), segment_ms=30_000)
