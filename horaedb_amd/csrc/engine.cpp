@@ -315,6 +315,7 @@ struct DevPlan {
     size_t sort_temp_cap = 0;
 
     hipStream_t stream = nullptr;
+    hipEvent_t ev[4] = {nullptr, nullptr, nullptr, nullptr};  // timing
 };
 
 static hx_status ensure_dev(void** p, size_t* cap, size_t need) {
@@ -963,6 +964,8 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
                         (void*)plan.d_counters,
                         (void*)plan.d_sset})
             if (p) hipFree(p);
+        for (auto ev : plan.ev)
+            if (ev) hipEventDestroy(ev);
         if (plan.stream) hipStreamDestroy(plan.stream);
     }
     delete P;
@@ -1036,11 +1039,17 @@ uint32_t kernel_ops(uint32_t ops) {
     return k;
 }
 
+static hx_status ensure_events(DevPlan& plan) {
+    for (auto& ev : plan.ev)
+        if (!ev) HIP_TRY(hipEventCreate(&ev));
+    return HX_OK;
+}
+
 hx_status ensure_decoded(DevPlan& plan, hipStream_t s) {
     if (plan.decoded) return HX_OK;
-    hipEvent_t d0, d1;
-    HIP_TRY(hipEventCreate(&d0));
-    HIP_TRY(hipEventCreate(&d1));
+    hx_status est = ensure_events(plan);
+    if (est != HX_OK) return est;
+    hipEvent_t d0 = plan.ev[2], d1 = plan.ev[3];
     HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
     HIP_TRY(hipEventRecord(d0, s));
     if (!plan.snappy_pages.empty())
@@ -1067,8 +1076,6 @@ hx_status ensure_decoded(DevPlan& plan, hipStream_t s) {
     unsigned long long decode_err = 0;
     HIP_TRY(hipMemcpy(&decode_err, plan.d_counters + 1, 8,
                       hipMemcpyDeviceToHost));
-    hipEventDestroy(d0);
-    hipEventDestroy(d1);
     if (decode_err)
         return fail(HX_ERR_FORMAT, "page decode failed (malformed snappy "
                                    "stream or delta page)");
@@ -1265,9 +1272,9 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.overflow = plan.d_counters + 1;
         A.matched = plan.d_counters + 2;
 
-        hipEvent_t e0, e1;
-        HIP_TRY(hipEventCreate(&e0));
-        HIP_TRY(hipEventCreate(&e1));
+        hx_status est = ensure_events(plan);
+        if (est != HX_OK) return est;
+        hipEvent_t e0 = plan.ev[0], e1 = plan.ev[1];
         HIP_TRY(hipEventRecord(e0, s));
         // the LDS-gang variant measured slower than the wave kernel at every
         // tested config (see profiles/README r01 notes); opt-in for further
@@ -1313,8 +1320,6 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipStreamSynchronize(s));
         float ms = 0;
         HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
-        hipEventDestroy(e0);
-        hipEventDestroy(e1);
         HIP_TRY(hipMemcpy(counters, plan.d_counters, 32, hipMemcpyDeviceToHost));
         if (counters[1] == 0) {  // no overflow
             *agg_kernel_ms = ms;
