@@ -69,7 +69,8 @@ def make_series_ids(n_series, seed):
 
 
 def write_sst(path, series, ts, value, seq, row_group=8192,
-              compression="none", ts_encoding="PLAIN", dict_columns=()):
+              compression="none", ts_encoding="PLAIN", dict_columns=(),
+              slim_builtins=False):
     """Write one SST from explicit row arrays (must be (series,ts)-sorted)."""
     import pyarrow as pa
     import pyarrow.parquet as pq
@@ -82,6 +83,12 @@ def write_sst(path, series, ts, value, seq, row_group=8192,
         "__reserved__": np.zeros(n, dtype=np.uint64),
     }, schema=_pa_schema())
     kw = {}
+    if slim_builtins and not dict_columns:
+        # __seq__/__reserved__ are constant per file and never read by the
+        # scan (seq comes from the file id); dictionary-encode them to cut
+        # ~40% of file size for the 8-GPU bench datasets. The three DATA
+        # columns keep their configured encodings.
+        dict_columns = ("__seq__", "__reserved__")
     if dict_columns:
         # RLE_DICTIONARY chunks (config.rs:54-75 with dictionaries on);
         # large limits keep one dictionary per chunk (no PLAIN fallback)
@@ -137,7 +144,8 @@ def _gen_one(args):
     seq = w + 1
     path = os.path.join(out_dir, "data", f"{seq}.sst")
     n = write_sst(path, series, ts, value, seq, row_group=row_group,
-                  compression=compression, ts_encoding=ts_encoding)
+                  compression=compression, ts_encoding=ts_encoding,
+                  slim_builtins=True)
     return {"seq": seq, "path": path, "rows": int(n),
             "ts_min": int(ts_start + t0 * step_ms),
             "ts_max": int(ts_start + (t1 - 1) * step_ms)}
