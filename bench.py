@@ -105,6 +105,30 @@ def cpu_baseline_leg(store_dir, m, ts_range, budget_s=20.0, cores=1):
     }
 
 
+def merge_bucket_partials(series_t, bucket_t, sum_t, cnt_t):
+    """Merge gathered (series,bucket) partial tables (torch tensors, any
+    device): stable sort by bucket then series, then segment-combine equal
+    keys. The config-5 combine step after the RCCL all-gather."""
+    import torch
+    if series_t.numel() == 0:
+        return series_t, bucket_t, sum_t, cnt_t
+    ob = torch.argsort(bucket_t, stable=True)
+    s1, b1, v1, c1 = series_t[ob], bucket_t[ob], sum_t[ob], cnt_t[ob]
+    os_ = torch.argsort(s1, stable=True)
+    s2, b2, v2, c2 = s1[os_], b1[os_], v1[os_], c1[os_]
+    new = torch.ones_like(s2, dtype=torch.bool)
+    new[1:] = (s2[1:] != s2[:-1]) | (b2[1:] != b2[:-1])
+    gid = torch.cumsum(new.to(torch.int64), 0) - 1
+    n = int(gid[-1].item()) + 1
+    out_s = s2[new]
+    out_b = b2[new]
+    out_v = torch.zeros(n, dtype=v2.dtype, device=v2.device)
+    out_c = torch.zeros(n, dtype=c2.dtype, device=c2.device)
+    out_v.scatter_add_(0, gid, v2)
+    out_c.scatter_add_(0, gid, c2)
+    return out_s, out_b, out_v, out_c
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -120,6 +144,10 @@ def main():
     p.add_argument("--bucket-ms", type=int, default=0)
     p.add_argument("--selectivity", type=float, default=0,
                    help="config 3: series-set predicate keeping this fraction")
+    p.add_argument("--config5", action="store_true",
+                   help="BASELINE config 5: ONE shared dataset, SSTs sharded "
+                        "round-robin across ranks, time_bucket partials "
+                        "merged via an RCCL/gloo all-gather + GPU combine")
     p.add_argument("--ops", default="sum,count")
     p.add_argument("--data-dir", default="/tmp/hx_bench_data")
     p.add_argument("--no-cpu-baseline", action="store_true")
@@ -153,7 +181,19 @@ def main():
         ops |= opmap[o.strip()]
 
     from tools.gen_ssts import middle_range
-    store_dir, m = get_dataset(args, rank)
+    if args.config5:
+        if args.bucket_ms <= 0:
+            args.bucket_ms = 60_000
+        # ONE shared dataset (same seed); rank 0 generates, others reuse
+        if dist and rank != 0:
+            dist.barrier()
+            store_dir, m = get_dataset(args, 0)
+        else:
+            store_dir, m = get_dataset(args, 0)
+            if dist:
+                dist.barrier()
+    else:
+        store_dir, m = get_dataset(args, rank)
     ts_range = middle_range(m, args.range_frac)
     series_in = None
     if args.selectivity > 0:
@@ -174,11 +214,49 @@ def main():
 
     store = Store(store_dir)
     t0 = time.time()
-    prep = store.prepare(ts_range, series_in=series_in, devices=[device])
+    sst_subset = None
+    if args.config5 and world > 1:
+        cat = store.find_ssts(ts_range)
+        sst_subset = [e for i, e in enumerate(sorted(cat, key=lambda x: x[1]))
+                      if i % world == rank]
+        log(f"rank{rank}: config5 shard = {len(sst_subset)} of {len(cat)} SSTs")
+    prep = store.prepare(ts_range, series_in=series_in, devices=[device],
+                         sst_subset=sst_subset)
     log(f"rank{rank}: staged in {time.time() - t0:.1f}s")
 
     def step(copy=False):
-        return prep.exec_agg(ops=ops, bucket_ms=args.bucket_ms, copy=copy)
+        if not (args.config5 and dist):
+            return prep.exec_agg(ops=ops, bucket_ms=args.bucket_ms, copy=copy)
+        # config 5: local bucket partials -> all-gather over RCCL (xGMI) ->
+        # combine on the GPU; part of the timed step
+        res = prep.exec_agg(ops=ops | 2, bucket_ms=args.bucket_ms, copy=True)
+        dev = torch.device(f"cuda:{device}") if backend == "nccl" \
+            else torch.device("cpu")
+        s_t = torch.from_numpy(res["series_id"].view("int64").copy()).to(dev)
+        b_t = torch.from_numpy(res["bucket"].copy()).to(dev)
+        v_t = torch.from_numpy(res["sum"].copy()).to(dev)
+        c_t = torch.from_numpy(res["count"].view("int64").copy()).to(dev)
+        n_local = torch.tensor([s_t.numel()], dtype=torch.int64, device=dev)
+        sizes = [torch.zeros_like(n_local) for _ in range(world)]
+        dist.all_gather(sizes, n_local)
+        n_max = int(max(int(x.item()) for x in sizes))
+        def pad(t, fill):
+            out = torch.full((n_max,), fill, dtype=t.dtype, device=dev)
+            out[: t.numel()] = t
+            return out
+        gathered = []
+        for t, fill in ((s_t, -1), (b_t, 0), (v_t, 0.0), (c_t, 0)):
+            bufs = [torch.empty(n_max, dtype=t.dtype, device=dev)
+                    for _ in range(world)]
+            dist.all_gather(bufs, pad(t, fill))
+            gathered.append(bufs)
+        ss, bb, vv, cc = (
+            torch.cat([g[r][: int(sizes[r].item())] for r in range(world)])
+            for g in gathered)
+        ms, mb, mv, mc = merge_bucket_partials(ss, bb, vv, cc)
+        if backend == "nccl":
+            torch.cuda.synchronize(device)
+        return {"n_groups": int(ms.numel()), "merged_count": int(mc.sum().item())}
 
     # warmup (one materialized run for the group count, the rest light)
     res = step(copy=True)
@@ -186,7 +264,8 @@ def main():
         step()
     st = prep.stats()
     log(f"rank{rank}: rows_scanned={st['rows_scanned']} "
-        f"matched={st['rows_matched']} groups={len(res['series_id'])} "
+        f"matched={st['rows_matched']} "
+        f"groups={res.get('n_groups', len(res.get('series_id', [])))} "
         f"stage_ms={st['stage_ms']:.0f} exec_ms={st['exec_ms']:.1f} "
         f"agg_kernel_ms={st['agg_kernel_ms']:.2f}")
 

@@ -277,8 +277,20 @@ class Store:
         self._keepalive += [ids, ds]
         return ds
 
-    def prepare(self, ts_range, series_in=None, devices=None):
+    def prepare(self, ts_range, series_in=None, devices=None,
+                sst_subset=None):
+        """sst_subset: list of (path, seq) to scan (hx_scan_spec.ssts — the
+        per-GPU shard of BASELINE configs 4/5); None = all overlapping."""
         spec = self._spec(ts_range, series_in)
+        if sst_subset is not None:
+            arr = (_SstDesc * len(sst_subset))()
+            enc = [p.encode() for p, _ in sst_subset]
+            for i, ((p, q), pe) in enumerate(zip(sst_subset, enc)):
+                arr[i].path = pe
+                arr[i].sequence = q
+            spec.ssts = arr
+            spec.n_ssts = len(sst_subset)
+            self._keepalive += [arr, enc]
         ds = self._devset(devices)
         out = C.c_void_p()
         _check(_lib.hx_prepare(self._h, C.byref(spec),
