@@ -97,3 +97,53 @@ def test_two_rank_sharded_aggregate(tmp_path):
         combined["sum"], rtol=1e-12)
     assert [merged[s][1] for s in combined["series_id"].tolist()] == \
         combined["count"].tolist()
+
+
+def test_merge_bucket_partials_math():
+    # the config-5 combine after the all-gather, vs the oracle on the union
+    import torch
+    from bench import merge_bucket_partials
+    import oracle
+    from oracle.scan import AGG_SUM, AGG_COUNT
+    from oracle import SstBatch
+
+    rng = np.random.default_rng(5)
+    # two "rank shards" with overlapping (series,bucket) keys
+    shards = []
+    all_ssts = []
+    for r in range(2):
+        n = 5000
+        series = rng.integers(0, 60, n).astype(np.uint64)
+        ts = rng.integers(0, 100_000, n).astype(np.int64)
+        vals = rng.random(n)
+        order = np.lexsort((ts, series))
+        sst = SstBatch([series[order], ts[order], vals[order]], r + 1)
+        all_ssts.append(sst)
+        part = oracle.scan_agg([sst], (0, 10**9), bucket_ms=7000,
+                               ops=AGG_SUM | AGG_COUNT)
+        shards.append(part)
+
+    s = torch.from_numpy(np.concatenate(
+        [p["series_id"].view("int64") for p in shards]).copy())
+    bkt = torch.from_numpy(np.concatenate([p["bucket"] for p in shards]).copy())
+    v = torch.from_numpy(np.concatenate([p["sum"] for p in shards]).copy())
+    c = torch.from_numpy(np.concatenate(
+        [p["count"].view("int64") for p in shards]).copy())
+    ms, mb, mv, mc = merge_bucket_partials(s, bkt, v, c)
+
+    # NOTE: partial merge is only key-wise addition — valid when shards hold
+    # DISJOINT row sets (config 5's SST sharding); dedup across shards is
+    # not the merge's job, so build the expectation without cross-shard dedup
+    exp_keys = {}
+    for p in shards:
+        for se, bu, su, ct in zip(p["series_id"], p["bucket"], p["sum"],
+                                  p["count"]):
+            k = (int(se), int(bu))
+            a0, b0 = exp_keys.get(k, (0.0, 0))
+            exp_keys[k] = (a0 + float(su), b0 + int(ct))
+    keys = sorted(exp_keys)
+    assert [(int(a), int(b)) for a, b in zip(ms.numpy().astype(np.uint64),
+                                             mb.numpy())] == keys
+    np.testing.assert_allclose(mv.numpy(), [exp_keys[k][0] for k in keys],
+                               rtol=1e-12)
+    assert mc.numpy().tolist() == [exp_keys[k][1] for k in keys]
