@@ -216,10 +216,23 @@ def main():
     t0 = time.time()
     sst_subset = None
     if args.config5 and world > 1:
+        # Shard whole ts-overlap clusters (dedup never crosses ranks; the
+        # cross-rank combine stays pure key-wise addition — DESIGN.md §6)
         cat = store.find_ssts(ts_range)
-        sst_subset = [e for i, e in enumerate(sorted(cat, key=lambda x: x[1]))
-                      if i % world == rank]
-        log(f"rank{rank}: config5 shard = {len(sst_subset)} of {len(cat)} SSTs")
+        spans = {e["seq"]: (e["ts_min"], e["ts_max"])
+                 for e in store.catalog()}
+        clusters = []
+        for path, seq in sorted(cat, key=lambda x: spans[x[1]][0]):
+            lo, hi = spans[seq]
+            if clusters and lo <= clusters[-1][0]:
+                clusters[-1] = (max(hi, clusters[-1][0]),
+                                clusters[-1][1] + [(path, seq)])
+            else:
+                clusters.append((hi, [(path, seq)]))
+        sst_subset = [e for i, (_, members) in enumerate(clusters)
+                      if i % world == rank for e in members]
+        log(f"rank{rank}: config5 shard = {len(sst_subset)} of {len(cat)} "
+            f"SSTs ({len(clusters)} ts-overlap clusters)")
     prep = store.prepare(ts_range, series_in=series_in, devices=[device],
                          sst_subset=sst_subset)
     log(f"rank{rank}: staged in {time.time() - t0:.1f}s")
