@@ -976,9 +976,63 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
 // ---------------------------------------------------------------------------
 namespace {
 
+// Pinned result buffers are recycled process-wide: hipHostMalloc/hipHostFree
+// page-lock and unlock the whole region, a multi-ms host cost per exec at
+// config-2 sizes (~240 MB per result). The pool keeps a few buffers alive
+// across hx_exec_agg/hx_result_free cycles. HX_HOSTPOOL=0 disables it (A/B).
+struct PinnedPool {
+    std::mutex mu;
+    std::vector<std::pair<void*, size_t>> bufs;  // free pinned buffers
+    bool enabled() {
+        const char* e = getenv("HX_HOSTPOOL");
+        return !(e && e[0] == '0');
+    }
+    void* acquire(size_t bytes, bool* pinned, size_t* cap) {
+        if (enabled()) {
+            std::lock_guard<std::mutex> g(mu);
+            // smallest free buffer that fits
+            size_t best = bufs.size();
+            for (size_t i = 0; i < bufs.size(); i++)
+                if (bufs[i].second >= bytes &&
+                    (best == bufs.size() || bufs[i].second < bufs[best].second))
+                    best = i;
+            if (best < bufs.size()) {
+                void* p = bufs[best].first;
+                *cap = bufs[best].second;
+                bufs.erase(bufs.begin() + best);
+                *pinned = true;
+                return p;
+            }
+        }
+        void* p = nullptr;
+        if (hipHostMalloc(&p, bytes, hipHostMallocDefault) == hipSuccess) {
+            *pinned = true;
+            *cap = bytes;
+            return p;
+        }
+        *pinned = false;
+        *cap = bytes;
+        return malloc(bytes);
+    }
+    void release(void* p, size_t cap, bool pinned) {
+        if (!p) return;
+        if (!pinned) { free(p); return; }
+        if (enabled()) {
+            std::lock_guard<std::mutex> g(mu);
+            if (bufs.size() < 4) {
+                bufs.emplace_back(p, cap);
+                return;
+            }
+        }
+        hipHostFree(p);
+    }
+};
+PinnedPool g_result_pool;
+
 struct HostTable {  // one device's sorted aggregate table, on host
     size_t n = 0;
     void* buf = nullptr;   // pinned when possible (single D2H copy)
+    size_t cap = 0;
     bool pinned = false;
     uint64_t* series = nullptr;
     int64_t* bucket = nullptr;
@@ -989,11 +1043,11 @@ struct HostTable {  // one device's sorted aggregate table, on host
     double* avg = nullptr;
     void release() {
         if (buf) {
-            if (pinned) hipHostFree(buf);
-            else free(buf);
+            g_result_pool.release(buf, cap, pinned);
             buf = nullptr;
         }
         n = 0;
+        cap = 0;
     }
 };
 
@@ -1442,13 +1496,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                                (double*)(d_dst + size_t(n_core) * ng), ng));
 
     const size_t bytes = size_t(ng) * 8 * n_total;
-    if (hipHostMalloc(&out.buf, bytes, hipHostMallocDefault) == hipSuccess) {
-        out.pinned = true;
-    } else {
-        out.buf = malloc(bytes);
-        out.pinned = false;
-        if (!out.buf) return fail(HX_ERR_IO, "result alloc failed");
-    }
+    out.buf = g_result_pool.acquire(bytes, &out.pinned, &out.cap);
+    if (!out.buf) return fail(HX_ERR_IO, "result alloc failed");
     HIP_TRY(hipMemcpyAsync(out.buf, d_dst, bytes, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
     uint64_t* hb = (uint64_t*)out.buf;
