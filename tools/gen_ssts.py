@@ -70,7 +70,7 @@ def make_series_ids(n_series, seed):
 
 def write_sst(path, series, ts, value, seq, row_group=8192,
               compression="none", ts_encoding="PLAIN", dict_columns=(),
-              slim_builtins=False):
+              slim_builtins=False, data_page_version="1.0"):
     """Write one SST from explicit row arrays (must be (series,ts)-sorted)."""
     import pyarrow as pa
     import pyarrow.parquet as pq
@@ -105,7 +105,7 @@ def write_sst(path, series, ts, value, seq, row_group=8192,
     pq.write_table(
         tbl, path, row_group_size=row_group,
         compression="NONE" if compression == "none" else compression.upper(),
-        data_page_version="1.0", column_encoding=enc,
+        data_page_version=data_page_version, column_encoding=enc,
         write_statistics=True, **kw)
     return n
 
