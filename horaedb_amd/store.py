@@ -328,10 +328,13 @@ class Store:
                                C.byref(ds) if ds else None, C.byref(out)))
         return out.value
 
-    def scan(self, ts_range, series_in=None, projection=None, devices=None):
+    def scan(self, ts_range, series_in=None, projection=None, devices=None,
+             batch_limit=None):
         """Streaming parity mode (hx_scan): the merged, deduplicated row
         stream itself — ColumnarStorage::scan semantics (storage.rs:335-370).
-        Returns dict of concatenated column arrays in stream order."""
+        Returns dict of concatenated column arrays in stream order.
+        batch_limit: stop the stream (callback returns nonzero — the
+        header's early-stop contract) after that many batches."""
         spec = self._spec(ts_range, series_in)
         if projection is not None:
             parr = (C.c_int32 * len(projection))(*projection)
@@ -353,6 +356,8 @@ class Store:
                 arr = np.ctypeslib.as_array(ptr, shape=(b.n_rows,)).copy()
                 cols.append(arr.view(dt))
             chunks.append(cols)
+            if batch_limit is not None and len(chunks) >= batch_limit:
+                return 1
             return 0
 
         def on_batch_safe(ctx, bp):

@@ -436,3 +436,91 @@ def test_negative_timestamps_buckets_and_segments(tmp_path):
     check_parity(store, (-50_000, 50_000), ops=OPS_ALL)
     # streaming mode with negative segments
     check_scan_rows(store, (-200_000, 200_000), segment_ms=30_000)
+
+def test_data_page_v2(tmp_path):
+    # DataPageHeaderV2 (REQUIRED flat columns => zero level bytes); the
+    # walker's v2 branch + payload offset math
+    from tools.gen_ssts import write_sst
+    store = str(tmp_path)
+    os.makedirs(os.path.join(store, "data"))
+    rng = np.random.default_rng(21)
+    n = 50_000
+    series = np.sort(rng.integers(0, 400, n).astype(np.uint64))
+    ts = np.arange(n, dtype=np.int64) * 9
+    vals = rng.random(n)
+    write_sst(os.path.join(store, "data", "1.sst"), series, ts, vals, 1,
+              data_page_version="2.0")
+    check_parity(store, (0, 10**9), ops=OPS_ALL)
+    check_parity(store, (50_000, 300_000), ops=AGG_SUM | AGG_COUNT)
+
+
+def test_data_page_v2_snappy(tmp_path):
+    # v2 + snappy: levels (0 bytes here) stay uncompressed, data compressed
+    from tools.gen_ssts import write_sst
+    store = str(tmp_path)
+    os.makedirs(os.path.join(store, "data"))
+    rng = np.random.default_rng(22)
+    n = 40_000
+    series = np.sort(rng.integers(0, 200, n).astype(np.uint64))
+    ts = np.arange(n, dtype=np.int64) * 5
+    vals = rng.random(n)
+    write_sst(os.path.join(store, "data", "1.sst"), series, ts, vals, 1,
+              compression="snappy", data_page_version="2.0")
+    check_parity(store, (0, 10**9), ops=AGG_SUM | AGG_COUNT | AGG_MIN)
+
+
+def test_unsupported_codec_fails_loudly(tmp_path):
+    # out-of-scope codec => HX_ERR_UNSUPPORTED, never silent wrong numbers
+    from horaedb_amd import Store, HxError
+    from tools.gen_ssts import write_sst
+    store = str(tmp_path)
+    os.makedirs(os.path.join(store, "data"))
+    n = 10_000
+    write_sst(os.path.join(store, "data", "1.sst"),
+              np.arange(n, dtype=np.uint64), np.arange(n, dtype=np.int64),
+              np.ones(n), 1, compression="zstd")
+    with Store(store) as st:
+        with pytest.raises(HxError) as ei:
+            st.scan_agg((0, 10**9), devices=[0])
+        assert ei.value.code == 3  # HX_ERR_UNSUPPORTED
+        assert "codec" in str(ei.value)
+
+
+def test_series_set_unsorted_with_absent_ids(ds_plain):
+    # predicate ids arrive unsorted, with duplicates and ids not in the data
+    out, m = ds_plain
+    ids = np.load(os.path.join(out, "series_ids.npy"))
+    sel = ids[::71].tolist()
+    sel = sel[::-1] + sel[:3] + [2**63 + 5, 12345]  # reversed + dups + absent
+    check_parity(out, middle_range(m), ops=AGG_SUM | AGG_COUNT,
+                 series_in=sel)
+
+
+def test_many_tiny_overlapping_ssts(tmp_path):
+    # 40 single-row-group SSTs over the same ts window: dedup across the
+    # whole cluster, newest file wins per PK
+    store = str(tmp_path)
+    rng = np.random.default_rng(23)
+    for seq in range(1, 41):
+        n = 50
+        series = rng.integers(0, 25, n).astype(np.uint64)
+        ts = rng.integers(0, 2_000, n).astype(np.int64)
+        vals = rng.random(n)
+        gen_sst_from_arrays(store, seq, series, ts, vals)
+    check_parity(store, (0, 10**9), ops=OPS_ALL)
+    check_parity(store, (500, 1_500), ops=AGG_SUM | AGG_COUNT, bucket_ms=250)
+
+
+def test_scan_early_stop_batch_limit(ds_plain):
+    # hx_scan's callback contract: nonzero return stops the stream cleanly
+    # (horaedb_hx.h hx_batch_cb; the reference consumer dropping the stream)
+    from horaedb_amd import Store
+    out, m = ds_plain
+    with Store(out) as st:
+        full = st.scan((0, 2**62), devices=[0])
+        part = st.scan((0, 2**62), devices=[0], batch_limit=2)
+    assert len(full["series_id"]) > 2 * 65536
+    assert len(part["series_id"]) == 2 * 65536
+    np.testing.assert_array_equal(part["series_id"],
+                                  full["series_id"][:2 * 65536])
+    np.testing.assert_array_equal(part["value"], full["value"][:2 * 65536])
