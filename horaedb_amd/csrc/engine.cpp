@@ -526,7 +526,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
 
             uint64_t data_off = j.dst_off;  // raw page bytes (post-codec)
             size_t raw_size = payload;
-            if (j.codec == hx::CODEC_SNAPPY) {
+            if (j.codec == hx::CODEC_SNAPPY && dp->is_compressed) {
                 std::lock_guard<std::mutex> g(mu);
                 hx::SnappyPageDesc sp{};
                 sp.src_off = j.dst_off;

@@ -162,6 +162,7 @@ std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
         ThriftReader r(buf + pos, len - pos);
         PageDesc pd{};
         pd.page_type = -1;
+        pd.is_compressed = 1;
         int32_t uncomp = 0, comp = 0;
         int16_t last = 0, id; uint8_t t;
         while (r.field(last, id, t)) {
@@ -199,6 +200,7 @@ std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
                             case 4: pd.encoding = int32_t(r.zigzag()); break;
                             case 5: pd.def_level_bytes = int32_t(r.zigzag()); break;
                             case 6: pd.def_level_bytes += int32_t(r.zigzag()); break;
+                            case 7: pd.is_compressed = (t2 == 1); break;
                             default: r.skip(t2);
                         }
                     }

@@ -70,6 +70,7 @@ struct PageDesc {
     int32_t compressed_size;  // payload bytes in file
     int32_t uncompressed_size;
     int32_t def_level_bytes;  // data page v2 only (levels precede payload)
+    int32_t is_compressed = 1; // data page v2 is_compressed flag (v1: always)
 };
 
 // Walk page headers of a column chunk (buf = the chunk's bytes; base_off =
