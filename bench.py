@@ -149,7 +149,7 @@ def main():
                         "round-robin across ranks, time_bucket partials "
                         "merged via an RCCL/gloo all-gather + GPU combine")
     p.add_argument("--ops", default="sum,count")
-    p.add_argument("--pipeline", type=int, default=1,
+    p.add_argument("--pipeline", type=int, default=3,
                    help="concurrent scans in flight (N prepared objects, N "
                         "host threads): overlaps one query result path "
                         "with the next query kernel, the server "
