@@ -524,3 +524,30 @@ def test_scan_early_stop_batch_limit(ds_plain):
     np.testing.assert_array_equal(part["series_id"],
                                   full["series_id"][:2 * 65536])
     np.testing.assert_array_equal(part["value"], full["value"][:2 * 65536])
+
+
+def test_concurrent_prepared_scans(ds_plain):
+    # the concurrency contract (horaedb_hx.h): scan-side calls from many
+    # threads of one handle, one call per prepared — two prepareds on two
+    # threads must produce identical, independent results (the server
+    # concurrent-scan pattern bench.py --pipeline relies on)
+    from concurrent.futures import ThreadPoolExecutor
+    from horaedb_amd import Store
+    out, m = ds_plain
+    with Store(out) as st:
+        preps = [st.prepare(middle_range(m), devices=[0]) for _ in range(3)]
+        try:
+            def run(p):
+                return p.exec_agg(ops=OPS_ALL)
+            for _ in range(3):  # a few rounds to shake out races
+                with ThreadPoolExecutor(3) as ex:
+                    rs = list(ex.map(run, preps))
+                for r in rs[1:]:
+                    assert r["series_id"].tolist() == rs[0]["series_id"].tolist()
+                    assert r["count"].tolist() == rs[0]["count"].tolist()
+                    np.testing.assert_array_equal(r["vmin"], rs[0]["vmin"])
+                    np.testing.assert_allclose(r["sum"], rs[0]["sum"],
+                                               rtol=1e-12)
+        finally:
+            for p in preps:
+                p.close()
