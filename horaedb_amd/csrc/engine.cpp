@@ -314,6 +314,18 @@ struct DevPlan {
     void* d_sort_temp = nullptr;
     size_t sort_temp_cap = 0;
 
+    // series-range mode (k_scan_agg_range)
+    std::vector<int32_t> h_sst_rgs;     // rg indices grouped per SST, row order
+    std::vector<int32_t> h_sst_rg_off;
+    std::vector<int32_t> h_sst_rg_cnt;
+    int32_t* d_sst_rgs = nullptr;
+    int32_t* d_sst_rg_off = nullptr;
+    int32_t* d_sst_rg_cnt = nullptr;
+    uint64_t* d_range_bounds = nullptr;   // n_blocks+1 series boundaries
+    uint64_t* d_bound_rows = nullptr;     // (n_blocks+1) x n_ssts packed
+    uint32_t range_nblocks = 0;
+    bool range_ready = false;
+
     hipStream_t stream = nullptr;
     hipEvent_t ev[4] = {nullptr, nullptr, nullptr, nullptr};  // timing
 };
@@ -731,6 +743,29 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
         plan.rgs.swap(reordered);
     }
 
+    // per-SST rg lists in row order (series-range kernel): the
+    // fractional reorder above scrambles plan.rgs, so the range kernel
+    // walks SSTs through these index lists instead
+    {
+        const size_t n = plan.rgs.size();
+        std::vector<uint32_t> idx(n);
+        for (size_t i = 0; i < n; i++) idx[i] = (uint32_t)i;
+        std::sort(idx.begin(), idx.end(), [&](uint32_t a, uint32_t b) {
+            if (plan.rgs[a].sst_id != plan.rgs[b].sst_id)
+                return plan.rgs[a].sst_id < plan.rgs[b].sst_id;
+            return plan.rgs[a].row_base < plan.rgs[b].row_base;
+        });
+        plan.h_sst_rgs.assign(idx.begin(), idx.end());
+        plan.h_sst_rg_off.assign(plan.ssts.size(), 0);
+        plan.h_sst_rg_cnt.assign(plan.ssts.size(), 0);
+        for (size_t i = 0; i < n; i++) {
+            uint32_t sid = plan.rgs[idx[i]].sst_id;
+            if (plan.h_sst_rg_cnt[sid] == 0)
+                plan.h_sst_rg_off[sid] = (int32_t)i;
+            plan.h_sst_rg_cnt[sid]++;
+        }
+    }
+
     // ---- upload ----------------------------------------------------------
     HIP_TRY(hipSetDevice(plan.device));
     if (!plan.stream) HIP_TRY(hipStreamCreate(&plan.stream));
@@ -757,6 +792,9 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     HIP_TRY(upload(plan.d_snappy, plan.snappy_pages));
     HIP_TRY(upload(plan.d_rledict, plan.rledict_pages));
     HIP_TRY(upload(plan.d_copies, plan.copies));
+    HIP_TRY(upload(plan.d_sst_rgs, plan.h_sst_rgs));
+    HIP_TRY(upload(plan.d_sst_rg_off, plan.h_sst_rg_off));
+    HIP_TRY(upload(plan.d_sst_rg_cnt, plan.h_sst_rg_cnt));
     HIP_TRY(hipMalloc((void**)&plan.d_counters, 4 * sizeof(unsigned long long)));
     HIP_TRY(hipStreamSynchronize(plan.stream));
     // staging buffer no longer needed once resident in HBM (frees up to
@@ -962,7 +1000,9 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
                         (void*)plan.t_max, (void*)plan.t_slab,
                         (void*)plan.t_rep, (void*)plan.t_bstore,
                         (void*)plan.d_counters,
-                        (void*)plan.d_sset})
+                        (void*)plan.d_sset, (void*)plan.d_sst_rgs,
+                        (void*)plan.d_sst_rg_off, (void*)plan.d_sst_rg_cnt,
+                        (void*)plan.d_range_bounds, (void*)plan.d_bound_rows})
             if (p) hipFree(p);
         for (auto ev : plan.ev)
             if (ev) hipEventDestroy(ev);
@@ -1185,6 +1225,74 @@ static hx::AggParams base_params(hx_prepared* P, DevPlan& plan) {
     return A;
 }
 
+// Series-range partition build, once per prepared: stride-512
+// device samples -> host sort -> distinct-count solve (u = x(1 - e^{-m/x})
+// inverted by bisection) -> equal-sample quantile boundaries -> cached
+// per-(block, sst) row bounds via k_range_bounds. Decode is deterministic,
+// so the cached row bounds stay valid under HX_REDECODE.
+static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
+    const uint32_t n_rgs = (uint32_t)plan.rgs.size();
+    const uint32_t n_ssts = (uint32_t)plan.ssts.size();
+    if (!n_rgs || !n_ssts) return HX_OK;  // leaves range_ready false
+    hipStream_t s = plan.stream;
+    const size_t n_samp = (size_t)n_rgs * 16;
+    uint64_t* d_samp = nullptr;
+    HIP_TRY(hipMalloc((void**)&d_samp, n_samp * 8));
+    hipError_t ke = hx::launch_sample_series(s, plan.d_rgs, n_rgs,
+                                             plan.d_blob, plan.d_dec, d_samp);
+    std::vector<uint64_t> samp(n_samp);
+    if (ke == hipSuccess)
+        ke = hipMemcpyAsync(samp.data(), d_samp, n_samp * 8,
+                            hipMemcpyDeviceToHost, s);
+    hipStreamSynchronize(s);
+    hipFree(d_samp);
+    if (ke != hipSuccess)
+        return fail(HX_ERR_HIP, std::string("range sampling failed: ") +
+                                    hipGetErrorString(ke));
+    std::sort(samp.begin(), samp.end());
+    while (!samp.empty() && samp.back() == ~0ull) samp.pop_back();
+    const size_t m = samp.size();
+    if (m == 0) return HX_OK;
+    size_t u = 1;
+    for (size_t i = 1; i < m; i++) u += samp[i] != samp[i - 1];
+    double staged = 0;
+    for (const auto& sd : plan.ssts) staged += (double)sd.n_staged;
+    // solve u = x (1 - e^{-m/x}) for the distinct-series count x
+    double x_lo = (double)u, x_hi = std::max<double>(staged, (double)u + 1);
+    for (int it = 0; it < 60; it++) {
+        double x = 0.5 * (x_lo + x_hi);
+        double g = x * (1.0 - std::exp(-(double)m / x));
+        (g < (double)u ? x_lo : x_hi) = x;
+    }
+    const double x_est = 0.5 * (x_lo + x_hi);
+    double target = 700.0;   // series per block (ne=2048, load ~0.34)
+    if (const char* e = getenv("HX_RANGE_TARGET")) target = atof(e);
+    uint32_t nb = 512;
+    while (nb < (uint32_t)std::min(1e9, x_est * 1.3 / target) &&
+           nb < (1u << 17))
+        nb <<= 1;
+    std::vector<uint64_t> bounds(nb + 1);
+    bounds[0] = 0;
+    for (uint32_t b = 1; b < nb; b++) bounds[b] = samp[(size_t)b * m / nb];
+    bounds[nb] = ~0ull;
+    HIP_TRY(hipMalloc((void**)&plan.d_range_bounds, (size_t)(nb + 1) * 8));
+    HIP_TRY(hipMemcpyAsync(plan.d_range_bounds, bounds.data(),
+                           (size_t)(nb + 1) * 8, hipMemcpyHostToDevice, s));
+    HIP_TRY(hipMalloc((void**)&plan.d_bound_rows,
+                      (size_t)(nb + 1) * n_ssts * 8));
+    hx::RangeAux R{plan.d_range_bounds, plan.d_bound_rows, plan.d_sst_rgs,
+                   plan.d_sst_rg_off,   plan.d_sst_rg_cnt, n_ssts,
+                   nb,                  2048};
+    ke = hx::launch_range_bounds(s, base, R, plan.d_bound_rows);
+    if (ke != hipSuccess)
+        return fail(HX_ERR_HIP, std::string("range bounds failed: ") +
+                                    hipGetErrorString(ke));
+    HIP_TRY(hipStreamSynchronize(s));
+    plan.range_nblocks = nb;
+    plan.range_ready = true;
+    return HX_OK;
+}
+
 hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                     HostTable& out, double* agg_kernel_ms,
                     unsigned long long* matched_out) {
@@ -1329,11 +1437,36 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         hx_status est = ensure_events(plan);
         if (est != HX_OK) return est;
         hipEvent_t e0 = plan.ev[0], e1 = plan.ev[1];
+        // series-range mode: series-only key-claim grouping
+        // routed through per-block LDS tables. Opt-in via HX_RANGE=1.
+        bool use_range = !bucket && key_claim && !n_buckets &&
+                         getenv("HX_RANGE") && atoi(getenv("HX_RANGE")) != 0;
+        if (use_range && !plan.range_ready) {
+            hx_status rs = ensure_range(plan, A);
+            if (rs != HX_OK) return rs;
+        }
+        if (use_range && !plan.range_ready) use_range = false;
         HIP_TRY(hipEventRecord(e0, s));
+        if (use_range) {
+            uint32_t ne = 2048;
+            if (const char* e = getenv("HX_RANGE_NE"))
+                ne = (uint32_t)strtoul(e, nullptr, 10);
+            hx::RangeAux R{plan.d_range_bounds, plan.d_bound_rows,
+                           plan.d_sst_rgs,     plan.d_sst_rg_off,
+                           plan.d_sst_rg_cnt,  (uint32_t)plan.ssts.size(),
+                           plan.range_nblocks, ne};
+            hipError_t re2 = hx::launch_scan_agg_range(
+                s, A, R, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0);
+            if (re2 != hipSuccess)
+                return fail(HX_ERR_HIP,
+                            std::string("range kernel launch failed: ") +
+                                hipGetErrorString(re2));
+        }
         // the LDS-gang variant measured slower than the wave kernel at every
         // tested config (see profiles/README r01 notes); opt-in for further
         // experiments
-        bool use_gang = !bucket && getenv("HX_GANG_ON") != nullptr;
+        bool use_gang = !use_range && !bucket &&
+                        getenv("HX_GANG_ON") != nullptr;
         if (use_gang) {
             // gang = number of SSTs in the dominant size class: one gang
             // then covers EXACTLY one aligned series window (the transposed
@@ -1369,7 +1502,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                             std::string("gang kernel launch failed: ") +
                                 hipGetErrorString(ge2));
         }
-        if (!use_gang) HIP_TRY(hx::launch_scan_agg(s, A, 0));
+        if (!use_range && !use_gang)
+            HIP_TRY(hx::launch_scan_agg(s, A, 0));
         HIP_TRY(hipEventRecord(e1, s));
         HIP_TRY(hipStreamSynchronize(s));
         float ms = 0;

@@ -101,6 +101,23 @@ struct AggParams {
     unsigned long long* matched;   // rows surviving filter+dedup
 };
 
+// Series-range partitioned aggregation (k_scan_agg_range): the
+// series space is split at equal-sample quantile boundaries; block b owns
+// series in [bounds[b], bounds[b+1]) across ALL SSTs, accumulating into an
+// LDS table flushed once — global-table RMWs drop from one per 64-row-
+// window run to one per (block, series). Row bounds per (block, sst) are
+// precomputed by k_range_bounds (64-ary parallel lower_bound) and cached.
+struct RangeAux {
+    const uint64_t* bounds;      // n_blocks+1 ascending series boundaries
+    const uint64_t* bound_rows;  // (n_blocks+1) x n_ssts packed (pos<<32|row)
+    const int32_t* sst_rgs;      // rg-desc indices grouped per SST, row order
+    const int32_t* sst_rg_off;   // per SST: offset into sst_rgs
+    const int32_t* sst_rg_cnt;   // per SST: count
+    uint32_t n_ssts;
+    uint32_t n_blocks;
+    uint32_t ne;                 // LDS hash slots (power of two)
+};
+
 // DELTA_BINARY_PACKED decode unit: one page -> dense i64 at dst_off (dec).
 struct DeltaPageDesc {
     uint64_t src_off;     // page payload (bit63: in dec blob, e.g. post-snappy)

@@ -32,6 +32,15 @@ struct GangParams;  // kernels.hip internal; caller provides param buffers
 hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
                                 uint32_t gang_size, bool minmax,
                                 GangParams* h_params, GangParams* d_params);
+// Series-range mode: sample series for quantile boundaries,
+// precompute per-(block,sst) row bounds, then the LDS-table range kernel.
+hipError_t launch_sample_series(hipStream_t s, const RgDesc* rgs,
+                                uint32_t n_rgs, const uint8_t* blob,
+                                const uint8_t* dec, uint64_t* out);
+hipError_t launch_range_bounds(hipStream_t s, const AggParams& p,
+                               const RangeAux& r, uint64_t* out);
+hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
+                                 const RangeAux& r, bool minmax);
 hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
                             uint32_t rg_first, uint32_t rg_last,
                             uint64_t* out_series, long long* out_ts,

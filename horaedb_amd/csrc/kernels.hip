@@ -724,6 +724,249 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
 }
 
 // ---------------------------------------------------------------------------
+// Series-range partitioned aggregation. Position-aligned LDS
+// gangs lose to series-position jitter (sqrt-scale fluctuations between
+// SSTs dwarf a window), so the partition here is by series VALUE: block b
+// owns [bounds[b], bounds[b+1]) across every SST; its series fit an LDS
+// table by construction, flushed once — global RMWs drop from one per
+// window-run (~rows/1.6) to one per (block, series).
+// ---------------------------------------------------------------------------
+
+// Stride-512 series samples per rg slice (16 fixed slots, ~0 padding): the
+// host sorts these into equal-sample quantile boundaries.
+extern "C" __global__ void __launch_bounds__(256)
+k_sample_series(const RgDesc* __restrict__ rgs, uint32_t n_rgs,
+                const uint8_t* __restrict__ blob,
+                const uint8_t* __restrict__ dec, uint64_t* __restrict__ out) {
+    const int lane = threadIdx.x & 63;
+    const uint32_t wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+    for (uint32_t g = wave_id; g < n_rgs; g += n_waves) {
+        const RgDesc rg = rgs[g];
+        const uint64_t* S =
+            (const uint64_t*)hx_ptr(blob, dec, rg.series_off);
+        if (lane < 16) {
+            uint32_t r = (uint32_t)lane * 512u;
+            out[(size_t)g * 16 + lane] = (r < rg.n_rows) ? S[r] : ~0ull;
+        }
+    }
+}
+
+// Wave-parallel 64-ary lower_bound: count of indices in [0,n) whose probed
+// key is < target (keys ascending). ~log64(n) rounds of one parallel load.
+template <typename Pred>
+__device__ __forceinline__ uint32_t lb64(uint32_t n, uint64_t target,
+                                         int lane, Pred key_at) {
+    uint32_t lo = 0, rem = n;
+    while (rem > 0) {
+        const uint32_t step = (rem + 63u) >> 6;
+        const uint32_t p = lo + (uint32_t)lane * step;
+        const bool in = (uint32_t)lane * step < rem;
+        const bool t = in && (key_at(p) < target);
+        const uint32_t k = (uint32_t)__popcll(__ballot(t));
+        if (k == 0) break;
+        const uint32_t adv = (k - 1) * step + 1;
+        lo += adv;
+        rem = (step - 1u) < (rem - adv) ? (step - 1u) : (rem - adv);
+    }
+    return lo;   // first index with key >= target (== n if none)
+}
+
+// Row bounds per (boundary, sst): first staged row with series >= bounds[b],
+// packed (rg-list position << 32 | row within slice). Computed once per
+// prepare (decode is deterministic, so bounds survive re-decode).
+extern "C" __global__ void __launch_bounds__(256)
+k_range_bounds(AggParams P, RangeAux R, uint64_t* __restrict__ out) {
+    const int lane = threadIdx.x & 63;
+    const uint32_t wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+    const uint32_t total = (R.n_blocks + 1) * R.n_ssts;
+    for (uint32_t w = wave_id; w < total; w += n_waves) {
+        const uint32_t b = w / R.n_ssts, si = w % R.n_ssts;
+        const uint64_t target = R.bounds[b];
+        const int32_t off = R.sst_rg_off[si];
+        const uint32_t cnt = (uint32_t)R.sst_rg_cnt[si];
+        // level 1: over slices by their first-row series
+        const uint32_t g = lb64(cnt, target, lane, [&](uint32_t p) {
+            const RgDesc rg = P.rgs[R.sst_rgs[off + (int32_t)p]];
+            return *(const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+        });
+        uint64_t packed;
+        if (g == 0) {
+            packed = 0;   // boundary at the very first row
+        } else {
+            // boundary row is inside slice g-1, or at the start of slice g
+            const RgDesc rg = P.rgs[R.sst_rgs[off + (int32_t)(g - 1)]];
+            const uint64_t* S =
+                (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+            const uint32_t r = lb64(rg.n_rows, target, lane,
+                                    [&](uint32_t p) { return S[p]; });
+            packed = (r == rg.n_rows) ? ((uint64_t)g << 32)
+                                      : (((uint64_t)(g - 1) << 32) | r);
+        }
+        if (lane == 0) out[w] = packed;
+    }
+}
+
+// 64-row window worker for the range kernel: like scan_window but per-wave
+// (lane-indexed rows), clamped to [base, hi) with successor dedup against
+// the slice's TRUE row count, and no global-table probe hint.
+__device__ __forceinline__ void scan_window_range(
+    const AggParams& P, const RgDesc& rg, const SstDev& sst,
+    const uint64_t* S, const int64_t* T, const double* V, uint32_t base,
+    uint32_t hi, uint32_t n_true, int lane, unsigned long long& my_matched,
+    WinResult& W) {
+    const uint32_t r = base + (uint32_t)lane;
+    const bool inb = r < hi;
+    const bool has_next = r + 1 < n_true;
+    uint64_t s = KEY_EMPTY;
+    int64_t t = 0;
+    double v = 0.0;
+    uint64_t s1 = 0;
+    int64_t t1 = 0;
+    bool alive = false;
+    if (inb) {
+        t = T[r];
+        s = S[r];
+        v = V[r];
+        if (has_next) {
+            s1 = S[r + 1];
+            t1 = T[r + 1];
+        }
+    }
+    if (inb) {
+        alive = (t >= P.ts_lo) & (t < P.ts_hi);
+        if (alive && P.use_sset) alive = sset_has(P, s);
+        if (alive && P.skip < 2) {
+            bool dup = false;
+            if (has_next) {
+                dup = (s1 == s) & (t1 == t);
+            } else if (rg.next_rg >= 0) {
+                const RgDesc nx = P.rgs[rg.next_rg];
+                uint64_t s2 = *(const uint64_t*)hx_ptr(P.blob, P.dec,
+                                                       nx.series_off);
+                int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
+                dup = (s2 == s) & (t2 == t);
+            }
+            if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+            alive = !dup;
+        }
+        if (!alive) v = 0.0;
+    }
+    unsigned long long c = alive ? 1ull : 0ull;
+    double vv = alive ? v : 0.0;
+    double mn = alive ? v : HUGE_VAL;
+    double mx = alive ? v : -HUGE_VAL;
+    my_matched += c;
+    const uint64_t sp = __shfl_up(s, 1, 64);
+    const bool head = (lane == 0) || sp != s;
+    bool done = false;
+    for (int d = 1; d < 64; d++) {
+        const uint64_t s2 = __shfl_down(s, d, 64);
+        const double v2 = __shfl_down(vv, d, 64);
+        const unsigned long long c2 = __shfl_down(c, d, 64);
+        const double mn2 = __shfl_down(mn, d, 64);
+        const double mx2 = __shfl_down(mx, d, 64);
+        done = done || (lane + d >= 64) || s2 != s;
+        if (head && !done) {
+            vv += v2;
+            c += c2;
+            mn = fmin(mn, mn2);
+            mx = fmax(mx, mx2);
+        }
+        if (__all(done)) break;
+    }
+    W.s = s;
+    W.b = 0;
+    W.vv = vv;
+    W.mn = mn;
+    W.mx = mx;
+    W.c = c;
+    W.head = head;
+}
+
+template <bool MM>
+__global__ void __launch_bounds__(256)
+k_scan_agg_range(AggParams P, RangeAux R) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const uint32_t ne = R.ne;
+    uint64_t* lkey = (uint64_t*)smem;
+    double* lsum = (double*)(smem + (size_t)ne * 8);
+    unsigned long long* lmin = (unsigned long long*)(smem + (size_t)ne * 16);
+    unsigned long long* lmax = (unsigned long long*)(smem + (size_t)ne * 24);
+    unsigned int* lcnt = (unsigned int*)(smem + (size_t)ne * (MM ? 32 : 16));
+    const int lane = threadIdx.x & 63;
+    const uint32_t wave = threadIdx.x >> 6;
+    const uint32_t n_waves = blockDim.x >> 6;
+    unsigned long long my_matched = 0;
+    for (uint32_t blk = blockIdx.x; blk < R.n_blocks; blk += gridDim.x) {
+        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+            lkey[i] = KEY_EMPTY;
+            lsum[i] = 0.0;
+            lcnt[i] = 0u;
+            if (MM) {
+                lmin[i] = ~0ull;
+                lmax[i] = 0ull;
+            }
+        }
+        __syncthreads();
+        for (uint32_t si = wave; si < R.n_ssts; si += n_waves) {
+            const int32_t loff = R.sst_rg_off[si];
+            const uint64_t pk0 = R.bound_rows[(size_t)blk * R.n_ssts + si];
+            const uint64_t pk1 =
+                R.bound_rows[(size_t)(blk + 1) * R.n_ssts + si];
+            uint32_t pos = (uint32_t)(pk0 >> 32);
+            uint32_t row = (uint32_t)pk0;
+            const uint32_t epos = (uint32_t)(pk1 >> 32);
+            const uint32_t erow = (uint32_t)pk1;
+            while (pos < epos || (pos == epos && row < erow)) {
+                const RgDesc rg = P.rgs[R.sst_rgs[loff + (int32_t)pos]];
+                const uint64_t* S =
+                    (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+                const int64_t* T =
+                    (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+                const double* V =
+                    (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+                const SstDev sst = P.ssts[rg.sst_id];
+                const uint32_t hi = (pos == epos) ? erow : rg.n_rows;
+                // dual 64-row windows: overlap the two streaming-load chains
+                for (uint32_t base = row; base < hi; base += 128) {
+                    WinResult A, B;
+                    B.c = 0;
+                    B.head = false;
+                    scan_window_range(P, rg, sst, S, T, V, base, hi,
+                                      rg.n_rows, lane, my_matched, A);
+                    if (base + 64 < hi)
+                        scan_window_range(P, rg, sst, S, T, V, base + 64, hi,
+                                          rg.n_rows, lane, my_matched, B);
+                    if (P.skip == 1) continue;
+                    if (A.head && A.c > 0)
+                        lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
+                                       A.s, A.vv, (uint32_t)A.c, A.mn, A.mx);
+                    if (B.head && B.c > 0)
+                        lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
+                                       B.s, B.vv, (uint32_t)B.c, B.mn, B.mx);
+                }
+                pos++;
+                row = 0;
+            }
+        }
+        __syncthreads();
+        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+            if (lkey[i] == KEY_EMPTY) continue;
+            agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
+                       MM ? ordered_f64(lmin[i]) : 0.0,
+                       MM ? ordered_f64(lmax[i]) : 0.0);
+        }
+        __syncthreads();
+    }
+    for (int off = 32; off > 0; off >>= 1)
+        my_matched += __shfl_down(my_matched, off, 64);
+    if ((threadIdx.x & 63) == 0 && my_matched)
+        atomicAdd(P.matched, my_matched);
+}
+
+// ---------------------------------------------------------------------------
 // RLE_DICTIONARY decode (parquet-format Encodings.md "RLE/bit-packed
 // hybrid"; the encoding config.rs:54-75 enables with dictionaries on).
 // One workgroup per page: lane-serial run-header walk into LDS (varints are
@@ -1481,6 +1724,49 @@ hipError_t launch_scan_agg_gang(hipStream_t s, const AggParams& p,
     else
         hipLaunchKernelGGL(k_scan_agg_gang<false>, dim3(grid), dim3(1024), lds,
                            s, d_params, p.blob, p.dec, p.rgs);
+    return hipGetLastError();
+}
+
+hipError_t launch_sample_series(hipStream_t s, const RgDesc* rgs,
+                                uint32_t n_rgs, const uint8_t* blob,
+                                const uint8_t* dec, uint64_t* out) {
+    uint32_t blocks = (n_rgs * 64 + 255) / 256;
+    if (blocks > 65535) blocks = 65535;
+    if (blocks == 0) blocks = 1;
+    hipLaunchKernelGGL(k_sample_series, dim3(blocks), dim3(256), 0, s, rgs,
+                       n_rgs, blob, dec, out);
+    return hipGetLastError();
+}
+
+hipError_t launch_range_bounds(hipStream_t s, const AggParams& p,
+                               const RangeAux& r, uint64_t* out) {
+    const uint64_t total_waves = (uint64_t)(r.n_blocks + 1) * r.n_ssts;
+    uint64_t blocks = (total_waves + 3) / 4;   // 4 waves per 256-thread block
+    if (blocks > 262144) blocks = 262144;
+    if (blocks == 0) blocks = 1;
+    hipLaunchKernelGGL(k_range_bounds, dim3((uint32_t)blocks), dim3(256), 0,
+                       s, p, r, out);
+    return hipGetLastError();
+}
+
+hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
+                                 const RangeAux& r, bool minmax) {
+    const size_t lds = (size_t)r.ne * (minmax ? 36 : 20);
+    const void* f = minmax
+                        ? reinterpret_cast<const void*>(&k_scan_agg_range<true>)
+                        : reinterpret_cast<const void*>(
+                              &k_scan_agg_range<false>);
+    // >64 KiB dynamic LDS requires the opt-in or the launch FAILS silently
+    // (measured the hard way on the gang kernel)
+    hipError_t e = hipFuncSetAttribute(
+        f, hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    if (e != hipSuccess) return e;
+    if (minmax)
+        hipLaunchKernelGGL(k_scan_agg_range<true>, dim3(r.n_blocks), dim3(256),
+                           lds, s, p, r);
+    else
+        hipLaunchKernelGGL(k_scan_agg_range<false>, dim3(r.n_blocks),
+                           dim3(256), lds, s, p, r);
     return hipGetLastError();
 }
 
