@@ -1438,9 +1438,11 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         if (est != HX_OK) return est;
         hipEvent_t e0 = plan.ev[0], e1 = plan.ev[1];
         // series-range mode: series-only key-claim grouping
-        // routed through per-block LDS tables. Opt-in via HX_RANGE=1.
-        bool use_range = !bucket && key_claim && !n_buckets &&
-                         getenv("HX_RANGE") && atoi(getenv("HX_RANGE")) != 0;
+        // routed through per-block LDS tables. Default ON (1.55x over the
+        // wave kernel same-box); HX_RANGE=0 falls back to k_scan_agg.
+        bool use_range = !bucket && key_claim && !n_buckets;
+        if (const char* renv = getenv("HX_RANGE"))
+            use_range = use_range && atoi(renv) != 0;
         if (use_range && !plan.range_ready) {
             hx_status rs = ensure_range(plan, A);
             if (rs != HX_OK) return rs;
