@@ -929,7 +929,9 @@ k_scan_agg_range(AggParams P, RangeAux R) {
                     (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
                 const SstDev sst = P.ssts[rg.sst_id];
                 const uint32_t hi = (pos == epos) ? erow : rg.n_rows;
-                // dual 64-row windows: overlap the two streaming-load chains
+                // dual 64-row windows: two independent load->reduce->LDS
+                // chains in flight per wave (quad measured equal-to-worse:
+                // the bound is not window-chain MLP)
                 for (uint32_t base = row; base < hi; base += 128) {
                     WinResult A, B;
                     B.c = 0;
