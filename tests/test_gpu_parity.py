@@ -551,3 +551,16 @@ def test_concurrent_prepared_scans(ds_plain):
         finally:
             for p in preps:
                 p.close()
+
+
+def test_wave_kernel_fallback_parity(ds_plain):
+    # HX_RANGE=0 must route through k_scan_agg (the wave kernel) with
+    # identical results — the fallback stays load-bearing for bucket/state
+    # paths and must not rot
+    out, m = ds_plain
+    os.environ["HX_RANGE"] = "0"
+    try:
+        check_parity(out, middle_range(m), ops=OPS_ALL)
+        check_parity(out, (0, 2**62), ops=AGG_SUM | AGG_COUNT)
+    finally:
+        del os.environ["HX_RANGE"]
