@@ -324,6 +324,7 @@ struct DevPlan {
     uint64_t* d_range_bounds = nullptr;   // n_blocks+1 series boundaries
     uint64_t* d_bound_rows = nullptr;     // (n_blocks+1) x n_ssts packed
     uint32_t range_nblocks = 0;
+    double range_xest = 0;   // distinct-series estimate (sizes the table)
     bool range_ready = false;
 
     hipStream_t stream = nullptr;
@@ -1265,6 +1266,7 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
         (g < (double)u ? x_lo : x_hi) = x;
     }
     const double x_est = 0.5 * (x_lo + x_hi);
+    plan.range_xest = x_est;
     double target = 700.0;   // series per block (ne=2048, load ~0.34)
     if (const char* e = getenv("HX_RANGE_TARGET")) target = atof(e);
     uint32_t nb = 512;
@@ -1356,6 +1358,22 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         ((!bucket || n_buckets) && P->key_claim_safe &&
          !getenv("HX_FORCE_STATE")) ? 1 : 0;
 
+    // series-range partition: built once per prepared, before
+    // table sizing so its distinct-series estimate can size the table
+    // (prevents a saturated first pass + retry)
+    bool use_range = !bucket && key_claim && !n_buckets;
+    if (const char* renv = getenv("HX_RANGE"))
+        use_range = use_range && atoi(renv) != 0;
+    if (use_range && !plan.range_ready) {
+        hx::AggParams base{};
+        base.rgs = plan.d_rgs;
+        base.blob = plan.d_blob;
+        base.dec = plan.d_dec;
+        hx_status rs = ensure_range(plan, base);
+        if (rs != HX_OK) return rs;
+    }
+    if (use_range && !plan.range_ready) use_range = false;
+
     // table size heuristic; grows on overflow
     uint32_t slots = plan.slots;
     if (!slots) {
@@ -1367,6 +1385,10 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                  // the direct-indexed path is keyed by series only
                  (uint64_t)plan.rows_scanned /
                      ((bucket && !n_buckets) ? 2 : 32)));
+        if (plan.range_xest > 0 && key_claim && !bucket) {
+            uint32_t rs = next_pow2_u32((uint64_t)(plan.range_xest * 1.4));
+            if (rs > slots) slots = rs;
+        }
         if (slots > (1u << 28)) slots = 1u << 28;
     }
 
@@ -1437,17 +1459,6 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         hx_status est = ensure_events(plan);
         if (est != HX_OK) return est;
         hipEvent_t e0 = plan.ev[0], e1 = plan.ev[1];
-        // series-range mode: series-only key-claim grouping
-        // routed through per-block LDS tables. Default ON (1.55x over the
-        // wave kernel same-box); HX_RANGE=0 falls back to k_scan_agg.
-        bool use_range = !bucket && key_claim && !n_buckets;
-        if (const char* renv = getenv("HX_RANGE"))
-            use_range = use_range && atoi(renv) != 0;
-        if (use_range && !plan.range_ready) {
-            hx_status rs = ensure_range(plan, A);
-            if (rs != HX_OK) return rs;
-        }
-        if (use_range && !plan.range_ready) use_range = false;
         HIP_TRY(hipEventRecord(e0, s));
         if (use_range) {
             uint32_t ne = 2048;

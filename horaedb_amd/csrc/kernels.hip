@@ -889,6 +889,7 @@ template <bool MM>
 __global__ void __launch_bounds__(256)
 k_scan_agg_range(AggParams P, RangeAux R) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
+    __shared__ int s_abort;   // global table saturated: stop, host retries
     const uint32_t ne = R.ne;
     uint64_t* lkey = (uint64_t*)smem;
     double* lsum = (double*)(smem + (size_t)ne * 8);
@@ -900,6 +901,7 @@ k_scan_agg_range(AggParams P, RangeAux R) {
     const uint32_t n_waves = blockDim.x >> 6;
     unsigned long long my_matched = 0;
     for (uint32_t blk = blockIdx.x; blk < R.n_blocks; blk += gridDim.x) {
+        if (threadIdx.x == 0) s_abort = 0;
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             lkey[i] = KEY_EMPTY;
             lsum[i] = 0.0;
@@ -911,6 +913,19 @@ k_scan_agg_range(AggParams P, RangeAux R) {
         }
         __syncthreads();
         for (uint32_t si = wave; si < R.n_ssts; si += n_waves) {
+            // saturation early-abort (the wave kernel's lesson: a doomed
+            // pass against a full table grinds 4096-probe scans); plain
+            // L1-served load, one lane per wave, once per (block, sst)
+            {
+                unsigned long long f = 0;
+                if (lane == 0)
+                    f = *(volatile const unsigned long long*)P.fill;
+                f = __shfl(f, 0, 64);
+                if (f > P.fill_limit) {
+                    s_abort = 1;
+                    break;
+                }
+            }
             const int32_t loff = R.sst_rg_off[si];
             const uint64_t pk0 = R.bound_rows[(size_t)blk * R.n_ssts + si];
             const uint64_t pk1 =
@@ -954,6 +969,11 @@ k_scan_agg_range(AggParams P, RangeAux R) {
             }
         }
         __syncthreads();
+        if (s_abort) {
+            if (threadIdx.x == 0)
+                __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+            break;   // host retries with a larger table; skip the flush
+        }
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             if (lkey[i] == KEY_EMPTY) continue;
             agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
