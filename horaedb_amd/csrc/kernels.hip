@@ -1779,10 +1779,15 @@ hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
                         : reinterpret_cast<const void*>(
                               &k_scan_agg_range<false>);
     // >64 KiB dynamic LDS requires the opt-in or the launch FAILS silently
-    // (measured the hard way on the gang kernel)
-    hipError_t e = hipFuncSetAttribute(
-        f, hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-    if (e != hipSuccess) return e;
+    // (measured the hard way on the gang kernel). Request only what this
+    // launch needs: the kernel's static shared (s_abort) counts against
+    // the 160 KiB block limit, so asking for the full 160 KiB is itself
+    // an invalid argument.
+    if (lds > 64 * 1024) {
+        hipError_t e = hipFuncSetAttribute(
+            f, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+        if (e != hipSuccess) return e;
+    }
     if (minmax)
         hipLaunchKernelGGL(k_scan_agg_range<true>, dim3(r.n_blocks), dim3(256),
                            lds, s, p, r);
