@@ -195,3 +195,28 @@ def test_schema_surface(store_dir):
         [True, True, False, False, False]
     assert [c["builtin"] for c in sch["columns"]] == \
         [False, False, False, True, True]
+
+
+def test_native_writer_odd_sizes(tmp_path):
+    # row counts straddling row-group boundaries; single row; row_group 1
+    import pyarrow.parquet as pq
+    from horaedb_amd.store import write_sst_native
+    rng = np.random.default_rng(7)
+    for i, (n, rg) in enumerate([(1, 8192), (8191, 8192), (8192, 8192),
+                                 (8193, 8192), (5, 1), (100, 7)]):
+        series = np.sort(rng.integers(0, 50, n).astype(np.uint64))
+        ts = np.arange(n, dtype=np.int64)
+        val = rng.random(n)
+        ddir = tmp_path / f"w{i}" / "data"
+        ddir.mkdir(parents=True)
+        p = str(ddir / "3.sst")
+        write_sst_native(p, series, ts, val, seq=3, row_group=rg)
+        t = pq.read_table(p)
+        assert t.num_rows == n
+        np.testing.assert_array_equal(t.column("series_id").to_numpy(),
+                                      series)
+        np.testing.assert_array_equal(t.column("value").to_numpy(), val)
+        pf = pq.ParquetFile(p)
+        assert pf.metadata.num_row_groups == (n + rg - 1) // rg
+        with Store(str(tmp_path / f"w{i}")) as st:
+            assert st.catalog()[0]["n_rows"] == n
