@@ -308,13 +308,18 @@ def main():
             agg_kernel_ms.append(prep.stats()["agg_kernel_ms"])
     else:
         # concurrent scans: ctypes releases the GIL inside hx_exec_agg,
-        # so N threads over N prepared objects genuinely overlap on GPU
+        # so N threads over N prepared objects genuinely overlap on GPU.
+        # Per-prep locks: the executor does not guarantee task i+N waits
+        # for task i, and one hx_prepared serves ONE call at a time.
+        import threading
         from concurrent.futures import ThreadPoolExecutor
+        locks = [threading.Lock() for _ in range(n_pipe)]
 
         def timed_step(i):
             p_ = preps[i % n_pipe]
-            step(prep=p_)
-            return p_.stats()["agg_kernel_ms"]
+            with locks[i % n_pipe]:
+                step(prep=p_)
+                return p_.stats()["agg_kernel_ms"]
 
         with ThreadPoolExecutor(max_workers=n_pipe) as ex:
             for k in ex.map(timed_step, range(args.steps)):

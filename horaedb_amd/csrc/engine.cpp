@@ -1060,7 +1060,10 @@ struct PinnedPool {
         if (!pinned) { free(p); return; }
         if (enabled()) {
             std::lock_guard<std::mutex> g(mu);
-            if (bufs.size() < 4) {
+            size_t held = 0;
+            for (auto& b : bufs) held += b.second;
+            // bound the pinned hoard: big (bucket-mode) results free normally
+            if (bufs.size() < 4 && held + cap <= (4ull << 30)) {
                 bufs.emplace_back(p, cap);
                 return;
             }
@@ -1277,6 +1280,8 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
     bounds[0] = 0;
     for (uint32_t b = 1; b < nb; b++) bounds[b] = samp[(size_t)b * m / nb];
     bounds[nb] = ~0ull;
+    if (plan.d_range_bounds) { hipFree(plan.d_range_bounds); plan.d_range_bounds = nullptr; }
+    if (plan.d_bound_rows) { hipFree(plan.d_bound_rows); plan.d_bound_rows = nullptr; }
     HIP_TRY(hipMalloc((void**)&plan.d_range_bounds, (size_t)(nb + 1) * 8));
     HIP_TRY(hipMemcpyAsync(plan.d_range_bounds, bounds.data(),
                            (size_t)(nb + 1) * 8, hipMemcpyHostToDevice, s));
