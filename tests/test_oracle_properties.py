@@ -64,12 +64,14 @@ def build_ssts(raw):
 
 @settings(max_examples=120, deadline=None)
 @given(raw=sst_strategy, lo=st.integers(-15, 15), width=st.integers(0, 25),
-       bucket=st.sampled_from([0, 3, 7]))
-def test_scan_agg_matches_brute_force(raw, lo, width, bucket):
+       bucket=st.sampled_from([0, 3, 7]),
+       sel=st.one_of(st.none(), st.sets(st.integers(0, 6), max_size=4)))
+def test_scan_agg_matches_brute_force(raw, lo, width, bucket, sel):
     ssts, sst_rows = build_ssts(raw)
     tr = (lo, lo + width)
-    res = oracle.scan_agg(ssts, tr, bucket_ms=bucket, ops=OPS_ALL)
-    exp = brute_force(sst_rows, tr, bucket_ms=bucket)
+    res = oracle.scan_agg(ssts, tr, bucket_ms=bucket, ops=OPS_ALL,
+                          series_set=sel)
+    exp = brute_force(sst_rows, tr, bucket_ms=bucket, series_set=sel)
     keys = list(zip(res["series_id"].tolist(),
                     res["bucket"].tolist())) if bucket else \
         [(s,) for s in res["series_id"].tolist()]
