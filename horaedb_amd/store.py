@@ -254,9 +254,11 @@ class Store:
         return [(out[i].path.decode(), out[i].sequence) for i in range(n.value)]
 
     def _spec(self, ts_range, series_in=None):
+        # keepalive is CALL-LOCAL (returned, held by the caller until the C
+        # call returns): a Store serves prepares from many threads
         spec = _ScanSpec()
         spec.range = _TimeRange(*ts_range)
-        self._keepalive = []
+        keep = []
         if series_in is not None:
             arr = np.ascontiguousarray(np.asarray(sorted(set(int(s) for s in series_in)),
                                                   dtype=np.uint64))
@@ -266,22 +268,22 @@ class Store:
             preds = (_Pred * 1)(pred)
             spec.preds = preds
             spec.n_preds = 1
-            self._keepalive += [arr, preds]
-        return spec
+            keep += [arr, preds]
+        return spec, keep
 
-    def _devset(self, devices):
+    def _devset(self, devices, keep):
         if devices is None:
             return None
         ids = (C.c_int32 * len(devices))(*devices)
         ds = _DeviceSet(device_ids=ids, n_devices=len(devices))
-        self._keepalive += [ids, ds]
+        keep += [ids, ds]
         return ds
 
     def prepare(self, ts_range, series_in=None, devices=None,
                 sst_subset=None):
         """sst_subset: list of (path, seq) to scan (hx_scan_spec.ssts — the
         per-GPU shard of BASELINE configs 4/5); None = all overlapping."""
-        spec = self._spec(ts_range, series_in)
+        spec, keep = self._spec(ts_range, series_in)
         if sst_subset is not None:
             arr = (_SstDesc * len(sst_subset))()
             enc = [p.encode() for p, _ in sst_subset]
@@ -290,11 +292,12 @@ class Store:
                 arr[i].sequence = q
             spec.ssts = arr
             spec.n_ssts = len(sst_subset)
-            self._keepalive += [arr, enc]
-        ds = self._devset(devices)
+            keep += [arr, enc]
+        ds = self._devset(devices, keep)
         out = C.c_void_p()
         _check(_lib.hx_prepare(self._h, C.byref(spec),
                                C.byref(ds) if ds else None, C.byref(out)))
+        del keep
         return Prepared(self, out)
 
     def scan_agg(self, ts_range, ops=AGG_SUM | AGG_COUNT, bucket_ms=0,
@@ -321,8 +324,7 @@ class Store:
         """ColumnarStorage::compact (storage.rs:76-89): GPU-merge the
         ts-overlap closure of the range's SSTs into one new SST. Returns the
         new file's sequence (0 = nothing to compact)."""
-        self._keepalive = []
-        ds = self._devset(devices)
+        ds = self._devset(devices, [])
         out = C.c_uint64()
         _check(_lib.hx_compact(self._h, _TimeRange(*ts_range),
                                C.byref(ds) if ds else None, C.byref(out)))
@@ -335,13 +337,13 @@ class Store:
         Returns dict of concatenated column arrays in stream order.
         batch_limit: stop the stream (callback returns nonzero — the
         header's early-stop contract) after that many batches."""
-        spec = self._spec(ts_range, series_in)
+        spec, keep = self._spec(ts_range, series_in)
         if projection is not None:
             parr = (C.c_int32 * len(projection))(*projection)
             spec.projection = parr
             spec.n_projection = len(projection)
-            self._keepalive.append(parr)
-        ds = self._devset(devices)
+            keep.append(parr)
+        ds = self._devset(devices, keep)
         chunks = []
 
         _BATCH = C.CFUNCTYPE(C.c_int32, C.c_void_p, C.POINTER(_ColBatch))
