@@ -115,3 +115,39 @@ def brute_force_winners(sst_rows, ts_range, series_set=None):
             if cur is None or (seq, i) > (cur[0], cur[1]):
                 winner[(s, t)] = (seq, i, float(v))
     return winner
+
+
+@settings(max_examples=60, deadline=None)
+@given(raw=st.lists(st.lists(st.tuples(
+           st.integers(0, 4),          # series
+           st.integers(0, 10),         # ts
+           st.integers(-99, 99),       # value
+           st.integers(1, 5)),         # per-row seq (test scaffolding)
+       min_size=0, max_size=20), min_size=1, max_size=3))
+def test_merge_scan_per_row_seq(raw):
+    # oracle's per-row __seq__ support (the reference's own test scaffolding
+    # uses per-row seq, read.rs:512-573): winner = max (seq, stream order)
+    from oracle.scan import merge_scan, MERGE_LAST
+    ssts = []
+    model_rows = []
+    for batch in raw:
+        # stream batches must be (pk..., seq)-sorted (merge input contract)
+        arr = sorted(batch, key=lambda r: (r[0], r[1], r[3]))
+        s = np.array([r[0] for r in arr], np.uint64)
+        t = np.array([r[1] for r in arr], np.int64)
+        v = np.array([float(r[2]) for r in arr], np.float64)
+        q = np.array([r[3] for r in arr], np.uint64)
+        ssts.append(SstBatch([s, t, v], q))
+        model_rows.append(arr)
+    out = merge_scan(ssts, num_primary_keys=2)
+    winner = {}
+    for bi, rows in enumerate(model_rows):
+        for i, (s, t, v, q) in enumerate(rows):
+            cur = winner.get((s, t))
+            # lexsort ties (same pk, same seq) resolve by stream order:
+            # batch index then row index
+            if cur is None or (q, bi, i) >= (cur[0], cur[1], cur[2]):
+                winner[(s, t)] = (q, bi, i, float(v))
+    want = sorted(((s, t, v) for (s, t), (_, _, _, v) in winner.items()))
+    got = list(zip(out[0].tolist(), out[1].tolist(), out[2].tolist()))
+    assert got == want
