@@ -858,6 +858,18 @@ __device__ __forceinline__ void scan_window_range(
     double mn = alive ? v : HUGE_VAL;
     double mx = alive ? v : -HUGE_VAL;
     my_matched += c;
+    if (P.skip == 4) {
+        // bisect mode: no wave pre-reduction — every alive lane updates the
+        // LDS table itself (short runs make the shfl loop VALU-heavy)
+        W.s = s;
+        W.b = 0;
+        W.vv = vv;
+        W.mn = mn;
+        W.mx = mx;
+        W.c = c;
+        W.head = alive;
+        return;
+    }
     const uint64_t sp = __shfl_up(s, 1, 64);
     const bool head = (lane == 0) || sp != s;
     bool done = false;
