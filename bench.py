@@ -412,7 +412,11 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": "config2_1b_rows_ts_range_sum_count",
+                "workload": ("config5_bucketed_sharded" if args.config5 else
+                             "config3_series_set_predicate"
+                             if args.selectivity > 0 else
+                             "config2_bucketed" if args.bucket_ms else
+                             "config2_1b_rows_ts_range_sum_count"),
                 "rows_per_gpu": args.rows,
                 "rows_scanned_per_gpu": int(prep.stats()["rows_scanned"]),
                 "series_per_gpu": args.series,
