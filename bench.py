@@ -362,7 +362,8 @@ def main():
     ms_per_step = elapsed * 1000.0 / args.steps
 
     if rank == 0:
-        # roofline for the dominant kernel (k_scan_agg): ALGORITHMIC bytes =
+        # roofline for the dominant kernel (k_scan_agg_range by default;
+        # k_scan_agg under HX_RANGE=0): ALGORITHMIC bytes =
         # 24 B/row (series u64 + ts i64 + value f64; DESIGN.md §8, SURVEY
         # §8(d)) per launch / HIP-event launch time, measured on the engine's
         # own stream. traffic = PMC-measured HBM bytes per launch, injected
