@@ -55,6 +55,7 @@ namespace {
 
 struct ChunkRef {
     int32_t codec = 0;
+    bool required = true;
     int64_t chunk_start = 0;
     int64_t comp_size = 0;
     int64_t num_values = 0;
@@ -148,6 +149,7 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
                 return fail(HX_ERR_FORMAT, path + ": column chunk missing");
             const auto& cc = rg.columns[ci[k]];
             cr.cols[k].codec = cc.codec;
+            cr.cols[k].required = m.columns[ci[k]].required;
             cr.cols[k].chunk_start = cc.chunk_start();
             cr.cols[k].comp_size = cc.total_compressed_size;
             cr.cols[k].num_values = cc.num_values;
@@ -382,6 +384,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
         int col;            // 0 series 1 ts 2 value
         int64_t chunk_start, comp_size, num_values;
         int32_t codec;
+        bool required;
         size_t dst_off;     // blob offset reserved (chunk_size upper bound)
         // filled after page walk:
         uint64_t final_off = 0;  // RgDesc offset value (blob or OFF_DEC)
@@ -430,6 +433,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 j.comp_size = cr.comp_size;
                 j.num_values = cr.num_values;
                 j.codec = cr.codec;
+                j.required = cr.required;
                 j.dst_off = blob_off;
                 blob_off = align64(blob_off + size_t(cr.comp_size));
                 jobs.push_back(j);
@@ -496,7 +500,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
             std::vector<hx::PageDesc> pages;
             try {
                 pages = hx::walk_pages(tmp.data(), tmp.size(), j.chunk_start,
-                                       j.num_values);
+                                       j.num_values, !j.required, j.codec);
             } catch (const std::exception& e) {
                 std::lock_guard<std::mutex> g(mu);
                 err_msg = j.ss->cat->path + ": " + e.what();
@@ -532,7 +536,20 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 err_flag = 1;
                 break;
             }
+            // size sanity BEFORE any copy: a corrupt header claiming
+            // def_level_bytes > compressed_size would underflow `payload`
+            // (size_t) and read/write out of bounds (walk_pages validates
+            // too; this guards the invariants this copy depends on)
             size_t in_chunk = size_t(dp->payload_off - j.chunk_start);
+            if (dp->def_level_bytes < 0 ||
+                dp->def_level_bytes > dp->compressed_size ||
+                dp->payload_off < j.chunk_start ||
+                in_chunk + size_t(dp->compressed_size) > size_t(j.comp_size)) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": page size fields out of bounds";
+                err_flag = 1;
+                break;
+            }
             size_t payload = size_t(dp->compressed_size) - size_t(dp->def_level_bytes);
             const uint8_t* src = tmp.data() + in_chunk + dp->def_level_bytes;
             std::memcpy(plan.h_blob + j.dst_off, src, payload);
@@ -796,7 +813,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     HIP_TRY(upload(plan.d_sst_rgs, plan.h_sst_rgs));
     HIP_TRY(upload(plan.d_sst_rg_off, plan.h_sst_rg_off));
     HIP_TRY(upload(plan.d_sst_rg_cnt, plan.h_sst_rg_cnt));
-    HIP_TRY(hipMalloc((void**)&plan.d_counters, 4 * sizeof(unsigned long long)));
+    HIP_TRY(hipMalloc((void**)&plan.d_counters, 6 * sizeof(unsigned long long)));
     HIP_TRY(hipStreamSynchronize(plan.stream));
     // staging buffer no longer needed once resident in HBM (frees up to
     // 12 GB of pinned host memory per rank for the 8-GPU runs)
@@ -827,7 +844,12 @@ extern "C" hx_status hx_prepare(hx_handle* h, const hx_scan_spec* spec,
 
     auto P = std::make_unique<hx_prepared>();
     P->h = h;
-    P->spec = *spec;
+    // retain ONLY the fields the exec paths read (the time range): the
+    // caller owns preds/ssts/projection and may free them right after this
+    // call — a shallow struct copy would dangle (sset keys are deep-copied
+    // into sset_keys below; the SST subset is resolved into `chosen` here)
+    P->spec = hx_scan_spec{};
+    P->spec.range = spec->range;
     for (size_t i = 0; i < spec->n_preds; i++) {
         const hx_pred& p = spec->preds[i];
         if (p.kind != HX_PRED_SERIES_IN)
@@ -1289,7 +1311,7 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
                       (size_t)(nb + 1) * n_ssts * 8));
     hx::RangeAux R{plan.d_range_bounds, plan.d_bound_rows, plan.d_sst_rgs,
                    plan.d_sst_rg_off,   plan.d_sst_rg_cnt, n_ssts,
-                   nb,                  2048};
+                   nb,                  2048,              0};
     ke = hx::launch_range_bounds(s, base, R, plan.d_bound_rows);
     if (ke != hipSuccess)
         return fail(HX_ERR_HIP, std::string("range bounds failed: ") +
@@ -1359,7 +1381,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
             }
         }
     }
-    const int32_t key_claim =
+    int32_t key_claim =
         ((!bucket || n_buckets) && P->key_claim_safe &&
          !getenv("HX_FORCE_STATE")) ? 1 : 0;
 
@@ -1397,7 +1419,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         if (slots > (1u << 28)) slots = 1u << 28;
     }
 
-    unsigned long long counters[4];
+    unsigned long long counters[6];
     for (int attempt = 0; attempt < 4; attempt++) {
         hx_status st = alloc_table(plan, slots, ops, bucket, key_claim);
         if (st != HX_OK) return st;
@@ -1428,7 +1450,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                 s, plan.t_slab, slots, plan.slab_stride,
                 (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0));
         }
-        HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 32, s));
+        HIP_TRY(hipMemsetAsync(plan.d_counters, 0, 48, s));
 
         hx::AggParams A{};
         A.rgs = plan.d_rgs;
@@ -1460,6 +1482,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         A.fill = plan.d_counters + 0;
         A.overflow = plan.d_counters + 1;
         A.matched = plan.d_counters + 2;
+        A.fallback = plan.d_counters + 3;
+        A.poll = getenv("HX_NO_POLL") ? 0 : 1;
 
         hx_status est = ensure_events(plan);
         if (est != HX_OK) return est;
@@ -1469,10 +1493,14 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
             uint32_t ne = 2048;
             if (const char* e = getenv("HX_RANGE_NE"))
                 ne = (uint32_t)strtoul(e, nullptr, 10);
+            uint32_t interp = 1;   // HX_INTERP=0 reverts to mix64 slots
+            if (const char* ie = getenv("HX_INTERP"))
+                interp = (uint32_t)atoi(ie);
             hx::RangeAux R{plan.d_range_bounds, plan.d_bound_rows,
                            plan.d_sst_rgs,     plan.d_sst_rg_off,
                            plan.d_sst_rg_cnt,  (uint32_t)plan.ssts.size(),
-                           plan.range_nblocks, ne};
+                           plan.range_nblocks, ne,
+                           interp};
             hipError_t re2 = hx::launch_scan_agg_range(
                 s, A, R, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0);
             if (re2 != hipSuccess)
@@ -1526,14 +1554,31 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         HIP_TRY(hipStreamSynchronize(s));
         float ms = 0;
         HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
-        HIP_TRY(hipMemcpy(counters, plan.d_counters, 32, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(counters, plan.d_counters, 48, hipMemcpyDeviceToHost));
+        if (getenv("HX_DEBUG"))
+            fprintf(stderr,
+                    "[hx] exec attempt=%d kernel=%s slots=%u nb=%u ne_env=%s "
+                    "xest=%.0f kernel_ms=%.2f fill=%llu overflow=%llu "
+                    "matched=%llu lds_fallback=%llu\n",
+                    attempt, use_range ? "range" : "wave", slots,
+                    plan.range_nblocks, getenv("HX_RANGE_NE") ?: "-",
+                    plan.range_xest, ms, counters[0], counters[1],
+                    counters[2], counters[3]);
         if (counters[1] == 0) {  // no overflow
             *agg_kernel_ms = ms;
             break;
         }
         if (attempt == 3)
             return fail(HX_ERR_HIP, "aggregate table overflow persisted");
-        slots = slots >= (1u << 28) ? slots : slots * 4;
+        if (n_buckets) {
+            // direct-indexed bucket overflow (e.g. footer ts stats under-
+            // reported the range): fall back to the generic hashed path
+            n_buckets = 0;
+            bstride = 0;
+            key_claim = 0;
+        } else {
+            slots = slots >= (1u << 28) ? slots : slots * 4;
+        }
         plan.slots = 0;  // force realloc
     }
     *matched_out = counters[2];
@@ -1971,18 +2016,16 @@ extern "C" hx_status hx_write(hx_handle* h, const uint64_t* series,
         return fail(HX_ERR_INVALID, "bad argument");
     *out_seq = 0;
     if (enable_check) {
-        // segment-crossing check (storage.rs:309-316)
+        // segment-crossing check (storage.rs:309-316). The reference uses
+        // Rust truncating division on start/seg vs (end-1)/seg, where end
+        // is the exclusive max; mirror that exactly (C++ `/` truncates the
+        // same way), so negative-timestamp writes agree with the reference.
         int64_t mn = ts[0], mx = ts[0];
         for (int64_t i = 1; i < n; i++) {
             mn = std::min(mn, ts[i]);
             mx = std::max(mx, ts[i]);
         }
-        auto seg = [&](int64_t t) {
-            int64_t q = t / h->segment_ms;
-            if ((t % h->segment_ms) != 0 && t < 0) q--;
-            return q;
-        };
-        if (seg(mn) != seg(mx))
+        if (mn / h->segment_ms != mx / h->segment_ms)
             return fail(HX_ERR_INVALID,
                         "write crosses a segment boundary (storage.rs:309-316)");
     }

@@ -99,6 +99,9 @@ struct AggParams {
     unsigned long long* fill;      // claimed slots
     unsigned long long* overflow;  // !=0 => rerun with a larger table
     unsigned long long* matched;   // rows surviving filter+dedup
+    unsigned long long* fallback;  // LDS-table misses routed to global RMWs
+    int32_t poll;                  // 0 = no saturation poll in range kernel
+    int32_t _pad2;
 };
 
 // Series-range partitioned aggregation (k_scan_agg_range): the
@@ -116,6 +119,12 @@ struct RangeAux {
     uint32_t n_ssts;
     uint32_t n_blocks;
     uint32_t ne;                 // LDS hash slots (power of two)
+    // 1 = interpolation slots: first probe at (s-lo)*ne/(hi-lo) within the
+    // block's series range — ascending heads probe ascending slots, so the
+    // wave's DS accesses land on consecutive banks (conflict-free) instead
+    // of random ones, and collisions stay rare (series values are hashes,
+    // ~uniform in value). 0 = mix64 hashing.
+    uint32_t interp;
 };
 
 // DELTA_BINARY_PACKED decode unit: one page -> dense i64 at dst_off (dec).

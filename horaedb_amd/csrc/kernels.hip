@@ -96,8 +96,15 @@ __device__ __forceinline__ void keycas_add(const AggParams& P, uint32_t i,
     uint8_t* slot;
     uint32_t off8, off16, off24, off32;
     if (P.n_buckets) {  // direct-indexed bucket row of this series slot
-        slot = P.bstore + ((size_t)i * P.n_buckets +
-                           (size_t)(b - P.lo_bucket)) * P.bstride;
+        // bounds check: lo_bucket/n_buckets derive from footer ts stats; an
+        // SST whose stats under-report the data range must not index out of
+        // the dense row (host sees overflow and retries the hashed path)
+        const uint64_t bi = (uint64_t)(b - P.lo_bucket);
+        if (bi >= (uint64_t)P.n_buckets) {
+            __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+            return;
+        }
+        slot = P.bstore + ((size_t)i * P.n_buckets + (size_t)bi) * P.bstride;
         off8 = 0; off16 = 8; off24 = 16; off32 = 24;
     } else if (P.table.rep) {
         uint32_t xcc;
@@ -495,10 +502,11 @@ __device__ __forceinline__ void lds_update(const AggParams& P, uint64_t* lkey,
                                            unsigned long long* lmin,
                                            unsigned long long* lmax,
                                            uint32_t ne, uint64_t sv, double v,
-                                           uint32_t cnt, double mn, double mx) {
-    uint32_t i = (uint32_t)mix64(sv) & (ne - 1);
+                                           uint32_t cnt, double mn, double mx,
+                                           uint32_t i0 = 0xFFFFFFFFu) {
+    uint32_t i = (i0 != 0xFFFFFFFFu) ? i0 : ((uint32_t)mix64(sv) & (ne - 1));
 #pragma unroll 1
-    for (int probes = 0; probes < 8; probes++) {
+    for (int probes = 0; probes < 16; probes++) {
         uint64_t kk = lkey[i];
         if (kk == KEY_EMPTY) {
             uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, sv);
@@ -515,7 +523,16 @@ __device__ __forceinline__ void lds_update(const AggParams& P, uint64_t* lkey,
         }
         i = (i + 1) & (ne - 1);
     }
-    // LDS table full: direct global update
+    // LDS table full: direct global update (count it — a non-trivial rate
+    // here turns the kernel into a global-RMW storm; wave-aggregated so the
+    // counter itself cannot serialize)
+    if (P.fallback) {
+        const unsigned long long mk = __ballot(1);
+        const int lane = (int)(threadIdx.x & 63);
+        if ((__ffsll(mk) - 1) == lane)
+            __hip_atomic_fetch_add(P.fallback,
+                                   (unsigned long long)__popcll(mk), RLX, AGT);
+    }
     agg_update(P, sv, 0, v, cnt, mn, mx);
 }
 
@@ -810,7 +827,10 @@ k_range_bounds(AggParams P, RangeAux R, uint64_t* __restrict__ out) {
 
 // 64-row window worker for the range kernel: like scan_window but per-wave
 // (lane-indexed rows), clamped to [base, hi) with successor dedup against
-// the slice's TRUE row count, and no global-table probe hint.
+// the slice's TRUE row count, and no global-table probe hint. Templated on
+// MM so the sum/count path never pays the min/max cross-lane shuffles
+// (each 64-bit __shfl is two ds_bpermutes — real LDS-array cycles).
+template <bool MM>
 __device__ __forceinline__ void scan_window_range(
     const AggParams& P, const RgDesc& rg, const SstDev& sst,
     const uint64_t* S, const int64_t* T, const double* V, uint32_t base,
@@ -877,14 +897,16 @@ __device__ __forceinline__ void scan_window_range(
         const uint64_t s2 = __shfl_down(s, d, 64);
         const double v2 = __shfl_down(vv, d, 64);
         const unsigned long long c2 = __shfl_down(c, d, 64);
-        const double mn2 = __shfl_down(mn, d, 64);
-        const double mx2 = __shfl_down(mx, d, 64);
+        const double mn2 = MM ? __shfl_down(mn, d, 64) : 0.0;
+        const double mx2 = MM ? __shfl_down(mx, d, 64) : 0.0;
         done = done || (lane + d >= 64) || s2 != s;
         if (head && !done) {
             vv += v2;
             c += c2;
-            mn = fmin(mn, mn2);
-            mx = fmax(mx, mx2);
+            if (MM) {
+                mn = fmin(mn, mn2);
+                mx = fmax(mx, mx2);
+            }
         }
         if (__all(done)) break;
     }
@@ -923,15 +945,32 @@ k_scan_agg_range(AggParams P, RangeAux R) {
                 lmax[i] = 0ull;
             }
         }
+        // interpolation slotting (R.interp): first probe at the key's
+        // fractional position within the block's series range. Heads arrive
+        // in ascending series order, so a wave's DS ops hit consecutive
+        // slots = consecutive banks (conflict-free); collisions fall back to
+        // linear probing as before.
+        double islope = 0.0;
+        uint64_t ilo = 0;
+        if (R.interp) {
+            ilo = R.bounds[blk];
+            const uint64_t ihi = R.bounds[blk + 1];
+            islope = (double)ne / ((double)(ihi - ilo) + 1.0);
+        }
         __syncthreads();
         for (uint32_t si = wave; si < R.n_ssts; si += n_waves) {
-            // saturation early-abort (the wave kernel's lesson: a doomed
-            // pass against a full table grinds 4096-probe scans); plain
-            // L1-served load, one lane per wave, once per (block, sst)
-            {
+            // saturation early-abort (a doomed pass against a full table
+            // grinds probe scans). NOT a coherence-point load: P.fill is
+            // updated by memory-side (agent-scope) atomics, and a volatile
+            // load of that one hot word from every (wave, sst) iteration
+            // serializes at ~88 loads/us chip-wide. A workgroup-scope
+            // atomic load is L1/L2-served (possibly stale — the abort is
+            // best-effort; the probe cap still bounds a doomed pass).
+            if (P.poll) {
                 unsigned long long f = 0;
                 if (lane == 0)
-                    f = *(volatile const unsigned long long*)P.fill;
+                    f = __hip_atomic_load(P.fill, RLX,
+                                          __HIP_MEMORY_SCOPE_WORKGROUP);
                 f = __shfl(f, 0, 64);
                 if (f > P.fill_limit) {
                     s_abort = 1;
@@ -963,18 +1002,28 @@ k_scan_agg_range(AggParams P, RangeAux R) {
                     WinResult A, B;
                     B.c = 0;
                     B.head = false;
-                    scan_window_range(P, rg, sst, S, T, V, base, hi,
-                                      rg.n_rows, lane, my_matched, A);
+                    scan_window_range<MM>(P, rg, sst, S, T, V, base, hi,
+                                          rg.n_rows, lane, my_matched, A);
                     if (base + 64 < hi)
-                        scan_window_range(P, rg, sst, S, T, V, base + 64, hi,
-                                          rg.n_rows, lane, my_matched, B);
+                        scan_window_range<MM>(P, rg, sst, S, T, V, base + 64,
+                                              hi, rg.n_rows, lane, my_matched,
+                                              B);
                     if (P.skip == 1) continue;
+                    uint32_t iA = 0xFFFFFFFFu, iB = 0xFFFFFFFFu;
+                    if (R.interp) {
+                        uint32_t a2 = (uint32_t)((double)(A.s - ilo) * islope);
+                        uint32_t b2 = (uint32_t)((double)(B.s - ilo) * islope);
+                        iA = a2 < ne ? a2 : ne - 1;
+                        iB = b2 < ne ? b2 : ne - 1;
+                    }
                     if (A.head && A.c > 0)
                         lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
-                                       A.s, A.vv, (uint32_t)A.c, A.mn, A.mx);
+                                       A.s, A.vv, (uint32_t)A.c, A.mn, A.mx,
+                                       iA);
                     if (B.head && B.c > 0)
                         lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
-                                       B.s, B.vv, (uint32_t)B.c, B.mn, B.mx);
+                                       B.s, B.vv, (uint32_t)B.c, B.mn, B.mx,
+                                       iB);
                 }
                 pos++;
                 row = 0;

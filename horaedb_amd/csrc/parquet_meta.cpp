@@ -153,7 +153,8 @@ FileMetadata parse_footer(const uint8_t* tail, size_t tail_len, int64_t file_siz
 }
 
 std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
-                                 int64_t base_off, int64_t num_values) {
+                                 int64_t base_off, int64_t num_values,
+                                 bool optional_col, int32_t codec) {
     std::vector<PageDesc> pages;
     size_t pos = 0;
     int64_t seen = 0;
@@ -170,12 +171,22 @@ std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
                 case 1: pd.page_type = int32_t(r.zigzag()); break;
                 case 2: uncomp = int32_t(r.zigzag()); break;
                 case 3: comp = int32_t(r.zigzag()); break;
-                case 5: {  // DataPageHeader
+                case 5: {  // DataPageHeader (v1)
                     int16_t l2 = 0, id2; uint8_t t2;
                     while (r.field(l2, id2, t2)) {
                         switch (id2) {
                             case 1: pd.num_values = int32_t(r.zigzag()); break;
                             case 2: pd.encoding = int32_t(r.zigzag()); break;
+                            case 5: {  // Statistics
+                                int16_t l3 = 0, id3; uint8_t t3;
+                                while (r.field(l3, id3, t3)) {
+                                    if (id3 == 3)
+                                        pd.null_count = r.zigzag();
+                                    else
+                                        r.skip(t3);
+                                }
+                                break;
+                            }
                             default: r.skip(t2);
                         }
                     }
@@ -212,9 +223,42 @@ std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
         if (comp < 0 || uncomp < 0)
             throw std::runtime_error("negative page size");
         size_t hdr_len = r.offset(buf + pos);
+        // every page's payload (including the last) must fit the chunk
+        if (pos + hdr_len > len || size_t(comp) > len - pos - hdr_len)
+            throw std::runtime_error("page payload exceeds chunk bounds");
+        // v2 def/rep level bytes precede the payload and must fit it
+        if (pd.def_level_bytes < 0 || pd.def_level_bytes > comp ||
+            pd.def_level_bytes > uncomp)
+            throw std::runtime_error("level bytes exceed page size");
         pd.payload_off = base_off + int64_t(pos + hdr_len);
         pd.compressed_size = comp;
         pd.uncompressed_size = uncomp;
+        if (pd.page_type == 0 && optional_col) {
+            // v1 data page of an OPTIONAL column: definition levels are a
+            // 4-byte-length-prefixed RLE block at the START of the payload
+            // (inside the compressed stream when the page is compressed —
+            // that case is not staged here). The metric schema's rows have
+            // no nulls in practice (arrow-rs writes def levels for nullable
+            // columns anyway); reject real nulls loudly.
+            if (pd.null_count > 0)
+                throw std::runtime_error(
+                    "v1 data page has nulls (null_count > 0): nullable "
+                    "values are outside the metric scan contract");
+            if (codec != CODEC_UNCOMPRESSED)
+                throw std::runtime_error(
+                    "v1 OPTIONAL column with compressed pages: def levels "
+                    "are inside the compressed stream (unsupported; write "
+                    "REQUIRED columns or uncompressed pages)");
+            if (size_t(comp) < 4)
+                throw std::runtime_error("v1 def-level prefix truncated");
+            const uint8_t* pl = buf + pos + hdr_len;
+            uint32_t rle_len = (uint32_t)pl[0] | ((uint32_t)pl[1] << 8) |
+                               ((uint32_t)pl[2] << 16) |
+                               ((uint32_t)pl[3] << 24);
+            if (4 + (int64_t)rle_len >= comp)
+                throw std::runtime_error("v1 def-level block exceeds page");
+            pd.def_level_bytes = int32_t(4 + rle_len);
+        }
         pos += hdr_len + size_t(comp);
         if (pd.page_type == 0 || pd.page_type == 3) seen += pd.num_values;
         pages.push_back(pd);

@@ -69,14 +69,20 @@ struct PageDesc {
     int64_t payload_off;      // absolute file offset of page payload
     int32_t compressed_size;  // payload bytes in file
     int32_t uncompressed_size;
-    int32_t def_level_bytes;  // data page v2 only (levels precede payload)
+    int32_t def_level_bytes;  // levels preceding payload (v2; v1 OPTIONAL
+                              // uncompressed: 4-byte-prefixed RLE block)
     int32_t is_compressed = 1; // data page v2 is_compressed flag (v1: always)
+    int64_t null_count = -1;  // v1 statistics null_count (-1 = absent)
 };
 
 // Walk page headers of a column chunk (buf = the chunk's bytes; base_off =
 // its absolute file offset). Stops after `num_values` data values seen.
+// optional_col + codec let v1 pages of OPTIONAL columns account for their
+// embedded definition-level block (nulls themselves are rejected).
 std::vector<PageDesc> walk_pages(const uint8_t* buf, size_t len,
-                                 int64_t base_off, int64_t num_values);
+                                 int64_t base_off, int64_t num_values,
+                                 bool optional_col = false,
+                                 int32_t codec = 0);
 
 // little-endian i64 from an 8-byte statistics value
 int64_t stat_i64(const std::string& s);
