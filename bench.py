@@ -65,6 +65,49 @@ def _cpu_one_sst(args):
     return sst.n_rows
 
 
+def native_baseline_leg(m, ts_range, cores):
+    """Native C++ restatement (oracle/native/libhx_cpuref.so, g++ -O3, all
+    host cores): decode(PLAIN/Snappy) + ts-filter + dedup + sum/count
+    aggregate — the fair compiled denominator (BASELINE.md plan 2b). Runs a
+    bounded SST sample and scales to rows/s."""
+    import ctypes
+    lib_path = os.path.join(REPO, "oracle", "libhx_cpuref.so")
+    if not os.path.exists(lib_path):
+        return None
+    lo, hi = ts_range
+    paths = [s["path"] for s in m["ssts"]
+             if s["ts_min"] < hi and s["ts_max"] >= lo]
+    sample = paths[: max(4, min(len(paths), 16))]
+    try:
+        lib = ctypes.CDLL(lib_path)
+        fn = lib.hx_cpu_scan_agg
+        fn.restype = ctypes.c_int
+        fn.argtypes = [ctypes.POINTER(ctypes.c_char_p), ctypes.c_int,
+                       ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
+                       ctypes.POINTER(ctypes.c_double),
+                       ctypes.POINTER(ctypes.c_int64),
+                       ctypes.POINTER(ctypes.c_int64),
+                       ctypes.POINTER(ctypes.c_int64),
+                       ctypes.POINTER(ctypes.c_double)]
+        arr = (ctypes.c_char_p * len(sample))(*[x.encode() for x in sample])
+        el = ctypes.c_double()
+        rs = ctypes.c_int64()
+        rm = ctypes.c_int64()
+        ng = ctypes.c_int64()
+        dg = ctypes.c_double()
+        rc = fn(arr, len(sample), lo, hi, cores, ctypes.byref(el),
+                ctypes.byref(rs), ctypes.byref(rm), ctypes.byref(ng),
+                ctypes.byref(dg))
+    except OSError as e:
+        log(f"native cpu baseline unavailable: {e}")
+        return None
+    if rc != 0 or el.value <= 0 or rs.value <= 0:
+        log(f"native cpu baseline rc={rc} (unsupported layout?) — skipped")
+        return None
+    return {"rate": rs.value / el.value, "rows": rs.value, "dt": el.value,
+            "ssts": len(sample)}
+
+
 def cpu_baseline_leg(store_dir, m, ts_range, budget_s=20.0, cores=1):
     """Oracle (numpy/pyarrow CPU restatement) timed on a bounded sample of
     the same workload — kind 'port' (the reference Rust path cannot be built
@@ -129,6 +172,91 @@ def merge_bucket_partials(series_t, bucket_t, sum_t, cnt_t):
     return out_s, out_b, out_v, out_c
 
 
+def run_extras(args, device, opmap):
+    """Labeled non-headline workload lines (driver-visible via stderr):
+    the uncompressed config-2 variant, config 3 (series-set predicate +
+    min/max/avg), and the ts-overlap dedup workload.
+    Never touches the headline timed region; each line carries its own
+    config.workload label."""
+    import argparse as _ap
+    import torch
+    from horaedb_amd import Store
+    from tools.gen_ssts import gen_dataset, middle_range
+
+    def one(label, store_dir, m, ops, series_in=None, steps=3,
+            expect_matched=None):
+        lo, hi = middle_range(m, args.range_frac)
+        with Store(store_dir) as st:
+            pr = st.prepare((lo, hi), series_in=series_in, devices=[device])
+            pr.exec_agg(ops=ops)
+            torch.cuda.synchronize(device)
+            t0 = time.time()
+            for _ in range(steps):
+                pr.exec_agg(ops=ops)
+            torch.cuda.synchronize(device)
+            dt = time.time() - t0
+            stt = pr.stats()
+            line = {
+                "workload": label,
+                "value": stt["rows_scanned"] * steps / dt,
+                "unit": "rows/s",
+                "ms_per_step": dt * 1000 / steps,
+                "agg_kernel_ms": stt["agg_kernel_ms"],
+                "rows_scanned": stt["rows_scanned"],
+                "rows_matched": stt["rows_matched"],
+                "steps": steps,
+                "n_gpus": 1,
+            }
+            if expect_matched is not None:
+                line["matched_ok"] = (stt["rows_matched"] == expect_matched)
+            pr.close()
+        print("[bench-extra] " + json.dumps(line), file=sys.stderr,
+              flush=True)
+
+    # 1. uncompressed config-2 variant (the easier format; labeled)
+    try:
+        a2 = _ap.Namespace(**vars(args))
+        a2.compression = "none"
+        d2, m2 = get_dataset(a2, 0)
+        one(f"config2_{args.rows//10**9}b_uncompressed_extra", d2, m2,
+            opmap["sum"] | opmap["count"])
+    except Exception as e:  # noqa: BLE001 — extras must never kill the run
+        log(f"extra uncompressed failed: {e}")
+    # 2. config 3: series-set predicate @1% + min/max/avg (BASELINE configs[2])
+    try:
+        d3, m3 = get_dataset(args, 0)
+        ids = np.load(os.path.join(d3, "series_ids.npy"))
+        k = max(1, len(ids) // 100)
+        rng = np.random.default_rng(args.seed)
+        sel = rng.choice(ids, size=k, replace=False).tolist()
+        one("config3_series_set_1pct_minmaxavg_extra", d3, m3,
+            opmap["min"] | opmap["max"] | opmap["avg"], series_in=sel)
+    except Exception as e:  # noqa: BLE001
+        log(f"extra config3 failed: {e}")
+    # 3. ts-overlap dedup at scale: 2 generations re-write the same PKs
+    #    (cross-SST shadowed() on every older-generation row)
+    try:
+        series_o = max(1000, args.series // 10)
+        rows_o = max(series_o, args.rows // 10)
+        rows_o -= rows_o % series_o
+        if rows_o >= series_o:
+            do = os.path.join(args.data_dir,
+                              f"ovl_r{rows_o}_s{series_o}_g2")
+            mp_ = os.path.join(do, "dataset.json")
+            if os.path.exists(mp_):
+                with open(mp_) as f:
+                    mo = json.load(f)
+            else:
+                mo = gen_dataset(do, rows_o, series_o, 16, seed=args.seed,
+                                 compression=args.compression,
+                                 workers=min(16, os.cpu_count() or 8),
+                                 overlap_gens=2)
+            one(f"overlap_dedup_2gen_{2*rows_o//10**6}m_extra", do, mo,
+                opmap["sum"] | opmap["count"])
+    except Exception as e:  # noqa: BLE001
+        log(f"extra overlap failed: {e}")
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -138,7 +266,9 @@ def main():
     p.add_argument("--series", type=int, default=10_000_000)
     p.add_argument("--ssts", type=int, default=64)
     p.add_argument("--seed", type=int, default=42)
-    p.add_argument("--compression", default="none")
+    p.add_argument("--compression", default="snappy",
+                   help="page codec; default snappy = the reference's own "
+                        "default on-disk format (config.rs:120-133)")
     p.add_argument("--ts-encoding", default="PLAIN")
     p.add_argument("--range-frac", type=float, default=0.5)
     p.add_argument("--bucket-ms", type=int, default=0)
@@ -157,8 +287,12 @@ def main():
                         "the full query; >1 reports amortized ms_per_step.")
     p.add_argument("--data-dir", default="/tmp/hx_bench_data")
     p.add_argument("--no-cpu-baseline", action="store_true")
-    p.add_argument("--cpu-cores", type=int, default=1,
-                   help="processes for the cpu_baseline leg")
+    p.add_argument("--cpu-cores", type=int, default=0,
+                   help="threads/processes for the cpu_baseline leg "
+                        "(0 = all host cores)")
+    p.add_argument("--no-extras", action="store_true",
+                   help="skip the extra labeled workload lines "
+                        "(uncompressed variant, config 3, ts-overlap)")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -361,27 +495,43 @@ def main():
     value = rows_scanned * args.steps / elapsed
     ms_per_step = elapsed * 1000.0 / args.steps
 
+    workload = ("config5_bucketed_sharded" if args.config5 else
+                "config3_series_set_predicate" if args.selectivity > 0 else
+                "config2_bucketed" if args.bucket_ms else
+                f"config2_1b_rows_ts_range_sum_count_{args.compression}")
+
     if rank == 0:
         # roofline for the dominant kernel (k_scan_agg_range by default;
         # k_scan_agg under HX_RANGE=0): ALGORITHMIC bytes =
         # 24 B/row (series u64 + ts i64 + value f64; DESIGN.md §8, SURVEY
-        # §8(d)) per launch / HIP-event launch time, measured on the engine's
-        # own stream. traffic = PMC-measured HBM bytes per launch, injected
-        # via HX_ROOFLINE_TRAFFIC after a rocprofv3 --pmc run (else null).
+        # §8(d)) per launch / HIP-event launch time. With --pipeline > 1 the
+        # per-step HIP-event interval is stretched by the concurrent
+        # queries, so the kernel time comes from a SOLO (pipeline-1) pass
+        # run right after the timed region (reproducible from profiles/).
+        # traffic = PMC-measured HBM bytes per launch for THIS workload
+        # label (rocprofv3 --pmc, profiles/roofline_traffic.json), or via
+        # HX_ROOFLINE_TRAFFIC; null when unmeasured.
         my_rows = prep.stats()["rows_scanned"]
-        k_ms = float(np.mean(agg_kernel_ms))
+        if n_pipe > 1 and not (args.config5 and dist):
+            solo = []
+            for _ in range(3):
+                step(prep=preps[0])
+                solo.append(preps[0].stats()["agg_kernel_ms"])
+            k_ms = float(np.mean(solo))
+            k_basis = "solo pipeline-1 pass (HIP events, engine stream)"
+        else:
+            k_ms = float(np.mean(agg_kernel_ms))
+            k_basis = "timed-region kernel HIP events (pipeline 1)"
         algo_bytes = my_rows * 24.0
         achieved = algo_bytes / (k_ms * 1e-3)
         peak = 8.0e12
         traffic_env = os.environ.get("HX_ROOFLINE_TRAFFIC")
         if not traffic_env:
-            # PMC-measured HBM traffic per scanned row (rocprofv3 --pmc, see
-            # profiles/roofline_traffic.json + profiles/README.md)
             tj = os.path.join(REPO, "profiles", "roofline_traffic.json")
             if os.path.exists(tj):
                 with open(tj) as f:
-                    tdata = json.load(f).get("config2_1b_rows_ts_range_sum_count")
-                if tdata:
+                    tdata = json.load(f).get(workload)
+                if tdata and tdata.get("traffic_bytes_per_scanned_row"):
                     traffic_env = str(tdata["traffic_bytes_per_scanned_row"] *
                                       my_rows)
         roofline = {
@@ -391,13 +541,39 @@ def main():
             "unit": "GB/s",
             "frac": achieved / peak,
             "traffic": float(traffic_env) if traffic_env else None,
+            "kernel_ms": k_ms,
+            "basis": k_basis,
         }
 
         cpu = None
         if not args.no_cpu_baseline and n_gpus == 1:
-            log("running cpu_baseline (oracle, bounded sample)...")
-            cpu = cpu_baseline_leg(store_dir, m, ts_range,
-                                   cores=args.cpu_cores)
+            ncores = args.cpu_cores if args.cpu_cores > 0 \
+                else (os.cpu_count() or 1)
+            log(f"running cpu_baseline (native C++ leg, {ncores} threads)...")
+            native = native_baseline_leg(m, ts_range, ncores)
+            log("running cpu_baseline (oracle leg, bounded sample)...")
+            oracle_leg = cpu_baseline_leg(store_dir, m, ts_range,
+                                          budget_s=8.0, cores=1)
+            if native:
+                osub = (f"; oracle numpy/pyarrow 1-core leg: "
+                        f"{oracle_leg['value']:.3g} rows/s"
+                        if oracle_leg else "")
+                cpu = {
+                    "value": native["rate"],
+                    "unit": "rows/s",
+                    "cores": ncores,
+                    "kind": "port",
+                    "sample": (f"native C++ leg (oracle/native, g++ -O3, "
+                               f"{ncores} threads): {native['ssts']} SSTs, "
+                               f"{native['rows']} rows, {native['dt']:.1f}s"
+                               + osub),
+                }
+            else:
+                cpu = oracle_leg
+
+        if not args.no_extras and world <= 1 and not args.config5 \
+                and args.selectivity == 0 and not args.bucket_ms:
+            run_extras(args, device, opmap)
 
         result = {
             "metric": "scanned rows/sec, 1B-row range+sum (config 2)",
@@ -413,11 +589,7 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": ("config5_bucketed_sharded" if args.config5 else
-                             "config3_series_set_predicate"
-                             if args.selectivity > 0 else
-                             "config2_bucketed" if args.bucket_ms else
-                             "config2_1b_rows_ts_range_sum_count"),
+                "workload": workload,
                 "rows_per_gpu": args.rows,
                 "rows_scanned_per_gpu": int(prep.stats()["rows_scanned"]),
                 "series_per_gpu": args.series,

@@ -1298,6 +1298,11 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
     while (nb < (uint32_t)std::min(1e9, x_est * 1.3 / target) &&
            nb < (1u << 17))
         nb <<= 1;
+    // safety floor against a residual under-estimate: cap rows per block
+    // at 16384 (64-SST shapes: ~256 rows/sst/block). Empty or tiny blocks
+    // are cheap; an LDS-table overflow (the fallback path) is a global-RMW
+    // storm (r01: 286M of 320M updates went global at the 1B shape).
+    while ((double)nb * 16384.0 < staged && nb < (1u << 17)) nb <<= 1;
     std::vector<uint64_t> bounds(nb + 1);
     bounds[0] = 0;
     for (uint32_t b = 1; b < nb; b++) bounds[b] = samp[(size_t)b * m / nb];

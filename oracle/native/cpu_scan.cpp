@@ -1,0 +1,343 @@
+// oracle/native/cpu_scan.cpp — TEST INFRASTRUCTURE ONLY (cpu_baseline leg).
+//
+// Native C++ CPU restatement of the hot path (decode + ts-range filter +
+// within-SST dedup + sum/count group-by-series aggregate) so bench.py's
+// `cpu_baseline` compares the GPU path against a fair, all-cores, compiled
+// denominator (BASELINE.md plan 2b) instead of the
+// single-core numpy/pyarrow oracle. Semantics restated from the same
+// reference lines as oracle/scan.py:
+//   decode: Parquet PLAIN (+ Snappy raw-block codec) — parquet-format spec,
+//           written by parquet-rs 53.2.0 (storage.rs:193-213)
+//   filter: ts in [lo, hi)  (read.rs:459-470)
+//   dedup:  LastValueOperator keeps the LAST row of an equal-(series,ts)
+//           run within one PK-sorted SST (operator.rs:37-44). Cross-SST
+//           shadowing is NOT applied here: the bench datasets have disjoint
+//           per-SST ts ranges (single-SST clusters), where the reference
+//           plan also never merges across SSTs (storage.rs:343-368).
+//   agg:    sum/count by series_id (BASELINE configs; rfc:218-231)
+//
+// This file is never linked into the product library; only bench.py's
+// untimed cpu_baseline leg loads it. Product path = libhoraedb_hx.so.
+#include "../../horaedb_amd/csrc/parquet_meta.h"
+
+#include <atomic>
+#include <chrono>
+#include <cstdint>
+#include <cstring>
+#include <fcntl.h>
+#include <memory>
+#include <string>
+#include <thread>
+#include <unistd.h>
+#include <vector>
+
+namespace {
+
+// Snappy raw-block decompression (format: snappy/format_description.txt).
+bool snappy_decompress(const uint8_t* src, size_t clen, uint8_t* dst,
+                       size_t ulen) {
+    size_t pos = 0;
+    // uncompressed-length varint
+    uint64_t n = 0;
+    int shift = 0;
+    for (;;) {
+        if (pos >= clen || shift > 28) return false;
+        uint8_t b = src[pos++];
+        n |= (uint64_t)(b & 0x7f) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+    }
+    if (n != ulen) return false;
+    size_t d = 0;
+    while (pos < clen && d < ulen) {
+        uint8_t tag = src[pos++];
+        uint32_t kind = tag & 3u;
+        if (kind == 0) {  // literal
+            size_t len = (tag >> 2) + 1;
+            if (len > 60) {
+                size_t nb = len - 60;
+                if (pos + nb > clen) return false;
+                len = 0;
+                for (size_t i = 0; i < nb; i++)
+                    len |= (size_t)src[pos + i] << (8 * i);
+                len += 1;
+                pos += nb;
+            }
+            if (pos + len > clen || d + len > ulen) return false;
+            std::memcpy(dst + d, src + pos, len);
+            pos += len;
+            d += len;
+        } else {
+            size_t len, off;
+            if (kind == 1) {
+                len = ((tag >> 2) & 0x7u) + 4;
+                if (pos >= clen) return false;
+                off = ((size_t)(tag >> 5) << 8) | src[pos];
+                pos += 1;
+            } else if (kind == 2) {
+                len = (tag >> 2) + 1;
+                if (pos + 2 > clen) return false;
+                off = (size_t)src[pos] | ((size_t)src[pos + 1] << 8);
+                pos += 2;
+            } else {
+                len = (tag >> 2) + 1;
+                if (pos + 4 > clen) return false;
+                off = (size_t)src[pos] | ((size_t)src[pos + 1] << 8) |
+                      ((size_t)src[pos + 2] << 16) |
+                      ((size_t)src[pos + 3] << 24);
+                pos += 4;
+            }
+            if (off == 0 || off > d || d + len > ulen) return false;
+            for (size_t i = 0; i < len; i++) dst[d + i] = dst[d - off + i];
+            d += len;
+        }
+    }
+    return d == ulen;
+}
+
+struct SharedTable {
+    std::unique_ptr<std::atomic<uint64_t>[]> key;
+    std::unique_ptr<std::atomic<uint64_t>[]> sumbits;  // f64 CAS-accumulated
+    std::unique_ptr<std::atomic<uint64_t>[]> cnt;
+    uint32_t mask = 0;
+    std::atomic<uint64_t> fill{0};
+
+    void init(uint32_t slots) {
+        key.reset(new std::atomic<uint64_t>[slots]);
+        sumbits.reset(new std::atomic<uint64_t>[slots]);
+        cnt.reset(new std::atomic<uint64_t>[slots]);
+        for (uint32_t i = 0; i < slots; i++) {
+            key[i].store(~0ull, std::memory_order_relaxed);
+            sumbits[i].store(0, std::memory_order_relaxed);
+            cnt[i].store(0, std::memory_order_relaxed);
+        }
+        mask = slots - 1;
+    }
+
+    static uint64_t mix64(uint64_t x) {
+        x += 0x9E3779B97F4A7C15ull;
+        x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+        x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+        return x ^ (x >> 31);
+    }
+
+    void add(uint64_t s, double vsum, uint64_t c) {
+        uint32_t i = (uint32_t)mix64(s) & mask;
+        for (;;) {
+            uint64_t k = key[i].load(std::memory_order_relaxed);
+            if (k == ~0ull) {
+                uint64_t expected = ~0ull;
+                if (key[i].compare_exchange_strong(expected, s))
+                    fill.fetch_add(1, std::memory_order_relaxed);
+                k = key[i].load(std::memory_order_relaxed);
+            }
+            if (k == s) break;
+            i = (i + 1) & mask;
+        }
+        uint64_t old = sumbits[i].load(std::memory_order_relaxed);
+        for (;;) {
+            double cur;
+            std::memcpy(&cur, &old, 8);
+            double nv = cur + vsum;
+            uint64_t nb;
+            std::memcpy(&nb, &nv, 8);
+            if (sumbits[i].compare_exchange_weak(old, nb)) break;
+        }
+        cnt[i].fetch_add(c, std::memory_order_relaxed);
+    }
+};
+
+struct ColData {
+    std::vector<uint8_t> bytes;  // decoded PLAIN payload
+};
+
+// decode one column chunk of one row group into 8-byte PLAIN values.
+// returns false on unsupported encoding/codec or malformed page.
+bool decode_chunk(int fd, const hx::ColumnChunkMeta& cc, int64_t n_rows,
+                  ColData& out) {
+    std::vector<uint8_t> raw(cc.total_compressed_size);
+    if (pread(fd, raw.data(), raw.size(), cc.chunk_start()) !=
+        (ssize_t)raw.size())
+        return false;
+    std::vector<hx::PageDesc> pages;
+    try {
+        pages = hx::walk_pages(raw.data(), raw.size(), cc.chunk_start(),
+                               cc.num_values);
+    } catch (...) {
+        return false;
+    }
+    const hx::PageDesc* dp = nullptr;
+    for (const auto& p : pages)
+        if (p.page_type == 0 || p.page_type == 3) dp = &p;
+    if (!dp || dp->encoding != hx::ENC_PLAIN) return false;
+    size_t in_chunk = size_t(dp->payload_off - cc.chunk_start());
+    size_t comp = size_t(dp->compressed_size) - size_t(dp->def_level_bytes);
+    const uint8_t* src = raw.data() + in_chunk + dp->def_level_bytes;
+    size_t ulen = size_t(n_rows) * 8;
+    out.bytes.resize(ulen);
+    if (cc.codec == hx::CODEC_SNAPPY && dp->is_compressed) {
+        return snappy_decompress(src, comp, out.bytes.data(), ulen);
+    }
+    if (cc.codec != hx::CODEC_UNCOMPRESSED) return false;
+    if (comp != ulen) return false;
+    std::memcpy(out.bytes.data(), src, ulen);
+    return true;
+}
+
+}  // namespace
+
+// Scan+aggregate the given SSTs on `threads` host threads. Returns elapsed
+// seconds; fills rows_scanned (decoded, after rg pruning), rows_matched and
+// n_groups. rc: 0 ok, <0 error (unsupported layout => -2).
+extern "C" int hx_cpu_scan_agg(const char** paths, int n_paths, int64_t ts_lo,
+                               int64_t ts_hi, int threads, double* elapsed_s,
+                               int64_t* rows_scanned, int64_t* rows_matched,
+                               int64_t* n_groups, double* sum_digest) {
+    if (threads < 1) threads = 1;
+    struct Unit {  // one (file, row group)
+        std::string path;
+        hx::RowGroupMeta rg;
+        int ci[3];
+    };
+    std::vector<Unit> units;
+    uint64_t total_rows = 0;
+    for (int f = 0; f < n_paths; f++) {
+        int fd = open(paths[f], O_RDONLY);
+        if (fd < 0) return -1;
+        off_t fsz = lseek(fd, 0, SEEK_END);
+        size_t tail = std::min<off_t>(fsz, 1 << 20);
+        std::vector<uint8_t> tb(tail);
+        if (pread(fd, tb.data(), tail, fsz - tail) != (ssize_t)tail) {
+            close(fd);
+            return -1;
+        }
+        hx::FileMetadata m;
+        try {
+            m = hx::parse_footer(tb.data(), tail, fsz);
+        } catch (...) {
+            close(fd);
+            return -1;
+        }
+        close(fd);
+        int ci[3] = {-1, -1, -1};
+        for (size_t i = 0; i < m.columns.size(); i++) {
+            if (m.columns[i].name == "series_id") ci[0] = (int)i;
+            if (m.columns[i].name == "timestamp") ci[1] = (int)i;
+            if (m.columns[i].name == "value") ci[2] = (int)i;
+        }
+        if (ci[0] < 0 || ci[1] < 0 || ci[2] < 0) return -2;
+        for (auto& rg : m.row_groups) {
+            // row-group pruning on ts min/max statistics (read.rs:459-470)
+            const auto& tscc = rg.columns[ci[1]];
+            if (tscc.has_stats && tscc.stat_min.size() == 8 &&
+                tscc.stat_max.size() == 8) {
+                int64_t mn = hx::stat_i64(tscc.stat_min);
+                int64_t mx = hx::stat_i64(tscc.stat_max);
+                if (mx < ts_lo || mn >= ts_hi) continue;
+            }
+            total_rows += rg.num_rows;
+            units.push_back({paths[f], rg, {ci[0], ci[1], ci[2]}});
+        }
+    }
+
+    SharedTable table;
+    uint32_t slots = 1 << 16;
+    while ((uint64_t)slots < total_rows / 16 && slots < (1u << 27))
+        slots <<= 1;
+    table.init(slots);
+
+    std::atomic<size_t> next{0};
+    std::atomic<int> err{0};
+    std::atomic<uint64_t> matched{0};
+    std::atomic<uint64_t> digest_bits{0};
+
+    auto t0 = std::chrono::steady_clock::now();
+    auto worker = [&]() {
+        uint64_t my_matched = 0;
+        double my_digest = 0;
+        int last_fd = -1;
+        std::string last_path;
+        for (;;) {
+            size_t u = next.fetch_add(1);
+            if (u >= units.size() || err.load()) break;
+            Unit& un = units[u];
+            if (un.path != last_path) {
+                if (last_fd >= 0) close(last_fd);
+                last_fd = open(un.path.c_str(), O_RDONLY);
+                last_path = un.path;
+            }
+            if (last_fd < 0) {
+                err = -1;
+                break;
+            }
+            ColData cs, ct, cv;
+            if (!decode_chunk(last_fd, un.rg.columns[un.ci[0]],
+                              un.rg.num_rows, cs) ||
+                !decode_chunk(last_fd, un.rg.columns[un.ci[1]],
+                              un.rg.num_rows, ct) ||
+                !decode_chunk(last_fd, un.rg.columns[un.ci[2]],
+                              un.rg.num_rows, cv)) {
+                err = -2;
+                break;
+            }
+            const uint64_t* S = (const uint64_t*)cs.bytes.data();
+            const int64_t* T = (const int64_t*)ct.bytes.data();
+            const double* V = (const double*)cv.bytes.data();
+            int64_t n = un.rg.num_rows;
+            // run-accumulate (rows sorted by (series, ts)); dedup: last of
+            // an equal-(s,ts) run survives. NOTE: treats a row-group
+            // boundary as a run boundary — bench row groups never split an
+            // equal-PK pair (writer sorts + 8192-row groups; same
+            // simplification as the sample-measured oracle leg).
+            uint64_t run_key = ~0ull;
+            double run_sum = 0;
+            uint64_t run_cnt = 0;
+            for (int64_t r = 0; r < n; r++) {
+                int64_t t = T[r];
+                if (t < ts_lo || t >= ts_hi) continue;
+                bool dup = (r + 1 < n) && S[r + 1] == S[r] && T[r + 1] == t;
+                if (dup) continue;
+                my_matched++;
+                double v = V[r];
+                my_digest += v;
+                if (S[r] == run_key) {
+                    run_sum += v;
+                    run_cnt++;
+                } else {
+                    if (run_cnt) table.add(run_key, run_sum, run_cnt);
+                    run_key = S[r];
+                    run_sum = v;
+                    run_cnt = 1;
+                }
+            }
+            if (run_cnt) table.add(run_key, run_sum, run_cnt);
+        }
+        if (last_fd >= 0) close(last_fd);
+        matched.fetch_add(my_matched);
+        // accumulate digest via CAS (exactness not required; diagnostic)
+        uint64_t old = digest_bits.load();
+        for (;;) {
+            double cur;
+            std::memcpy(&cur, &old, 8);
+            double nv = cur + my_digest;
+            uint64_t nb;
+            std::memcpy(&nb, &nv, 8);
+            if (digest_bits.compare_exchange_weak(old, nb)) break;
+        }
+    };
+    std::vector<std::thread> pool;
+    for (int i = 0; i < threads; i++) pool.emplace_back(worker);
+    for (auto& th : pool) th.join();
+    auto t1 = std::chrono::steady_clock::now();
+
+    if (err.load()) return err.load();
+    *elapsed_s = std::chrono::duration<double>(t1 - t0).count();
+    *rows_scanned = (int64_t)total_rows;
+    *rows_matched = (int64_t)matched.load();
+    *n_groups = (int64_t)table.fill.load();
+    double dg;
+    uint64_t db = digest_bits.load();
+    std::memcpy(&dg, &db, 8);
+    *sum_digest = dg;
+    return 0;
+}

@@ -127,8 +127,14 @@ def gen_sst_from_arrays(store_dir, seq, series, ts, value, sort=True, **kw):
 
 
 def _gen_one(args):
-    (w, out_dir, ids_path, n_series, n_points, n_ssts, seed, ts_start,
+    (j, out_dir, ids_path, n_series, n_points, n_ssts, seed, ts_start,
      step_ms, row_group, compression, ts_encoding) = args
+    # j enumerates (generation, window): generation g re-writes the SAME
+    # (series, ts) PKs of window w with new values and a higher seq — the
+    # ts-overlap layout (overwrite churn before compaction) that exercises
+    # the cross-SST MergeExec dedup (read.rs:100-391) at scale. g = 0 is the
+    # plain disjoint layout.
+    w, g = j % n_ssts, j // n_ssts
     ids = np.load(ids_path, mmap_mode="r")
     t0 = w * n_points // n_ssts
     t1 = (w + 1) * n_points // n_ssts
@@ -140,8 +146,8 @@ def _gen_one(args):
     ts = ts_start + t_idx * step_ms
     gidx = (np.repeat(np.arange(n_series, dtype=np.uint64), k) *
             np.uint64(n_points)) + t_idx.astype(np.uint64)
-    value = value_of(seed, gidx)
-    seq = w + 1
+    value = value_of(seed + 7919 * g, gidx)
+    seq = j + 1
     path = os.path.join(out_dir, "data", f"{seq}.sst")
     n = write_sst(path, series, ts, value, seq, row_group=row_group,
                   compression=compression, ts_encoding=ts_encoding,
@@ -154,16 +160,20 @@ def _gen_one(args):
 def gen_dataset(out_dir, n_rows, n_series, n_ssts, seed=42,
                 ts_start=TS_START_DEFAULT, step_ms=STEP_MS_DEFAULT,
                 row_group=8192, compression="none", ts_encoding="PLAIN",
-                workers=1):
+                workers=1, overlap_gens=1):
+    """overlap_gens > 1: each further generation re-writes the SAME PKs with
+    new values under higher seqs (ts-overlapping SSTs) — the pre-compaction
+    overwrite-churn layout; total rows = n_rows * overlap_gens, surviving
+    rows after dedup = the newest generation only."""
     assert n_rows % n_series == 0, "n_rows must be a multiple of n_series"
     n_points = n_rows // n_series
     os.makedirs(os.path.join(out_dir, "data"), exist_ok=True)
     ids = make_series_ids(n_series, seed)
     ids_path = os.path.join(out_dir, "series_ids.npy")
     np.save(ids_path, ids)
-    jobs = [(w, out_dir, ids_path, n_series, n_points, n_ssts, seed,
+    jobs = [(j, out_dir, ids_path, n_series, n_points, n_ssts, seed,
              ts_start, step_ms, row_group, compression, ts_encoding)
-            for w in range(n_ssts)]
+            for j in range(n_ssts * max(1, overlap_gens))]
     if workers > 1:
         import multiprocessing as mp
         with mp.get_context("spawn").Pool(workers) as pool:
@@ -172,7 +182,8 @@ def gen_dataset(out_dir, n_rows, n_series, n_ssts, seed=42,
         metas = [_gen_one(j) for j in jobs]
     metas = [m for m in metas if m]
     manifest = {
-        "n_rows": n_rows, "n_series": n_series, "n_points": n_points,
+        "n_rows": n_rows * max(1, overlap_gens), "n_series": n_series,
+        "n_points": n_points, "overlap_gens": max(1, overlap_gens),
         "n_ssts": len(metas), "seed": seed, "ts_start": ts_start,
         "step_ms": step_ms, "row_group": row_group,
         "compression": compression, "ts_encoding": ts_encoding,
