@@ -1302,7 +1302,8 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
     // at 16384 (64-SST shapes: ~256 rows/sst/block). Empty or tiny blocks
     // are cheap; an LDS-table overflow (the fallback path) is a global-RMW
     // storm (r01: 286M of 320M updates went global at the 1B shape).
-    while ((double)nb * 16384.0 < staged && nb < (1u << 17)) nb <<= 1;
+    if (!getenv("HX_RANGE_TARGET"))   // explicit target: trust the caller
+        while ((double)nb * 16384.0 < staged && nb < (1u << 17)) nb <<= 1;
     std::vector<uint64_t> bounds(nb + 1);
     bounds[0] = 0;
     for (uint32_t b = 1; b < nb; b++) bounds[b] = samp[(size_t)b * m / nb];
@@ -1506,8 +1507,17 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                            plan.d_sst_rg_cnt,  (uint32_t)plan.ssts.size(),
                            plan.range_nblocks, ne,
                            interp};
-            hipError_t re2 = hx::launch_scan_agg_range(
-                s, A, R, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0);
+            // HX_RANGE2: pair-load variant (16 B/lane dwordx4, in-lane
+            // pair merge, no cross-lane pre-reduce)
+            const bool r2k = [] {
+                const char* e = getenv("HX_RANGE2");
+                return e ? atoi(e) != 0 : true;
+            }();
+            hipError_t re2 =
+                r2k ? hx::launch_scan_agg_range2(
+                          s, A, R, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0)
+                    : hx::launch_scan_agg_range(
+                          s, A, R, (ops & (HX_AGG_MIN | HX_AGG_MAX)) != 0);
             if (re2 != hipSuccess)
                 return fail(HX_ERR_HIP,
                             std::string("range kernel launch failed: ") +

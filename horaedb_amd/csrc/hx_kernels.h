@@ -39,6 +39,8 @@ hipError_t launch_sample_series(hipStream_t s, const RgDesc* rgs,
                                 const uint8_t* dec, uint64_t* out);
 hipError_t launch_range_bounds(hipStream_t s, const AggParams& p,
                                const RangeAux& r, uint64_t* out);
+hipError_t launch_scan_agg_range2(hipStream_t s, const AggParams& p,
+                                  const RangeAux& r, bool minmax);
 hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
                                  const RangeAux& r, bool minmax);
 hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,

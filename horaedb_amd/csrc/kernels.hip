@@ -944,7 +944,20 @@ k_scan_agg_range(AggParams P, RangeAux R) {
     const uint32_t n_waves = blockDim.x >> 6;
     unsigned long long my_matched = 0;
     for (uint32_t blk = blockIdx.x; blk < R.n_blocks; blk += gridDim.x) {
-        if (threadIdx.x == 0) s_abort = 0;
+        // saturation early-abort, ONCE PER BLOCK: P.fill's line is evicted
+        // chip-wide by every agent-scope claim, so ANY load of it is served
+        // at the coherence point (~88/us single-word) — polling per
+        // (wave, sst) serialized the whole kernel (+23 ms at the 1B shape).
+        // A stale read only delays the abort; the probe cap still bounds a
+        // doomed pass.
+        if (threadIdx.x == 0) {
+            s_abort = 0;
+            if (P.poll && (blk & 7u) == 0) {
+                unsigned long long f = __hip_atomic_load(
+                    P.fill, RLX, __HIP_MEMORY_SCOPE_WORKGROUP);
+                if (f > P.fill_limit) s_abort = 1;
+            }
+        }
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             lkey[i] = KEY_EMPTY;
             lsum[i] = 0.0;
@@ -967,25 +980,12 @@ k_scan_agg_range(AggParams P, RangeAux R) {
             islope = (double)ne / ((double)(ihi - ilo) + 1.0);
         }
         __syncthreads();
+        if (s_abort) {
+            if (threadIdx.x == 0)
+                __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
+            break;   // host retries with a larger table
+        }
         for (uint32_t si = wave; si < R.n_ssts; si += n_waves) {
-            // saturation early-abort (a doomed pass against a full table
-            // grinds probe scans). NOT a coherence-point load: P.fill is
-            // updated by memory-side (agent-scope) atomics, and a volatile
-            // load of that one hot word from every (wave, sst) iteration
-            // serializes at ~88 loads/us chip-wide. A workgroup-scope
-            // atomic load is L1/L2-served (possibly stale — the abort is
-            // best-effort; the probe cap still bounds a doomed pass).
-            if (P.poll) {
-                unsigned long long f = 0;
-                if (lane == 0)
-                    f = __hip_atomic_load(P.fill, RLX,
-                                          __HIP_MEMORY_SCOPE_WORKGROUP);
-                f = __shfl(f, 0, 64);
-                if (f > P.fill_limit) {
-                    s_abort = 1;
-                    break;
-                }
-            }
             const int32_t loff = R.sst_rg_off[si];
             const uint64_t pk0 = R.bound_rows[(size_t)blk * R.n_ssts + si];
             const uint64_t pk1 =
@@ -1039,11 +1039,206 @@ k_scan_agg_range(AggParams P, RangeAux R) {
             }
         }
         __syncthreads();
+        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+            if (lkey[i] == KEY_EMPTY) continue;
+            agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
+                       MM ? ordered_f64(lmin[i]) : 0.0,
+                       MM ? ordered_f64(lmax[i]) : 0.0);
+        }
+        __syncthreads();
+    }
+    for (int off = 32; off > 0; off >>= 1)
+        my_matched += __shfl_down(my_matched, off, 64);
+    if ((threadIdx.x & 63) == 0 && my_matched)
+        atomicAdd(P.matched, my_matched);
+}
+
+// ---------------------------------------------------------------------------
+// Pair-load series-range kernel (round-2 redesign of k_scan_agg_range).
+// The r01 kernel was 95% SQ_WAIT_ANY at ~2% LDS-array utilization: pure
+// latency exposure, not throughput. This variant removes latency sources:
+//   - 16 B/lane dwordx4 loads (the calibrated wide-stream width): each lane
+//     owns TWO consecutive rows, halving load instructions per row;
+//   - the successor row for dedup comes from the register pair / one
+//     cross-lane shuffle instead of a second global read of S and T;
+//   - NO cross-lane run pre-reduction (measured ±3%): each lane merges its
+//     own pair in registers (36% of adjacent rows share a series at the
+//     headline shape) and issues its 1-2 LDS updates directly — the shuffle
+//     loop's ~7 dependent ds_bpermute rounds per window disappear.
+// Table layout, partitioning, flush and dedup semantics are identical to
+// k_scan_agg_range.
+// ---------------------------------------------------------------------------
+template <bool MM>
+__global__ void __launch_bounds__(256)
+k_scan_agg_range2(AggParams P, RangeAux R) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    __shared__ int s_abort;
+    const uint32_t ne = R.ne;
+    uint64_t* lkey = (uint64_t*)smem;
+    double* lsum = (double*)(smem + (size_t)ne * 8);
+    unsigned long long* lmin = (unsigned long long*)(smem + (size_t)ne * 16);
+    unsigned long long* lmax = (unsigned long long*)(smem + (size_t)ne * 24);
+    unsigned int* lcnt = (unsigned int*)(smem + (size_t)ne * (MM ? 32 : 16));
+    const int lane = threadIdx.x & 63;
+    const uint32_t wave = threadIdx.x >> 6;
+    const uint32_t n_waves = blockDim.x >> 6;
+    unsigned long long my_matched = 0;
+    for (uint32_t blk = blockIdx.x; blk < R.n_blocks; blk += gridDim.x) {
+        if (threadIdx.x == 0) {
+            s_abort = 0;
+            if (P.poll && (blk & 7u) == 0) {
+                unsigned long long f = __hip_atomic_load(
+                    P.fill, RLX, __HIP_MEMORY_SCOPE_WORKGROUP);
+                if (f > P.fill_limit) s_abort = 1;
+            }
+        }
+        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+            lkey[i] = KEY_EMPTY;
+            lsum[i] = 0.0;
+            lcnt[i] = 0u;
+            if (MM) {
+                lmin[i] = ~0ull;
+                lmax[i] = 0ull;
+            }
+        }
+        double islope = 0.0;
+        uint64_t ilo = 0;
+        if (R.interp) {
+            ilo = R.bounds[blk];
+            const uint64_t ihi = R.bounds[blk + 1];
+            islope = (double)ne / ((double)(ihi - ilo) + 1.0);
+        }
+        __syncthreads();
         if (s_abort) {
             if (threadIdx.x == 0)
                 __hip_atomic_fetch_add(P.overflow, 1ull, RLX, AGT);
-            break;   // host retries with a larger table; skip the flush
+            break;
         }
+        for (uint32_t si = wave; si < R.n_ssts; si += n_waves) {
+            const int32_t loff = R.sst_rg_off[si];
+            const uint64_t pk0 = R.bound_rows[(size_t)blk * R.n_ssts + si];
+            const uint64_t pk1 =
+                R.bound_rows[(size_t)(blk + 1) * R.n_ssts + si];
+            uint32_t pos = (uint32_t)(pk0 >> 32);
+            uint32_t row = (uint32_t)pk0;
+            const uint32_t epos = (uint32_t)(pk1 >> 32);
+            const uint32_t erow = (uint32_t)pk1;
+            while (pos < epos || (pos == epos && row < erow)) {
+                const RgDesc rg = P.rgs[R.sst_rgs[loff + (int32_t)pos]];
+                const uint64_t* S =
+                    (const uint64_t*)hx_ptr(P.blob, P.dec, rg.series_off);
+                const int64_t* T =
+                    (const int64_t*)hx_ptr(P.blob, P.dec, rg.ts_off);
+                const double* V =
+                    (const double*)hx_ptr(P.blob, P.dec, rg.val_off);
+                const SstDev sst = P.ssts[rg.sst_id];
+                const uint32_t hi = (pos == epos) ? erow : rg.n_rows;
+                const uint32_t nt = rg.n_rows;   // true slice rows
+                for (uint32_t base = row & ~1u; base < hi; base += 128) {
+                    const uint32_t r0 = base + 2u * (uint32_t)lane;
+                    const uint32_t r1 = r0 + 1u;
+                    // load the pair (dwordx4); mask by the ARRAY bound nt —
+                    // rows in [hi, nt) load safely and are filtered below
+                    uint64_t s0 = KEY_EMPTY, s1 = KEY_EMPTY;
+                    int64_t t0 = 0, t1 = 0;
+                    double v0 = 0.0, v1 = 0.0;
+                    if (r1 < nt) {
+                        const ulonglong2 sp =
+                            *(const ulonglong2*)(S + r0);
+                        const longlong2 tp = *(const longlong2*)(T + r0);
+                        const double2 vp = *(const double2*)(V + r0);
+                        s0 = sp.x;
+                        s1 = sp.y;
+                        t0 = tp.x;
+                        t1 = tp.y;
+                        v0 = vp.x;
+                        v1 = vp.y;
+                    } else if (r0 < nt) {
+                        s0 = S[r0];
+                        t0 = T[r0];
+                        v0 = V[r0];
+                    }
+                    // window validity: [max(row, base), hi)
+                    bool a0 = r0 >= row && r0 < hi;
+                    bool a1 = r1 >= row && r1 < hi;
+                    a0 = a0 && (t0 >= P.ts_lo) & (t0 < P.ts_hi);
+                    a1 = a1 && (t1 >= P.ts_lo) & (t1 < P.ts_hi);
+                    if (P.use_sset) {
+                        if (a0) a0 = sset_has(P, s0);
+                        if (a1) a1 = sset_has(P, s1);
+                    }
+                    // successor of r1 = next lane's r0 (one 64-bit shuffle
+                    // pair); lane 63 / slice-tail fall back to scalar loads
+                    const uint64_t sn = __shfl_down(s0, 1, 64);
+                    const int64_t tn = __shfl_down((long long)t0, 1, 64);
+                    if (a0 && P.skip < 2) {
+                        bool dup;
+                        if (r1 < nt) {
+                            dup = (s1 == s0) & (t1 == t0);
+                        } else if (rg.next_rg >= 0) {
+                            const RgDesc nx = P.rgs[rg.next_rg];
+                            dup = (*(const uint64_t*)hx_ptr(
+                                       P.blob, P.dec, nx.series_off) == s0) &
+                                  (*(const int64_t*)hx_ptr(
+                                       P.blob, P.dec, nx.ts_off) == t0);
+                        } else {
+                            dup = false;
+                        }
+                        if (!dup && sst.cluster >= 0)
+                            dup = shadowed(P, sst, s0, t0);
+                        a0 = !dup;
+                    }
+                    if (a1 && P.skip < 2) {
+                        bool dup;
+                        const uint32_t r2 = r1 + 1u;
+                        if (r2 < nt) {
+                            if (lane < 63) {
+                                dup = (sn == s1) & (tn == t1);
+                            } else {
+                                dup = (S[r2] == s1) & (T[r2] == t1);
+                            }
+                        } else if (rg.next_rg >= 0) {
+                            const RgDesc nx = P.rgs[rg.next_rg];
+                            dup = (*(const uint64_t*)hx_ptr(
+                                       P.blob, P.dec, nx.series_off) == s1) &
+                                  (*(const int64_t*)hx_ptr(
+                                       P.blob, P.dec, nx.ts_off) == t1);
+                        } else {
+                            dup = false;
+                        }
+                        if (!dup && sst.cluster >= 0)
+                            dup = shadowed(P, sst, s1, t1);
+                        a1 = !dup;
+                    }
+                    my_matched += (a0 ? 1u : 0u) + (a1 ? 1u : 0u);
+                    if (P.skip == 1) continue;
+                    // in-lane pair merge, then direct LDS updates
+                    const bool merge = a0 && a1 && (s0 == s1);
+                    uint32_t i0 = 0xFFFFFFFFu, i1 = 0xFFFFFFFFu;
+                    if (R.interp) {
+                        uint32_t x0 = (uint32_t)((double)(s0 - ilo) * islope);
+                        uint32_t x1 = (uint32_t)((double)(s1 - ilo) * islope);
+                        i0 = x0 < ne ? x0 : ne - 1;
+                        i1 = x1 < ne ? x1 : ne - 1;
+                    }
+                    if (merge) {
+                        lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
+                                       s0, v0 + v1, 2u, fmin(v0, v1),
+                                       fmax(v0, v1), i0);
+                    } else {
+                        if (a0)
+                            lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax,
+                                           ne, s0, v0, 1u, v0, v0, i0);
+                        if (a1)
+                            lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax,
+                                           ne, s1, v1, 1u, v1, v1, i1);
+                    }
+                }
+                pos++;
+                row = 0;
+            }
+        }
+        __syncthreads();
         for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
             if (lkey[i] == KEY_EMPTY) continue;
             agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
@@ -1838,6 +2033,27 @@ hipError_t launch_range_bounds(hipStream_t s, const AggParams& p,
     if (blocks == 0) blocks = 1;
     hipLaunchKernelGGL(k_range_bounds, dim3((uint32_t)blocks), dim3(256), 0,
                        s, p, r, out);
+    return hipGetLastError();
+}
+
+hipError_t launch_scan_agg_range2(hipStream_t s, const AggParams& p,
+                                  const RangeAux& r, bool minmax) {
+    const size_t lds = (size_t)r.ne * (minmax ? 36 : 20);
+    const void* f = minmax
+                        ? reinterpret_cast<const void*>(&k_scan_agg_range2<true>)
+                        : reinterpret_cast<const void*>(
+                              &k_scan_agg_range2<false>);
+    if (lds > 64 * 1024) {
+        hipError_t e = hipFuncSetAttribute(
+            f, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+        if (e != hipSuccess) return e;
+    }
+    if (minmax)
+        hipLaunchKernelGGL(k_scan_agg_range2<true>, dim3(r.n_blocks),
+                           dim3(256), lds, s, p, r);
+    else
+        hipLaunchKernelGGL(k_scan_agg_range2<false>, dim3(r.n_blocks),
+                           dim3(256), lds, s, p, r);
     return hipGetLastError();
 }
 

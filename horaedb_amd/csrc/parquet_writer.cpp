@@ -82,10 +82,10 @@ void schema_element(TW& w, const ColSpec& c) {
     w.stop();
 }
 
-void statistics(TW& w, const void* mn, const void* mx) {
+void statistics(TW& w, const std::string& mn, const std::string& mx) {
     int16_t last = 0;
-    w.binary(last, 5, mx, 8);  // max_value
-    w.binary(last, 6, mn, 8);  // min_value
+    w.binary(last, 5, mx.data(), mx.size());  // max_value
+    w.binary(last, 6, mn.data(), mn.size());  // min_value
     w.stop();
 }
 
@@ -93,30 +93,34 @@ struct ChunkMeta {
     int64_t data_page_offset;
     int64_t total_size;
     int64_t num_values;
-    uint8_t mn[8], mx[8];
+    std::string mn, mx;
+    bool has_stats = true;
 };
 
-void column_chunk(TW& w, const ColSpec& c, const ChunkMeta& m) {
+void column_chunk(TW& w, const char* name, int32_t physical,
+                  const ChunkMeta& m) {
     int16_t last = 0;
     w.i64f(last, 2, m.data_page_offset);  // file_offset (deprecated, required)
     // meta_data struct (field 3)
     w.field(last, 3, 12);
     {
         int16_t l2 = 0;
-        w.i32(l2, 1, c.physical);                      // type
+        w.i32(l2, 1, physical);                        // type
         w.list_header(l2, 2, 5, 1);                    // encodings: [PLAIN]
         w.zigzag(0);
         w.list_header(l2, 3, 8, 1);                    // path_in_schema
-        w.varint(strlen(c.name));
-        w.buf.insert(w.buf.end(), (const uint8_t*)c.name,
-                     (const uint8_t*)c.name + strlen(c.name));
+        w.varint(strlen(name));
+        w.buf.insert(w.buf.end(), (const uint8_t*)name,
+                     (const uint8_t*)name + strlen(name));
         w.i32(l2, 4, 0);                               // codec UNCOMPRESSED
         w.i64f(l2, 5, m.num_values);
         w.i64f(l2, 6, m.total_size);                   // uncompressed
         w.i64f(l2, 7, m.total_size);                   // compressed
         w.i64f(l2, 9, m.data_page_offset);
-        w.field(l2, 12, 12);                           // statistics
-        statistics(w, m.mn, m.mx);
+        if (m.has_stats) {
+            w.field(l2, 12, 12);                       // statistics
+            statistics(w, m.mn, m.mx);
+        }
         w.stop();
     }
     w.stop();
@@ -142,14 +146,14 @@ std::vector<uint8_t> page_header(int32_t n_values, int32_t payload) {
 }
 
 template <typename T>
-void minmax_bytes(const T* v, int64_t n, uint8_t* mn, uint8_t* mx) {
+void minmax_bytes(const T* v, int64_t n, std::string& mn, std::string& mx) {
     T lo = v[0], hi = v[0];
     for (int64_t i = 1; i < n; i++) {
         if (v[i] < lo) lo = v[i];
         if (v[i] > hi) hi = v[i];
     }
-    memcpy(mn, &lo, 8);
-    memcpy(mx, &hi, 8);
+    mn.assign((const char*)&lo, 8);
+    mx.assign((const char*)&hi, 8);
 }
 
 }  // namespace
@@ -203,6 +207,7 @@ std::string write_metric_sst(const std::string& path, const uint64_t* series,
                     minmax_bytes((const double*)data[c], rows, m.mn, m.mx);
                     break;
             }
+            m.has_stats = true;
             if (fwrite(hdr.data(), 1, hdr.size(), f) != hdr.size())
                 return fail("write failed");
             if (fwrite(data[c], 8, rows, f) != size_t(rows))
@@ -231,7 +236,8 @@ std::string write_metric_sst(const std::string& path, const uint64_t* series,
         w.list_header(l2, 1, 12, 5);
         int64_t total = 0;
         for (int c = 0; c < 5; c++) total += rg.cols[c].total_size;
-        for (int c = 0; c < 5; c++) column_chunk(w, kCols[c], rg.cols[c]);
+        for (int c = 0; c < 5; c++)
+            column_chunk(w, kCols[c].name, kCols[c].physical, rg.cols[c]);
         w.i64f(l2, 2, total);        // total_byte_size
         w.i64f(l2, 3, rg.num_rows);  // num_rows
         w.stop();

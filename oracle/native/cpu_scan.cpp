@@ -205,9 +205,16 @@ extern "C" int hx_cpu_scan_agg(const char** paths, int n_paths, int64_t ts_lo,
         int fd = open(paths[f], O_RDONLY);
         if (fd < 0) return -1;
         off_t fsz = lseek(fd, 0, SEEK_END);
-        size_t tail = std::min<off_t>(fsz, 1 << 20);
+        if (fsz < 12) { close(fd); return -1; }
+        uint8_t trailer[8];
+        if (pread(fd, trailer, 8, fsz - 8) != 8) { close(fd); return -1; }
+        uint32_t flen;
+        std::memcpy(&flen, trailer, 4);
+        // footer + its 8-byte trailer (1B-row SSTs carry ~2k row groups —
+        // a fixed 1 MB tail was too small and failed with rc=-1)
+        size_t tail = std::min<off_t>(fsz, (off_t)flen + 8);
         std::vector<uint8_t> tb(tail);
-        if (pread(fd, tb.data(), tail, fsz - tail) != (ssize_t)tail) {
+        if (pread(fd, tb.data(), tail, fsz - (off_t)tail) != (ssize_t)tail) {
             close(fd);
             return -1;
         }
