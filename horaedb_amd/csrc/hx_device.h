@@ -164,4 +164,31 @@ struct CopyDesc {
     uint32_t _pad;
 };
 
+// ---- inverted-index query units (rfc:86-137 index table) ------------------
+// One PLAIN BYTE_ARRAY page (u32 length + bytes per value): walked into a
+// packed (offset << 20 | len) per row at out[first_row..first_row+n).
+struct BaPageDesc {
+    uint64_t src_off;     // page payload offset in the index blob
+    uint64_t src_len;
+    uint32_t n_values;
+    uint32_t _pad;
+    int64_t first_row;    // global row index of this page's first value
+};
+
+// Tag-equality filter over decoded (key, value) offset arrays + tsid column.
+struct TagFilterParams {
+    const uint8_t* blob;          // index page blob
+    const uint64_t* key_offlen;   // packed per-row (off << 20 | len)
+    const uint64_t* val_offlen;
+    const uint64_t* tsid;         // dense per-row tsids
+    int64_t n_rows;
+    const uint8_t* pred_key;      // predicate bytes (device)
+    uint32_t pred_key_len;
+    uint32_t pred_val_len;
+    const uint8_t* pred_val;
+    uint64_t* out;                // matching tsids (unordered)
+    unsigned long long* cursor;
+    unsigned long long cap;
+};
+
 }  // namespace hx

@@ -73,6 +73,22 @@ hipError_t launch_seg_keys(hipStream_t s, const long long* ts, long long seg_ms,
                            unsigned long long* keys, uint32_t n);
 hipError_t launch_xor_sign(hipStream_t s, unsigned long long* buf, uint32_t n);
 
+// inverted-index query kernels (rfc:86-137)
+hipError_t launch_ba_offsets(hipStream_t s, const uint8_t* blob,
+                             const BaPageDesc* pages, uint32_t n_pages,
+                             uint64_t* out, unsigned long long* err_flag);
+hipError_t launch_tag_filter(hipStream_t s, const TagFilterParams& f);
+hipError_t launch_tsid_intersect(hipStream_t s, const uint64_t* a,
+                                 unsigned long long n_a, const uint64_t* b,
+                                 unsigned long long n_b, uint64_t* out,
+                                 unsigned long long* cursor);
+hipError_t launch_unique_u64(hipStream_t s, const uint64_t* in,
+                             unsigned long long n, uint64_t* out,
+                             unsigned long long* cursor);
+hipError_t sort_keys_u64(hipStream_t s, const uint64_t* keys_in,
+                         uint64_t* keys_out, size_t n, void** d_temp,
+                         size_t* temp_bytes);
+
 // rocPRIM stable LSD radix sort: sorts values (u32 perm) by u64 keys.
 // temp buffer managed internally on the stream (hipMallocAsync-free impl).
 hipError_t sort_pairs_u64(hipStream_t s, const uint64_t* keys_in,

@@ -1424,6 +1424,142 @@ k_scan_rows(ScanRowsParams R) {
 }
 
 // ---------------------------------------------------------------------------
+// Inverted-index query kernels (rfc:86-137): PLAIN BYTE_ARRAY offset walk,
+// tag-equality postings filter, sorted-set intersection / adjacent-unique.
+// All HBM-bound byte work; no MFMA.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+k_ba_offsets(const uint8_t* __restrict__ blob,
+             const BaPageDesc* __restrict__ pages, uint32_t n_pages,
+             uint64_t* __restrict__ out, unsigned long long* err_flag) {
+    // the length-prefixed layout is a serial chain per page; pages decode in
+    // parallel (one thread walks one page: ~8192 values)
+    const uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const uint32_t n_threads = gridDim.x * blockDim.x;
+    for (uint32_t pg = tid; pg < n_pages; pg += n_threads) {
+        const BaPageDesc pd = pages[pg];
+        uint64_t pos = 0;
+        bool bad = false;
+        for (uint32_t i = 0; i < pd.n_values; i++) {
+            if (pos + 4 > pd.src_len) { bad = true; break; }
+            const uint8_t* p = blob + pd.src_off + pos;
+            uint32_t len = (uint32_t)p[0] | ((uint32_t)p[1] << 8) |
+                           ((uint32_t)p[2] << 16) | ((uint32_t)p[3] << 24);
+            pos += 4;
+            if (len >= (1u << 20) || pos + len > pd.src_len) {
+                bad = true;
+                break;
+            }
+            out[pd.first_row + i] = ((pd.src_off + pos) << 20) | len;
+            pos += len;
+        }
+        if (bad) atomicAdd(err_flag, 1ull);
+    }
+}
+
+__device__ __forceinline__ bool bytes_eq(const uint8_t* a, const uint8_t* b,
+                                         uint32_t n) {
+    uint32_t i = 0;
+    for (; i + 8 <= n; i += 8) {
+        uint64_t x, y;
+        __builtin_memcpy(&x, a + i, 8);
+        __builtin_memcpy(&y, b + i, 8);
+        if (x != y) return false;
+    }
+    for (; i < n; i++)
+        if (a[i] != b[i]) return false;
+    return true;
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_tag_filter(TagFilterParams F) {
+    const int lane = threadIdx.x & 63;
+    for (int64_t r = blockIdx.x * blockDim.x + threadIdx.x;
+         __any(r < F.n_rows); r += (int64_t)blockDim.x * gridDim.x) {
+        bool hit = false;
+        uint64_t id = 0;
+        if (r < F.n_rows) {
+            const uint64_t ko = F.key_offlen[r];
+            const uint32_t klen = (uint32_t)(ko & 0xFFFFFu);
+            if (klen == F.pred_key_len &&
+                bytes_eq(F.blob + (ko >> 20), F.pred_key, klen)) {
+                const uint64_t vo = F.val_offlen[r];
+                const uint32_t vlen = (uint32_t)(vo & 0xFFFFFu);
+                if (vlen == F.pred_val_len &&
+                    bytes_eq(F.blob + (vo >> 20), F.pred_val, vlen)) {
+                    hit = true;
+                    id = F.tsid[r];
+                }
+            }
+        }
+        const unsigned long long mask = __ballot(hit);
+        if (!mask) continue;
+        const int leader = __ffsll((unsigned long long)mask) - 1;
+        unsigned long long base = 0;
+        if (lane == leader)
+            base = atomicAdd(F.cursor, (unsigned long long)__popcll(mask));
+        base = __shfl(base, leader, 64);
+        if (hit) {
+            const unsigned long long j =
+                base + __popcll(mask & ((1ull << lane) - 1ull));
+            if (j < F.cap) F.out[j] = id;
+        }
+    }
+}
+
+// sorted-set intersection: keep a[i] iff it appears in sorted b[0..n_b)
+extern "C" __global__ void __launch_bounds__(256)
+k_tsid_intersect(const uint64_t* __restrict__ a, unsigned long long n_a,
+                 const uint64_t* __restrict__ b, unsigned long long n_b,
+                 uint64_t* __restrict__ out, unsigned long long* cursor) {
+    const int lane = threadIdx.x & 63;
+    for (unsigned long long r = blockIdx.x * blockDim.x + threadIdx.x;
+         __any(r < n_a); r += (unsigned long long)blockDim.x * gridDim.x) {
+        bool hit = false;
+        uint64_t v = 0;
+        if (r < n_a) {
+            v = a[r];
+            unsigned long long lo = 0, hi = n_b;
+            while (lo < hi) {
+                unsigned long long mid = (lo + hi) >> 1;
+                if (b[mid] < v) lo = mid + 1;
+                else hi = mid;
+            }
+            hit = lo < n_b && b[lo] == v;
+        }
+        const unsigned long long mask = __ballot(hit);
+        if (!mask) continue;
+        const int leader = __ffsll((unsigned long long)mask) - 1;
+        unsigned long long base = 0;
+        if (lane == leader)
+            base = atomicAdd(cursor, (unsigned long long)__popcll(mask));
+        base = __shfl(base, leader, 64);
+        if (hit)
+            out[base + __popcll(mask & ((1ull << lane) - 1ull))] = v;
+    }
+}
+
+// adjacent-unique over a SORTED array
+extern "C" __global__ void __launch_bounds__(256)
+k_unique_u64(const uint64_t* __restrict__ in, unsigned long long n,
+             uint64_t* __restrict__ out, unsigned long long* cursor) {
+    const int lane = threadIdx.x & 63;
+    for (unsigned long long r = blockIdx.x * blockDim.x + threadIdx.x;
+         __any(r < n); r += (unsigned long long)blockDim.x * gridDim.x) {
+        const bool keep = (r < n) && (r == 0 || in[r] != in[r - 1]);
+        const unsigned long long mask = __ballot(keep);
+        if (!mask) continue;
+        const int leader = __ffsll((unsigned long long)mask) - 1;
+        unsigned long long base = 0;
+        if (lane == leader)
+            base = atomicAdd(cursor, (unsigned long long)__popcll(mask));
+        base = __shfl(base, leader, 64);
+        if (keep)
+            out[base + __popcll(mask & ((1ull << lane) - 1ull))] = in[r];
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Result compaction: live slots -> dense arrays (unsorted; host sorts with
 // rocPRIM then gathers).
 // ---------------------------------------------------------------------------
@@ -1898,6 +2034,7 @@ k_decode_delta_i64(const uint8_t* blob, uint8_t* dec,
 // host-side launchers (hx_kernels.h)
 // ---------------------------------------------------------------------------
 #include "hx_kernels.h"
+#include <algorithm>
 #include <cstring>
 #include <cstdlib>
 #include <rocprim/device/device_radix_sort.hpp>
@@ -2204,6 +2341,57 @@ hipError_t launch_init_rep(hipStream_t s, uint8_t* rep, size_t total_slots,
 hipError_t launch_iota(hipStream_t s, uint32_t* out, uint32_t n) {
     hipLaunchKernelGGL(k_iota, dim3(grid_for(n, 256)), dim3(256), 0, s, out, n);
     return hipGetLastError();
+}
+
+hipError_t launch_ba_offsets(hipStream_t s, const uint8_t* blob,
+                             const BaPageDesc* pages, uint32_t n_pages,
+                             uint64_t* out, unsigned long long* err_flag) {
+    hipLaunchKernelGGL(k_ba_offsets, dim3(grid_for(n_pages, 256)), dim3(256),
+                       0, s, blob, pages, n_pages, out, err_flag);
+    return hipGetLastError();
+}
+
+hipError_t launch_tag_filter(hipStream_t s, const TagFilterParams& f) {
+    uint32_t work = (uint32_t)std::min<int64_t>(f.n_rows, 0x7FFFFFFF);
+    hipLaunchKernelGGL(k_tag_filter, dim3(grid_for(work, 256)), dim3(256), 0,
+                       s, f);
+    return hipGetLastError();
+}
+
+hipError_t launch_tsid_intersect(hipStream_t s, const uint64_t* a,
+                                 unsigned long long n_a, const uint64_t* b,
+                                 unsigned long long n_b, uint64_t* out,
+                                 unsigned long long* cursor) {
+    uint32_t work = (uint32_t)std::min<unsigned long long>(n_a, 0x7FFFFFFF);
+    hipLaunchKernelGGL(k_tsid_intersect, dim3(grid_for(work, 256)), dim3(256),
+                       0, s, a, n_a, b, n_b, out, cursor);
+    return hipGetLastError();
+}
+
+hipError_t launch_unique_u64(hipStream_t s, const uint64_t* in,
+                             unsigned long long n, uint64_t* out,
+                             unsigned long long* cursor) {
+    uint32_t work = (uint32_t)std::min<unsigned long long>(n, 0x7FFFFFFF);
+    hipLaunchKernelGGL(k_unique_u64, dim3(grid_for(work, 256)), dim3(256), 0,
+                       s, in, n, out, cursor);
+    return hipGetLastError();
+}
+
+hipError_t sort_keys_u64(hipStream_t s, const uint64_t* keys_in,
+                         uint64_t* keys_out, size_t n, void** d_temp,
+                         size_t* temp_bytes) {
+    size_t need = 0;
+    hipError_t e = rocprim::radix_sort_keys(nullptr, need, keys_in, keys_out,
+                                            n, 0, 64, s);
+    if (e != hipSuccess) return e;
+    if (need > *temp_bytes) {
+        if (*d_temp) (void)hipFree(*d_temp);
+        e = hipMalloc(d_temp, need);
+        if (e != hipSuccess) { *temp_bytes = 0; *d_temp = nullptr; return e; }
+        *temp_bytes = need;
+    }
+    return rocprim::radix_sort_keys(*d_temp, *temp_bytes, keys_in, keys_out,
+                                    n, 0, 64, s);
 }
 
 hipError_t sort_pairs_u64(hipStream_t s, const uint64_t* keys_in,

@@ -262,4 +262,160 @@ std::string write_metric_sst(const std::string& path, const uint64_t* series,
     return "";
 }
 
+// ---------------------------------------------------------------------------
+// General flat-table writer (BYTE_ARRAY-capable) for the RFC's auxiliary
+// tables (rfc:86-137) and per-row-seq compaction outputs. Same page/footer
+// layout as write_metric_sst: one PLAIN uncompressed data page v1 per
+// chunk, row groups of `row_group` rows, min/max statistics (omitted for
+// byte values longer than 64 B), thrift-compact footer with column_orders.
+// ---------------------------------------------------------------------------
+std::string write_table_sst(const std::string& path, const WriterCol* cols,
+                            int32_t n_cols, int64_t n, int64_t row_group) {
+    if (n <= 0 || n_cols <= 0) return "write_table_sst: no rows/cols";
+    FILE* f = fopen(path.c_str(), "wb");
+    if (!f) return "write_table_sst: cannot open " + path;
+    auto fail = [&](const char* m) {
+        fclose(f);
+        remove(path.c_str());
+        return std::string(m);
+    };
+    for (int c = 0; c < n_cols; c++) {
+        const WriterCol& wc = cols[c];
+        if (wc.physical != 2 && wc.physical != 5 && wc.physical != 6)
+            return fail("write_table_sst: unsupported physical type");
+        if (wc.physical == 6 && !wc.offsets)
+            return fail("write_table_sst: BYTE_ARRAY needs offsets");
+    }
+    if (fwrite("PAR1", 1, 4, f) != 4) return fail("write failed");
+    int64_t off = 4;
+
+    struct RG {
+        int64_t num_rows;
+        std::vector<ChunkMeta> cols;
+    };
+    std::vector<RG> rgs;
+    std::vector<uint8_t> payload;
+    for (int64_t base = 0; base < n; base += row_group) {
+        int64_t rows = std::min<int64_t>(row_group, n - base);
+        RG rg;
+        rg.num_rows = rows;
+        rg.cols.resize(n_cols);
+        for (int c = 0; c < n_cols; c++) {
+            const WriterCol& wc = cols[c];
+            ChunkMeta& m = rg.cols[c];
+            m.num_values = rows;
+            m.data_page_offset = off;
+            const uint8_t* body = nullptr;
+            size_t body_len = 0;
+            if (wc.physical == 6) {
+                const uint8_t* bytes = (const uint8_t*)wc.data;
+                payload.clear();
+                m.has_stats = true;
+                std::string mn, mx;
+                for (int64_t r = base; r < base + rows; r++) {
+                    int64_t b0 = wc.offsets[r], b1 = wc.offsets[r + 1];
+                    if (b1 < b0) return fail("write_table_sst: bad offsets");
+                    uint32_t len = (uint32_t)(b1 - b0);
+                    payload.insert(payload.end(), (const uint8_t*)&len,
+                                   (const uint8_t*)&len + 4);
+                    payload.insert(payload.end(), bytes + b0, bytes + b1);
+                    std::string v((const char*)bytes + b0, (size_t)len);
+                    if (r == base || v < mn) mn = v;
+                    if (r == base || v > mx) mx = v;
+                }
+                if (mn.size() > 64 || mx.size() > 64) m.has_stats = false;
+                m.mn = mn;
+                m.mx = mx;
+                body = payload.data();
+                body_len = payload.size();
+            } else {
+                body = (const uint8_t*)wc.data + base * 8;
+                body_len = size_t(rows) * 8;
+                if (wc.physical == 5)
+                    minmax_bytes((const double*)wc.data + base, rows, m.mn,
+                                 m.mx);
+                else if (wc.converted == 14)
+                    minmax_bytes((const uint64_t*)wc.data + base, rows, m.mn,
+                                 m.mx);
+                else
+                    minmax_bytes((const int64_t*)wc.data + base, rows, m.mn,
+                                 m.mx);
+            }
+            auto hdr = page_header((int32_t)rows, (int32_t)body_len);
+            m.total_size = int64_t(hdr.size()) + int64_t(body_len);
+            if (fwrite(hdr.data(), 1, hdr.size(), f) != hdr.size())
+                return fail("write failed");
+            if (body_len &&
+                fwrite(body, 1, body_len, f) != body_len)
+                return fail("write failed");
+            off += m.total_size;
+        }
+        rgs.push_back(std::move(rg));
+    }
+
+    TW w;
+    int16_t last = 0;
+    w.i32(last, 1, 1);  // version
+    w.list_header(last, 2, 12, size_t(n_cols) + 1);
+    {
+        int16_t l2 = 0;
+        w.str(l2, 4, "schema");
+        w.i32(l2, 5, n_cols);
+        w.stop();
+    }
+    for (int c = 0; c < n_cols; c++) {
+        const WriterCol& wc = cols[c];
+        int16_t l2 = 0;
+        w.i32(l2, 1, wc.physical);
+        w.i32(l2, 3, 0);  // REQUIRED
+        w.str(l2, 4, wc.name);
+        if (wc.converted >= 0) w.i32(l2, 6, wc.converted);
+        w.stop();
+    }
+    w.i64f(last, 3, n);
+    w.list_header(last, 4, 12, rgs.size());
+    for (const auto& rg : rgs) {
+        int16_t l2 = 0;
+        w.list_header(l2, 1, 12, n_cols);
+        int64_t total = 0;
+        for (int c = 0; c < n_cols; c++) total += rg.cols[c].total_size;
+        for (int c = 0; c < n_cols; c++)
+            column_chunk(w, cols[c].name, cols[c].physical, rg.cols[c]);
+        w.i64f(l2, 2, total);
+        w.i64f(l2, 3, rg.num_rows);
+        w.stop();
+    }
+    w.str(last, 6, "horaedb-amd hx table writer");
+    w.list_header(last, 7, 12, n_cols);
+    for (int c = 0; c < n_cols; c++) {
+        int16_t l2 = 0;
+        w.field(l2, 1, 12);
+        w.stop();
+        w.stop();
+    }
+    w.stop();
+
+    uint32_t flen = (uint32_t)w.buf.size();
+    if (fwrite(w.buf.data(), 1, flen, f) != flen) return fail("write failed");
+    if (fwrite(&flen, 4, 1, f) != 1) return fail("write failed");
+    if (fwrite("PAR1", 1, 4, f) != 4) return fail("write failed");
+    if (fclose(f) != 0) return "write_table_sst: close failed";
+    return "";
+}
+
+std::string write_metric_sst_seqs(const std::string& path,
+                                  const uint64_t* series, const int64_t* ts,
+                                  const double* value, const uint64_t* seqs,
+                                  int64_t n, int64_t row_group) {
+    std::vector<uint64_t> zeros(n, 0);
+    WriterCol cols[5] = {
+        {"series_id", 2, 14, series, nullptr},
+        {"timestamp", 2, -1, ts, nullptr},
+        {"value", 5, -1, value, nullptr},
+        {"__seq__", 2, 14, seqs, nullptr},
+        {"__reserved__", 2, 14, zeros.data(), nullptr},
+    };
+    return write_table_sst(path, cols, 5, n, row_group);
+}
+
 }  // namespace hx

@@ -208,6 +208,38 @@ hx_status hx_write_sst(const char* path, const uint64_t* series,
                        const int64_t* ts, const double* value, uint64_t seq,
                        int64_t n_rows, int64_t row_group);
 
+/* ---- inverted index (RFC docs/rfcs/20240827-metric-engine.md:86-137) ----
+ * The reference's planned index tables (its metric_engine index module is
+ * an uncompiled skeleton; semantics pinned by the RFC): an `index` table
+ * with rows (MetricID u64, TagKey bytes, TagValue bytes, TSID u64), stored
+ * as Parquet SSTs under {store}/index/, rows sorted by
+ * (tag_key, tag_value, tsid). A tag-equality predicate resolves to the
+ * union of its postings across index SSTs; multiple predicates AND
+ * (intersect) or OR (union). The resulting TSID set feeds the scan's
+ * series-membership predicate (hx_scan_spec preds / hx_prepare series set),
+ * closing the rfc "tag filter -> TSID -> data" query path on GPU. */
+typedef struct {
+    const char* tag_key;      /* NUL-terminated byte strings */
+    const char* tag_value;
+} hx_tag_pred;
+
+/* Append one index SST (rows need not be pre-sorted; the writer sorts by
+ * (tag_key, tag_value, tsid) — the RFC's index PK order). */
+hx_status hx_index_write(hx_handle*, const uint64_t* metric_id,
+                         const char* const* tag_keys,
+                         const char* const* tag_values,
+                         const uint64_t* tsids, int64_t n_rows,
+                         uint64_t* out_seq);
+
+/* Evaluate tag-equality predicates on `device`: GPU BYTE_ARRAY decode +
+ * postings filter + set combine (AND = sorted intersection, OR = union).
+ * Returns the sorted distinct TSID set (callee-allocated; free with
+ * hx_tsids_free). */
+hx_status hx_index_query(hx_handle*, const hx_tag_pred* preds, size_t n_preds,
+                         int combine_and, int device, uint64_t** out_tsids,
+                         size_t* n_out);
+void hx_tsids_free(uint64_t* tsids);
+
 /* ---- introspection for bench/tests ------------------------------------ */
 typedef struct {
     double  exec_ms;          /* wall of last hx_exec_agg (HIP events)      */
