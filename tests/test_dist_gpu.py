@@ -148,6 +148,7 @@ def _rank5(rank, world, shared_dir, q):
         "count": mc.tolist(),
         "ts_range": list(ts_range),
         "n_ssts_mine": len(mine),
+        "n_ssts_cat": len(cat),
     }))
     dist.destroy_process_group()
 
@@ -167,7 +168,10 @@ def test_two_rank_config5_library(tmp_path):
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
-    assert outs[0]["n_ssts_mine"] + outs[1]["n_ssts_mine"] == 6
+    # shards partition exactly the SSTs overlapping the scan range
+    assert outs[0]["n_ssts_cat"] == outs[1]["n_ssts_cat"] > 0
+    assert (outs[0]["n_ssts_mine"] + outs[1]["n_ssts_mine"]
+            == outs[0]["n_ssts_cat"])
 
     import oracle
     from oracle.scan import AGG_SUM, AGG_COUNT
