@@ -157,7 +157,9 @@ def merge_bucket_partials(series_t, bucket_t, sum_t, cnt_t):
         return series_t, bucket_t, sum_t, cnt_t
     ob = torch.argsort(bucket_t, stable=True)
     s1, b1, v1, c1 = series_t[ob], bucket_t[ob], sum_t[ob], cnt_t[ob]
-    os_ = torch.argsort(s1, stable=True)
+    # series are u64 carried in int64 tensors: bias the sign bit so the
+    # sort order is UNSIGNED (the engine's/reference's PK order)
+    os_ = torch.argsort(s1 ^ (-2**63), stable=True)
     s2, b2, v2, c2 = s1[os_], b1[os_], v1[os_], c1[os_]
     new = torch.ones_like(s2, dtype=torch.bool)
     new[1:] = (s2[1:] != s2[:-1]) | (b2[1:] != b2[:-1])
@@ -222,15 +224,21 @@ def run_extras(args, device, opmap):
             opmap["sum"] | opmap["count"])
     except Exception as e:  # noqa: BLE001 — extras must never kill the run
         log(f"extra uncompressed failed: {e}")
-    # 2. config 3: series-set predicate @1% + min/max/avg (BASELINE configs[2])
+    # 2. config 3 driven by a TAG through the inverted index (rfc:86-137):
+    #    dc=dc7 -> 1% of the series -> min/max/avg scan (BASELINE configs[2])
     try:
+        from tools.gen_ssts import gen_tag_index
         d3, m3 = get_dataset(args, 0)
-        ids = np.load(os.path.join(d3, "series_ids.npy"))
-        k = max(1, len(ids) // 100)
-        rng = np.random.default_rng(args.seed)
-        sel = rng.choice(ids, size=k, replace=False).tolist()
-        one("config3_series_set_1pct_minmaxavg_extra", d3, m3,
-            opmap["min"] | opmap["max"] | opmap["avg"], series_in=sel)
+        gen_tag_index(d3, n_dc=100)
+        from horaedb_amd import Store
+        with Store(d3) as st_idx:
+            t_i = time.time()
+            tsids = st_idx.index_query([("dc", "dc7")], device=device)
+            index_ms = (time.time() - t_i) * 1000
+        log(f"index query dc=dc7 -> {len(tsids)} TSIDs in {index_ms:.1f}ms")
+        one("config3_tag_dc7_1pct_minmaxavg_extra", d3, m3,
+            opmap["min"] | opmap["max"] | opmap["avg"],
+            series_in=tsids.tolist())
     except Exception as e:  # noqa: BLE001
         log(f"extra config3 failed: {e}")
     # 3. ts-overlap dedup at scale: 2 generations re-write the same PKs

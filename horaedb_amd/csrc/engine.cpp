@@ -27,6 +27,7 @@
 #include <vector>
 
 #include "hx_kernels.h"
+#include "hx_internal.h"
 
 // ---------------------------------------------------------------------------
 // error plumbing
@@ -86,6 +87,13 @@ struct hx_handle {
     std::vector<CatSst> ssts;            // ascending seq
     std::vector<hx_sst_desc> find_out;   // scratch for hx_find_ssts
 };
+
+namespace hx_int {
+hx_status set_error(hx_status code, const std::string& msg) {
+    return fail(code, msg);
+}
+const std::string& store_path(hx_handle* h) { return h->store; }
+}  // namespace hx_int
 
 static hx_status read_file_meta(const std::string& path, uint64_t seq,
                                 CatSst& out) {
@@ -1292,7 +1300,11 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
     }
     const double x_est = 0.5 * (x_lo + x_hi);
     plan.range_xest = x_est;
-    double target = 700.0;   // series per block (ne=2048, load ~0.34)
+    double target = 1300.0;  // series per block (ne=1024: LDS 20.5 KB/block
+                             // => 7 blocks/CU = 28 waves; the kernel is
+                             // latency-bound, so occupancy beats table
+                             // headroom — measured 26 vs 38 ms at the 1B
+                             // shape vs ne=2048's 16 waves)
     if (const char* e = getenv("HX_RANGE_TARGET")) target = atof(e);
     uint32_t nb = 512;
     while (nb < (uint32_t)std::min(1e9, x_est * 1.3 / target) &&
@@ -1303,7 +1315,7 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
     // are cheap; an LDS-table overflow (the fallback path) is a global-RMW
     // storm (r01: 286M of 320M updates went global at the 1B shape).
     if (!getenv("HX_RANGE_TARGET"))   // explicit target: trust the caller
-        while ((double)nb * 16384.0 < staged && nb < (1u << 17)) nb <<= 1;
+        while ((double)nb * 32768.0 < staged && nb < (1u << 17)) nb <<= 1;
     std::vector<uint64_t> bounds(nb + 1);
     bounds[0] = 0;
     for (uint32_t b = 1; b < nb; b++) bounds[b] = samp[(size_t)b * m / nb];
@@ -1496,7 +1508,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
         hipEvent_t e0 = plan.ev[0], e1 = plan.ev[1];
         HIP_TRY(hipEventRecord(e0, s));
         if (use_range) {
-            uint32_t ne = 2048;
+            uint32_t ne = 1024;
             if (const char* e = getenv("HX_RANGE_NE"))
                 ne = (uint32_t)strtoul(e, nullptr, 10);
             uint32_t interp = 1;   // HX_INTERP=0 reverts to mix64 slots

@@ -503,11 +503,17 @@ __device__ __forceinline__ void lds_update(const AggParams& P, uint64_t* lkey,
                                            unsigned long long* lmax,
                                            uint32_t ne, uint64_t sv, double v,
                                            uint32_t cnt, double mn, double mx,
-                                           uint32_t i0 = 0xFFFFFFFFu) {
+                                           uint32_t i0 = 0xFFFFFFFFu,
+                                           uint64_t pre = 0,
+                                           bool has_pre = false) {
     uint32_t i = (i0 != 0xFFFFFFFFu) ? i0 : ((uint32_t)mix64(sv) & (ne - 1));
 #pragma unroll 1
     for (int probes = 0; probes < 16; probes++) {
-        uint64_t kk = lkey[i];
+        // probe prefetch: the caller read slot i0 right after the row loads,
+        // so the common-case key check pays no fresh LDS round-trip on the
+        // update critical path. A stale EMPTY only re-routes through the
+        // CAS (keys transition EMPTY->key exactly once), never corrupts.
+        uint64_t kk = (has_pre && probes == 0) ? pre : lkey[i];
         if (kk == KEY_EMPTY) {
             uint64_t old = atomicCAS(&lkey[i], KEY_EMPTY, sv);
             kk = (old == KEY_EMPTY) ? sv : old;
@@ -1158,6 +1164,20 @@ k_scan_agg_range2(AggParams P, RangeAux R) {
                         t0 = T[r0];
                         v0 = V[r0];
                     }
+                    // slot + probe prefetch: issue the LDS probe reads NOW
+                    // so their ~50-cycle latency hides under filter/dedup
+                    uint32_t iA, iB;
+                    if (R.interp) {
+                        uint32_t x0 = (uint32_t)((double)(s0 - ilo) * islope);
+                        uint32_t x1 = (uint32_t)((double)(s1 - ilo) * islope);
+                        iA = x0 < ne ? x0 : ne - 1;
+                        iB = x1 < ne ? x1 : ne - 1;
+                    } else {
+                        iA = (uint32_t)mix64(s0) & (ne - 1);
+                        iB = (uint32_t)mix64(s1) & (ne - 1);
+                    }
+                    const uint64_t preA = lkey[iA];
+                    const uint64_t preB = lkey[iB];
                     // window validity: [max(row, base), hi)
                     bool a0 = r0 >= row && r0 < hi;
                     bool a1 = r1 >= row && r1 < hi;
@@ -1214,24 +1234,19 @@ k_scan_agg_range2(AggParams P, RangeAux R) {
                     if (P.skip == 1) continue;
                     // in-lane pair merge, then direct LDS updates
                     const bool merge = a0 && a1 && (s0 == s1);
-                    uint32_t i0 = 0xFFFFFFFFu, i1 = 0xFFFFFFFFu;
-                    if (R.interp) {
-                        uint32_t x0 = (uint32_t)((double)(s0 - ilo) * islope);
-                        uint32_t x1 = (uint32_t)((double)(s1 - ilo) * islope);
-                        i0 = x0 < ne ? x0 : ne - 1;
-                        i1 = x1 < ne ? x1 : ne - 1;
-                    }
                     if (merge) {
                         lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
                                        s0, v0 + v1, 2u, fmin(v0, v1),
-                                       fmax(v0, v1), i0);
+                                       fmax(v0, v1), iA, preA, true);
                     } else {
                         if (a0)
                             lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax,
-                                           ne, s0, v0, 1u, v0, v0, i0);
+                                           ne, s0, v0, 1u, v0, v0, iA, preA,
+                                           true);
                         if (a1)
                             lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax,
-                                           ne, s1, v1, 1u, v1, v1, i1);
+                                           ne, s1, v1, 1u, v1, v1, iB, preB,
+                                           true);
                     }
                 }
                 pos++;

@@ -186,6 +186,10 @@ class Prepared:
         self.close()
 
 
+class _TagPred(C.Structure):
+    _fields_ = [("tag_key", C.c_char_p), ("tag_value", C.c_char_p)]
+
+
 class Store:
     """MI355X-native ColumnarStorage (scan side). storage.rs:76-89."""
 
@@ -305,6 +309,52 @@ class Store:
         """One-shot scan+aggregate (prepare + exec + release)."""
         with self.prepare(ts_range, series_in, devices) as p:
             return p.exec_agg(ops=ops, bucket_ms=bucket_ms)
+
+    def index_write(self, tag_keys, tag_values, tsids, metric_id=None):
+        """Append one inverted-index SST (rfc:86-137 index table): rows
+        (metric_id, tag_key, tag_value, tsid), writer-sorted by
+        (tag_key, tag_value, tsid). Returns the index file sequence."""
+        n = len(tsids)
+        assert len(tag_keys) == n and len(tag_values) == n
+        tk = (C.c_char_p * n)(*[k.encode() if isinstance(k, str) else k
+                                for k in tag_keys])
+        tv = (C.c_char_p * n)(*[v.encode() if isinstance(v, str) else v
+                                for v in tag_values])
+        tsids = np.ascontiguousarray(tsids, dtype=np.uint64)
+        mid = None
+        midp = None
+        if metric_id is not None:
+            mid = np.ascontiguousarray(metric_id, dtype=np.uint64)
+            midp = mid.ctypes.data_as(C.POINTER(C.c_uint64))
+        seq = C.c_uint64()
+        _check(_lib.hx_index_write(
+            self._h, midp, tk, tv,
+            tsids.ctypes.data_as(C.POINTER(C.c_uint64)), n, C.byref(seq)))
+        return int(seq.value)
+
+    def index_query(self, preds, combine="and", device=0):
+        """Tag-equality predicates -> sorted distinct TSID set (GPU
+        postings filter + set combine). preds: [(key, value), ...]."""
+        n = len(preds)
+        arr = (_TagPred * n)()
+        keep = []
+        for i, (k, v) in enumerate(preds):
+            kb = k.encode() if isinstance(k, str) else k
+            vb = v.encode() if isinstance(v, str) else v
+            keep += [kb, vb]
+            arr[i].tag_key = kb
+            arr[i].tag_value = vb
+        out = C.POINTER(C.c_uint64)()
+        n_out = C.c_size_t()
+        _check(_lib.hx_index_query(self._h, arr, n,
+                                   1 if combine == "and" else 0, device,
+                                   C.byref(out), C.byref(n_out)))
+        try:
+            res = np.array(_np(out, n_out.value, np.uint64), copy=True) \
+                if n_out.value else np.empty(0, np.uint64)
+        finally:
+            _lib.hx_tsids_free(out)
+        return res
 
     def write(self, series, ts, value, enable_check=True):
         """ColumnarStorage::write (storage.rs:76-89): stable PK sort + new

@@ -194,6 +194,54 @@ def gen_dataset(out_dir, n_rows, n_series, n_ssts, seed=42,
     return manifest
 
 
+def gen_tag_index(store_dir, n_dc=100):
+    """Write {store}/index/1.sst (the RFC index table, rfc:86-137) for the
+    dataset's series: tags dc=dc{i%n_dc} (1/n_dc selectivity each) and
+    env in {prod, dev}. Vectorized pyarrow write (PLAIN uncompressed,
+    row-group 8192, stats) matching the native reader's layout contract;
+    rows sorted by (tag_key, tag_value, tsid)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    idx_path = os.path.join(store_dir, "index", "1.sst")
+    if os.path.exists(idx_path):
+        return idx_path
+    os.makedirs(os.path.dirname(idx_path), exist_ok=True)
+    ids = np.load(os.path.join(store_dir, "series_ids.npy"))
+    n = len(ids)
+    # dc block: values dc0..dc{n_dc-1} in STRING sort order, tsids ascending
+    dc_strings = sorted(f"dc{k}" for k in range(n_dc))
+    rank = np.empty(n_dc, np.int64)
+    for r, sname in enumerate(dc_strings):
+        rank[int(sname[2:])] = r
+    codes = rank[np.arange(n) % n_dc]
+    order = np.lexsort((ids, codes))   # ids already ascending per group
+    dc_vals = np.array([f"dc{k}" for k in range(n_dc)], dtype=object)
+    dc_value_col = dc_vals[np.arange(n) % n_dc][order]
+    dc_tsid = ids[order]
+    # env block: dev (odd i) then prod (even i), tsids ascending per value
+    env_is_prod = (np.arange(n) % 2) == 0
+    env_order = np.lexsort((ids, env_is_prod))  # dev(False=0) first? no:
+    # lexsort ascending: False(0) < True(1) -> dev rows first... but string
+    # order is "dev" < "prod" and dev rows are the odd ones (is_prod False)
+    env_value_col = np.where(env_is_prod[env_order], "prod", "dev")
+    env_tsid = ids[env_order]
+    tbl = pa.table({
+        "metric_id": pa.array(np.zeros(2 * n, np.uint64), pa.uint64()),
+        "tag_key": pa.array([b"dc"] * n + [b"env"] * n, pa.binary()),
+        "tag_value": pa.array(
+            [x.encode() for x in dc_value_col] +
+            [x.encode() for x in env_value_col], pa.binary()),
+        "tsid": pa.array(np.concatenate([dc_tsid, env_tsid]), pa.uint64()),
+    })
+    pq.write_table(tbl, idx_path, row_group_size=8192, compression="NONE",
+                   use_dictionary=False, data_page_version="1.0",
+                   column_encoding={c: "PLAIN" for c in
+                                    ("metric_id", "tag_key", "tag_value",
+                                     "tsid")},
+                   write_statistics=True)
+    return idx_path
+
+
 def middle_range(manifest, frac=0.5):
     """The benchmark's ts-range: middle `frac` of the dataset span."""
     span = manifest["ts_end"] - manifest["ts_start"]
