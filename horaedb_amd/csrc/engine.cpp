@@ -30,6 +30,38 @@
 #include "hx_internal.h"
 
 // ---------------------------------------------------------------------------
+// Zstd page codec (config.rs:84 includes Zstd): decompressed on the HOST at
+// staging time into the page blob — the same place SURVEY §7 allows for
+// inherently serial codecs — so the kernels see raw PLAIN pages. The image
+// ships libzstd.so.1 without a dev header; dlopen with self-declared
+// prototypes (stable ZSTD_* ABI). Fails loudly when the library is absent.
+// ---------------------------------------------------------------------------
+#include <dlfcn.h>
+
+namespace {
+typedef size_t (*zstd_decompress_fn)(void*, size_t, const void*, size_t);
+typedef unsigned (*zstd_iserror_fn)(size_t);
+
+struct ZstdLib {
+    zstd_decompress_fn decompress = nullptr;
+    zstd_iserror_fn is_error = nullptr;
+    bool ok = false;
+    ZstdLib() {
+        void* h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_LOCAL);
+        if (!h) return;
+        decompress = (zstd_decompress_fn)dlsym(h, "ZSTD_decompress");
+        is_error = (zstd_iserror_fn)dlsym(h, "ZSTD_isError");
+        ok = decompress && is_error;
+    }
+};
+
+const ZstdLib& zstd() {
+    static ZstdLib z;
+    return z;
+}
+}  // namespace
+
+// ---------------------------------------------------------------------------
 // error plumbing
 // ---------------------------------------------------------------------------
 static thread_local std::string g_last_error;
@@ -59,6 +91,7 @@ struct ChunkRef {
     bool required = true;
     int64_t chunk_start = 0;
     int64_t comp_size = 0;
+    int64_t uncomp_size = 0;
     int64_t num_values = 0;
 };
 
@@ -160,6 +193,7 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
             cr.cols[k].required = m.columns[ci[k]].required;
             cr.cols[k].chunk_start = cc.chunk_start();
             cr.cols[k].comp_size = cc.total_compressed_size;
+            cr.cols[k].uncomp_size = cc.total_uncompressed_size;
             cr.cols[k].num_values = cc.num_values;
         }
         const auto& secc = rg.columns[ci[0]];
@@ -390,7 +424,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
         StagedSst* ss;
         int rg_cat;         // catalog rg index
         int col;            // 0 series 1 ts 2 value
-        int64_t chunk_start, comp_size, num_values;
+        int64_t chunk_start, comp_size, uncomp_size, num_values;
         int32_t codec;
         bool required;
         size_t dst_off;     // blob offset reserved (chunk_size upper bound)
@@ -439,11 +473,17 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 j.col = c;
                 j.chunk_start = cr.chunk_start;
                 j.comp_size = cr.comp_size;
+                j.uncomp_size = cr.uncomp_size;
                 j.num_values = cr.num_values;
                 j.codec = cr.codec;
                 j.required = cr.required;
                 j.dst_off = blob_off;
-                blob_off = align64(blob_off + size_t(cr.comp_size));
+                // Zstd chunks are decompressed in place at staging: reserve
+                // the larger of the on-disk and decompressed sizes
+                size_t reserve = size_t(cr.comp_size);
+                if (cr.codec == hx::CODEC_ZSTD)
+                    reserve = std::max(reserve, size_t(cr.uncomp_size));
+                blob_off = align64(blob_off + reserve);
                 jobs.push_back(j);
             }
             row_base += rg.n_rows;
@@ -537,10 +577,18 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 break;
             }
             if (j.codec != hx::CODEC_UNCOMPRESSED &&
-                j.codec != hx::CODEC_SNAPPY) {
+                j.codec != hx::CODEC_SNAPPY &&
+                j.codec != hx::CODEC_ZSTD) {
                 std::lock_guard<std::mutex> g(mu);
-                err_msg = j.ss->cat->path + ": codec unsupported in round 1 "
-                          "(uncompressed and Snappy only)";
+                err_msg = j.ss->cat->path + ": codec unsupported "
+                          "(uncompressed, Snappy and Zstd)";
+                err_flag = 1;
+                break;
+            }
+            if (j.codec == hx::CODEC_ZSTD && !zstd().ok) {
+                std::lock_guard<std::mutex> g(mu);
+                err_msg = j.ss->cat->path + ": Zstd pages but libzstd.so.1 "
+                          "is not loadable";
                 err_flag = 1;
                 break;
             }
@@ -560,10 +608,24 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
             }
             size_t payload = size_t(dp->compressed_size) - size_t(dp->def_level_bytes);
             const uint8_t* src = tmp.data() + in_chunk + dp->def_level_bytes;
-            std::memcpy(plan.h_blob + j.dst_off, src, payload);
-
             uint64_t data_off = j.dst_off;  // raw page bytes (post-codec)
             size_t raw_size = payload;
+            if (j.codec == hx::CODEC_ZSTD && dp->is_compressed) {
+                // host-side decompress into the blob (staging, untimed)
+                size_t want = size_t(dp->uncompressed_size) -
+                              size_t(dp->def_level_bytes);
+                size_t got = zstd().decompress(plan.h_blob + j.dst_off, want,
+                                               src, payload);
+                if (zstd().is_error(got) || got != want) {
+                    std::lock_guard<std::mutex> g(mu);
+                    err_msg = j.ss->cat->path + ": Zstd page decode failed";
+                    err_flag = 1;
+                    break;
+                }
+                raw_size = want;
+            } else {
+                std::memcpy(plan.h_blob + j.dst_off, src, payload);
+            }
             if (j.codec == hx::CODEC_SNAPPY && dp->is_compressed) {
                 std::lock_guard<std::mutex> g(mu);
                 hx::SnappyPageDesc sp{};
@@ -610,18 +672,38 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 hx::RleDictPageDesc rd{};
                 size_t dict_in_chunk = size_t(dictp->payload_off - j.chunk_start);
                 size_t dict_payload = size_t(dictp->compressed_size);
-                // re-copy the dictionary payload right after the data payload
-                size_t dict_dst = j.dst_off + ((payload + 15) & ~size_t(15));
-                if (dict_dst + dict_payload >
-                    j.dst_off + size_t(j.comp_size) + 64) {
+                // re-copy the dictionary payload right after the data
+                // payload (raw_size >= payload when Zstd decompressed it)
+                size_t reserve = std::max(size_t(j.comp_size),
+                                          j.codec == hx::CODEC_ZSTD
+                                              ? size_t(j.uncomp_size)
+                                              : size_t(0));
+                size_t dict_dst = j.dst_off + ((raw_size + 15) & ~size_t(15));
+                size_t dict_need = j.codec == hx::CODEC_ZSTD
+                                       ? size_t(dictp->uncompressed_size)
+                                       : dict_payload;
+                if (dict_dst + dict_need > j.dst_off + reserve + 64) {
                     err_msg = j.ss->cat->path + ": dictionary staging overflow";
                     err_flag = 1;
                     break;
                 }
-                std::memcpy(plan.h_blob + dict_dst, tmp.data() + dict_in_chunk,
-                            dict_payload);
                 uint64_t dict_data_off = dict_dst;
-                uint32_t dict_raw = (uint32_t)dict_payload;
+                uint32_t dict_raw;
+                if (j.codec == hx::CODEC_ZSTD && dictp->is_compressed) {
+                    size_t got = zstd().decompress(
+                        plan.h_blob + dict_dst, dict_need,
+                        tmp.data() + dict_in_chunk, dict_payload);
+                    if (zstd().is_error(got) || got != dict_need) {
+                        err_msg = j.ss->cat->path + ": Zstd dict decode failed";
+                        err_flag = 1;
+                        break;
+                    }
+                    dict_raw = (uint32_t)dict_need;
+                } else {
+                    std::memcpy(plan.h_blob + dict_dst,
+                                tmp.data() + dict_in_chunk, dict_payload);
+                    dict_raw = (uint32_t)dict_payload;
+                }
                 if (j.codec == hx::CODEC_SNAPPY) {
                     hx::SnappyPageDesc sp{};
                     sp.src_off = dict_dst;

@@ -478,12 +478,24 @@ def test_unsupported_codec_fails_loudly(tmp_path):
     n = 10_000
     write_sst(os.path.join(store, "data", "1.sst"),
               np.arange(n, dtype=np.uint64), np.arange(n, dtype=np.int64),
-              np.ones(n), 1, compression="zstd")
+              np.ones(n), 1, compression="gzip")
     with Store(store) as st:
         with pytest.raises(HxError) as ei:
             st.scan_agg((0, 10**9), devices=[0])
         assert ei.value.code == 3  # HX_ERR_UNSUPPORTED
         assert "codec" in str(ei.value)
+
+
+def test_zstd_pages_parity(tmp_path):
+    # Zstd (config.rs:84): pages are decompressed on the HOST at staging
+    # (DESIGN §2 — the codec is inherently serial; staging is untimed),
+    # kernels then scan the raw pages. Full parity vs the oracle.
+    from tools.gen_ssts import gen_dataset, middle_range
+    out = str(tmp_path / "z")
+    m = gen_dataset(out, n_rows=60_000, n_series=600, n_ssts=3, seed=33,
+                    compression="zstd")
+    check_parity(out, middle_range(m),
+                 ops=AGG_SUM | AGG_COUNT | AGG_MIN | AGG_MAX)
 
 
 def test_series_set_unsorted_with_absent_ids(ds_plain):

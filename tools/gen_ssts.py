@@ -257,7 +257,7 @@ def main():
     p.add_argument("--series", type=int, default=100_000)
     p.add_argument("--ssts", type=int, default=1)
     p.add_argument("--seed", type=int, default=42)
-    p.add_argument("--compression", default="none", choices=["none", "snappy"])
+    p.add_argument("--compression", default="none", choices=["none", "snappy", "zstd"])
     p.add_argument("--ts-encoding", default="PLAIN",
                    choices=["PLAIN", "DELTA_BINARY_PACKED"])
     p.add_argument("--workers", type=int, default=1)
