@@ -1785,11 +1785,22 @@ __device__ __forceinline__ uint32_t snappy_varint(const uint8_t* p,
     }
 }
 
+// Per-wave LDS mirror of the most recent output bytes: back-references with
+// offset <= SNAP_MIRROR read the mirror (LDS, wave-ordered) instead of the
+// global dst — the old kernel drained ALL outstanding stores
+// (`s_waitcnt vmcnt(0)`, ~600-900 cycles) before EVERY match, and the
+// ts/series pages of PLAIN metric data are nothing but small periodic
+// matches (measured 94 ms/step at the 1B snappy shape, ~4x the aggregate
+// kernel). Offsets beyond the mirror still take the drained global path.
+#define SNAP_MIRROR 4096u
+
 extern "C" __global__ void __launch_bounds__(256)
 k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     const SnappyPageDesc* __restrict__ pages, uint32_t n_pages,
                     unsigned long long* err_flag) {
+    __shared__ uint8_t mirror_all[4][SNAP_MIRROR];
     const uint32_t lane = threadIdx.x & 63;
+    uint8_t* const mir = mirror_all[(threadIdx.x >> 6) & 3];
     const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
     for (uint32_t pg = wave; pg < n_pages; pg += n_waves) {
@@ -1820,8 +1831,15 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     pos += nb;
                 }
                 if (pos + len > clen || d + len > ulen) { err = 1; break; }
-                for (uint32_t i = lane; i < len; i += 64)
-                    dst[d + i] = src[pos + i];
+                // only the LAST <= SNAP_MIRROR bytes matter for the ring
+                const uint32_t mstart = len > SNAP_MIRROR
+                                            ? len - SNAP_MIRROR : 0u;
+                for (uint32_t i = lane; i < len; i += 64) {
+                    const uint8_t b = src[pos + i];
+                    dst[d + i] = b;
+                    if (i >= mstart)
+                        mir[(d + i) & (SNAP_MIRROR - 1)] = b;
+                }
                 pos += len;
                 d += len;
             } else {
@@ -1845,15 +1863,35 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     pos += 4;
                 }
                 if (off == 0 || off > d || d + len > ulen) { err = 1; break; }
-                // drain our own outstanding stores so lanes can read bytes
-                // written by other lanes of this wave (L2-coherent)
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-                if (off >= len) {
-                    for (uint32_t i = lane; i < len; i += 64)
-                        dst[d + i] = dst[d - off + i];
+                if (off + 64u <= SNAP_MIRROR) {
+                    // source entirely within the mirror ring (the 64-byte
+                    // margin keeps this element's ring writes from aliasing
+                    // its own source slots across strided iterations — max
+                    // match len is 64). The (i % off) form reads only
+                    // PRE-element bytes, so there is no intra-element
+                    // hazard; prior elements' ds_writes are ordered before
+                    // these ds_reads within the wave (drain lgkmcnt to keep
+                    // the compiler from reordering them).
+                    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                    for (uint32_t i = lane; i < len; i += 64) {
+                        const uint32_t sidx = (off >= len)
+                                                  ? d - off + i
+                                                  : d - off + (i % off);
+                        const uint8_t b = mir[sidx & (SNAP_MIRROR - 1)];
+                        dst[d + i] = b;
+                        mir[(d + i) & (SNAP_MIRROR - 1)] = b;
+                    }
                 } else {
-                    for (uint32_t i = lane; i < len; i += 64)
-                        dst[d + i] = dst[d - off + (i % off)];
+                    // far back-reference: drain outstanding stores so lanes
+                    // can read bytes written by other lanes (L2-coherent)
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                    for (uint32_t i = lane; i < len; i += 64) {
+                        const uint8_t b = (off >= len)
+                                              ? dst[d - off + i]
+                                              : dst[d - off + (i % off)];
+                        dst[d + i] = b;
+                        mir[(d + i) & (SNAP_MIRROR - 1)] = b;
+                    }
                 }
                 d += len;
             }
