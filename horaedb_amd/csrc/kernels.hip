@@ -850,7 +850,8 @@ __device__ __forceinline__ void scan_window_range(
     const AggParams& P, const RgDesc& rg, const SstDev& sst,
     const uint64_t* S, const int64_t* T, const double* V, uint32_t base,
     uint32_t hi, uint32_t n_true, int lane, unsigned long long& my_matched,
-    WinResult& W) {
+    WinResult& W, const uint64_t* lkey, uint32_t ne, uint32_t interp,
+    uint64_t ilo, double islope) {
     const uint32_t r = base + (uint32_t)lane;
     const bool inb = r < hi;
     const bool has_next = r + 1 < n_true;
@@ -868,6 +869,19 @@ __device__ __forceinline__ void scan_window_range(
             s1 = S[r + 1];
             t1 = T[r + 1];
         }
+    }
+    // probe prefetch: the head lane's LDS slot is known from `s` alone —
+    // issue the probe read NOW so its latency hides under filter/dedup
+    {
+        uint32_t slot;
+        if (interp) {
+            uint32_t x = (uint32_t)((double)(s - ilo) * islope);
+            slot = x < ne ? x : ne - 1;
+        } else {
+            slot = (uint32_t)mix64(s) & (ne - 1);
+        }
+        W.hint_i = slot;
+        W.hint_k = lkey[slot];
     }
     if (inb) {
         alive = (t >= P.ts_lo) & (t < P.ts_hi);
@@ -1018,27 +1032,22 @@ k_scan_agg_range(AggParams P, RangeAux R) {
                     B.c = 0;
                     B.head = false;
                     scan_window_range<MM>(P, rg, sst, S, T, V, base, hi,
-                                          rg.n_rows, lane, my_matched, A);
+                                          rg.n_rows, lane, my_matched, A,
+                                          lkey, ne, R.interp, ilo, islope);
                     if (base + 64 < hi)
                         scan_window_range<MM>(P, rg, sst, S, T, V, base + 64,
                                               hi, rg.n_rows, lane, my_matched,
-                                              B);
+                                              B, lkey, ne, R.interp, ilo,
+                                              islope);
                     if (P.skip == 1) continue;
-                    uint32_t iA = 0xFFFFFFFFu, iB = 0xFFFFFFFFu;
-                    if (R.interp) {
-                        uint32_t a2 = (uint32_t)((double)(A.s - ilo) * islope);
-                        uint32_t b2 = (uint32_t)((double)(B.s - ilo) * islope);
-                        iA = a2 < ne ? a2 : ne - 1;
-                        iB = b2 < ne ? b2 : ne - 1;
-                    }
                     if (A.head && A.c > 0)
                         lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
                                        A.s, A.vv, (uint32_t)A.c, A.mn, A.mx,
-                                       iA);
+                                       A.hint_i, A.hint_k, true);
                     if (B.head && B.c > 0)
                         lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
                                        B.s, B.vv, (uint32_t)B.c, B.mn, B.mx,
-                                       iB);
+                                       B.hint_i, B.hint_k, true);
                 }
                 pos++;
                 row = 0;
