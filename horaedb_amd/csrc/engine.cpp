@@ -99,7 +99,8 @@ struct CatRg {
     int64_t n_rows = 0;
     int64_t ts_min = 0, ts_max = 0;
     bool has_ts_stats = false;
-    ChunkRef cols[3];  // series, ts, value
+    ChunkRef cols[4];  // series, ts, value, __seq__ (staged only if mixed)
+    bool seq_mixed = false;  // __seq__ stats absent or min != max
 };
 
 struct CatSst {
@@ -109,6 +110,7 @@ struct CatSst {
     int64_t ts_min = 0, ts_max = 0;
     bool series_stats_ok = true;    // every chunk has series stats
     uint64_t series_max = 0;        // max over chunks (unsigned order)
+    bool seq_mixed = false;         // any rg with non-constant __seq__
     std::vector<CatRg> rgs;
 };
 
@@ -164,12 +166,13 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
 
     // schema contract (types.rs:150-240): series_id/timestamp/value present,
     // INT64/INT64/DOUBLE physical
-    int ci[3] = {-1, -1, -1};
+    int ci[4] = {-1, -1, -1, -1};
     for (size_t i = 0; i < m.columns.size(); i++) {
         const auto& c = m.columns[i];
         if (c.name == "series_id") ci[0] = (int)i;
         else if (c.name == "timestamp") ci[1] = (int)i;
         else if (c.name == "value") ci[2] = (int)i;
+        else if (c.name == "__seq__") ci[3] = (int)i;
     }
     if (ci[0] < 0 || ci[1] < 0 || ci[2] < 0)
         return fail(HX_ERR_SCHEMA, path + ": metric schema columns missing");
@@ -185,7 +188,8 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
     for (const auto& rg : m.row_groups) {
         CatRg cr;
         cr.n_rows = rg.num_rows;
-        for (int k = 0; k < 3; k++) {
+        const int n_cols_here = ci[3] >= 0 ? 4 : 3;
+        for (int k = 0; k < n_cols_here; k++) {
             if ((size_t)ci[k] >= rg.columns.size())
                 return fail(HX_ERR_FORMAT, path + ": column chunk missing");
             const auto& cc = rg.columns[ci[k]];
@@ -195,6 +199,15 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
             cr.cols[k].comp_size = cc.total_compressed_size;
             cr.cols[k].uncomp_size = cc.total_uncompressed_size;
             cr.cols[k].num_values = cc.num_values;
+        }
+        // per-row __seq__ detection (keep_builtin compaction outputs,
+        // executor.rs:155-222): constant iff stats prove min == max
+        if (ci[3] >= 0) {
+            const auto& qcc = rg.columns[ci[3]];
+            cr.seq_mixed = !(qcc.has_stats && qcc.stat_min.size() == 8 &&
+                             qcc.stat_max.size() == 8 &&
+                             qcc.stat_min == qcc.stat_max);
+            if (cr.seq_mixed) out.seq_mixed = true;
         }
         const auto& secc = rg.columns[ci[0]];
         if (secc.has_stats && secc.stat_max.size() == 8) {
@@ -441,12 +454,21 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
         sd.rank = ss->rank;
         sd.dense_series = 0;
         sd.dense_ts = 0;
+        sd.seq = ss->cat->seq;
+        sd.dense_seq = 0;
         sd.n_staged = ss->staged_rows;
         bool need_dense = ss->cluster >= 0;
+        // per-row seqs are only consulted by the cross-SST dedup, so they
+        // are staged only for overlap-cluster members with mixed seqs
+        bool need_seq = need_dense && ss->cat->seq_mixed;
         if (need_dense) {
             sd.dense_series = hx::OFF_DEC | dec_off;
             dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
             sd.dense_ts = hx::OFF_DEC | dec_off;
+            dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
+        }
+        if (need_seq) {
+            sd.dense_seq = hx::OFF_DEC | dec_off;
             dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
         }
         ss->dev_slot = (int)plan.ssts.size();
@@ -465,7 +487,8 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
             if (prev_rg_desc >= 0) plan.rgs[prev_rg_desc].next_rg = this_desc;
             prev_rg_desc = this_desc;
             plan.rgs.push_back(rd);
-            for (int c = 0; c < 3; c++) {
+            const int n_stage_cols = need_seq ? 4 : 3;
+            for (int c = 0; c < n_stage_cols; c++) {
                 const ChunkRef& cr = rg.cols[c];
                 PageJob j;
                 j.ss = ss;
@@ -752,14 +775,24 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     size_t rg_i = 0;
     for (StagedSst* ss : members) {
         hx::SstDev& sd = plan.ssts[ss->dev_slot];
+        const int n_staged_cols = sd.dense_seq ? 4 : 3;
         int64_t row_base = 0;
         for (size_t k = 0; k < ss->rg_idx.size(); k++, rg_i++) {
             hx::RgDesc& rd = plan.rgs[rg_i];
-            uint64_t offs[3];
-            for (int c = 0; c < 3; c++, job_i++) offs[c] = jobs[job_i].final_off;
+            uint64_t offs[4] = {0, 0, 0, 0};
+            for (int c = 0; c < n_staged_cols; c++, job_i++)
+                offs[c] = jobs[job_i].final_off;
             rd.series_off = offs[0];
             rd.ts_off = offs[1];
             rd.val_off = offs[2];
+            if (sd.dense_seq) {
+                hx::CopyDesc cq{};
+                cq.src_off = offs[3];
+                cq.dst_off = (sd.dense_seq & hx::OFF_MASK) +
+                             uint64_t(row_base) * 8;
+                cq.n_values = rd.n_rows;
+                plan.copies.push_back(cq);
+            }
             if (ss->cluster >= 0) {
                 // dense (series, ts) arrays for binary-search dedup
                 hx::CopyDesc cs{};
@@ -1601,11 +1634,14 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                            plan.d_sst_rg_cnt,  (uint32_t)plan.ssts.size(),
                            plan.range_nblocks, ne,
                            interp};
-            // HX_RANGE2: pair-load variant (16 B/lane dwordx4, in-lane
-            // pair merge, no cross-lane pre-reduce)
+            // HX_RANGE2=1: pair-load variant (16 B/lane dwordx4, in-lane
+            // pair merge, no cross-lane pre-reduce). Default OFF: the
+            // shuffle-pre-reduce kernel measured faster at the 1B shape
+            // (23.6-24.2 vs 25.4 ms same-box) — fewer LDS update calls
+            // beat the cheaper scan (its skip1 floor is 5.7 vs 2.5 ms).
             const bool r2k = [] {
                 const char* e = getenv("HX_RANGE2");
-                return e ? atoi(e) != 0 : true;
+                return e ? atoi(e) != 0 : false;
             }();
             hipError_t re2 =
                 r2k ? hx::launch_scan_agg_range2(
@@ -2309,6 +2345,147 @@ extern "C" hx_status hx_compact(hx_handle* h, hx_time_range range,
     for (size_t i = 0; i < h->ssts.size(); i++) {
         if (in_set[i]) {
             unlink(h->ssts[i].path.c_str());  // best-effort, like the reference
+            continue;
+        }
+        kept.push_back(std::move(h->ssts[i]));
+    }
+    kept.push_back(std::move(fresh));
+    std::sort(kept.begin(), kept.end(),
+              [](const CatSst& a, const CatSst& b) { return a.seq < b.seq; });
+    h->ssts = std::move(kept);
+    *out_new_seq = new_seq;
+    return HX_OK;
+}
+
+extern "C" hx_status hx_compact_files(hx_handle* h,
+                                      const uint64_t* input_seqs,
+                                      size_t n_inputs,
+                                      const hx_device_set* devs,
+                                      uint64_t* out_new_seq) {
+    // Executor general case (executor.rs:155-222, Task{inputs}
+    // compaction/mod.rs:26-36): re-run the scan over ONLY the named input
+    // files (keep_builtin=true) and rewrite them as one SST that preserves
+    // per-row __seq__ — correct for ANY input set, because rows shadowed by
+    // files outside the set keep losing future merges through their
+    // retained sequences.
+    if (!h || !input_seqs || n_inputs == 0 || !out_new_seq)
+        return fail(HX_ERR_INVALID, "bad argument");
+    *out_new_seq = 0;
+    std::vector<char> in_set(h->ssts.size(), 0);
+    for (size_t k = 0; k < n_inputs; k++) {
+        bool found = false;
+        for (size_t i = 0; i < h->ssts.size(); i++)
+            if (h->ssts[i].seq == input_seqs[k]) {
+                in_set[i] = 1;
+                found = true;
+            }
+        if (!found)
+            return fail(HX_ERR_INVALID,
+                        "input seq not in catalog: " +
+                            std::to_string(input_seqs[k]));
+    }
+    std::vector<hx_sst_desc> inputs;
+    uint64_t max_seq = 0;
+    for (size_t i = 0; i < h->ssts.size(); i++) {
+        if (in_set[i]) inputs.push_back({h->ssts[i].path.c_str(),
+                                         h->ssts[i].seq});
+        max_seq = std::max(max_seq, h->ssts[i].seq);
+    }
+
+    hx_scan_spec spec{};
+    spec.range = {INT64_MIN, INT64_MAX};
+    spec.ssts = inputs.data();
+    spec.n_ssts = inputs.size();
+    int32_t dev0 = (devs && devs->device_ids && devs->n_devices > 0)
+                       ? devs->device_ids[0] : 0;
+    hx_device_set one{&dev0, 1};
+    hx_prepared* P = nullptr;
+    hx_status st = hx_prepare(h, &spec, &one, &P);
+    if (st != HX_OK) return st;
+    std::unique_ptr<hx_prepared, void (*)(hx_prepared*)> guard(
+        P, hx_prepared_free);
+    DevPlan& plan = P->plans[0];
+    HIP_TRY(hipSetDevice(plan.device));
+    hipStream_t s = plan.stream;
+    st = ensure_decoded(plan, s);
+    if (st != HX_OK) return st;
+
+    const uint64_t cap = (uint64_t)plan.rows_scanned;
+    if (cap == 0) return HX_OK;
+    if (cap > 0xFFFFFFFFull)
+        return fail(HX_ERR_UNSUPPORTED,
+                    "row streams above 2^32 rows per call: split the range");
+    size_t need = cap * 8 * 6 + cap * 4 * 3 + 64;
+    st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
+    if (st != HX_OK) return st;
+    uint8_t* base = (uint8_t*)plan.d_scratch;
+    auto carve8 = [&](size_t count) {
+        uint8_t* p = base;
+        base += (count * 8 + 7) & ~size_t(7);
+        return p;
+    };
+    uint64_t* d_series = (uint64_t*)carve8(cap);
+    long long* d_ts = (long long*)carve8(cap);
+    double* d_val = (double*)carve8(cap);
+    uint64_t* d_seq = (uint64_t*)carve8(cap);
+    uint64_t* d_keys = (uint64_t*)carve8(cap);
+    uint64_t* d_keys_out = (uint64_t*)carve8(cap);
+    unsigned long long* d_cursor = (unsigned long long*)carve8(1);
+    uint32_t* perm_a = (uint32_t*)base; base += cap * 4;
+    uint32_t* perm_b = (uint32_t*)base; base += cap * 4;
+    uint32_t* perm_c = (uint32_t*)base; base += cap * 4;
+
+    HIP_TRY(hipMemsetAsync(d_cursor, 0, 8, s));
+    hx::AggParams A = base_params(P, plan);
+    HIP_TRY(hx::launch_scan_rows(s, A, 0, A.n_rgs, d_series, d_ts, d_val,
+                                 d_cursor, cap, d_seq));
+    unsigned long long n64 = 0;
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(&n64, d_cursor, 8, hipMemcpyDeviceToHost));
+    if (n64 > cap) return fail(HX_ERR_HIP, "compact row buffer overflow");
+    const uint32_t n = (uint32_t)n64;
+    if (n == 0) return HX_OK;
+
+    HIP_TRY(hx::launch_iota(s, perm_a, n));
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_ts, perm_a,
+                                  (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_a, perm_b, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_series,
+                                  perm_b, (unsigned long long*)d_keys, n));
+    HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_b, perm_c, n,
+                               &plan.d_sort_temp, &plan.sort_temp_cap));
+    const uint32_t* perm = perm_c;
+
+    std::vector<uint64_t> host(4 * size_t(n));
+    unsigned long long* d_dst = nullptr;
+    HIP_TRY(hipMalloc((void**)&d_dst, 4 * size_t(n) * 8));
+    const unsigned long long* srcs[4] = {
+        (const unsigned long long*)d_series, (const unsigned long long*)d_ts,
+        (const unsigned long long*)d_val, (const unsigned long long*)d_seq};
+    HIP_TRY(hx::launch_gather_multi(s, srcs, 4, perm, d_dst, n));
+    HIP_TRY(hipMemcpyAsync(host.data(), d_dst, 4 * size_t(n) * 8,
+                           hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    (void)hipFree(d_dst);
+
+    const uint64_t new_seq = max_seq + 1;  // fresh file id (sst.rs:39-46)
+    std::string out_path = h->store + "/data/" + std::to_string(new_seq) +
+                           ".sst";
+    std::string werr = hx::write_metric_sst_seqs(
+        out_path, host.data(), (const int64_t*)(host.data() + n),
+        (const double*)(host.data() + 2 * size_t(n)),
+        host.data() + 3 * size_t(n), n, 8192);
+    if (!werr.empty()) return fail(HX_ERR_IO, werr);
+
+    CatSst fresh;
+    st = read_file_meta(out_path, new_seq, fresh);
+    if (st != HX_OK) return st;
+    std::vector<CatSst> kept;
+    for (size_t i = 0; i < h->ssts.size(); i++) {
+        if (in_set[i]) {
+            unlink(h->ssts[i].path.c_str());
             continue;
         }
         kept.push_back(std::move(h->ssts[i]));

@@ -38,6 +38,11 @@ struct SstDev {
     uint64_t dense_series;   // byte offset (dec blob) of staged series array
     uint64_t dense_ts;       // byte offset (dec blob) of staged ts array
     int64_t  n_staged;       // staged rows (dense array length)
+    // per-row sequences (keep_builtin compaction outputs,
+    // executor.rs:155-222): a file whose __seq__ column is NOT constant
+    // carries dense_seq (staged per-row seqs); 0 = constant (seq below).
+    uint64_t seq;            // the file sequence (constant-seq files)
+    uint64_t dense_seq;      // byte offset (dec blob) of per-row seqs, or 0
 };
 
 // Overlap cluster: member SSTs sorted by ascending seq; members[] indexes

@@ -47,7 +47,8 @@ hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
                             uint32_t rg_first, uint32_t rg_last,
                             uint64_t* out_series, long long* out_ts,
                             double* out_value, unsigned long long* cursor,
-                            unsigned long long cap);
+                            unsigned long long cap,
+                            uint64_t* out_seq = nullptr);
 hipError_t launch_gather_multi(hipStream_t s,
                                const unsigned long long* const* srcs,
                                uint32_t n_arrays, const uint32_t* perm,

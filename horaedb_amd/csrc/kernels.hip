@@ -58,15 +58,20 @@ __device__ __forceinline__ bool sset_has(const AggParams& P, uint64_t s) {
     return false;
 }
 
-// Cross-SST dedup: is row (s,t) of SST `me` shadowed by an
-// equal PK in a higher-seq SST of the same ts-overlap cluster? SSTs are
-// PK-sorted (storage.rs:244-256), so a binary search per higher-seq member.
+// Cross-SST dedup: is row (s,t) of SST `me` (row sequence
+// my_seq) shadowed by an equal PK with a HIGHER sequence in another SST of
+// the same ts-overlap cluster? SSTs are PK-sorted (storage.rs:244-256), so
+// a binary search per candidate member. Constant-seq members below my_seq
+// are pruned outright; mixed-seq members (keep_builtin compaction outputs)
+// compare the matched row's own __seq__ (read.rs:289-343 orders by
+// (pk..., __seq__) read from the row).
 __device__ bool shadowed(const AggParams& P, const SstDev& me,
-                         uint64_t s, int64_t t) {
+                         uint64_t my_seq, uint64_t s, int64_t t) {
     ClusterDev c = P.clusters[me.cluster];
     for (int32_t j = c.first; j < c.first + c.n; ++j) {
         const SstDev o = P.ssts[P.cluster_members[j]];
-        if (o.rank <= me.rank) continue;
+        if (o.rank == me.rank) continue;
+        if (!o.dense_seq && o.seq <= my_seq) continue;
         const uint64_t* S = (const uint64_t*)(P.dec + (o.dense_series & OFF_MASK));
         const int64_t* T = (const int64_t*)(P.dec + (o.dense_ts & OFF_MASK));
         int64_t lo = 0, hi = o.n_staged;
@@ -76,9 +81,24 @@ __device__ bool shadowed(const AggParams& P, const SstDev& me,
             if (sm < s || (sm == s && T[mid] < t)) lo = mid + 1;
             else hi = mid;
         }
-        if (lo < o.n_staged && S[lo] == s && T[lo] == t) return true;
+        if (lo < o.n_staged && S[lo] == s && T[lo] == t) {
+            uint64_t o_seq = o.dense_seq
+                ? ((const uint64_t*)(P.dec + (o.dense_seq & OFF_MASK)))[lo]
+                : o.seq;
+            if (o_seq > my_seq) return true;
+        }
     }
     return false;
+}
+
+// row sequence of staged row (rg.row_base + r) of `sst`
+__device__ __forceinline__ uint64_t row_seq(const AggParams& P,
+                                            const SstDev& sst,
+                                            const RgDesc& rg, uint32_t r) {
+    return sst.dense_seq
+        ? ((const uint64_t*)(P.dec + (sst.dense_seq & OFF_MASK)))
+              [rg.row_base + r]
+        : sst.seq;
 }
 
 // Open-addressing claim + update. state: 0 empty / 1 claiming / 2 ready.
@@ -299,7 +319,8 @@ __device__ __forceinline__ bool row_alive(const AggParams& P, const RgDesc& rg,
             int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
             dup = (s2 == s) & (t2 == t);
         }
-        if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+        if (!dup && sst.cluster >= 0)
+            dup = shadowed(P, sst, row_seq(P, sst, rg, r), s, t);
         alive = !dup;
     }
     return alive;
@@ -366,7 +387,8 @@ __device__ __forceinline__ void scan_window(const AggParams& P,
                 int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
                 dup = (s2 == s) & (t2 == t);
             }
-            if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+            if (!dup && sst.cluster >= 0)
+            dup = shadowed(P, sst, row_seq(P, sst, rg, r), s, t);
             alive = !dup;
         }
         if (!alive) v = 0.0;
@@ -688,8 +710,14 @@ k_scan_agg_gang(const GangParams* __restrict__ gp,
                         }
                         if (!dup) {
                             const SstDev sst = ssts_all[sstA];
-                            if (sst.cluster >= 0)
-                                dup = shadowed(*Pm, sst, sv, tv);
+                            if (sst.cluster >= 0) {
+                                uint64_t msq = sst.dense_seq
+                                    ? ((const uint64_t*)(dec +
+                                           (sst.dense_seq & OFF_MASK)))
+                                          [ldesc[u].row_base + r]
+                                    : sst.seq;
+                                dup = shadowed(*Pm, sst, msq, sv, tv);
+                            }
                         }
                         alive = !dup;
                     }
@@ -897,7 +925,8 @@ __device__ __forceinline__ void scan_window_range(
                 int64_t t2 = *(const int64_t*)hx_ptr(P.blob, P.dec, nx.ts_off);
                 dup = (s2 == s) & (t2 == t);
             }
-            if (!dup && sst.cluster >= 0) dup = shadowed(P, sst, s, t);
+            if (!dup && sst.cluster >= 0)
+            dup = shadowed(P, sst, row_seq(P, sst, rg, r), s, t);
             alive = !dup;
         }
         if (!alive) v = 0.0;
@@ -1214,7 +1243,8 @@ k_scan_agg_range2(AggParams P, RangeAux R) {
                             dup = false;
                         }
                         if (!dup && sst.cluster >= 0)
-                            dup = shadowed(P, sst, s0, t0);
+                            dup = shadowed(P, sst, row_seq(P, sst, rg, r0),
+                                           s0, t0);
                         a0 = !dup;
                     }
                     if (a1 && P.skip < 2) {
@@ -1236,7 +1266,8 @@ k_scan_agg_range2(AggParams P, RangeAux R) {
                             dup = false;
                         }
                         if (!dup && sst.cluster >= 0)
-                            dup = shadowed(P, sst, s1, t1);
+                            dup = shadowed(P, sst, row_seq(P, sst, rg, r1),
+                                           s1, t1);
                         a1 = !dup;
                     }
                     my_matched += (a0 ? 1u : 0u) + (a1 ? 1u : 0u);
@@ -1403,6 +1434,7 @@ struct ScanRowsParams {
     uint64_t* out_series;
     long long* out_ts;
     double* out_value;
+    uint64_t* out_seq;        // per-row __seq__ of survivors (nullable)
     unsigned long long* cursor;
     unsigned long long cap;
 };
@@ -1441,6 +1473,7 @@ k_scan_rows(ScanRowsParams R) {
                     R.out_series[j] = s;
                     R.out_ts[j] = t;
                     R.out_value[j] = V[r];
+                    if (R.out_seq) R.out_seq[j] = row_seq(P, sst, rg, r);
                 }
             }
         }
@@ -1803,6 +1836,50 @@ __device__ __forceinline__ uint32_t snappy_varint(const uint8_t* p,
 // kernel). Offsets beyond the mirror still take the drained global path.
 #define SNAP_MIRROR 4096u
 
+// Register bitstream over the compressed page: the tag/length/offset fields
+// are parsed out of a 24-byte register window (three aligned u64 words, one
+// prefetched ahead) with pure ALU — the previous kernel issued 2-3
+// DEPENDENT global byte loads per element, and at ~7k elements per
+// ts/series page that serial chain dominated the whole decode (94-116 ms
+// per 1B-row step). The page payload is 64-byte aligned in the blob, so
+// aligned u64 loads are safe.
+struct SnapStream {
+    const uint64_t* words;   // aligned view of the page payload
+    uint64_t w0, w1, w2;     // words [wi, wi+3)
+    uint32_t wi;             // aligned word index of w0
+    uint32_t n_words;        // ceil(payload/8) — loads are clamped
+
+    __device__ void init(const uint8_t* src, uint32_t clen) {
+        words = (const uint64_t*)src;
+        n_words = (clen + 7u) >> 3;
+        wi = 0;
+        w0 = n_words > 0 ? words[0] : 0;
+        w1 = n_words > 1 ? words[1] : 0;
+        w2 = n_words > 2 ? words[2] : 0;
+    }
+    __device__ __forceinline__ void advance_to(uint32_t pos) {
+        const uint32_t target = pos >> 3;
+        if (target >= wi + 3) {  // long jump (after a big literal): reseat
+            wi = target;
+            w0 = wi < n_words ? words[wi] : 0;
+            w1 = wi + 1 < n_words ? words[wi + 1] : 0;
+            w2 = wi + 2 < n_words ? words[wi + 2] : 0;
+            return;
+        }
+        while (target > wi) {
+            wi++;
+            w0 = w1;
+            w1 = w2;
+            w2 = (wi + 2 < n_words) ? words[wi + 2] : 0;
+        }
+    }
+    // 8 bytes starting at byte `pos` (pos in [wi*8, wi*8+8))
+    __device__ __forceinline__ uint64_t peek8(uint32_t pos) {
+        const uint32_t sh = (pos & 7u) * 8u;
+        return sh ? ((w0 >> sh) | (w1 << (64 - sh))) : w0;
+    }
+};
+
 extern "C" __global__ void __launch_bounds__(256)
 k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     const SnappyPageDesc* __restrict__ pages, uint32_t n_pages,
@@ -1818,29 +1895,48 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
         uint8_t* dst = dec + (pd.dst_off & OFF_MASK);
         const uint32_t clen = pd.comp_len;
         int err = 0;
+        SnapStream st;
+        st.init(src, clen);
+        // uncompressed-length varint via the window
         uint32_t pos = 0;
-        const uint32_t ulen = snappy_varint(src, pos, clen, &err);
+        uint64_t v = st.peek8(0);
+        uint32_t ulen = 0;
+        {
+            int sh = 0;
+            for (;;) {
+                if (pos >= clen || sh > 28) { err = 1; break; }
+                uint8_t b = (uint8_t)(v >> (8 * pos));
+                if (pos >= 7) { err = 1; break; }  // varint fits 5 bytes
+                ulen |= (uint32_t)(b & 0x7f) << sh;
+                pos++;
+                if (!(b & 0x80)) break;
+                sh += 7;
+            }
+        }
         if (err || ulen != pd.uncomp_len) {
             if (lane == 0) atomicAdd(err_flag, 1ull);
             continue;
         }
         uint32_t d = 0;
         while (pos < clen && d < ulen && !err) {
-            const uint8_t tag = src[pos++];
+            st.advance_to(pos);
+            v = st.peek8(pos);
+            const uint32_t tag = (uint32_t)(v & 0xFFu);
             const uint32_t kind = tag & 3u;
             if (kind == 0) {  // literal
                 uint32_t len = (tag >> 2) + 1;
+                uint32_t hdr = 1;
                 if (len > 60) {
                     const uint32_t nb = len - 60;
-                    if (pos + nb > clen) { err = 1; break; }
-                    len = 0;
-                    for (uint32_t i = 0; i < nb; i++)
-                        len |= (uint32_t)src[pos + i] << (8 * i);
+                    if (pos + 1 + nb > clen) { err = 1; break; }
+                    len = (uint32_t)((v >> 8) &
+                                     ((nb >= 4) ? 0xFFFFFFFFull
+                                                : ((1ull << (8 * nb)) - 1)));
                     len += 1;
-                    pos += nb;
+                    hdr = 1 + nb;
                 }
+                pos += hdr;
                 if (pos + len > clen || d + len > ulen) { err = 1; break; }
-                // only the LAST <= SNAP_MIRROR bytes matter for the ring
                 const uint32_t mstart = len > SNAP_MIRROR
                                             ? len - SNAP_MIRROR : 0u;
                 for (uint32_t i = lane; i < len; i += 64) {
@@ -1852,36 +1948,32 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                 pos += len;
                 d += len;
             } else {
-                uint32_t len, off;
+                uint32_t len, off, hdr;
                 if (kind == 1) {
                     len = ((tag >> 2) & 0x7u) + 4;
-                    if (pos >= clen) { err = 1; break; }
-                    off = ((uint32_t)(tag >> 5) << 8) | src[pos];
-                    pos += 1;
+                    off = (uint32_t)((tag >> 5) << 8) |
+                          (uint32_t)((v >> 8) & 0xFFu);
+                    hdr = 2;
                 } else if (kind == 2) {
                     len = (tag >> 2) + 1;
-                    if (pos + 2 > clen) { err = 1; break; }
-                    off = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
-                    pos += 2;
+                    off = (uint32_t)((v >> 8) & 0xFFFFu);
+                    hdr = 3;
                 } else {
                     len = (tag >> 2) + 1;
-                    if (pos + 4 > clen) { err = 1; break; }
-                    off = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8) |
-                          ((uint32_t)src[pos + 2] << 16) |
-                          ((uint32_t)src[pos + 3] << 24);
-                    pos += 4;
+                    off = (uint32_t)((v >> 8) & 0xFFFFFFFFull);
+                    hdr = 5;
                 }
+                if (pos + hdr > clen) { err = 1; break; }
+                pos += hdr;
                 if (off == 0 || off > d || d + len > ulen) { err = 1; break; }
                 if (off + 64u <= SNAP_MIRROR) {
-                    // source entirely within the mirror ring (the 64-byte
-                    // margin keeps this element's ring writes from aliasing
-                    // its own source slots across strided iterations — max
-                    // match len is 64). The (i % off) form reads only
-                    // PRE-element bytes, so there is no intra-element
-                    // hazard; prior elements' ds_writes are ordered before
-                    // these ds_reads within the wave (drain lgkmcnt to keep
-                    // the compiler from reordering them).
-                    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                    // source within the mirror ring (64-byte margin: this
+                    // element's ring writes cannot alias its own source
+                    // slots across strided iterations — max match len 64).
+                    // The (i % off) form reads only pre-element bytes; ds
+                    // ops of one wave execute in order (asm barrier stops
+                    // compiler reordering).
+                    asm volatile("" ::: "memory");
                     for (uint32_t i = lane; i < len; i += 64) {
                         const uint32_t sidx = (off >= len)
                                                   ? d - off + i
@@ -2286,7 +2378,7 @@ hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
                             uint32_t rg_first, uint32_t rg_last,
                             uint64_t* out_series, long long* out_ts,
                             double* out_value, unsigned long long* cursor,
-                            unsigned long long cap) {
+                            unsigned long long cap, uint64_t* out_seq) {
     ScanRowsParams R;
     R.P = p;
     R.rg_first = rg_first;
@@ -2294,6 +2386,7 @@ hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
     R.out_series = out_series;
     R.out_ts = out_ts;
     R.out_value = out_value;
+    R.out_seq = out_seq;
     R.cursor = cursor;
     R.cap = cap;
     uint32_t n = rg_last - rg_first;

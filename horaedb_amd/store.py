@@ -370,6 +370,19 @@ class Store:
             1 if enable_check else 0, C.byref(out)))
         return out.value
 
+    def compact_files(self, input_seqs, devices=None):
+        """General compaction of an EXPLICIT input set (executor
+        Task{inputs}, keep_builtin): dedups among the named files only;
+        the output SST retains per-row __seq__ values."""
+        arr = (C.c_uint64 * len(input_seqs))(*input_seqs)
+        ds, keep = self._devset(devices, [])
+        seq = C.c_uint64()
+        _check(_lib.hx_compact_files(self._h, arr, len(input_seqs),
+                                     C.byref(ds) if ds else None,
+                                     C.byref(seq)))
+        del keep
+        return int(seq.value)
+
     def compact(self, ts_range, devices=None):
         """ColumnarStorage::compact (storage.rs:76-89): GPU-merge the
         ts-overlap closure of the range's SSTs into one new SST. Returns the

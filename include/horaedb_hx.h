@@ -193,6 +193,18 @@ hx_status hx_scan(hx_handle*, const hx_scan_spec*, const hx_device_set*,
 hx_status hx_compact(hx_handle*, hx_time_range range, const hx_device_set*,
                      uint64_t* out_new_seq);
 
+/* General compaction of an EXPLICIT input set (the executor's Task{inputs},
+ * compaction/mod.rs:26-36 + executor.rs:155-222 keep_builtin path): merges
+ * and deduplicates ONLY among the named files and writes one SST whose rows
+ * RETAIN their per-row __seq__ values, so later merges against files
+ * outside the set still order correctly (MergeStream reads __seq__ from the
+ * row, read.rs:289-343). No closure precondition. The scan path detects
+ * such mixed-seq files from the __seq__ column statistics and compares
+ * per-row sequences during cross-SST dedup. */
+hx_status hx_compact_files(hx_handle*, const uint64_t* input_seqs,
+                           size_t n_inputs, const hx_device_set*,
+                           uint64_t* out_new_seq);
+
 /* ColumnarStorage::write (storage.rs:76-89, :307-333): stable PK sort of
  * the batch (sort_batch, storage.rs:244-256), file id allocation
  * (= sequence), one new SST via the native writer, catalog add. Ingest-side
