@@ -190,11 +190,11 @@ def run_extras(args, device, opmap):
         lo, hi = middle_range(m, args.range_frac)
         with Store(store_dir) as st:
             pr = st.prepare((lo, hi), series_in=series_in, devices=[device])
-            pr.exec_agg(ops=ops)
+            pr.exec_agg(ops=ops, copy=False)
             torch.cuda.synchronize(device)
             t0 = time.time()
             for _ in range(steps):
-                pr.exec_agg(ops=ops)
+                pr.exec_agg(ops=ops, copy=False)
             torch.cuda.synchronize(device)
             dt = time.time() - t0
             stt = pr.stats()
