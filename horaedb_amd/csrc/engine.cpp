@@ -111,6 +111,9 @@ struct CatSst {
     bool series_stats_ok = true;    // every chunk has series stats
     uint64_t series_max = 0;        // max over chunks (unsigned order)
     bool seq_mixed = false;         // any rg with non-constant __seq__
+    bool bytes_value = false;       // value column is BYTE_ARRAY (Binary
+                                    // value schema, BytesMergeOperator /
+                                    // Append stores — operator.rs:47-111)
     std::vector<CatRg> rgs;
 };
 
@@ -119,6 +122,9 @@ struct CatSst {
 struct hx_handle {
     std::string store;
     int64_t segment_ms = 0;
+    int32_t update_mode = 0;             // 0 Overwrite / 1 Append
+                                         // (config.rs:166-172)
+    bool bytes_value = false;            // value column is BYTE_ARRAY
     std::vector<CatSst> ssts;            // ascending seq
     std::vector<hx_sst_desc> find_out;   // scratch for hx_find_ssts
 };
@@ -178,8 +184,10 @@ static hx_status read_file_meta(const std::string& path, uint64_t seq,
         return fail(HX_ERR_SCHEMA, path + ": metric schema columns missing");
     if (m.columns[ci[0]].physical_type != hx::PT_INT64 ||
         m.columns[ci[1]].physical_type != hx::PT_INT64 ||
-        m.columns[ci[2]].physical_type != hx::PT_DOUBLE)
+        (m.columns[ci[2]].physical_type != hx::PT_DOUBLE &&
+         m.columns[ci[2]].physical_type != hx::PT_BYTE_ARRAY))
         return fail(HX_ERR_SCHEMA, path + ": unexpected physical types");
+    out.bytes_value = m.columns[ci[2]].physical_type == hx::PT_BYTE_ARRAY;
 
     out.path = path;
     out.seq = seq;
@@ -262,6 +270,13 @@ extern "C" hx_status hx_open(const char* store_path, int64_t segment_duration_ms
         CatSst c;
         hx_status st = read_file_meta(path, seq, c);
         if (st != HX_OK) return st;
+        if (h->ssts.empty()) {
+            h->bytes_value = c.bytes_value;
+        } else if (c.bytes_value != h->bytes_value) {
+            return fail(HX_ERR_SCHEMA,
+                        path + ": value column type differs from the rest "
+                               "of the store");
+        }
         h->ssts.push_back(std::move(c));
     }
     *out = h.release();
@@ -269,6 +284,20 @@ extern "C" hx_status hx_open(const char* store_path, int64_t segment_duration_ms
 }
 
 extern "C" void hx_close(hx_handle* h) { delete h; }
+
+// UpdateMode (config.rs:166-172): selects the MergeOperator the scan plugs
+// in (read.rs:482-492): Overwrite -> LastValueOperator, Append ->
+// BytesMergeOperator (requires a Binary value column, operator.rs:47-111).
+extern "C" hx_status hx_set_update_mode(hx_handle* h, int32_t mode) {
+    if (!h || (mode != 0 && mode != 1))
+        return fail(HX_ERR_INVALID, "mode: 0 Overwrite / 1 Append");
+    if (mode == 1 && !h->bytes_value && !h->ssts.empty())
+        return fail(HX_ERR_UNSUPPORTED,
+                    "Append mode needs a Binary value column "
+                    "(BytesMergeOperator, operator.rs:47-111)");
+    h->update_mode = mode;
+    return HX_OK;
+}
 
 // TimeRange::overlaps (types.rs:125-127): [start,end) vs file [min,max]
 static bool overlaps(const CatSst& s, hx_time_range r) {
@@ -284,9 +313,14 @@ extern "C" hx_status hx_schema(hx_handle* h, const hx_col_desc** out,
         {"value", 2, 0, 0},       {"__seq__", 0, 0, 1},
         {"__reserved__", 0, 0, 1},
     };
+    static const hx_col_desc kSchemaBytes[5] = {
+        {"series_id", 0, 1, 0},   {"timestamp", 1, 1, 0},
+        {"value", 3, 0, 0},       {"__seq__", 0, 0, 1},
+        {"__reserved__", 0, 0, 1},
+    };
     if (!h || !out || !n_out || !n_primary_keys)
         return fail(HX_ERR_INVALID, "null argument");
-    *out = kSchema;
+    *out = h->bytes_value ? kSchemaBytes : kSchema;
     *n_out = 5;
     *n_primary_keys = 2;
     return HX_OK;

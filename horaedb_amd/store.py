@@ -375,7 +375,8 @@ class Store:
         Task{inputs}, keep_builtin): dedups among the named files only;
         the output SST retains per-row __seq__ values."""
         arr = (C.c_uint64 * len(input_seqs))(*input_seqs)
-        ds, keep = self._devset(devices, [])
+        keep = []
+        ds = self._devset(devices, keep)
         seq = C.c_uint64()
         _check(_lib.hx_compact_files(self._h, arr, len(input_seqs),
                                      C.byref(ds) if ds else None,
