@@ -350,6 +350,7 @@ struct DevPlan {
     std::vector<hx::ClusterDev> clusters;
     std::vector<int32_t> cluster_members;
     std::vector<hx::DeltaPageDesc> delta_pages;
+    std::vector<hx::BaPageDesc> ba_pages;   // byte-value stores: handle walk
     std::vector<hx::SnappyPageDesc> snappy_pages;
     std::vector<hx::RleDictPageDesc> rledict_pages;
     std::vector<hx::CopyDesc> copies;
@@ -369,6 +370,7 @@ struct DevPlan {
     hx::ClusterDev* d_clusters = nullptr;
     int32_t* d_members = nullptr;
     hx::DeltaPageDesc* d_delta = nullptr;
+    hx::BaPageDesc* d_ba = nullptr;
     hx::SnappyPageDesc* d_snappy = nullptr;
     hx::RleDictPageDesc* d_rledict = nullptr;
     hx::CopyDesc* d_copies = nullptr;
@@ -440,6 +442,8 @@ struct StagedSst {  // host-side bookkeeping per prepared SST
     int32_t rank = 0;
     int dev_slot = -1;         // index into DevPlan.ssts
     int plan = -1;             // which DevPlan
+    uint64_t bytes_handles = 0;  // dec offset of the value-handle array
+                                 // (byte-value stores only)
 };
 
 }  // namespace
@@ -466,12 +470,14 @@ static size_t align64(size_t x) { return (x + 63) & ~size_t(63); }
 
 static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                               std::vector<StagedSst*>& members) {
+    const bool bytes_val = P->h->bytes_value;
     // ---- layout pass ----------------------------------------------------
     struct PageJob {  // one (rg, col) data page to read+pack
         StagedSst* ss;
         int rg_cat;         // catalog rg index
         int col;            // 0 series 1 ts 2 value
         int64_t chunk_start, comp_size, uncomp_size, num_values;
+        int64_t row_base;
         int32_t codec;
         bool required;
         size_t dst_off;     // blob offset reserved (chunk_size upper bound)
@@ -505,6 +511,10 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
             sd.dense_seq = hx::OFF_DEC | dec_off;
             dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
         }
+        if (P->h->bytes_value) {
+            ss->bytes_handles = dec_off;
+            dec_off = align64(dec_off + size_t(ss->staged_rows) * 8);
+        }
         ss->dev_slot = (int)plan.ssts.size();
         plan.ssts.push_back(sd);
 
@@ -534,6 +544,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 j.num_values = cr.num_values;
                 j.codec = cr.codec;
                 j.required = cr.required;
+                j.row_base = row_base;
                 j.dst_off = blob_off;
                 // Zstd chunks are decompressed in place at staging: reserve
                 // the larger of the on-disk and decompressed sizes
@@ -697,7 +708,32 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
                 raw_size = sp.uncomp_len;
             }
 
-            if (dp->encoding == hx::ENC_PLAIN) {
+            if (bytes_val && j.col == 2) {
+                // Binary value column (BytesMergeOperator stores,
+                // operator.rs:47-111): PLAIN BYTE_ARRAY payload stays in
+                // the blob; per-row handles (off << 20 | len) are walked by
+                // k_ba_offsets into the SST's handle array at decode time.
+                // Snappy value pages would decode into the dec blob, which
+                // the handle packing cannot address — rejected loudly
+                // (uncompressed and Zstd, which decompresses in place, OK).
+                if (dp->encoding != hx::ENC_PLAIN ||
+                    (j.codec == hx::CODEC_SNAPPY && dp->is_compressed)) {
+                    std::lock_guard<std::mutex> g(mu);
+                    err_msg = j.ss->cat->path + ": byte value columns "
+                              "support PLAIN uncompressed/Zstd pages only";
+                    err_flag = 1;
+                    break;
+                }
+                std::lock_guard<std::mutex> g(mu);
+                hx::BaPageDesc bp{};
+                bp.src_off = data_off;   // blob byte offset (no flag bit)
+                bp.src_len = raw_size;
+                bp.n_values = (uint32_t)j.num_values;
+                bp.first_row = int64_t(j.ss->bytes_handles / 8) + j.row_base;
+                plan.ba_pages.push_back(bp);
+                j.final_off = hx::OFF_DEC |
+                              (j.ss->bytes_handles + uint64_t(j.row_base) * 8);
+            } else if (dp->encoding == hx::ENC_PLAIN) {
                 if (raw_size != size_t(j.num_values) * 8) {
                     std::lock_guard<std::mutex> g(mu);
                     err_msg = j.ss->cat->path + ": PLAIN payload size mismatch";
@@ -964,6 +1000,7 @@ static hx_status stage_device(hx_prepared* P, DevPlan& plan,
     HIP_TRY(upload(plan.d_clusters, plan.clusters));
     HIP_TRY(upload(plan.d_members, plan.cluster_members));
     HIP_TRY(upload(plan.d_delta, plan.delta_pages));
+    HIP_TRY(upload(plan.d_ba, plan.ba_pages));
     HIP_TRY(upload(plan.d_snappy, plan.snappy_pages));
     HIP_TRY(upload(plan.d_rledict, plan.rledict_pages));
     HIP_TRY(upload(plan.d_copies, plan.copies));
@@ -1173,6 +1210,7 @@ extern "C" void hx_prepared_free(hx_prepared* P) {
         for (void* p : {(void*)plan.d_blob, (void*)plan.d_dec, (void*)plan.d_rgs,
                         (void*)plan.d_ssts, (void*)plan.d_clusters,
                         (void*)plan.d_members, (void*)plan.d_delta,
+                        (void*)plan.d_ba,
                         (void*)plan.d_snappy, (void*)plan.d_rledict,
                         (void*)plan.d_copies, (void*)plan.t_series,
                         (void*)plan.t_bucket, (void*)plan.t_state,
@@ -1342,6 +1380,13 @@ hx_status ensure_decoded(DevPlan& plan, hipStream_t s) {
                                         plan.d_delta,
                                         (uint32_t)plan.delta_pages.size(),
                                         plan.d_counters + 1));
+    if (!plan.ba_pages.empty())
+        // byte-value stores: walk the PLAIN BYTE_ARRAY length prefixes of
+        // the value pages into per-row handles (off << 20 | len) in dec
+        HIP_TRY(hx::launch_ba_offsets(s, plan.d_blob, plan.d_ba,
+                                      (uint32_t)plan.ba_pages.size(),
+                                      (uint64_t*)plan.d_dec,
+                                      plan.d_counters + 1));
     if (!plan.copies.empty())
         HIP_TRY(hx::launch_copy_u64(s, plan.d_blob, plan.d_dec, plan.d_copies,
                                     (uint32_t)plan.copies.size()));

@@ -79,6 +79,9 @@ hipError_t launch_ba_offsets(hipStream_t s, const uint8_t* blob,
                              const BaPageDesc* pages, uint32_t n_pages,
                              uint64_t* out, unsigned long long* err_flag);
 hipError_t launch_tag_filter(hipStream_t s, const TagFilterParams& f);
+hipError_t launch_copy_bytes(hipStream_t s, const uint8_t* blob,
+                             const uint64_t* handles, const int64_t* dst_off,
+                             uint8_t* out, uint32_t n);
 hipError_t launch_tsid_intersect(hipStream_t s, const uint64_t* a,
                                  unsigned long long n_a, const uint64_t* b,
                                  unsigned long long n_b, uint64_t* out,

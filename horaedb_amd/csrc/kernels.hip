@@ -1564,6 +1564,28 @@ k_tag_filter(TagFilterParams F) {
     }
 }
 
+// gather variable-length byte values: row i copies its bytes (handle =
+// blob_off << 20 | len, from k_ba_offsets) to out[dst_off[i]..dst_off[i+1])
+struct CopyBytesParams {
+    const uint8_t* blob;
+    const uint64_t* handles;    // per OUTPUT row, already permuted
+    const int64_t* dst_off;     // n+1 prefix offsets into out
+    uint8_t* out;
+    uint32_t n;
+};
+
+extern "C" __global__ void __launch_bounds__(256)
+k_copy_bytes(CopyBytesParams C) {
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < C.n;
+         r += blockDim.x * gridDim.x) {
+        const uint64_t h = C.handles[r];
+        const uint8_t* src = C.blob + (h >> 20);
+        const uint32_t len = (uint32_t)(h & 0xFFFFFu);
+        uint8_t* dst = C.out + C.dst_off[r];
+        for (uint32_t i = 0; i < len; i++) dst[i] = src[i];
+    }
+}
+
 // sorted-set intersection: keep a[i] iff it appears in sorted b[0..n_b)
 extern "C" __global__ void __launch_bounds__(256)
 k_tsid_intersect(const uint64_t* __restrict__ a, unsigned long long n_a,
@@ -1834,19 +1856,19 @@ __device__ __forceinline__ uint32_t snappy_varint(const uint8_t* p,
 // ts/series pages of PLAIN metric data are nothing but small periodic
 // matches (measured 94 ms/step at the 1B snappy shape, ~4x the aggregate
 // kernel). Offsets beyond the mirror still take the drained global path.
-// Snappy decode, third design. ISA inspection of the previous kernels
-// showed the copy loops compiled to global_load_ubyte -> s_waitcnt
-// vmcnt(0) -> global_store_byte — a full HBM/L2 round trip per element,
-// and ts/series pages are thousands of tiny elements (94-116 ms/step).
-// This version decodes through a 16 KB per-wave LDS OUTPUT RING:
-//   - every output byte is written to ring[d & 16383] (LDS, ~50 cy);
-//   - matches read the ring (offsets are always <= 16 KB - 64 here; the
-//     rare far reference flushes, drains, and reads global dst);
-//   - completed 8 KB spans are flushed to dst with coalesced 4-byte
-//     stores; huge literals (>= 16 KB, the incompressible value pages)
-//     are copied src->dst directly and only their tail re-primes the ring.
-// The tag stream itself is parsed from a 24-byte register window (ALU).
-#define SNAP_RING 16384u
+// Snappy decode, fourth design. The decode loop is VALU-ISSUE-BOUND: a
+// ts/series page is thousands of tiny matches, and each one paid ~500
+// instructions of copy setup, exec-mask bookkeeping and flush checks (the
+// memory-level rewrites v2/v3 moved the bytes into LDS and changed almost
+// nothing). Two levers here:
+//   1. FUSE runs of same-offset matches at parse time: periodic metric
+//      columns compress into long chains of off=8/16 matches whose fused
+//      form is ONE long periodic copy — the per-tag cost drops to the bare
+//      parse (~40 instructions), the copy amortizes over the fused length;
+//   2. 4 KB LDS output ring per wave (16 KB per 256-thread block => full
+//      32-wave occupancy), flushed to HBM in coalesced spans; matches read
+//      the ring, never global dst (no vmcnt round trips).
+#define SNAP_RING 4096u
 
 struct SnapStream {
     const uint64_t* words;   // aligned view of the page payload
@@ -1949,8 +1971,8 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
         while (pos < clen && d < ulen && !err) {
             st.advance_to(pos);
             v = st.peek8(pos);
-            const uint32_t tag = (uint32_t)(v & 0xFFu);
-            const uint32_t kind = tag & 3u;
+            uint32_t tag = (uint32_t)(v & 0xFFu);
+            uint32_t kind = tag & 3u;
             if (kind == 0) {  // literal
                 uint32_t len = (tag >> 2) + 1;
                 uint32_t hdr = 1;
@@ -1972,17 +1994,14 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
                     for (uint32_t i = lane; i < len; i += 64)
                         dst[d + i] = src[pos + i];
-                    const uint32_t tail = SNAP_RING;
-                    for (uint32_t i = lane; i < tail; i += 64)
-                        ring[(d + len - tail + i) & (SNAP_RING - 1)] =
-                            src[pos + len - tail + i];
+                    for (uint32_t i = lane; i < SNAP_RING; i += 64)
+                        ring[(d + len - SNAP_RING + i) & (SNAP_RING - 1)] =
+                            src[pos + len - SNAP_RING + i];
                     d += len;
                     flushed = d;
                 } else {
                     if (d + len > flushed + SNAP_RING) {
-                        // make room BEFORE writing: never overwrite
-                        // unflushed ring bytes (keep last 8 KB unflushed)
-                        uint32_t want = d + len - 8192u;
+                        uint32_t want = d + len - (SNAP_RING / 2);
                         uint32_t take = want > flushed ? want - flushed : 0;
                         if (take > d - flushed) take = d - flushed;
                         snap_flush(dst, ring, flushed, take, lane);
@@ -1994,58 +2013,115 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                 }
                 pos += len;
             } else {
-                uint32_t len, off, hdr;
+                uint32_t len, off;
                 if (kind == 1) {
                     len = ((tag >> 2) & 0x7u) + 4;
                     off = (uint32_t)((tag >> 5) << 8) |
                           (uint32_t)((v >> 8) & 0xFFu);
-                    hdr = 2;
+                    pos += 2;
                 } else if (kind == 2) {
                     len = (tag >> 2) + 1;
                     off = (uint32_t)((v >> 8) & 0xFFFFu);
-                    hdr = 3;
+                    pos += 3;
                 } else {
                     len = (tag >> 2) + 1;
                     off = (uint32_t)((v >> 8) & 0xFFFFFFFFull);
-                    hdr = 5;
+                    pos += 5;
                 }
-                if (pos + hdr > clen) { err = 1; break; }
-                pos += hdr;
-                if (off == 0 || off > d || d + len > ulen) { err = 1; break; }
-                if (d + len > flushed + SNAP_RING) {
-                    uint32_t want = d + len - 8192u;
-                    uint32_t take = want > flushed ? want - flushed : 0;
-                    if (take > d - flushed) take = d - flushed;
-                    snap_flush(dst, ring, flushed, take, lane);
-                    flushed += take;
-                }
-                if (off + 64u <= SNAP_RING) {
-                    // ring-resident source (64-byte margin: this element's
-                    // writes cannot alias its own source slots across
-                    // strided iterations — max match len 64). (i % off)
-                    // reads only pre-element bytes; one wave's ds ops
-                    // execute in order.
-                    for (uint32_t i = lane; i < len; i += 64) {
-                        const uint32_t sidx = (off >= len)
-                                                  ? d - off + i
-                                                  : d - off + (i % off);
-                        ring[(d + i) & (SNAP_RING - 1)] =
-                            ring[sidx & (SNAP_RING - 1)];
+                if (pos > clen || off == 0 || off > d) { err = 1; break; }
+                // FUSE adjacent same-offset matches: copy k covers
+                // [d+L, d+L+len_k) from [d+L-off, ...) — the continuation
+                // of one periodic/shifted copy. The fused run executes as
+                // a single cooperative loop; per-tag cost = the parse.
+                for (;;) {
+                    st.advance_to(pos);
+                    const uint64_t nv = st.peek8(pos);
+                    const uint32_t ntag = (uint32_t)(nv & 0xFFu);
+                    const uint32_t nkind = ntag & 3u;
+                    uint32_t nlen, noff, nhdr;
+                    if (nkind == 1) {
+                        nlen = ((ntag >> 2) & 0x7u) + 4;
+                        noff = (uint32_t)((ntag >> 5) << 8) |
+                               (uint32_t)((nv >> 8) & 0xFFu);
+                        nhdr = 2;
+                    } else if (nkind == 2) {
+                        nlen = (ntag >> 2) + 1;
+                        noff = (uint32_t)((nv >> 8) & 0xFFFFu);
+                        nhdr = 3;
+                    } else {
+                        break;   // literal or 4-byte-offset copy: stop
                     }
-                } else {
-                    // far back-reference (not produced for this data shape;
-                    // correctness path): flush, drain, read global
+                    if (noff != off || pos + nhdr > clen ||
+                        d + len + nlen > ulen)
+                        break;
+                    len += nlen;
+                    pos += nhdr;
+                }
+                if (d + len > ulen) { err = 1; break; }
+                if (off > SNAP_RING / 2) {
+                    // far back-reference (not produced for this data
+                    // shape): flush + drain once, then chunked global
+                    // reads of the (flushed, stable) pre-run bytes
                     snap_flush(dst, ring, flushed, d - flushed, lane);
                     flushed = d;
                     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-                    for (uint32_t i = lane; i < len; i += 64) {
-                        const uint8_t b = (off >= len)
-                                              ? dst[d - off + i]
-                                              : dst[d - off + (i % off)];
-                        ring[(d + i) & (SNAP_RING - 1)] = b;
+                    const uint32_t D = d;
+                    uint32_t done = 0;
+                    while (done < len) {
+                        const uint32_t chunk =
+                            min(len - done, SNAP_RING / 2);
+                        if (d + chunk > flushed + SNAP_RING) {
+                            uint32_t want = d + chunk - (SNAP_RING / 2);
+                            uint32_t take =
+                                want > flushed ? want - flushed : 0;
+                            if (take > d - flushed) take = d - flushed;
+                            snap_flush(dst, ring, flushed, take, lane);
+                            flushed += take;
+                        }
+                        for (uint32_t i = lane; i < chunk; i += 64) {
+                            const uint32_t j = done + i;
+                            const uint8_t b =
+                                (off >= len) ? dst[D - off + j]
+                                             : dst[D - off + (j % off)];
+                            ring[(d + i) & (SNAP_RING - 1)] = b;
+                        }
+                        d += chunk;
+                        done += chunk;
+                    }
+                } else {
+                    // ring-resident source, period-doubling: after `done`
+                    // bytes of the run, everything in [d - done - off, d)
+                    // repeats with period (done + off), so each chunk can
+                    // copy with shift P = min(done + off, 2048) — sources
+                    // are all pre-chunk and at most ~2 KB behind the
+                    // cursor (always ring-resident). Chunks grow
+                    // geometrically: a 64 KB off=8 run takes ~40 rounds.
+                    uint32_t done = 0;
+                    while (done < len) {
+                        // shift P must stay a MULTIPLE of off (periodicity)
+                        // and <= RING/2 (so P + chunk <= RING: this chunk's
+                        // writes never clobber its sources). done + off is
+                        // a multiple of off by construction (done sums
+                        // previous P's).
+                        uint32_t P = done + off;
+                        if (P > SNAP_RING / 2)
+                            P = off * ((SNAP_RING / 2) / off);
+                        const uint32_t chunk = min(len - done, P);
+                        if (d + chunk > flushed + SNAP_RING) {
+                            uint32_t want = d + chunk - (SNAP_RING / 2);
+                            uint32_t take =
+                                want > flushed ? want - flushed : 0;
+                            if (take > d - flushed) take = d - flushed;
+                            snap_flush(dst, ring, flushed, take, lane);
+                            flushed += take;
+                        }
+                        for (uint32_t i = lane; i < chunk; i += 64)
+                            ring[(d + i) & (SNAP_RING - 1)] =
+                                ring[(d - P + i) & (SNAP_RING - 1)];
+                        d += chunk;
+                        done += chunk;
                     }
                 }
-                d += len;
             }
         }
         if (!err && d == ulen) {
@@ -2550,6 +2626,15 @@ hipError_t launch_init_rep(hipStream_t s, uint8_t* rep, size_t total_slots,
 
 hipError_t launch_iota(hipStream_t s, uint32_t* out, uint32_t n) {
     hipLaunchKernelGGL(k_iota, dim3(grid_for(n, 256)), dim3(256), 0, s, out, n);
+    return hipGetLastError();
+}
+
+hipError_t launch_copy_bytes(hipStream_t s, const uint8_t* blob,
+                             const uint64_t* handles, const int64_t* dst_off,
+                             uint8_t* out, uint32_t n) {
+    CopyBytesParams C{blob, handles, dst_off, out, n};
+    hipLaunchKernelGGL(k_copy_bytes, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       C);
     return hipGetLastError();
 }
 

@@ -122,11 +122,20 @@ typedef struct hx_result_table {
  * Columns follow the projection order; builtin __seq__/__reserved__ are
  * stripped as MergeStream does (read.rs:330-343). Column buffers are valid
  * only during the callback. */
+/* variable-length byte column payload (Binary value schema): for a column
+ * of type 3, the batch's cols[i] points at ONE hx_bytes_col; row r's value
+ * is bytes[offsets[r] .. offsets[r+1]). */
+typedef struct {
+    const int64_t* offsets;       /* n_rows + 1 */
+    const uint8_t* bytes;
+} hx_bytes_col;
+
 typedef struct {
     size_t   n_rows;
     size_t   n_cols;
-    const void* const* cols;      /* col i: n_rows × 8B elements            */
-    const int32_t*     col_types; /* 0=u64, 1=i64(ts ms), 2=f64             */
+    const void* const* cols;      /* col i: n_rows × 8B elements, or ONE
+                                     hx_bytes_col when col_types[i] == 3    */
+    const int32_t*     col_types; /* 0=u64, 1=i64(ts ms), 2=f64, 3=bytes    */
 } hx_col_batch;
 typedef int32_t (*hx_batch_cb)(void* ctx, const hx_col_batch* batch); /* nonzero => stop */
 
@@ -204,6 +213,12 @@ hx_status hx_compact(hx_handle*, hx_time_range range, const hx_device_set*,
 hx_status hx_compact_files(hx_handle*, const uint64_t* input_seqs,
                            size_t n_inputs, const hx_device_set*,
                            uint64_t* out_new_seq);
+
+/* UpdateMode (config.rs:166-172): 0 = Overwrite (LastValueOperator), 1 =
+ * Append (BytesMergeOperator — requires a Binary value column,
+ * operator.rs:47-111). Append changes hx_scan's merge: equal-PK rows'
+ * value bytes CONCATENATE in ascending __seq__ order. */
+hx_status hx_set_update_mode(hx_handle*, int32_t mode);
 
 /* ColumnarStorage::write (storage.rs:76-89, :307-333): stable PK sort of
  * the batch (sort_batch, storage.rs:244-256), file id allocation
