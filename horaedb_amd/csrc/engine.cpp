@@ -1537,6 +1537,10 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
 hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                     HostTable& out, double* agg_kernel_ms,
                     unsigned long long* matched_out) {
+    if (P->h->bytes_value)
+        return fail(HX_ERR_UNSUPPORTED,
+                    "aggregates over Binary value columns are undefined "
+                    "(scan the rows with hx_scan instead)");
     HIP_TRY(hipSetDevice(plan.device));
     hipStream_t s = plan.stream;
     const uint32_t ops = kernel_ops(agg->ops);
@@ -2097,6 +2101,15 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
     // read.rs:429-494). GPU: filter/dedup append + radix sorts; single
     // device.
     if (!h || !spec || !cb) return fail(HX_ERR_INVALID, "null argument");
+    const bool bytesv = h->bytes_value;
+    // Append mode (UpdateMode config.rs:166-172 -> BytesMergeOperator,
+    // operator.rs:47-111): every filtered row is kept (no last-wins dedup)
+    // and equal-PK groups concatenate their value bytes in ascending
+    // __seq__ order (MergeStream group order, read.rs:289-343).
+    const bool append = h->update_mode == 1;
+    if (append && !bytesv)
+        return fail(HX_ERR_UNSUPPORTED,
+                    "Append mode needs a Binary value column");
     int32_t dev0 = (devs && devs->device_ids && devs->n_devices > 0)
                        ? devs->device_ids[0] : 0;
     hx_device_set one{&dev0, 1};
@@ -2118,8 +2131,8 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
     if (cap > 0xFFFFFFFFull)
         return fail(HX_ERR_UNSUPPORTED,
                     "row streams above 2^32 rows per call: split the range");
-    // device buffers: 3 columns + sort keys/scratch + 3 perms
-    size_t need = cap * 8 * 5 + cap * 4 * 3 + 64;
+    // device buffers: 3-4 columns + sort keys/scratch + 3 perms
+    size_t need = cap * 8 * (append ? 6 : 5) + cap * 4 * 3 + 64;
     st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
     if (st != HX_OK) return st;
     uint8_t* base = (uint8_t*)plan.d_scratch;
@@ -2131,6 +2144,7 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
     uint64_t* d_series = (uint64_t*)carve8(cap);
     long long* d_ts = (long long*)carve8(cap);
     double* d_val = (double*)carve8(cap);
+    uint64_t* d_seq = append ? (uint64_t*)carve8(cap) : nullptr;
     uint64_t* d_keys = (uint64_t*)carve8(cap);
     uint64_t* d_keys_out = (uint64_t*)carve8(cap);
     unsigned long long* d_cursor = (unsigned long long*)carve8(1);
@@ -2140,8 +2154,16 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
 
     HIP_TRY(hipMemsetAsync(d_cursor, 0, 8, s));
     hx::AggParams A = base_params(P, plan);
+    if (append) {
+        A.skip = 2;   // keep every row: groups concatenate
+        // the packed (seq << 32 | row) sort key needs seq < 2^32
+        for (const auto& c : h->ssts)
+            if (c.seq >= (1ull << 32))
+                return fail(HX_ERR_UNSUPPORTED,
+                            "append stores need file sequences < 2^32");
+    }
     HIP_TRY(hx::launch_scan_rows(s, A, 0, A.n_rgs, d_series, d_ts, d_val,
-                                 d_cursor, cap));
+                                 d_cursor, cap, d_seq, append ? 1 : 0));
     unsigned long long n64 = 0;
     HIP_TRY(hipStreamSynchronize(s));
     HIP_TRY(hipMemcpy(&n64, d_cursor, 8, hipMemcpyDeviceToHost));
@@ -2150,8 +2172,18 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
     const uint32_t n = (uint32_t)n64;
     if (n == 0) return HX_OK;
 
-    // LSD-stable ordering: ts, then series, then time segment
+    // LSD-stable ordering: [seq,] ts, then series, then time segment
     HIP_TRY(hx::launch_iota(s, perm_a, n));
+    if (append) {
+        // least-significant pass first: group rows end up in ascending
+        // __seq__ order (the BytesMergeOperator concat order)
+        HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_seq,
+                                      perm_a, (unsigned long long*)d_keys,
+                                      n));
+        HIP_TRY(hx::sort_pairs_u64(s, d_keys, d_keys_out, perm_a, perm_b, n,
+                                   &plan.d_sort_temp, &plan.sort_temp_cap));
+        std::swap(perm_a, perm_b);
+    }
     HIP_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_ts, perm_a,
                                   (unsigned long long*)d_keys, n));
     HIP_TRY(hx::launch_xor_sign(s, (unsigned long long*)d_keys, n));
@@ -2182,7 +2214,65 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
     HIP_TRY(hipMemcpyAsync(host.data(), d_dst, 3 * size_t(n) * 8,
                            hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
-    hipFree(d_dst);
+
+    // Binary value columns: the gathered "value" words are per-row handles
+    // (blob_off << 20 | len, from k_ba_offsets). Materialize the output
+    // byte stream on the GPU (k_copy_bytes moves the bytes; the host does
+    // only the O(n) offset arithmetic), then batch groups to the caller.
+    std::vector<uint8_t> hbytes;
+    std::vector<int64_t> group_off;      // per emitted row (or group)
+    std::vector<uint64_t> gseries;
+    std::vector<int64_t> gts;
+    uint32_t n_out_rows = n;
+    if (bytesv) {
+        const uint64_t* hnd = host.data() + 2 * size_t(n);
+        const uint64_t* hs = host.data();
+        const int64_t* ht = (const int64_t*)(host.data() + n);
+        std::vector<int64_t> row_off(n + 1);
+        row_off[0] = 0;
+        for (uint32_t i = 0; i < n; i++)
+            row_off[i + 1] = row_off[i] + int64_t(hnd[i] & 0xFFFFFu);
+        // group layout: Append concatenates equal-PK runs (already
+        // adjacent and seq-ordered); Overwrite has one row per group
+        group_off.clear();
+        gseries.clear();
+        gts.clear();
+        if (append) {
+            for (uint32_t i = 0; i < n; i++) {
+                if (i == 0 || hs[i] != hs[i - 1] || ht[i] != ht[i - 1]) {
+                    gseries.push_back(hs[i]);
+                    gts.push_back(ht[i]);
+                    group_off.push_back(row_off[i]);
+                }
+            }
+            group_off.push_back(row_off[n]);
+            n_out_rows = (uint32_t)gseries.size();
+        } else {
+            n_out_rows = n;
+        }
+        const int64_t total = row_off[n];
+        int64_t* d_offs = nullptr;
+        uint8_t* d_bytes = nullptr;
+        HIP_TRY(hipMalloc((void**)&d_offs, (size_t(n) + 1) * 8));
+        HIP_TRY(hipMalloc((void**)&d_bytes, size_t(std::max<int64_t>(
+                                                total, 1))));
+        HIP_TRY(hipMemcpyAsync(d_offs, row_off.data(), (size_t(n) + 1) * 8,
+                               hipMemcpyHostToDevice, s));
+        HIP_TRY(hx::launch_copy_bytes(s, plan.d_blob,
+                                      (const uint64_t*)(d_dst + 2 * size_t(n)),
+                                      d_offs, d_bytes, n));
+        hbytes.resize(size_t(std::max<int64_t>(total, 0)));
+        if (total > 0)
+            HIP_TRY(hipMemcpyAsync(hbytes.data(), d_bytes, size_t(total),
+                                   hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        (void)hipFree(d_offs);
+        (void)hipFree(d_bytes);
+        if (!append) {
+            group_off = std::move(row_off);
+        }
+    }
+    (void)hipFree(d_dst);
 
     // projection over user columns (storage.rs:65-70; builtins stripped as
     // MergeStream does, read.rs:330-343)
@@ -2197,17 +2287,36 @@ extern "C" hx_status hx_scan(hx_handle* h, const hx_scan_spec* spec,
     } else {
         proj = {0, 1, 2};
     }
-    static const int32_t kTypes[3] = {0, 1, 2};  // u64, i64, f64
+    const int32_t val_type = bytesv ? 3 : 2;
+    const int32_t kTypes[3] = {0, 1, val_type};  // u64, i64, f64|bytes
     const uint64_t* cols_base[3] = {host.data(), host.data() + n,
                                     host.data() + 2 * size_t(n)};
+    if (bytesv && append) {
+        // group-level fixed columns replace the row-level ones
+        cols_base[0] = gseries.data();
+        cols_base[1] = (const uint64_t*)gts.data();
+    }
     const uint32_t BATCH = 65536;
     std::vector<const void*> colptrs(proj.size());
     std::vector<int32_t> coltypes(proj.size());
+    std::vector<int64_t> rebased;
     for (size_t i = 0; i < proj.size(); i++) coltypes[i] = kTypes[proj[i]];
-    for (uint32_t off = 0; off < n; off += BATCH) {
-        uint32_t len = std::min(BATCH, n - off);
-        for (size_t i = 0; i < proj.size(); i++)
-            colptrs[i] = (const void*)(cols_base[proj[i]] + off);
+    for (uint32_t off = 0; off < n_out_rows; off += BATCH) {
+        uint32_t len = std::min(BATCH, n_out_rows - off);
+        hx_bytes_col bc{};
+        for (size_t i = 0; i < proj.size(); i++) {
+            if (proj[i] == 2 && bytesv) {
+                rebased.assign(group_off.begin() + off,
+                               group_off.begin() + off + len + 1);
+                const int64_t base0 = rebased[0];
+                for (auto& x : rebased) x -= base0;
+                bc.offsets = rebased.data();
+                bc.bytes = hbytes.data() + base0;
+                colptrs[i] = (const void*)&bc;
+            } else {
+                colptrs[i] = (const void*)(cols_base[proj[i]] + off);
+            }
+        }
         hx_col_batch b{len, proj.size(), colptrs.data(), coltypes.data()};
         if (cb(ctx, &b)) break;
     }

@@ -48,7 +48,8 @@ hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
                             uint64_t* out_series, long long* out_ts,
                             double* out_value, unsigned long long* cursor,
                             unsigned long long cap,
-                            uint64_t* out_seq = nullptr);
+                            uint64_t* out_seq = nullptr,
+                            int32_t seq_rowidx = 0);
 hipError_t launch_gather_multi(hipStream_t s,
                                const unsigned long long* const* srcs,
                                uint32_t n_arrays, const uint32_t* perm,

@@ -1435,6 +1435,11 @@ struct ScanRowsParams {
     long long* out_ts;
     double* out_value;
     uint64_t* out_seq;        // per-row __seq__ of survivors (nullable)
+    int32_t seq_rowidx;       // 1: out_seq = (seq << 32) | staged row index
+                              // — the Append sort key: equal-PK equal-seq
+                              // rows keep FILE ROW ORDER (the writer's
+                              // stable sort; MergeStream stream order)
+    int32_t _pad;
     unsigned long long* cursor;
     unsigned long long cap;
 };
@@ -1473,7 +1478,13 @@ k_scan_rows(ScanRowsParams R) {
                     R.out_series[j] = s;
                     R.out_ts[j] = t;
                     R.out_value[j] = V[r];
-                    if (R.out_seq) R.out_seq[j] = row_seq(P, sst, rg, r);
+                    if (R.out_seq) {
+                        uint64_t q = row_seq(P, sst, rg, r);
+                        R.out_seq[j] = R.seq_rowidx
+                            ? (q << 32) |
+                                  (uint64_t)(uint32_t)(rg.row_base + r)
+                            : q;
+                    }
                 }
             }
         }
@@ -2509,7 +2520,8 @@ hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
                             uint32_t rg_first, uint32_t rg_last,
                             uint64_t* out_series, long long* out_ts,
                             double* out_value, unsigned long long* cursor,
-                            unsigned long long cap, uint64_t* out_seq) {
+                            unsigned long long cap, uint64_t* out_seq,
+                            int32_t seq_rowidx) {
     ScanRowsParams R;
     R.P = p;
     R.rg_first = rg_first;
@@ -2518,6 +2530,8 @@ hipError_t launch_scan_rows(hipStream_t s, const AggParams& p,
     R.out_ts = out_ts;
     R.out_value = out_value;
     R.out_seq = out_seq;
+    R.seq_rowidx = seq_rowidx;
+    R._pad = 0;
     R.cursor = cursor;
     R.cap = cap;
     uint32_t n = rg_last - rg_first;

@@ -190,14 +190,25 @@ class _TagPred(C.Structure):
     _fields_ = [("tag_key", C.c_char_p), ("tag_value", C.c_char_p)]
 
 
+class _BytesCol(C.Structure):
+    _fields_ = [("offsets", C.POINTER(C.c_int64)),
+                ("bytes", C.POINTER(C.c_uint8))]
+
+
 class Store:
     """MI355X-native ColumnarStorage (scan side). storage.rs:76-89."""
 
-    def __init__(self, store_path, segment_duration_ms=0):
+    def __init__(self, store_path, segment_duration_ms=0,
+                 update_mode="overwrite"):
         h = C.c_void_p()
+        self._update_mode = update_mode
         _check(_lib.hx_open(store_path.encode(), segment_duration_ms,
                             C.byref(h)))
         self._h = h
+        if update_mode == "append":
+            _check(_lib.hx_set_update_mode(self._h, 1))
+        elif update_mode != "overwrite":
+            raise ValueError("update_mode: 'overwrite' or 'append'")
 
     def close(self):
         if self._h:
@@ -417,6 +428,20 @@ class Store:
             cols = []
             for i in range(b.n_cols):
                 t = b.col_types[i]
+                if t == 3:  # Binary value column (hx_bytes_col)
+                    bcp = C.cast(b.cols[i], C.POINTER(_BytesCol)).contents
+                    offs = np.ctypeslib.as_array(
+                        bcp.offsets, shape=(b.n_rows + 1,)).copy()
+                    total = int(offs[-1])
+                    data = (np.ctypeslib.as_array(
+                        bcp.bytes, shape=(total,)).copy()
+                        if total else np.empty(0, np.uint8))
+                    vals = np.empty(b.n_rows, dtype=object)
+                    db = data.tobytes()
+                    for r in range(b.n_rows):
+                        vals[r] = db[offs[r]:offs[r + 1]]
+                    cols.append(vals)
+                    continue
                 dt = {0: np.uint64, 1: np.int64, 2: np.float64}[t]
                 ptr = C.cast(b.cols[i], C.POINTER(C.c_uint64))
                 arr = np.ctypeslib.as_array(ptr, shape=(b.n_rows,)).copy()

@@ -250,9 +250,25 @@ def read_sst(path, with_builtins=False):
             raise ValueError(f"{path}: missing column {nm} (schema contract)")
     s = t.column("series_id").to_numpy().astype(np.uint64)
     ts = t.column("timestamp").to_numpy().astype(np.int64)
-    v = t.column("value").to_numpy().astype(np.float64)
+    import pyarrow as pa
+    vcol = t.column("value")
+    if pa.types.is_binary(vcol.type) or pa.types.is_large_binary(vcol.type) \
+            or pa.types.is_string(vcol.type):
+        # Binary value schema (BytesMergeOperator stores, operator.rs:47-111)
+        v = np.empty(t.num_rows, dtype=object)
+        for i, x in enumerate(vcol.to_pylist()):
+            v[i] = x if isinstance(x, bytes) else x.encode()
+    else:
+        v = vcol.to_numpy().astype(np.float64)
     if "__seq__" in names and t.num_rows > 0:
-        seq = int(t.column("__seq__")[0].as_py())
+        sq = t.column("__seq__").to_numpy().astype(np.uint64)
+        if (sq == sq[0]).all():
+            seq = int(sq[0])
+        else:
+            # keep_builtin compaction output (executor.rs:155-222): the
+            # file carries PER-ROW sequences; MergeStream orders by the
+            # row's own __seq__ (read.rs:289-343)
+            seq = sq
     else:
         seq = int(os.path.splitext(os.path.basename(path))[0])
     cols = [s, ts, v]

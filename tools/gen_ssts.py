@@ -194,6 +194,40 @@ def gen_dataset(out_dir, n_rows, n_series, n_ssts, seed=42,
     return manifest
 
 
+def write_bytes_sst(path, series, ts, values, seq, row_group=8192,
+                    sort=True):
+    """One Binary-value SST (BytesMergeOperator stores, operator.rs:47-111):
+    (series u64 PK, ts i64 PK, value binary, builtins), PLAIN uncompressed,
+    writer-sorted by (series, ts) stable (equal PKs keep input order)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    series = np.asarray(series, dtype=np.uint64)
+    ts = np.asarray(ts, dtype=np.int64)
+    values = list(values)
+    if sort:
+        order = np.lexsort((ts, series))
+        series = series[order]
+        ts = ts[order]
+        values = [values[i] for i in order]
+    n = len(series)
+    tbl = pa.table({
+        "series_id": pa.array(series, pa.uint64()),
+        "timestamp": pa.array(ts, pa.int64()),
+        "value": pa.array([v if isinstance(v, bytes) else v.encode()
+                           for v in values], pa.binary()),
+        "__seq__": pa.array(np.full(n, seq, np.uint64), pa.uint64()),
+        "__reserved__": pa.array(np.zeros(n, np.uint64), pa.uint64()),
+    })
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    pq.write_table(tbl, path, row_group_size=row_group, compression="NONE",
+                   use_dictionary=False, data_page_version="1.0",
+                   column_encoding={c: "PLAIN" for c in
+                                    ("series_id", "timestamp", "value",
+                                     "__seq__", "__reserved__")},
+                   write_statistics=True)
+    return path
+
+
 def gen_tag_index(store_dir, n_dc=100):
     """Write {store}/index/1.sst (the RFC index table, rfc:86-137) for the
     dataset's series: tags dc=dc{i%n_dc} (1/n_dc selectivity each) and
