@@ -1979,11 +1979,114 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
         }
         uint32_t d = 0;        // output cursor
         uint32_t flushed = 0;  // dst bytes already written
+        // Register sliding window (v5 fast path): series pages are 8k
+        // ALTERNATING {7-byte literal, off=8 match} elements per page —
+        // per-element cost must be tens of instructions, not hundreds.
+        // tail0 = output[d-8, d), tail1 = output[d-16, d-8) (LE-packed);
+        // pend accumulates the aligned 8-byte word at (d & ~7), written
+        // with ONE ds_write_b64 per 8 output bytes. Writing the full pend
+        // word early is safe: the slop bytes at [d, dA+8) sit beyond the
+        // cursor in the ring and are rewritten before any flush reads them.
+        uint64_t tail0 = 0, tail1 = 0, pend = 0;
+        const uint64_t* const ring64 = (const uint64_t*)ring;
+        // append n (1..8) bytes (LE in low bytes of nv) to the output
+        auto rwin_append = [&](uint64_t nv, uint32_t n) {
+            const uint32_t sh = 8u * n;
+            if (n == 8) {
+                tail1 = tail0;
+                tail0 = nv;
+            } else {
+                tail1 = (tail1 >> sh) | (tail0 << (64 - sh));
+                tail0 = (tail0 >> sh) | (nv << (64 - sh));
+            }
+            const uint32_t a = d & 7u;
+            pend |= nv << (8u * a);   // a == 0 shifts by 0
+            if (a + n >= 8) {
+                if (lane == 0)
+                    *(uint64_t*)(ring + ((d - a) & (SNAP_RING - 1) & ~7u)) =
+                        pend;
+                pend = a ? (nv >> (8u * (8u - a))) : 0;
+            }
+            d += n;
+        };
+        // spill the pending partial word (slop-safe) before any path that
+        // reads/writes the ring directly or flushes
+        auto rwin_spill = [&]() {
+            if ((d & 7u) && lane == 0)
+                *(uint64_t*)(ring + (d & (SNAP_RING - 1) & ~7u)) = pend;
+        };
+        // reload tail/pend from the ring after direct-ring paths
+        auto rwin_reload = [&]() {
+            const uint32_t a = d & 7u;
+            const uint32_t base = (d - a) & (SNAP_RING - 1);
+            const uint64_t w0 = ring64[base >> 3];
+            const uint64_t w1 = ring64[((base + SNAP_RING - 8) &
+                                        (SNAP_RING - 1)) >> 3];
+            const uint64_t w2 = ring64[((base + SNAP_RING - 16) &
+                                        (SNAP_RING - 1)) >> 3];
+            if (a == 0) {
+                pend = 0;
+                tail0 = w1;
+                tail1 = w2;
+            } else {
+                const uint32_t sh = 8u * a;
+                pend = w0 & ((1ull << sh) - 1u);
+                tail0 = (w1 >> sh) | (w0 << (64 - sh));
+                tail1 = (w2 >> sh) | (w1 << (64 - sh));
+            }
+        };
         while (pos < clen && d < ulen && !err) {
             st.advance_to(pos);
             v = st.peek8(pos);
             uint32_t tag = (uint32_t)(v & 0xFFu);
             uint32_t kind = tag & 3u;
+            // ---- v5 fast path: tiny literal / pow2-offset short match ----
+            if (kind == 0) {
+                // len <= 7: the literal's bytes are exactly the upper 7
+                // bytes of the peek window (tag occupies byte 0)
+                const uint32_t len = (tag >> 2) + 1;
+                if (len <= 7 && pos + 1 + len <= clen && d + len <= ulen &&
+                    d + len + 8 <= flushed + SNAP_RING) {
+                    const uint64_t nv = (v >> 8) & ((1ull << (8 * len)) - 1);
+                    rwin_append(nv, len);
+                    pos += 1 + len;
+                    continue;
+                }
+            } else if (kind == 1 || kind == 2) {
+                uint32_t len, off, hdr;
+                if (kind == 1) {
+                    len = ((tag >> 2) & 0x7u) + 4;
+                    off = (uint32_t)((tag >> 5) << 8) |
+                          (uint32_t)((v >> 8) & 0xFFu);
+                    hdr = 2;
+                } else {
+                    len = (tag >> 2) + 1;
+                    off = (uint32_t)((v >> 8) & 0xFFFFu);
+                    hdr = 3;
+                }
+                if ((off == 1 || off == 2 || off == 4 || off == 8) &&
+                    off <= d && pos + hdr <= clen && d + len <= ulen &&
+                    d + len + 8 <= flushed + SNAP_RING) {
+                    // pow2 period divides 8: the replicated pattern word is
+                    // constant across every emitted 8-byte chunk
+                    uint64_t rep = tail0 >> (64 - 8 * off);
+                    if (off < 8) rep |= rep << (8 * off);
+                    if (off < 4) rep |= rep << 16;
+                    if (off < 2) rep |= rep << 32;
+                    uint32_t rem = len;
+                    while (rem >= 8) {
+                        rwin_append(rep, 8);
+                        rem -= 8;
+                    }
+                    if (rem)
+                        rwin_append(rep & ((1ull << (8 * rem)) - 1), rem);
+                    pos += hdr;
+                    continue;
+                }
+            }
+            // ---- general paths (ring-direct): sync the register window --
+            rwin_spill();
+            bool slow = true;
             if (kind == 0) {  // literal
                 uint32_t len = (tag >> 2) + 1;
                 uint32_t hdr = 1;
@@ -2011,7 +2114,7 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     d += len;
                     flushed = d;
                 } else {
-                    if (d + len > flushed + SNAP_RING) {
+                    if (d + len + 8 > flushed + SNAP_RING) {
                         uint32_t want = d + len - (SNAP_RING / 2);
                         uint32_t take = want > flushed ? want - flushed : 0;
                         if (take > d - flushed) take = d - flushed;
@@ -2081,7 +2184,7 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     while (done < len) {
                         const uint32_t chunk =
                             min(len - done, SNAP_RING / 2);
-                        if (d + chunk > flushed + SNAP_RING) {
+                        if (d + chunk + 8 > flushed + SNAP_RING) {
                             uint32_t want = d + chunk - (SNAP_RING / 2);
                             uint32_t take =
                                 want > flushed ? want - flushed : 0;
@@ -2118,7 +2221,7 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                         if (P > SNAP_RING / 2)
                             P = off * ((SNAP_RING / 2) / off);
                         const uint32_t chunk = min(len - done, P);
-                        if (d + chunk > flushed + SNAP_RING) {
+                        if (d + chunk + 8 > flushed + SNAP_RING) {
                             uint32_t want = d + chunk - (SNAP_RING / 2);
                             uint32_t take =
                                 want > flushed ? want - flushed : 0;
@@ -2134,7 +2237,9 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     }
                 }
             }
+            if (slow) rwin_reload();
         }
+        rwin_spill();
         if (!err && d == ulen) {
             snap_flush(dst, ring, flushed, d - flushed, lane);
         } else if (lane == 0) {
