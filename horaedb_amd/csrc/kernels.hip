@@ -307,7 +307,7 @@ __device__ __forceinline__ bool row_alive(const AggParams& P, const RgDesc& rg,
                                           uint32_t n, uint64_t s, int64_t t) {
     bool alive = (t >= P.ts_lo) & (t < P.ts_hi);
     if (alive && P.use_sset) alive = sset_has(P, s);
-    if (alive) {
+    if (alive && P.skip < 2) {   // skip>=2: Append keeps every row
         // within-SST dedup: the LAST row of an equal-PK run survives
         // (LastValueOperator, operator.rs:37-44; plan order read.rs:456-480)
         bool dup = false;
