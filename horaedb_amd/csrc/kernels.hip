@@ -2040,6 +2040,64 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
             v = st.peek8(pos);
             uint32_t tag = (uint32_t)(v & 0xFFu);
             uint32_t kind = tag & 3u;
+            // ---- v6 BULK-PAIR path: series pages are runs of the exact
+            // 10-byte pair {lit7 (tag 0x18), match off=8 len=9 (tag 0x15,
+            // off byte 0x08)} — 16 output bytes per pair, and the pair's
+            // output is [b0..b6, p, b0..b6, p] where p = the byte at d-1,
+            // INVARIANT across the run (the off-8 source telescopes).
+            // 64 lanes decode 64 pairs per wave iteration: the serial
+            // element loop costs ~500 instructions per element; this path
+            // costs ~2.
+            if (tag == 0x18u && d >= 1 && pos + 10 <= clen) {
+                rwin_spill();
+                const uint8_t pbyte =
+                    ring[(d - 1) & (SNAP_RING - 1)];
+                uint32_t consumed = 0;
+                for (;;) {
+                    const uint32_t pp = pos + 10u * lane;
+                    bool ok = pp + 10 <= clen &&
+                              d + 16u * (lane + 1u) <= ulen;
+                    uint8_t b[7];
+                    if (ok) {
+                        ok = src[pp] == 0x18u && src[pp + 8] == 0x15u &&
+                             src[pp + 9] == 0x08u;
+                        if (ok)
+                            for (int k = 0; k < 7; k++) b[k] = src[pp + 1 + k];
+                    }
+                    const unsigned long long bm = __ballot(ok);
+                    const uint32_t m =
+                        (~bm == 0ull) ? 64u
+                                      : (uint32_t)(__ffsll((long long)~bm) - 1);
+                    if (m == 0) break;
+                    // ring room for 16*m bytes (+8 slop margin)
+                    if (d + 16u * m + 8u > flushed + SNAP_RING) {
+                        uint32_t want = d + 16u * m - (SNAP_RING / 2);
+                        uint32_t take = want > flushed ? want - flushed : 0;
+                        if (take > d - flushed) take = d - flushed;
+                        snap_flush(dst, ring, flushed, take, lane);
+                        flushed += take;
+                    }
+                    if (lane < m) {
+                        const uint32_t base = d + 16u * lane;
+                        for (int k = 0; k < 7; k++) {
+                            ring[(base + k) & (SNAP_RING - 1)] = b[k];
+                            ring[(base + 8 + k) & (SNAP_RING - 1)] = b[k];
+                        }
+                        ring[(base + 7) & (SNAP_RING - 1)] = pbyte;
+                        ring[(base + 15) & (SNAP_RING - 1)] = pbyte;
+                    }
+                    d += 16u * m;
+                    pos += 10u * m;
+                    consumed += m;
+                    if (m < 64u) break;
+                }
+                if (consumed) {
+                    rwin_reload();
+                    continue;
+                }
+                // zero pairs matched: fall through to the per-element paths
+                // (the register window is still consistent after the spill)
+            }
             // ---- v5 fast path: tiny literal / pow2-offset short match ----
             if (kind == 0) {
                 // len <= 7: the literal's bytes are exactly the upper 7
