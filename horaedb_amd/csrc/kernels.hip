@@ -977,8 +977,8 @@ __device__ __forceinline__ void scan_window_range(
     W.head = head;
 }
 
-template <bool MM>
-__global__ void __launch_bounds__(256)
+template <bool MM, int MINW>
+__global__ void __launch_bounds__(256, MINW)
 k_scan_agg_range(AggParams P, RangeAux R) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     __shared__ int s_abort;   // global table saturated: stop, host retries
@@ -2656,10 +2656,21 @@ hipError_t launch_scan_agg_range2(hipStream_t s, const AggParams& p,
 hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
                                  const RangeAux& r, bool minmax) {
     const size_t lds = (size_t)r.ne * (minmax ? 36 : 20);
-    const void* f = minmax
-                        ? reinterpret_cast<const void*>(&k_scan_agg_range<true>)
-                        : reinterpret_cast<const void*>(
-                              &k_scan_agg_range<false>);
+    // occupancy: the unconstrained build allocates 118 VGPRs => 4 waves/
+    // SIMD (16/CU). MINW forces the register budget down
+    // (5 => <=96, 6 => <=80) — the kernel is latency-bound, so resident
+    // waves are the lever; HX_MINW picks (default 6, measured fastest).
+    int minw = 6;
+    if (const char* e = getenv("HX_MINW")) minw = atoi(e);
+    const void* f;
+    if (minmax)
+        f = minw >= 6 ? (const void*)&k_scan_agg_range<true, 6>
+            : minw == 5 ? (const void*)&k_scan_agg_range<true, 5>
+                        : (const void*)&k_scan_agg_range<true, 1>;
+    else
+        f = minw >= 6 ? (const void*)&k_scan_agg_range<false, 6>
+            : minw == 5 ? (const void*)&k_scan_agg_range<false, 5>
+                        : (const void*)&k_scan_agg_range<false, 1>;
     // >64 KiB dynamic LDS requires the opt-in or the launch FAILS silently
     // (measured the hard way on the gang kernel). Request only what this
     // launch needs: the kernel's static shared (s_abort) counts against
@@ -2670,12 +2681,10 @@ hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
             f, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
         if (e != hipSuccess) return e;
     }
-    if (minmax)
-        hipLaunchKernelGGL(k_scan_agg_range<true>, dim3(r.n_blocks), dim3(256),
-                           lds, s, p, r);
-    else
-        hipLaunchKernelGGL(k_scan_agg_range<false>, dim3(r.n_blocks),
-                           dim3(256), lds, s, p, r);
+    hipLaunchKernelGGL(
+        reinterpret_cast<void (*)(AggParams, RangeAux)>(
+            const_cast<void*>(f)),
+        dim3(r.n_blocks), dim3(256), lds, s, p, r);
     return hipGetLastError();
 }
 
