@@ -529,6 +529,18 @@ __device__ __forceinline__ void lds_update(const AggParams& P, uint64_t* lkey,
                                            uint64_t pre = 0,
                                            bool has_pre = false) {
     uint32_t i = (i0 != 0xFFFFFFFFu) ? i0 : ((uint32_t)mix64(sv) & (ne - 1));
+    // bisect modes (wrong results; cost decomposition only):
+    // skip=5 probe-read only, no atomics; skip=6 atomics to the probe slot
+    // without key verification
+    if (P.skip == 5) {
+        if (lkey[i] == sv) return;
+        return;
+    }
+    if (P.skip == 6) {
+        atomicAdd(&lsum[i], v);
+        atomicAdd(&lcnt[i], cnt);
+        return;
+    }
 #pragma unroll 1
     for (int probes = 0; probes < 16; probes++) {
         // probe prefetch: the caller read slot i0 right after the row loads,
