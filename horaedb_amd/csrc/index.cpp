@@ -236,10 +236,13 @@ extern "C" hx_status hx_index_query(hx_handle* h, const hx_tag_pred* preds,
                     close(fd);
                     return set_error(HX_ERR_IO, path + ": chunk read");
                 }
+                const int col_i = which == 0 ? ck : which == 1 ? cv : ct;
                 std::vector<hx::PageDesc> pages;
                 try {
                     pages = hx::walk_pages(chunk.data(), chunk.size(),
-                                           cc.chunk_start(), cc.num_values);
+                                           cc.chunk_start(), cc.num_values,
+                                           !m.columns[col_i].required,
+                                           cc.codec);
                 } catch (const std::exception& e) {
                     close(fd);
                     return set_error(HX_ERR_FORMAT, path + ": " + e.what());
@@ -257,8 +260,10 @@ extern "C" hx_status hx_index_query(hx_handle* h, const hx_tag_pred* preds,
                                      path + ": expected one data page per "
                                             "index chunk");
                 }
-                size_t in_chunk = size_t(dp->payload_off - cc.chunk_start());
-                size_t plen = size_t(dp->compressed_size);
+                size_t in_chunk = size_t(dp->payload_off - cc.chunk_start()) +
+                                  size_t(dp->def_level_bytes);
+                size_t plen = size_t(dp->compressed_size) -
+                              size_t(dp->def_level_bytes);
                 if (in_chunk + plen > chunk.size()) {
                     close(fd);
                     return set_error(HX_ERR_FORMAT, path + ": page bounds");

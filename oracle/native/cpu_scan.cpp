@@ -20,6 +20,7 @@
 // untimed cpu_baseline leg loads it. Product path = libhoraedb_hx.so.
 #include "../../horaedb_amd/csrc/parquet_meta.h"
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <cstdint>
@@ -244,6 +245,20 @@ extern "C" int hx_cpu_scan_agg(const char** paths, int n_paths, int64_t ts_lo,
             }
             total_rows += rg.num_rows;
             units.push_back({paths[f], rg, {ci[0], ci[1], ci[2]}});
+        }
+    }
+
+    // Decorrelate concurrent workers: units in (file, rg) order put every
+    // thread in the SAME series window at once (SSTs share one sorted
+    // series universe), and the shared table's hot lines ping-pong across
+    // cores — measured 28 Mrows/s at 256 threads. A deterministic shuffle
+    // spreads concurrent units over the whole table.
+    {
+        uint64_t h = 0x9E3779B97F4A7C15ull;
+        for (size_t i = units.size(); i > 1; i--) {
+            h ^= h >> 12; h ^= h << 25; h ^= h >> 27;
+            size_t j = (size_t)((h * 0x2545F4914F6CDD1Dull >> 33) % i);
+            std::swap(units[i - 1], units[j]);
         }
     }
 

@@ -210,6 +210,13 @@ def write_bytes_sst(path, series, ts, values, seq, row_group=8192,
         ts = ts[order]
         values = [values[i] for i in order]
     n = len(series)
+    schema = pa.schema([
+        pa.field("series_id", pa.uint64(), nullable=False),
+        pa.field("timestamp", pa.int64(), nullable=False),
+        pa.field("value", pa.binary(), nullable=False),
+        pa.field("__seq__", pa.uint64(), nullable=False),
+        pa.field("__reserved__", pa.uint64(), nullable=False),
+    ])
     tbl = pa.table({
         "series_id": pa.array(series, pa.uint64()),
         "timestamp": pa.array(ts, pa.int64()),
@@ -217,7 +224,7 @@ def write_bytes_sst(path, series, ts, values, seq, row_group=8192,
                            for v in values], pa.binary()),
         "__seq__": pa.array(np.full(n, seq, np.uint64), pa.uint64()),
         "__reserved__": pa.array(np.zeros(n, np.uint64), pa.uint64()),
-    })
+    }, schema=schema)
     os.makedirs(os.path.dirname(path), exist_ok=True)
     pq.write_table(tbl, path, row_group_size=row_group, compression="NONE",
                    use_dictionary=False, data_page_version="1.0",
@@ -259,6 +266,12 @@ def gen_tag_index(store_dir, n_dc=100):
     # order is "dev" < "prod" and dev rows are the odd ones (is_prod False)
     env_value_col = np.where(env_is_prod[env_order], "prod", "dev")
     env_tsid = ids[env_order]
+    schema = pa.schema([
+        pa.field("metric_id", pa.uint64(), nullable=False),
+        pa.field("tag_key", pa.binary(), nullable=False),
+        pa.field("tag_value", pa.binary(), nullable=False),
+        pa.field("tsid", pa.uint64(), nullable=False),
+    ])   # REQUIRED columns: v1 pages carry no def-level prefix
     tbl = pa.table({
         "metric_id": pa.array(np.zeros(2 * n, np.uint64), pa.uint64()),
         "tag_key": pa.array([b"dc"] * n + [b"env"] * n, pa.binary()),
@@ -266,7 +279,7 @@ def gen_tag_index(store_dir, n_dc=100):
             [x.encode() for x in dc_value_col] +
             [x.encode() for x in env_value_col], pa.binary()),
         "tsid": pa.array(np.concatenate([dc_tsid, env_tsid]), pa.uint64()),
-    })
+    }, schema=schema)
     pq.write_table(tbl, idx_path, row_group_size=8192, compression="NONE",
                    use_dictionary=False, data_page_version="1.0",
                    column_encoding={c: "PLAIN" for c in
