@@ -340,6 +340,15 @@ def main():
             store_dir, m = get_dataset(args, 0)
             if dist:
                 dist.barrier()
+    elif dist and world > 1:
+        # stagger per-rank dataset generation: 8 concurrent 24 GB writes
+        # can blow /tmp before the post-staging cleanup runs; token-ring
+        # order bounds peak disk to ~2 datasets
+        store_dir = m = None
+        for r in range(world):
+            if r == rank:
+                store_dir, m = get_dataset(args, rank)
+            dist.barrier()
     else:
         store_dir, m = get_dataset(args, rank)
     ts_range = middle_range(m, args.range_frac)
