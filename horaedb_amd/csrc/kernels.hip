@@ -1081,14 +1081,40 @@ k_scan_agg_range(AggParams P, RangeAux R) {
                                               B, lkey, ne, R.interp, ilo,
                                               islope);
                     if (P.skip == 1) continue;
-                    if (A.head && A.c > 0)
-                        lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
-                                       A.s, A.vv, (uint32_t)A.c, A.mn, A.mx,
-                                       A.hint_i, A.hint_k, true);
-                    if (B.head && B.c > 0)
-                        lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax, ne,
-                                       B.s, B.vv, (uint32_t)B.c, B.mn, B.mx,
-                                       B.hint_i, B.hint_k, true);
+                    // verified-hit short circuit: the probe read happened at
+                    // load time; hint_k == key PROVES the slot (keys are
+                    // write-once), so the adds go straight to LDS with no
+                    // fresh read and no probe-loop control flow — the loop
+                    // structure alone measured ~19.5 ms of the 23.9 ms
+                    // kernel (skip6 bisect)
+                    if (A.head && A.c > 0) {
+                        if (A.hint_k == A.s) {
+                            atomicAdd(&lsum[A.hint_i], A.vv);
+                            atomicAdd(&lcnt[A.hint_i], (uint32_t)A.c);
+                            if (MM) {
+                                atomicMin(&lmin[A.hint_i], f64_ordered(A.mn));
+                                atomicMax(&lmax[A.hint_i], f64_ordered(A.mx));
+                            }
+                        } else {
+                            lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax,
+                                           ne, A.s, A.vv, (uint32_t)A.c,
+                                           A.mn, A.mx, A.hint_i);
+                        }
+                    }
+                    if (B.head && B.c > 0) {
+                        if (B.hint_k == B.s) {
+                            atomicAdd(&lsum[B.hint_i], B.vv);
+                            atomicAdd(&lcnt[B.hint_i], (uint32_t)B.c);
+                            if (MM) {
+                                atomicMin(&lmin[B.hint_i], f64_ordered(B.mn));
+                                atomicMax(&lmax[B.hint_i], f64_ordered(B.mx));
+                            }
+                        } else {
+                            lds_update<MM>(P, lkey, lsum, lcnt, lmin, lmax,
+                                           ne, B.s, B.vv, (uint32_t)B.c,
+                                           B.mn, B.mx, B.hint_i);
+                        }
+                    }
                 }
                 pos++;
                 row = 0;
