@@ -2335,6 +2335,72 @@ extern "C" hx_status hx_write_sst(const char* path, const uint64_t* series,
     return HX_OK;
 }
 
+// GPU ingest sort (stable by (series, ts); equal PKs keep input order):
+// two stable LSD radix passes over a u32 permutation — ts (sign-biased),
+// then series — then a device gather and one D2H. Returns false on any HIP
+// error so the caller falls back to the host sort.
+static bool gpu_sort_batch(const uint64_t* series, const int64_t* ts,
+                           const double* value, int64_t n,
+                           std::vector<uint64_t>& s2, std::vector<int64_t>& t2,
+                           std::vector<double>& v2) {
+    hipStream_t s = nullptr;
+    uint8_t* dev = nullptr;
+    void* d_temp = nullptr;
+    size_t temp_cap = 0;
+    const size_t N = (size_t)n;
+    // layout: series, ts, value, keys_a, keys_b, dst(3N), perm a/b/c
+    size_t bytes = N * 8 * 8 + N * 4 * 3 + 256;
+    bool ok = hipMalloc((void**)&dev, bytes) == hipSuccess;
+    auto fail_out = [&]() {
+        if (d_temp) (void)hipFree(d_temp);
+        if (dev) (void)hipFree(dev);
+        return false;
+    };
+    if (!ok) return fail_out();
+    uint8_t* p = dev;
+    auto carve = [&](size_t cnt, size_t w) {
+        uint8_t* q = p;
+        p += (cnt * w + 7) & ~size_t(7);
+        return q;
+    };
+    uint64_t* d_series = (uint64_t*)carve(N, 8);
+    int64_t* d_ts = (int64_t*)carve(N, 8);
+    double* d_val = (double*)carve(N, 8);
+    uint64_t* d_ka = (uint64_t*)carve(N, 8);
+    uint64_t* d_kb = (uint64_t*)carve(N, 8);
+    unsigned long long* d_dst = (unsigned long long*)carve(N * 3, 8);
+    uint32_t* pa = (uint32_t*)carve(N, 4);
+    uint32_t* pb = (uint32_t*)carve(N, 4);
+    uint32_t* pc = (uint32_t*)carve(N, 4);
+#define GS_TRY(e) do { if ((e) != hipSuccess) return fail_out(); } while (0)
+    GS_TRY(hipMemcpyAsync(d_series, series, N * 8, hipMemcpyHostToDevice, s));
+    GS_TRY(hipMemcpyAsync(d_ts, ts, N * 8, hipMemcpyHostToDevice, s));
+    GS_TRY(hipMemcpyAsync(d_val, value, N * 8, hipMemcpyHostToDevice, s));
+    GS_TRY(hx::launch_iota(s, pa, (uint32_t)N));
+    GS_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_ts, pa,
+                                 (unsigned long long*)d_ka, (uint32_t)N));
+    GS_TRY(hx::launch_xor_sign(s, (unsigned long long*)d_ka, (uint32_t)N));
+    GS_TRY(hx::sort_pairs_u64(s, d_ka, d_kb, pa, pb, N, &d_temp, &temp_cap));
+    GS_TRY(hx::launch_gather_u64(s, (const unsigned long long*)d_series, pb,
+                                 (unsigned long long*)d_ka, (uint32_t)N));
+    GS_TRY(hx::sort_pairs_u64(s, d_ka, d_kb, pb, pc, N, &d_temp, &temp_cap));
+    const unsigned long long* srcs[3] = {
+        (const unsigned long long*)d_series, (const unsigned long long*)d_ts,
+        (const unsigned long long*)d_val};
+    GS_TRY(hx::launch_gather_multi(s, srcs, 3, pc, d_dst, (uint32_t)N));
+    std::vector<uint64_t> host(3 * N);
+    GS_TRY(hipMemcpyAsync(host.data(), d_dst, 3 * N * 8,
+                          hipMemcpyDeviceToHost, s));
+    GS_TRY(hipStreamSynchronize(s));
+#undef GS_TRY
+    std::memcpy(s2.data(), host.data(), N * 8);
+    std::memcpy(t2.data(), host.data() + N, N * 8);
+    std::memcpy(v2.data(), host.data() + 2 * N, N * 8);
+    if (d_temp) (void)hipFree(d_temp);
+    (void)hipFree(dev);
+    return true;
+}
+
 extern "C" hx_status hx_write(hx_handle* h, const uint64_t* series,
                               const int64_t* ts, const double* value,
                               int64_t n, int32_t enable_check,
@@ -2363,22 +2429,32 @@ extern "C" hx_status hx_write(hx_handle* h, const uint64_t* series,
                         "write crosses a segment boundary (storage.rs:309-316)");
     }
     // stable sort by (series, ts) — equal PKs keep batch order
-    // (LastValueOperator's last-wins depends on it, operator.rs:37-44)
-    std::vector<uint32_t> order(n);
-    for (int64_t i = 0; i < n; i++) order[i] = (uint32_t)i;
-    std::stable_sort(order.begin(), order.end(),
-                     [&](uint32_t a2, uint32_t b2) {
-                         if (series[a2] != series[b2])
-                             return series[a2] < series[b2];
-                         return ts[a2] < ts[b2];
-                     });
+    // (LastValueOperator's last-wins depends on it, operator.rs:37-44).
+    // Large batches sort on the GPU (rocPRIM LSD radix, stable): the
+    // ingest-side analog of sort_batch's SortExec (storage.rs:244-256) —
+    // SURVEY §8(f) row 3's sort half. Small batches (or no device) stay
+    // on the host.
     std::vector<uint64_t> s2(n);
     std::vector<int64_t> t2(n);
     std::vector<double> v2(n);
-    for (int64_t i = 0; i < n; i++) {
-        s2[i] = series[order[i]];
-        t2[i] = ts[order[i]];
-        v2[i] = value[order[i]];
+    bool sorted_on_gpu = false;
+    if (n >= (1 << 16) && hip_device_count() > 0) {
+        sorted_on_gpu = gpu_sort_batch(series, ts, value, n, s2, t2, v2);
+    }
+    if (!sorted_on_gpu) {
+        std::vector<uint32_t> order(n);
+        for (int64_t i = 0; i < n; i++) order[i] = (uint32_t)i;
+        std::stable_sort(order.begin(), order.end(),
+                         [&](uint32_t a2, uint32_t b2) {
+                             if (series[a2] != series[b2])
+                                 return series[a2] < series[b2];
+                             return ts[a2] < ts[b2];
+                         });
+        for (int64_t i = 0; i < n; i++) {
+            s2[i] = series[order[i]];
+            t2[i] = ts[order[i]];
+            v2[i] = value[order[i]];
+        }
     }
     uint64_t max_seq = 0;
     for (const auto& s : h->ssts) max_seq = std::max(max_seq, s.seq);

@@ -119,3 +119,32 @@ def _oracle_rows(d):
             for p in sorted(glob.glob(os.path.join(d, "data", "*.sst")),
                             key=lambda p: int(os.path.basename(p)[:-4]))]
     return oracle.scan_rows(ssts, (-10**15, 10**15))
+
+
+def test_gpu_ingest_sort_large_batch(tmp_path):
+    # hx_write's GPU radix ingest sort (SURVEY §8 f3's sort half) kicks in
+    # at >= 65536 rows: PK order + equal-PK input-order stability must
+    # match the host path (sort_batch semantics, storage.rs:244-256)
+    import pyarrow.parquet as pq
+    from horaedb_amd import Store
+    d = str(tmp_path / "store")
+    os.makedirs(os.path.join(d, "data"))
+    # seed file so the store opens
+    from tools.gen_ssts import gen_sst_from_arrays
+    gen_sst_from_arrays(d, 1, [1], [0], [0.0])
+    rng = np.random.default_rng(12)
+    n = 200_000
+    series = rng.integers(0, 500, n).astype(np.uint64)   # heavy duplicates
+    ts = rng.integers(0, 50, n).astype(np.int64) * 1000
+    value = np.arange(n, dtype=np.float64)               # encodes input order
+    with Store(d) as st:
+        seq = st.write(series, ts, value, enable_check=False)
+    t = pq.read_table(os.path.join(d, "data", f"{seq}.sst"))
+    gs = np.array(t.column("series_id").to_pylist(), dtype=np.uint64)
+    gt = np.array(t.column("timestamp").to_pylist(), dtype=np.int64)
+    gv = np.array(t.column("value").to_pylist())
+    # expected: numpy stable lexsort by (series, ts)
+    order = np.lexsort((ts, series))
+    np.testing.assert_array_equal(gs, series[order])
+    np.testing.assert_array_equal(gt, ts[order])
+    np.testing.assert_array_equal(gv, value[order])  # equal-PK stability
