@@ -157,7 +157,8 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
                                                   unsigned long long cnt,
                                                   double mn, double mx,
                                                   uint32_t hint_i = 0xFFFFFFFFu,
-                                                  uint64_t hint_k = 0) {
+                                                  uint64_t hint_k = 0,
+                                                  bool count_fill = true) {
     const uint32_t stride = P.table.stride;
     uint32_t i = (uint32_t)mix64(s) & P.table.mask;
     if (hint_i == i && hint_k == s) {  // prefetched probe already matched
@@ -172,7 +173,13 @@ __device__ __forceinline__ void agg_update_keycas(const AggParams& P, uint64_t s
             uint64_t expected = KEY_EMPTY;
             if (__hip_atomic_compare_exchange_strong((uint64_t*)slot,
                     &expected, s, RLX, RLX, AGT)) {
-                __hip_atomic_fetch_add(P.fill, 1ull, RLX, AGT);
+                // P.fill is ONE word: per-claim increments serialize at the
+                // coherence point and backpressure the whole vmcnt pipeline
+                // (measured ~17 ms of the 24 ms launch). Callers that count
+                // claims in bulk (the range kernels' flush) pass
+                // count_fill = false and add per-block totals instead.
+                if (count_fill)
+                    __hip_atomic_fetch_add(P.fill, 1ull, RLX, AGT);
                 k = s;
             } else {
                 k = expected;
@@ -242,9 +249,11 @@ __device__ __forceinline__ void agg_update(const AggParams& P, uint64_t s,
                                            unsigned long long cnt,
                                            double mn, double mx,
                                            uint32_t hint_i = 0xFFFFFFFFu,
-                                           uint64_t hint_k = 0) {
+                                           uint64_t hint_k = 0,
+                                           bool count_fill = true) {
     if (P.key_claim) {
-        agg_update_keycas(P, s, b, vsum, cnt, mn, mx, hint_i, hint_k);
+        agg_update_keycas(P, s, b, vsum, cnt, mn, mx, hint_i, hint_k,
+                          count_fill);
         return;
     }
     // AoS slot: {state u32, pad u32, series u64, bucket i64, sum, cnt
@@ -1121,11 +1130,21 @@ k_scan_agg_range(AggParams P, RangeAux R) {
             }
         }
         __syncthreads();
-        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
-            if (lkey[i] == KEY_EMPTY) continue;
-            agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
-                       MM ? ordered_f64(lmin[i]) : 0.0,
-                       MM ? ordered_f64(lmax[i]) : 0.0);
+        {
+            unsigned long long live = 0;
+            for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+                if (lkey[i] == KEY_EMPTY) continue;
+                live++;
+                agg_update(P, lkey[i], 0, lsum[i],
+                           (unsigned long long)lcnt[i],
+                           MM ? ordered_f64(lmin[i]) : 0.0,
+                           MM ? ordered_f64(lmax[i]) : 0.0, 0xFFFFFFFFu, 0,
+                           false);
+            }
+            for (int off = 32; off > 0; off >>= 1)
+                live += __shfl_down(live, off, 64);
+            if ((threadIdx.x & 63) == 0 && live)
+                __hip_atomic_fetch_add(P.fill, live, RLX, AGT);
         }
         __syncthreads();
     }
@@ -1332,11 +1351,21 @@ k_scan_agg_range2(AggParams P, RangeAux R) {
             }
         }
         __syncthreads();
-        for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
-            if (lkey[i] == KEY_EMPTY) continue;
-            agg_update(P, lkey[i], 0, lsum[i], (unsigned long long)lcnt[i],
-                       MM ? ordered_f64(lmin[i]) : 0.0,
-                       MM ? ordered_f64(lmax[i]) : 0.0);
+        {
+            unsigned long long live = 0;
+            for (uint32_t i = threadIdx.x; i < ne; i += blockDim.x) {
+                if (lkey[i] == KEY_EMPTY) continue;
+                live++;
+                agg_update(P, lkey[i], 0, lsum[i],
+                           (unsigned long long)lcnt[i],
+                           MM ? ordered_f64(lmin[i]) : 0.0,
+                           MM ? ordered_f64(lmax[i]) : 0.0, 0xFFFFFFFFu, 0,
+                           false);
+            }
+            for (int off = 32; off > 0; off >>= 1)
+                live += __shfl_down(live, off, 64);
+            if ((threadIdx.x & 63) == 0 && live)
+                __hip_atomic_fetch_add(P.fill, live, RLX, AGT);
         }
         __syncthreads();
     }

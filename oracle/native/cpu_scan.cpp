@@ -103,15 +103,27 @@ struct SharedTable {
     uint32_t mask = 0;
     std::atomic<uint64_t> fill{0};
 
-    void init(uint32_t slots) {
+    void init(uint32_t slots, int threads) {
         key.reset(new std::atomic<uint64_t>[slots]);
         sumbits.reset(new std::atomic<uint64_t>[slots]);
         cnt.reset(new std::atomic<uint64_t>[slots]);
-        for (uint32_t i = 0; i < slots; i++) {
-            key[i].store(~0ull, std::memory_order_relaxed);
-            sumbits[i].store(0, std::memory_order_relaxed);
-            cnt[i].store(0, std::memory_order_relaxed);
-        }
+        // parallel first-touch: a single-threaded init pins every table
+        // page to one NUMA node and its memory controller then caps the
+        // whole scan (measured 124 Mrows/s at 16 threads collapsing to
+        // 32 M at 256)
+        std::vector<std::thread> ts;
+        int nt = std::max(1, std::min(threads, 64));
+        for (int t = 0; t < nt; t++)
+            ts.emplace_back([&, t]() {
+                uint32_t lo = (uint32_t)((uint64_t)slots * t / nt);
+                uint32_t hi = (uint32_t)((uint64_t)slots * (t + 1) / nt);
+                for (uint32_t i = lo; i < hi; i++) {
+                    key[i].store(~0ull, std::memory_order_relaxed);
+                    sumbits[i].store(0, std::memory_order_relaxed);
+                    cnt[i].store(0, std::memory_order_relaxed);
+                }
+            });
+        for (auto& th : ts) th.join();
         mask = slots - 1;
     }
 
@@ -122,14 +134,14 @@ struct SharedTable {
         return x ^ (x >> 31);
     }
 
-    void add(uint64_t s, double vsum, uint64_t c) {
+    void add(uint64_t s, double vsum, uint64_t c, uint64_t& my_fill) {
         uint32_t i = (uint32_t)mix64(s) & mask;
         for (;;) {
             uint64_t k = key[i].load(std::memory_order_relaxed);
             if (k == ~0ull) {
                 uint64_t expected = ~0ull;
                 if (key[i].compare_exchange_strong(expected, s))
-                    fill.fetch_add(1, std::memory_order_relaxed);
+                    my_fill++;   // one hot counter would serialize all cores
                 k = key[i].load(std::memory_order_relaxed);
             }
             if (k == s) break;
@@ -266,7 +278,7 @@ extern "C" int hx_cpu_scan_agg(const char** paths, int n_paths, int64_t ts_lo,
     uint32_t slots = 1 << 16;
     while ((uint64_t)slots < total_rows / 16 && slots < (1u << 27))
         slots <<= 1;
-    table.init(slots);
+    table.init(slots, threads);
 
     std::atomic<size_t> next{0};
     std::atomic<int> err{0};
@@ -276,6 +288,7 @@ extern "C" int hx_cpu_scan_agg(const char** paths, int n_paths, int64_t ts_lo,
     auto t0 = std::chrono::steady_clock::now();
     auto worker = [&]() {
         uint64_t my_matched = 0;
+        uint64_t my_fill = 0;
         double my_digest = 0;
         int last_fd = -1;
         std::string last_path;
@@ -326,15 +339,17 @@ extern "C" int hx_cpu_scan_agg(const char** paths, int n_paths, int64_t ts_lo,
                     run_sum += v;
                     run_cnt++;
                 } else {
-                    if (run_cnt) table.add(run_key, run_sum, run_cnt);
+                    if (run_cnt)
+                        table.add(run_key, run_sum, run_cnt, my_fill);
                     run_key = S[r];
                     run_sum = v;
                     run_cnt = 1;
                 }
             }
-            if (run_cnt) table.add(run_key, run_sum, run_cnt);
+            if (run_cnt) table.add(run_key, run_sum, run_cnt, my_fill);
         }
         if (last_fd >= 0) close(last_fd);
+        table.fill.fetch_add(my_fill, std::memory_order_relaxed);
         matched.fetch_add(my_matched);
         // accumulate digest via CAS (exactness not required; diagnostic)
         uint64_t old = digest_bits.load();
