@@ -328,6 +328,11 @@ def main():
     for o in args.ops.split(","):
         ops |= opmap[o.strip()]
 
+    device = int(os.environ.get("HX_DEV_OVERRIDE", local_rank))
+    if torch.cuda.is_available():
+        # set BEFORE any collective: nccl barriers bind to the current device
+        torch.cuda.set_device(device)
+
     from tools.gen_ssts import middle_range
     if args.config5:
         if args.bucket_ms <= 0:
@@ -365,9 +370,6 @@ def main():
         # decode/decompress must run in EVERY timed step (no cached decode)
         os.environ["HX_REDECODE"] = "1"
         log("workload has a decode stage: HX_REDECODE=1 (per-step decode)")
-
-    device = int(os.environ.get("HX_DEV_OVERRIDE", local_rank))
-    torch.cuda.set_device(device)
 
     store = Store(store_dir)
     t0 = time.time()
