@@ -1494,11 +1494,11 @@ static hx_status ensure_range(DevPlan& plan, const hx::AggParams& base) {
     }
     const double x_est = 0.5 * (x_lo + x_hi);
     plan.range_xest = x_est;
-    double target = 1300.0;  // series per block (ne=1024: LDS 20.5 KB/block
-                             // => 7 blocks/CU = 28 waves; the kernel is
-                             // latency-bound, so occupancy beats table
-                             // headroom — measured 26 vs 38 ms at the 1B
-                             // shape vs ne=2048's 16 waves)
+    double target = 650.0;   // series per block: ne=1024 at load ~0.3.
+                             // Load 0.6 put ~0.3% of heads through the
+                             // global fallback whose per-claim fill adds
+                             // re-serialize at the coherence point —
+                             // 13.5 ms vs 7.5 ms at the 1B shape.
     if (const char* e = getenv("HX_RANGE_TARGET")) target = atof(e);
     uint32_t nb = 512;
     while (nb < (uint32_t)std::min(1e9, x_est * 1.3 / target) &&

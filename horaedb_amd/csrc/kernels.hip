@@ -2231,8 +2231,37 @@ k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     // ring, copy src->dst directly, re-prime the ring tail
                     snap_flush(dst, ring, flushed, d - flushed, lane);
                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-                    for (uint32_t i = lane; i < len; i += 64)
-                        dst[d + i] = src[pos + i];
+                    if ((d & 7u) == 0) {
+                        // dst 8-aligned (whole-page literals start at 0):
+                        // copy 8 output bytes per lane from two aligned
+                        // source words + funnel shift — 8x fewer
+                        // instructions than the byte loop
+                        const uint64_t* sw =
+                            (const uint64_t*)(src + ((pos) & ~7u));
+                        const uint32_t sh = 8u * (pos & 7u);
+                        // sh != 0 reads sw[w+1]: stop a word early so the
+                        // last aligned read never crosses the blob slack
+                        const uint32_t words =
+                            sh ? ((len >> 3) ? (len >> 3) - 1 : 0)
+                               : (len >> 3);
+                        for (uint32_t w = lane; w < words; w += 64) {
+                            uint64_t lo = sw[w];
+                            uint64_t x;
+                            if (sh) {
+                                uint64_t hi2 = sw[w + 1];
+                                x = (lo >> sh) | (hi2 << (64 - sh));
+                            } else {
+                                x = lo;
+                            }
+                            *(uint64_t*)(dst + d + 8 * w) = x;
+                        }
+                        for (uint32_t i = (words << 3) + lane; i < len;
+                             i += 64)
+                            dst[d + i] = src[pos + i];
+                    } else {
+                        for (uint32_t i = lane; i < len; i += 64)
+                            dst[d + i] = src[pos + i];
+                    }
                     for (uint32_t i = lane; i < SNAP_RING; i += 64)
                         ring[(d + len - SNAP_RING + i) & (SNAP_RING - 1)] =
                             src[pos + len - SNAP_RING + i];
