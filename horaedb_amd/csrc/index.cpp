@@ -475,8 +475,14 @@ extern "C" hx_status hx_index_query(hx_handle* h, const hx_tag_pred* preds,
     IHIP_TRY(hipStreamSynchronize(s));
     uint64_t* host = (uint64_t*)malloc(std::max<size_t>(1, n_acc * 8));
     if (!host) return set_error(HX_ERR_IO, "hx_index_query: oom");
-    if (n_acc)
-        IHIP_TRY(hipMemcpy(host, d_acc.p, n_acc * 8, hipMemcpyDeviceToHost));
+    if (n_acc) {
+        hipError_t ce = hipMemcpy(host, d_acc.p, n_acc * 8,
+                                  hipMemcpyDeviceToHost);
+        if (ce != hipSuccess) {
+            free(host);
+            return set_error(HX_ERR_HIP, hipGetErrorString(ce));
+        }
+    }
     *out_tsids = host;
     *n_out = (size_t)n_acc;
     return HX_OK;
