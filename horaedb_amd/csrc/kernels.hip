@@ -2753,10 +2753,11 @@ hipError_t launch_scan_agg_range(hipStream_t s, const AggParams& p,
                                  const RangeAux& r, bool minmax) {
     const size_t lds = (size_t)r.ne * (minmax ? 36 : 20);
     // occupancy: the unconstrained build allocates 118 VGPRs => 4 waves/
-    // SIMD (16/CU). MINW forces the register budget down
-    // (5 => <=96, 6 => <=80) — the kernel is latency-bound, so resident
-    // waves are the lever; HX_MINW picks (default 6, measured fastest).
-    int minw = 6;
+    // SIMD (16/CU). MINW 5/6 force the register budget down (<=96/<=80)
+    // at 80/128 B of spill; after the fill-counter fix all three measure
+    // within ~3% (7.1-7.4 ms) — default 1 (no spill) keeps the PMC
+    // traffic at the algorithmic 1.1-1.3x instead of +7 GB of scratch.
+    int minw = 1;
     if (const char* e = getenv("HX_MINW")) minw = atoi(e);
     const void* f;
     if (minmax)
