@@ -1717,14 +1717,15 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                            plan.d_sst_rg_cnt,  (uint32_t)plan.ssts.size(),
                            plan.range_nblocks, ne,
                            interp};
-            // HX_RANGE2=1: pair-load variant (16 B/lane dwordx4, in-lane
-            // pair merge, no cross-lane pre-reduce). Default OFF: the
-            // shuffle-pre-reduce kernel measured faster at the 1B shape
-            // (23.6-24.2 vs 25.4 ms same-box) — fewer LDS update calls
-            // beat the cheaper scan (its skip1 floor is 5.7 vs 2.5 ms).
+            // HX_RANGE2: pair-load variant (16 B/lane dwordx4 loads, the
+            // dedup successor from registers/one shuffle, in-lane pair
+            // merge, direct LDS updates). Default ON since the fill-counter
+            // fix: with updates cheap its lighter scan wins — 5.3 ms vs the
+            // shuffle-pre-reduce kernel's 7.4 at the 1B shape (before the
+            // fix the relation was reversed: 25.4 vs 23.6).
             const bool r2k = [] {
                 const char* e = getenv("HX_RANGE2");
-                return e ? atoi(e) != 0 : false;
+                return e ? atoi(e) != 0 : true;
             }();
             hipError_t re2 =
                 r2k ? hx::launch_scan_agg_range2(
