@@ -2009,7 +2009,8 @@ __device__ __forceinline__ void snap_flush(uint8_t* __restrict__ dst,
     }
 }
 
-extern "C" __global__ void __launch_bounds__(256)
+template <int MINW>
+__global__ void __launch_bounds__(256, MINW)
 k_snappy_decompress(const uint8_t* __restrict__ blob, uint8_t* __restrict__ dec,
                     const SnappyPageDesc* __restrict__ pages, uint32_t n_pages,
                     unsigned long long* err_flag) {
@@ -2645,8 +2646,20 @@ hipError_t launch_snappy(hipStream_t s, const uint8_t* blob, uint8_t* dec,
     uint32_t blocks = (waves_needed + 3) / 4;  // 4 waves per 256-thread block
     if (blocks > 8192) blocks = 8192;
     if (blocks == 0) blocks = 1;
-    hipLaunchKernelGGL(k_snappy_decompress, dim3(blocks), dim3(256), 0, s,
-                       blob, dec, pages, n_pages, err_flag);
+    // decode is 31% issue-busy / 62% parked at the unconstrained 111 VGPRs
+    // (16 waves/CU) — unlike the aggregate it benefits from occupancy;
+    // HX_SNAPPY_MINW selects (default 5: 20 waves/CU).
+    int minw = 5;
+    if (const char* e = getenv("HX_SNAPPY_MINW")) minw = atoi(e);
+    if (minw >= 6)
+        hipLaunchKernelGGL(k_snappy_decompress<6>, dim3(blocks), dim3(256),
+                           0, s, blob, dec, pages, n_pages, err_flag);
+    else if (minw == 5)
+        hipLaunchKernelGGL(k_snappy_decompress<5>, dim3(blocks), dim3(256),
+                           0, s, blob, dec, pages, n_pages, err_flag);
+    else
+        hipLaunchKernelGGL(k_snappy_decompress<1>, dim3(blocks), dim3(256),
+                           0, s, blob, dec, pages, n_pages, err_flag);
     return hipGetLastError();
 }
 
