@@ -2648,8 +2648,11 @@ hipError_t launch_snappy(hipStream_t s, const uint8_t* blob, uint8_t* dec,
     if (blocks == 0) blocks = 1;
     // decode is 31% issue-busy / 62% parked at the unconstrained 111 VGPRs
     // (16 waves/CU) — unlike the aggregate it benefits from occupancy;
-    // HX_SNAPPY_MINW selects (default 5: 20 waves/CU).
-    int minw = 5;
+    // HX_SNAPPY_MINW selects. Default 6 (24 waves/CU, 440 B scratch):
+    // measured best in two same-box headline A/Bs (26.1 vs 27.0/27.1 ms
+    // per step against MINW 1/5 at pipeline 3; 30.4 vs 31.0/31.3 solo) —
+    // the extra waves buy more than the spill costs.
+    int minw = 6;
     if (const char* e = getenv("HX_SNAPPY_MINW")) minw = atoi(e);
     if (minw >= 6)
         hipLaunchKernelGGL(k_snappy_decompress<6>, dim3(blocks), dim3(256),
