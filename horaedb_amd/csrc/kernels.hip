@@ -2654,7 +2654,13 @@ hipError_t launch_snappy(hipStream_t s, const uint8_t* blob, uint8_t* dec,
     // the extra waves buy more than the spill costs.
     int minw = 6;
     if (const char* e = getenv("HX_SNAPPY_MINW")) minw = atoi(e);
-    if (minw >= 6)
+    if (minw >= 8)
+        hipLaunchKernelGGL(k_snappy_decompress<8>, dim3(blocks), dim3(256),
+                           0, s, blob, dec, pages, n_pages, err_flag);
+    else if (minw == 7)
+        hipLaunchKernelGGL(k_snappy_decompress<7>, dim3(blocks), dim3(256),
+                           0, s, blob, dec, pages, n_pages, err_flag);
+    else if (minw == 6)
         hipLaunchKernelGGL(k_snappy_decompress<6>, dim3(blocks), dim3(256),
                            0, s, blob, dec, pages, n_pages, err_flag);
     else if (minw == 5)
