@@ -1840,7 +1840,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     // scratch: compact arrays (n_core) + sort keys in/out (2) + dst (n_total)
     // + 3 perm arrays + counter
     size_t need = size_t(n) * 8 * (n_core + 2 + n_total) +
-                  size_t(n) * 4 * 3 + 256;
+                  size_t(n) * 4 * 3 + 256 + 2 * 2048 * 4;
     hx_status st = ensure_dev(&plan.d_scratch, &plan.scratch_cap, need);
     if (st != HX_OK) return st;
     uint8_t* base = (uint8_t*)plan.d_scratch;
@@ -1859,6 +1859,7 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
     uint64_t* keys_out = (uint64_t*)carve8(n);
     unsigned long long* d_dst = (unsigned long long*)carve8(size_t(n) * n_total);
     unsigned long long* d_nout = (unsigned long long*)carve8(1);
+    uint32_t* compact_scratch = (uint32_t*)carve8(2048);  // 2x2048 u32
     uint32_t* perm_a = (uint32_t*)base; base += size_t(n) * 4;
     uint32_t* perm_b = (uint32_t*)base; base += size_t(n) * 4;
     uint32_t* perm_c = (uint32_t*)base; base += size_t(n) * 4;
@@ -1871,7 +1872,8 @@ hx_status exec_plan(hx_prepared* P, DevPlan& plan, const hx_agg_spec* agg,
                    plan.rep_stride};
     HIP_TRY(hx::launch_compact(s, T, plan.slots, ops, key_claim,
                                bucket ? agg->bucket_ms : 0, lo_bucket,
-                               n_buckets, bstride, plan.t_bstore, co));
+                               n_buckets, bstride, plan.t_bstore, co,
+                               compact_scratch));
     unsigned long long n_groups64 = 0;
     HIP_TRY(hipStreamSynchronize(s));
     HIP_TRY(hipMemcpy(&n_groups64, d_nout, 8, hipMemcpyDeviceToHost));

@@ -54,11 +54,13 @@ hipError_t launch_gather_multi(hipStream_t s,
                                const unsigned long long* const* srcs,
                                uint32_t n_arrays, const uint32_t* perm,
                                unsigned long long* dst, uint32_t n);
+// scratch: device buffer of 2*grid_for(n_slots,256) u32 (two-phase compact
+// counts+bases); null routes to the single-pass kernel.
 hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
                           uint32_t ops, int32_t key_claim, int64_t bucket_ms,
                           int64_t lo_bucket, uint32_t n_buckets,
                           uint32_t bstride, const uint8_t* bstore,
-                          const CompactOut& o);
+                          const CompactOut& o, uint32_t* scratch);
 hipError_t launch_gather_u64(hipStream_t s, const unsigned long long* in,
                              const uint32_t* perm, unsigned long long* out,
                              uint32_t n);

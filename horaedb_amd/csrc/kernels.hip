@@ -1854,6 +1854,148 @@ k_compact(CompactParams C) {
     }
 }
 
+// ---- two-phase compaction (non-bucket slab/state tables) ------------------
+// The single-pass k_compact reserves output positions with one agent-scope
+// atomicAdd per live wave on ONE counter word. At headline fill (10M live in
+// a 16M-slot table) that is ~250k serialized hot-word RMWs ≈ 2.8 ms — the
+// same coherence-point storm the scan kernel's fill counter had (§9c). The
+// two-phase form removes the hot word entirely: per-block live counts
+// (no atomics), a one-block exclusive scan, then an atomic-free write pass
+// positioned by block base + LDS-local offset. Sweep traffic is paid twice
+// but is only ~0.5 GB at headline size.
+
+__device__ __forceinline__ bool compact_live(const CompactParams& C,
+                                             uint32_t i) {
+    const uint8_t* slot = C.table.slab + (size_t)i * C.table.stride;
+    return C.key_claim ? (*(const uint64_t*)slot != KEY_EMPTY)
+                       : (*(const uint32_t*)slot == 2u);
+}
+
+__device__ __forceinline__ void compact_emit(const CompactParams& C,
+                                             uint32_t i,
+                                             unsigned long long j) {
+    const uint8_t* slot = C.table.slab + (size_t)i * C.table.stride;
+    if (C.key_claim && C.table.rep) {
+        C.out_series[j] = *(const uint64_t*)slot;
+        double sum = 0, mnv = HUGE_VAL, mxv = -HUGE_VAL;
+        unsigned long long cntv = 0;
+        for (uint32_t x = 0; x < 8; x++) {
+            const uint8_t* r2 =
+                C.table.rep +
+                ((size_t)x * (C.table.mask + 1ull) + i) * C.table.rep_stride;
+            sum += *(const double*)r2;
+            cntv += *(const unsigned long long*)(r2 + 8);
+            if (C.out_min)
+                mnv = fmin(mnv,
+                           ordered_f64(*(const unsigned long long*)(r2 + 16)));
+            if (C.out_max)
+                mxv = fmax(mxv,
+                           ordered_f64(*(const unsigned long long*)(r2 + 24)));
+        }
+        if (C.out_sum) C.out_sum[j] = sum;
+        if (C.out_cnt) C.out_cnt[j] = cntv;
+        if (C.out_min) C.out_min[j] = mnv;
+        if (C.out_max) C.out_max[j] = mxv;
+    } else if (C.key_claim) {
+        C.out_series[j] = *(const uint64_t*)slot;
+        if (C.out_sum) C.out_sum[j] = *(const double*)(slot + 8);
+        if (C.out_cnt) C.out_cnt[j] = *(const unsigned long long*)(slot + 16);
+        if (C.out_min)
+            C.out_min[j] =
+                ordered_f64(*(const unsigned long long*)(slot + 24));
+        if (C.out_max)
+            C.out_max[j] =
+                ordered_f64(*(const unsigned long long*)(slot + 32));
+    } else {
+        C.out_series[j] = *(const uint64_t*)(slot + 8);
+        if (C.bucket_ms) C.out_bucket[j] = *(const long long*)(slot + 16);
+        if (C.out_sum) C.out_sum[j] = *(const double*)(slot + 24);
+        if (C.out_cnt) C.out_cnt[j] = *(const unsigned long long*)(slot + 32);
+        if (C.out_min)
+            C.out_min[j] =
+                ordered_f64(*(const unsigned long long*)(slot + 40));
+        if (C.out_max)
+            C.out_max[j] =
+                ordered_f64(*(const unsigned long long*)(slot + 48));
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_compact_count(CompactParams C, uint32_t* __restrict__ counts) {
+    __shared__ uint32_t wsum[4];
+    const uint32_t chunk = (C.n_slots + gridDim.x - 1) / gridDim.x;
+    const uint32_t beg = blockIdx.x * chunk;
+    const uint32_t end = beg + chunk < C.n_slots ? beg + chunk : C.n_slots;
+    uint32_t cnt = 0;
+    for (uint32_t i0 = beg; i0 < end; i0 += blockDim.x) {
+        const uint32_t i = i0 + threadIdx.x;
+        const bool live = i < end && compact_live(C, i);
+        const unsigned long long m = __ballot(live);
+        if ((threadIdx.x & 63) == 0) cnt += (uint32_t)__popcll(m);
+    }
+    if ((threadIdx.x & 63) == 0) wsum[threadIdx.x >> 6] = cnt;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        counts[blockIdx.x] = wsum[0] + wsum[1] + wsum[2] + wsum[3];
+}
+
+// One-block exclusive scan of <=2048 per-block counts -> bases + total.
+extern "C" __global__ void __launch_bounds__(256)
+k_scan_counts(const uint32_t* __restrict__ counts, uint32_t nb,
+              uint32_t* __restrict__ bases,
+              unsigned long long* __restrict__ n_out) {
+    __shared__ uint32_t tsum[256];
+    const uint32_t per = (nb + 255) / 256;
+    const uint32_t b0 = threadIdx.x * per;
+    const uint32_t b1 = b0 + per < nb ? b0 + per : nb;
+    uint32_t s = 0;
+    for (uint32_t j = b0; j < b1; j++) s += counts[j];
+    tsum[threadIdx.x] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t acc = 0;
+        for (int t = 0; t < 256; t++) {
+            const uint32_t v = tsum[t];
+            tsum[t] = acc;
+            acc += v;
+        }
+        *n_out = acc;
+    }
+    __syncthreads();
+    uint32_t acc = tsum[threadIdx.x];
+    for (uint32_t j = b0; j < b1; j++) {
+        bases[j] = acc;
+        acc += counts[j];
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_compact_write(CompactParams C, const uint32_t* __restrict__ bases) {
+    __shared__ uint32_t blk_off;
+    if (threadIdx.x == 0) blk_off = 0;
+    __syncthreads();
+    const uint32_t chunk = (C.n_slots + gridDim.x - 1) / gridDim.x;
+    const uint32_t beg = blockIdx.x * chunk;
+    const uint32_t end = beg + chunk < C.n_slots ? beg + chunk : C.n_slots;
+    const unsigned long long base = bases[blockIdx.x];
+    const int lane = threadIdx.x & 63;
+    for (uint32_t i0 = beg; i0 < end; i0 += blockDim.x) {
+        const uint32_t i = i0 + threadIdx.x;
+        const bool live = i < end && compact_live(C, i);
+        const unsigned long long mask = __ballot(live);
+        if (!mask) continue;
+        const int leader = __ffsll((unsigned long long)mask) - 1;
+        uint32_t wave_base = 0;
+        if (lane == leader)
+            wave_base = atomicAdd(&blk_off, (uint32_t)__popcll(mask));
+        wave_base = __shfl(wave_base, leader, 64);
+        if (live)
+            compact_emit(C, i,
+                         base + wave_base +
+                             __popcll(mask & ((1ull << lane) - 1ull)));
+    }
+}
+
 // Gather 8-byte elements by permutation (applies the sort order).
 struct GatherMulti {
     const unsigned long long* src[8];
@@ -2849,7 +2991,7 @@ hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
                           uint32_t ops, int32_t key_claim, int64_t bucket_ms,
                           int64_t lo_bucket, uint32_t n_buckets,
                           uint32_t bstride, const uint8_t* bstore,
-                          const CompactOut& o) {
+                          const CompactOut& o, uint32_t* scratch) {
     CompactParams C;
     C.table = t;
     C.n_slots = n_slots;
@@ -2867,7 +3009,26 @@ hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
     C.out_min = o.vmin;
     C.out_max = o.vmax;
     C.n_out = o.n_out;
-    hipLaunchKernelGGL(k_compact, dim3(grid_for(n_slots, 256)), dim3(256), 0, s, C);
+    const uint32_t grid = grid_for(n_slots, 256);
+    // non-bucket tables: two-phase compaction (count -> scan -> write)
+    // avoids the single hot n_out counter (~250k serialized agent-scope
+    // RMWs = ~3 ms at headline fill). scratch = 2*grid u32 (counts, bases).
+    static const bool legacy = [] {
+        const char* e = getenv("HX_COMPACT_LEGACY");
+        return e && atoi(e) != 0;
+    }();
+    if (n_buckets == 0 && scratch && !legacy) {
+        uint32_t* counts = scratch;
+        uint32_t* bases = scratch + grid;
+        hipLaunchKernelGGL(k_compact_count, dim3(grid), dim3(256), 0, s,
+                           C, counts);
+        hipLaunchKernelGGL(k_scan_counts, dim3(1), dim3(256), 0, s,
+                           counts, grid, bases, C.n_out);
+        hipLaunchKernelGGL(k_compact_write, dim3(grid), dim3(256), 0, s,
+                           C, bases);
+        return hipGetLastError();
+    }
+    hipLaunchKernelGGL(k_compact, dim3(grid), dim3(256), 0, s, C);
     return hipGetLastError();
 }
 
