@@ -1865,15 +1865,40 @@ k_compact(CompactParams C) {
 // but is only ~0.5 GB at headline size.
 
 __device__ __forceinline__ bool compact_live(const CompactParams& C,
-                                             uint32_t i) {
-    const uint8_t* slot = C.table.slab + (size_t)i * C.table.stride;
+                                             size_t p) {
+    if (C.n_buckets) {  // direct-indexed bucket mode: p = (slot, bucket)
+        const uint32_t slot_i = (uint32_t)(p / C.n_buckets);
+        const uint64_t key = *(const uint64_t*)(
+            C.table.slab + (size_t)slot_i * C.table.stride);
+        return key != KEY_EMPTY &&
+               *(const unsigned long long*)(C.bstore + p * C.bstride + 8) > 0;
+    }
+    const uint8_t* slot = C.table.slab + p * C.table.stride;
     return C.key_claim ? (*(const uint64_t*)slot != KEY_EMPTY)
                        : (*(const uint32_t*)slot == 2u);
 }
 
 __device__ __forceinline__ void compact_emit(const CompactParams& C,
-                                             uint32_t i,
+                                             size_t p,
                                              unsigned long long j) {
+    if (C.n_buckets) {
+        const uint32_t slot_i = (uint32_t)(p / C.n_buckets);
+        const uint8_t* brow = C.bstore + p * C.bstride;
+        C.out_series[j] = *(const uint64_t*)(
+            C.table.slab + (size_t)slot_i * C.table.stride);
+        C.out_bucket[j] = C.lo_bucket + (long long)(p % C.n_buckets);
+        if (C.out_sum) C.out_sum[j] = *(const double*)brow;
+        if (C.out_cnt)
+            C.out_cnt[j] = *(const unsigned long long*)(brow + 8);
+        if (C.out_min)
+            C.out_min[j] =
+                ordered_f64(*(const unsigned long long*)(brow + 16));
+        if (C.out_max)
+            C.out_max[j] =
+                ordered_f64(*(const unsigned long long*)(brow + 24));
+        return;
+    }
+    const uint32_t i = (uint32_t)p;
     const uint8_t* slot = C.table.slab + (size_t)i * C.table.stride;
     if (C.key_claim && C.table.rep) {
         C.out_series[j] = *(const uint64_t*)slot;
@@ -1920,15 +1945,20 @@ __device__ __forceinline__ void compact_emit(const CompactParams& C,
     }
 }
 
+__device__ __forceinline__ size_t compact_total(const CompactParams& C) {
+    return C.n_buckets ? (size_t)C.n_slots * C.n_buckets : (size_t)C.n_slots;
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 k_compact_count(CompactParams C, uint32_t* __restrict__ counts) {
     __shared__ uint32_t wsum[4];
-    const uint32_t chunk = (C.n_slots + gridDim.x - 1) / gridDim.x;
-    const uint32_t beg = blockIdx.x * chunk;
-    const uint32_t end = beg + chunk < C.n_slots ? beg + chunk : C.n_slots;
+    const size_t total = compact_total(C);
+    const size_t chunk = (total + gridDim.x - 1) / gridDim.x;
+    const size_t beg = blockIdx.x * chunk;
+    const size_t end = beg + chunk < total ? beg + chunk : total;
     uint32_t cnt = 0;
-    for (uint32_t i0 = beg; i0 < end; i0 += blockDim.x) {
-        const uint32_t i = i0 + threadIdx.x;
+    for (size_t i0 = beg; i0 < end; i0 += blockDim.x) {
+        const size_t i = i0 + threadIdx.x;
         const bool live = i < end && compact_live(C, i);
         const unsigned long long m = __ballot(live);
         if ((threadIdx.x & 63) == 0) cnt += (uint32_t)__popcll(m);
@@ -1974,13 +2004,14 @@ k_compact_write(CompactParams C, const uint32_t* __restrict__ bases) {
     __shared__ uint32_t blk_off;
     if (threadIdx.x == 0) blk_off = 0;
     __syncthreads();
-    const uint32_t chunk = (C.n_slots + gridDim.x - 1) / gridDim.x;
-    const uint32_t beg = blockIdx.x * chunk;
-    const uint32_t end = beg + chunk < C.n_slots ? beg + chunk : C.n_slots;
+    const size_t total = compact_total(C);
+    const size_t chunk = (total + gridDim.x - 1) / gridDim.x;
+    const size_t beg = blockIdx.x * chunk;
+    const size_t end = beg + chunk < total ? beg + chunk : total;
     const unsigned long long base = bases[blockIdx.x];
     const int lane = threadIdx.x & 63;
-    for (uint32_t i0 = beg; i0 < end; i0 += blockDim.x) {
-        const uint32_t i = i0 + threadIdx.x;
+    for (size_t i0 = beg; i0 < end; i0 += blockDim.x) {
+        const size_t i = i0 + threadIdx.x;
         const bool live = i < end && compact_live(C, i);
         const unsigned long long mask = __ballot(live);
         if (!mask) continue;
@@ -3009,15 +3040,20 @@ hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
     C.out_min = o.vmin;
     C.out_max = o.vmax;
     C.n_out = o.n_out;
-    const uint32_t grid = grid_for(n_slots, 256);
-    // non-bucket tables: two-phase compaction (count -> scan -> write)
-    // avoids the single hot n_out counter (~250k serialized agent-scope
-    // RMWs = ~3 ms at headline fill). scratch = 2*grid u32 (counts, bases).
+    const size_t total =
+        n_buckets ? (size_t)n_slots * n_buckets : (size_t)n_slots;
+    const uint32_t grid =
+        total > (size_t)2048 * 256
+            ? 2048u
+            : (uint32_t)((total + 255) / 256 ? (total + 255) / 256 : 1);
+    // two-phase compaction (count -> scan -> write) avoids the single hot
+    // n_out counter (~250k serialized agent-scope RMWs = ~3 ms at headline
+    // fill). scratch = 2*grid u32 (counts, bases).
     static const bool legacy = [] {
         const char* e = getenv("HX_COMPACT_LEGACY");
         return e && atoi(e) != 0;
     }();
-    if (n_buckets == 0 && scratch && !legacy) {
+    if (scratch && !legacy) {
         uint32_t* counts = scratch;
         uint32_t* bases = scratch + grid;
         hipLaunchKernelGGL(k_compact_count, dim3(grid), dim3(256), 0, s,
