@@ -3048,12 +3048,16 @@ hipError_t launch_compact(hipStream_t s, const AggTable& t, uint32_t n_slots,
             : (uint32_t)((total + 255) / 256 ? (total + 255) / 256 : 1);
     // two-phase compaction (count -> scan -> write) avoids the single hot
     // n_out counter (~250k serialized agent-scope RMWs = ~3 ms at headline
-    // fill). scratch = 2*grid u32 (counts, bases).
+    // fill). scratch = 2*grid u32 (counts, bases). NON-BUCKET ONLY: the
+    // bucket sweep walks slots x n_buckets positions at low live fraction,
+    // and doubling that sweep measured ~5% SLOWER than the single-pass
+    // kernel's atomics (201.8 vs 192.9 ms/step same-box) — the storm is
+    // proportional to LIVE waves, which bucket sweeps have few of.
     static const bool legacy = [] {
         const char* e = getenv("HX_COMPACT_LEGACY");
         return e && atoi(e) != 0;
     }();
-    if (scratch && !legacy) {
+    if (n_buckets == 0 && scratch && !legacy) {
         uint32_t* counts = scratch;
         uint32_t* bases = scratch + grid;
         hipLaunchKernelGGL(k_compact_count, dim3(grid), dim3(256), 0, s,
